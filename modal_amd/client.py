@@ -1,0 +1,176 @@
+"""Client: the seam between user-facing handles and the control plane.
+
+Parity: the reference's ``_Client`` singleton (/root/reference/py/modal/client.py:78,
+``from_env`` :209) that owns the gRPC stub. Here there is no network:
+
+* in the client process, ``_Client.svc`` IS the in-process ``Scheduler`` —
+  zero-serialization method calls;
+* in worker/sandbox processes, ``svc`` is a ``SchedulerProxy`` speaking the
+  Unix-socket RPC protocol, giving user code inside workers the same
+  Queue/Dict/Volume/Function semantics (reference CLIENT_TYPE_CONTAINER,
+  client.py:238).
+
+Post-fork safety: the singleton is keyed by pid (reference client.py:356-369).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import os
+from typing import Any, Optional
+
+from ._sync import synchronize_api
+from .exception import (
+    AlreadyExistsError,
+    ClientClosed,
+    Error,
+    ExecutionError,
+    InvalidError,
+    NotFoundError,
+    QueueEmptyError,
+    QueueFullError,
+    RemoteError,
+)
+from .scheduler.transport import Connection, RemoteRPCError
+
+HEARTBEAT_INTERVAL = 15.0  # parity: reference client.py:29
+
+_ERROR_BY_NAME = {
+    "NotFoundError": NotFoundError,
+    "AlreadyExistsError": AlreadyExistsError,
+    "InvalidError": InvalidError,
+    "ExecutionError": ExecutionError,
+    "QueueEmptyError": QueueEmptyError,
+    "QueueFullError": QueueFullError,
+    "TimeoutError": TimeoutError,
+}
+
+
+def map_remote_error(exc: RemoteRPCError) -> Exception:
+    cls = _ERROR_BY_NAME.get(exc.code or "", RemoteError)
+    return cls(str(exc))
+
+
+class SchedulerProxy:
+    """Socket-backed scheduler: async methods forwarded as RPC frames."""
+
+    def __init__(self, conn: Connection):
+        self._conn = conn
+
+    def __getattr__(self, name: str) -> Any:
+        if name.startswith("_"):
+            raise AttributeError(name)
+
+        async def call(**kwargs: Any) -> Any:
+            try:
+                return await self._conn.call(name, kwargs)
+            except RemoteRPCError as exc:
+                raise map_remote_error(exc) from None
+
+        call.__name__ = name
+        return call
+
+
+class _Client:
+    """Process-wide access point to the control plane."""
+
+    _singleton: Optional["_Client"] = None
+    _singleton_pid: Optional[int] = None
+
+    def __init__(self, svc: Any, client_type: str = "client", run_dir: Optional[str] = None):
+        self.svc = svc
+        self.client_type = client_type  # "client" | "container"
+        self._closed = False
+        self._run_dir = run_dir
+
+    @property
+    def is_container_client(self) -> bool:
+        return self.client_type == "container"
+
+    @property
+    def run_dir(self) -> Optional[str]:
+        if self._run_dir:
+            return self._run_dir
+        sched = self.svc
+        return getattr(sched, "run_dir", None)
+
+    @property
+    def blob_store(self) -> Any:
+        """The shared content-addressed store (same filesystem on all sides)."""
+        if getattr(self, "_blob_store", None) is None:
+            svc = self.svc
+            if hasattr(svc, "blob_store"):
+                self._blob_store = svc.blob_store
+            else:
+                import os as _os
+
+                from .scheduler.blobs import BlobStore
+
+                run_dir = self.run_dir
+                if run_dir is None:
+                    return None
+                self._blob_store = BlobStore(_os.path.join(run_dir, "blobs"))
+        return self._blob_store
+
+    @classmethod
+    async def from_env(cls) -> "_Client":
+        """The default client for this process (singleton, pid-keyed)."""
+        pid = os.getpid()
+        if cls._singleton is not None and cls._singleton_pid == pid and not cls._singleton._closed:
+            return cls._singleton
+        # worker processes install their client explicitly before user code runs;
+        # reaching here in a worker means env-based attach to the host scheduler.
+        socket_path = os.environ.get("MODAL_AMD_ATTACH_SOCKET")
+        if socket_path:
+            client = await cls.connect(socket_path)
+        else:
+            from .scheduler.core import Scheduler
+
+            scheduler = Scheduler()
+            await scheduler.start()
+            client = cls(scheduler, "client")
+        cls._singleton = client
+        cls._singleton_pid = pid
+        return client
+
+    @classmethod
+    async def connect(cls, socket_path: str, client_type: str = "client") -> "_Client":
+        """Attach to another process's scheduler over its Unix socket."""
+        reader, writer = await asyncio.open_unix_connection(socket_path)
+
+        async def handler(msg: dict) -> None:
+            pass
+
+        conn = Connection(reader, writer, handler)
+        conn.start()
+        await conn.send({"t": "hello", "role": "client"})
+        return cls(SchedulerProxy(conn), client_type, run_dir=os.path.dirname(socket_path))
+
+    @classmethod
+    def set_default(cls, client: "_Client") -> None:
+        cls._singleton = client
+        cls._singleton_pid = os.getpid()
+
+    async def close(self) -> None:
+        if self._closed:
+            return
+        self._closed = True
+        svc = self.svc
+        if hasattr(svc, "_conn"):
+            await svc._conn.close()
+        elif hasattr(svc, "stop"):
+            await svc.stop()
+        if _Client._singleton is self:
+            _Client._singleton = None
+
+    async def __aenter__(self) -> "_Client":
+        return self
+
+    async def __aexit__(self, *exc: Any) -> None:
+        await self.close()
+
+    def __repr__(self) -> str:
+        return f"<Client {self.client_type} svc={type(self.svc).__name__}>"
+
+
+Client = synchronize_api(_Client, "Client")

@@ -1,0 +1,575 @@
+"""Function objects + the invocation engine.
+
+Parity: /root/reference/py/modal/_functions.py —
+``_Function`` (:598, from_local :666), the unary invocation state machine
+``_Invocation`` (:124, create :141 with pipelined inputs, run_function
+:286-316), output processing (function_utils.py:527-583), and
+``FunctionCall`` (:2121-2214).
+
+MI355X-native shape: an invocation is a direct async call into the in-process
+scheduler (zero serialization of control messages), inputs land on per-GPU
+worker queues over a Unix socket, and the 55 s long-poll collapses to awaiting
+an asyncio future. Blob offload kicks in above the 2 MiB inline limit
+(parity: blob_utils.py:36).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import inspect
+import time
+from typing import Any, AsyncGenerator, Callable, Optional
+
+from ._object import _Object, live_method
+from ._serialization import (
+    DataFormat,
+    GeneratorDone,
+    deserialize,
+    serialize,
+)
+from ._sync import synchronize_api
+from .exception import (
+    Error,
+    ExecutionError,
+    FunctionTimeoutError,
+    InternalFailure,
+    InvalidError,
+    NotFoundError,
+    RemoteError,
+)
+from .scheduler.blobs import INLINE_LIMIT
+from .scheduler.calls import (
+    GENERIC_STATUS_FAILURE,
+    GENERIC_STATUS_INTERNAL_FAILURE,
+    GENERIC_STATUS_SUCCESS,
+    GENERIC_STATUS_TERMINATED,
+    GENERIC_STATUS_TIMEOUT,
+)
+
+
+class FunctionCallCancelledError(Error):
+    """The function call was cancelled before producing an output."""
+
+
+def make_payload_item(client: Any, args: tuple, kwargs: dict) -> dict:
+    """Serialize one input; offload to the CAS above the inline limit."""
+    payload = serialize(("P", (args, kwargs)))
+    if len(payload) > INLINE_LIMIT:
+        store = client.blob_store
+        if store is not None:
+            return {"payload": b"", "payload_blob": store.put(payload)}
+    return {"payload": payload}
+
+
+def process_output_item(item: dict, client: Any) -> Any:
+    """Decode one completed output: return the value or raise the remote error
+    (parity: _process_result, reference function_utils.py:527-583)."""
+    data = item.get("data")
+    if item.get("data_blob"):
+        data = client.blob_store.get(item["data_blob"])
+    status = item.get("status")
+    if status == GENERIC_STATUS_SUCCESS:
+        if item.get("format") == DataFormat.GENERATOR_DONE:
+            from ._serialization import deserialize_data_format
+
+            return deserialize_data_format(data, DataFormat.GENERATOR_DONE)
+        return deserialize(data) if data is not None else None
+    if status == GENERIC_STATUS_TIMEOUT:
+        raise FunctionTimeoutError(item.get("exc") or "Function call timed out")
+    if status == GENERIC_STATUS_TERMINATED:
+        raise FunctionCallCancelledError(item.get("exc") or "Function call was cancelled")
+    if status == GENERIC_STATUS_INTERNAL_FAILURE:
+        raise InternalFailure(item.get("exc") or "internal failure")
+    if status == GENERIC_STATUS_FAILURE:
+        if data is not None:
+            try:
+                exc = deserialize(data)
+            except Exception:
+                raise RemoteError(item.get("exc") or "remote exception (undeserializable)") from None
+            if isinstance(exc, BaseException):
+                raise exc
+        raise RemoteError(item.get("exc") or "remote exception")
+    raise ExecutionError(f"Unknown output status {status}")
+
+
+def _is_inproc(svc: Any) -> bool:
+    return hasattr(svc, "pool")
+
+
+async def await_output_item(
+    client: Any, call_id: str, idx: int = 0, timeout: Optional[float] = None
+) -> dict:
+    """Wait for one input's final output.
+
+    In-process: await the input's future directly (the 55 s
+    FunctionGetOutputs long-poll of the reference collapses to this).
+    Over the socket: poll function_get_outputs.
+    """
+    svc = client.svc
+    if _is_inproc(svc):
+        rec = await svc.function_wait_output(call_id, idx, timeout)
+        return {
+            "idx": idx,
+            "status": rec.status,
+            "data": rec.output,
+            "data_blob": rec.output_blob,
+            "format": rec.output_format,
+            "exc": rec.exc_repr,
+        }
+    deadline = None if timeout is None else time.monotonic() + timeout
+    while True:
+        remaining = 55.0 if deadline is None else min(55.0, deadline - time.monotonic())
+        if remaining <= 0:
+            raise TimeoutError(f"Timed out waiting for output of {call_id}")
+        outs = await svc.function_get_outputs(
+            function_call_id=call_id, max_values=16, timeout=remaining
+        )
+        for out in outs:
+            if out["idx"] == idx:
+                return out
+
+
+class _Invocation:
+    """Single-call state machine (parity: reference _functions.py:124)."""
+
+    def __init__(self, client: Any, call_id: str):
+        self.client = client
+        self.call_id = call_id
+
+    @classmethod
+    async def create(
+        cls, fn: "_Function", args: tuple, kwargs: dict, kind: str = "unary"
+    ) -> "_Invocation":
+        client = fn._client
+        item = make_payload_item(client, args, kwargs)
+        if fn._method_name:
+            item["method"] = fn._method_name
+        resp = await client.svc.function_map(
+            function_id=fn.object_id, kind=kind, pipelined_inputs=[item]
+        )
+        call_id = resp["function_call_id"]
+        await client.svc.function_finish_inputs(function_call_id=call_id)
+        return cls(client, call_id)
+
+    async def run_function(self, timeout: Optional[float] = None) -> Any:
+        item = await await_output_item(self.client, self.call_id, 0, timeout)
+        return process_output_item(item, self.client)
+
+    async def run_generator(self) -> AsyncGenerator[Any, None]:
+        """Consume the generator data channel until the done marker
+        (parity: generator merge, reference _functions.py:339)."""
+        svc = self.client.svc
+        next_index = 0
+        buffered: dict[int, Any] = {}
+        done = False
+        while not done:
+            entries = await svc.generator_poll(
+                function_call_id=self.call_id, idx=0, timeout=5.0
+            )
+            if not entries:
+                # no data: check whether the call already finished (error path)
+                info = await svc.function_call_info(function_call_id=self.call_id)
+                if info.get("completed", 0) >= 1:
+                    item = await await_output_item(self.client, self.call_id, 0, 1.0)
+                    process_output_item(item, self.client)  # raises on failure
+                    return
+                continue
+            for index, data, fmt, is_done in entries:
+                if is_done:
+                    done = True
+                    continue
+                buffered[index] = deserialize(data)
+                while next_index in buffered:
+                    yield buffered.pop(next_index)
+                    next_index += 1
+        # drain the final output so exceptions surface
+        item = await await_output_item(self.client, self.call_id, 0, None)
+        result = process_output_item(item, self.client)
+        if isinstance(result, GeneratorDone):
+            return
+
+
+class _Function(_Object, type_kind="function"):
+    """Handle for a registered function (``fu-``)."""
+
+    _raw_f: Optional[Callable]
+    _app: Any
+    _options: dict
+    _is_generator: bool
+    _method_name: str
+
+    def _init_attrs(self) -> None:
+        self._raw_f = None
+        self._app = None
+        self._options = {}
+        self._is_generator = False
+        self._method_name = ""
+        self._web_url: Optional[str] = None
+
+    # -- constructors ----------------------------------------------------
+    @classmethod
+    def from_local(cls, raw_f: Callable, app: Any, options: dict) -> "_Function":
+        is_generator = bool(
+            options.get("is_generator")
+            or inspect.isgeneratorfunction(raw_f)
+            or inspect.isasyncgenfunction(raw_f)
+        )
+        options = dict(options)
+        options["is_generator"] = is_generator
+
+        async def _load(obj: "_Function", resolver: Any, existing: Any) -> None:
+            definition = serialize(raw_f)
+            opts = dict(options)
+            opts["is_generator"] = is_generator
+            resp = await resolver.client.svc.function_create(
+                app_id=resolver.app_id,
+                name=options.get("name") or raw_f.__name__,
+                definition=definition,
+                options=opts,
+            )
+            obj._hydrate(resp["function_id"], resolver.client, resp["metadata"])
+
+        obj = cls._from_loader(_load, rep=f"Function({raw_f.__qualname__})")
+        obj._raw_f = raw_f
+        obj._app = app
+        obj._options = options
+        obj._is_generator = is_generator
+        return obj
+
+    @classmethod
+    def from_name(
+        cls, app_name: str, name: str, *, environment_name: str = "", namespace: Any = None
+    ) -> "_Function":
+        """Reference a function on a previously deployed app (lazy lookup;
+        parity: reference Function.from_name)."""
+
+        async def _load(obj: "_Function", resolver: Any, existing: Any) -> None:
+            resp = await resolver.client.svc.function_lookup(
+                app_name=app_name, name=name, environment=environment_name
+            )
+            obj._hydrate(resp["function_id"], resolver.client, resp["metadata"])
+
+        return cls._from_loader(_load, rep=f"Function.from_name({app_name}/{name})")
+
+    @classmethod
+    async def lookup(
+        cls, app_name: str, name: str, *, environment_name: str = ""
+    ) -> "_Function":
+        obj = cls.from_name(app_name, name, environment_name=environment_name)
+        return await obj.hydrate()
+
+    # -- metadata --------------------------------------------------------
+    def _hydrate_metadata(self, metadata: dict) -> None:
+        self._is_generator = bool(metadata.get("is_generator"))
+        self._web_url = metadata.get("web_url")
+
+    def _get_metadata(self) -> dict:
+        return {
+            "is_generator": self._is_generator,
+            "function_name": self._options.get("name") if self._options else None,
+        }
+
+    @property
+    def is_generator(self) -> bool:
+        return self._is_generator
+
+    @property
+    def web_url(self) -> Optional[str]:
+        return self._web_url
+
+    def get_raw_f(self) -> Callable:
+        if self._raw_f is None:
+            raise InvalidError("This function handle has no local definition")
+        return self._raw_f
+
+    def _serialize_definition(self) -> bytes:
+        """Cloudpickle the executable payload for workers. Class services
+        provide a spec dict through _definition_provider (see cls.py)."""
+        provider = self._options.get("_definition_provider")
+        if provider is not None:
+            return serialize(provider())
+        return serialize(self.get_raw_f())
+
+    @property
+    def info(self) -> dict:
+        return dict(self._options)
+
+    # -- invocation ------------------------------------------------------
+    @live_method
+    async def remote(self, *args: Any, **kwargs: Any) -> Any:
+        """Execute remotely and wait (parity: reference _functions.py:1825)."""
+        if self._is_generator:
+            raise InvalidError("Use .remote_gen() for generator functions")
+        invocation = await _Invocation.create(self, args, kwargs, "unary")
+        return await invocation.run_function()
+
+    async def remote_gen(self, *args: Any, **kwargs: Any) -> AsyncGenerator[Any, None]:
+        """Execute a generator function remotely, streaming items
+        (parity: reference _functions.py:1846)."""
+        if not self._is_hydrated:
+            await self.hydrate()
+        if not self._is_generator:
+            raise InvalidError(".remote_gen() requires a generator function")
+        invocation = await _Invocation.create(self, args, kwargs, "unary")
+        async for value in invocation.run_generator():
+            yield value
+
+    def local(self, *args: Any, **kwargs: Any) -> Any:
+        """Run the underlying function in-process (reference :1883)."""
+        return self.get_raw_f()(*args, **kwargs)
+
+    @live_method
+    async def spawn(self, *args: Any, **kwargs: Any) -> "_FunctionCall":
+        """Start the call without waiting; returns a FunctionCall
+        (parity: reference _functions.py:1984)."""
+        invocation = await _Invocation.create(self, args, kwargs, "spawn")
+        fc = _FunctionCall._new_hydrated(invocation.call_id, self._client, None)
+        fc._is_generator = self._is_generator
+        return fc
+
+    # -- fan-out (delegates to the map engine) ----------------------------
+    async def _map_inner(
+        self,
+        input_iter: Any,
+        kwargs: dict,
+        order_outputs: bool,
+        return_exceptions: bool,
+        wrap_returned_exceptions: bool,
+    ) -> AsyncGenerator[Any, None]:
+        from .parallel.map import map_invocation
+
+        if not self._is_hydrated:
+            await self.hydrate()
+        async for out in map_invocation(
+            self, input_iter, kwargs, order_outputs, return_exceptions, wrap_returned_exceptions
+        ):
+            yield out
+
+    async def map_async(
+        self,
+        *input_iterators: Any,
+        kwargs: Optional[dict] = None,
+        order_outputs: bool = True,
+        return_exceptions: bool = False,
+        wrap_returned_exceptions: bool = True,
+    ) -> AsyncGenerator[Any, None]:
+        def gen_args() -> Any:
+            for combo in zip(*input_iterators):
+                yield (combo, {})
+
+        async for out in self._map_inner(
+            gen_args(), kwargs or {}, order_outputs, return_exceptions, wrap_returned_exceptions
+        ):
+            yield out
+
+    async def starmap_async(
+        self,
+        input_iterator: Any,
+        *,
+        kwargs: Optional[dict] = None,
+        order_outputs: bool = True,
+        return_exceptions: bool = False,
+        wrap_returned_exceptions: bool = True,
+    ) -> AsyncGenerator[Any, None]:
+        def gen_args() -> Any:
+            for item in input_iterator:
+                args = tuple(item) if isinstance(item, (list, tuple)) else (item,)
+                yield (args, {})
+
+        async for out in self._map_inner(
+            gen_args(), kwargs or {}, order_outputs, return_exceptions, wrap_returned_exceptions
+        ):
+            yield out
+
+    async def for_each_async(
+        self, *input_iterators: Any, kwargs: Optional[dict] = None, ignore_exceptions: bool = False
+    ) -> None:
+        async for _ in self.map_async(
+            *input_iterators,
+            kwargs=kwargs,
+            order_outputs=False,
+            return_exceptions=ignore_exceptions,
+            wrap_returned_exceptions=False,
+        ):
+            pass
+
+    @live_method
+    async def spawn_map(self, *input_iterators: Any, kwargs: Optional[dict] = None) -> "_FunctionCall":
+        """Enqueue the whole fan-out without consuming outputs
+        (parity: reference parallel_map.py:1227)."""
+        from .parallel.map import spawn_map_invocation
+
+        items = [(combo, {}) for combo in zip(*input_iterators)]
+        call_id = await spawn_map_invocation(self, items, kwargs or {})
+        fc = _FunctionCall._new_hydrated(call_id, self._client, None)
+        return fc
+
+    # -- management ------------------------------------------------------
+    @live_method
+    async def update_autoscaler(
+        self,
+        *,
+        min_containers: Optional[int] = None,
+        max_containers: Optional[int] = None,
+        buffer_containers: Optional[int] = None,
+        scaledown_window: Optional[float] = None,
+    ) -> None:
+        await self._client.svc.function_update_autoscaler(
+            function_id=self.object_id,
+            min_containers=min_containers,
+            max_containers=max_containers,
+            buffer_containers=buffer_containers,
+            scaledown_window=scaledown_window,
+        )
+
+    @live_method
+    async def keep_warm(self, warm_pool_size: int) -> None:
+        await self._client.svc.function_update_autoscaler(
+            function_id=self.object_id, min_containers=warm_pool_size
+        )
+
+    @live_method
+    async def get_current_stats(self) -> dict:
+        return await self._client.svc.function_get_current_stats(function_id=self.object_id)
+
+    def with_options(
+        self,
+        *,
+        timeout: Optional[float] = None,
+        retries: Any = None,
+        max_containers: Optional[int] = None,
+        **extra: Any,
+    ) -> "_Function":
+        """A new handle with overridden options (parity: reference :1577)."""
+        new_options = dict(self._options)
+        if timeout is not None:
+            new_options["timeout"] = timeout
+        if retries is not None:
+            new_options["retries"] = retries._to_policy_dict() if hasattr(retries, "_to_policy_dict") else retries
+        if max_containers is not None:
+            new_options["max_containers"] = max_containers
+        new_options.update(extra)
+        if self._raw_f is not None and self._app is not None:
+            return _Function.from_local(self._raw_f, self._app, new_options)
+        raise InvalidError("with_options requires a locally defined function")
+
+    def __call__(self, *args: Any, **kwargs: Any) -> Any:
+        raise InvalidError(
+            f"Functions are invoked with `.remote()`, `.local()`, `.map()` etc. "
+            f"— not called directly (tried to call {self._rep})"
+        )
+
+
+class _FunctionCall(_Object, type_kind="function_call"):
+    """Handle for an in-flight or completed call (``fc-``);
+    parity: reference _functions.py:2121."""
+
+    def _init_attrs(self) -> None:
+        self._is_generator = False
+        self._cached: Any = None
+        self._has_cached = False
+
+    @live_method
+    async def get(self, timeout: Optional[float] = None) -> Any:
+        if self._has_cached:
+            return self._cached
+        item = await await_output_item(self._client, self.object_id, 0, timeout)
+        value = process_output_item(item, self._client)
+        self._cached = value
+        self._has_cached = True
+        return value
+
+    async def __aiter__(self) -> AsyncGenerator[Any, None]:
+        """Iterate generator outputs of a spawned generator call."""
+        inv = _Invocation(self._client, self.object_id)
+        async for value in inv.run_generator():
+            yield value
+
+    @live_method
+    async def get_call_graph(self) -> list:
+        info = await self._client.svc.function_call_info(function_call_id=self.object_id)
+        return [info]
+
+    @live_method
+    async def cancel(self, terminate_containers: bool = False) -> None:
+        await self._client.svc.function_call_cancel(
+            function_call_id=self.object_id, terminate_containers=terminate_containers
+        )
+
+    @classmethod
+    def from_id(cls, function_call_id: str, client: Any = None) -> "_FunctionCall":
+        async def _load(obj: "_FunctionCall", resolver: Any, existing: Any) -> None:
+            obj._hydrate(function_call_id, resolver.client, None)
+
+        obj = cls._from_loader(_load, rep=f"FunctionCall({function_call_id})")
+        if client is not None:
+            obj._hydrate(function_call_id, client, None)
+        return obj
+
+    @staticmethod
+    async def gather(*function_calls: "_FunctionCall") -> list:
+        return list(await asyncio.gather(*(fc.get() for fc in function_calls)))
+
+
+Function = synchronize_api(_Function, "Function")
+FunctionCall = synchronize_api(_FunctionCall, "FunctionCall")
+
+
+def _install_sync_map_methods() -> None:
+    """Attach .map/.starmap/.for_each with dual sync/async forms.
+
+    The reference exposes ``fn.map`` as a sync generator whose ``.aio`` twin
+    is async (reference parallel_map.py:1032-1088); our generic wrapper
+    machinery handles methods, so these iterator-valued hybrids are wired
+    explicitly.
+    """
+    from ._sync import synchronizer, unwrap, wrap
+
+    def make(name: str, async_name: str) -> Any:
+        class _MapDescriptor:
+            def __get__(self, obj: Any, objtype: Any = None) -> Any:
+                if obj is None:
+                    return self
+                impl = obj._impl
+
+                def blocking(*args: Any, **kwargs: Any) -> Any:
+                    agen = getattr(impl, async_name)(*unwrap(args), **unwrap(kwargs))
+                    for item in synchronizer.run_generator_sync(agen):
+                        yield wrap(item)
+
+                async def aio(*args: Any, **kwargs: Any) -> Any:
+                    agen = getattr(impl, async_name)(*unwrap(args), **unwrap(kwargs))
+                    async for item in synchronizer.run_generator_async(agen):
+                        yield wrap(item)
+
+                from ._sync import _AioCallable
+
+                return _AioCallable(blocking, aio, getattr(impl, async_name))
+
+        return _MapDescriptor()
+
+    setattr(Function, "map", make("map", "map_async"))
+    setattr(Function, "starmap", make("starmap", "starmap_async"))
+
+    class _ForEachDescriptor:
+        def __get__(self, obj: Any, objtype: Any = None) -> Any:
+            if obj is None:
+                return self
+            impl = obj._impl
+
+            def blocking(*args: Any, **kwargs: Any) -> None:
+                return synchronizer.run(impl.for_each_async(*unwrap(args), **unwrap(kwargs)))
+
+            async def aio(*args: Any, **kwargs: Any) -> None:
+                return await synchronizer.run_async(
+                    impl.for_each_async(*unwrap(args), **unwrap(kwargs))
+                )
+
+            from ._sync import _AioCallable
+
+            return _AioCallable(blocking, aio, impl.for_each_async)
+
+    setattr(Function, "for_each", _ForEachDescriptor())
+
+
+_install_sync_map_methods()

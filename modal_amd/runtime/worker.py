@@ -1,0 +1,604 @@
+"""Worker process: the per-GPU execution runtime.
+
+Re-implementation of the reference's container entrypoint + IO manager
+(/root/reference/py/modal/_container_entrypoint.py:477,
+/root/reference/py/modal/_runtime/container_io_manager.py:531) as a
+single-node worker: one process per MI355X GPU (``HIP_VISIBLE_DEVICES``
+pinned by the pool), connected to the scheduler over a Unix socket.
+
+Differences from the reference, by design:
+* inputs are *pushed* (credit-bounded) instead of long-polled;
+* sync user functions run on a daemonized thread pool, async ones on the
+  worker's event loop (parity: _container_entrypoint.py:193-266);
+* dynamic batching transposes args and splits outputs per item
+  (parity: container_io_manager.py:196-262);
+* generator outputs stream over a data channel with a GeneratorDone tail
+  (parity: container_io_manager.py:759-846).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import contextvars
+import inspect
+import os
+import sys
+import time
+import traceback
+from concurrent.futures import ThreadPoolExecutor
+from typing import Any, Optional
+
+from .._serialization import (
+    DataFormat,
+    GeneratorDone,
+    deserialize,
+    deserialize_payload,
+    serialize,
+    serialize_data_format,
+    set_client_context,
+)
+from ..scheduler.blobs import INLINE_LIMIT, BlobStore
+from ..scheduler.calls import (
+    GENERIC_STATUS_FAILURE,
+    GENERIC_STATUS_INTERNAL_FAILURE,
+    GENERIC_STATUS_SUCCESS,
+    GENERIC_STATUS_TERMINATED,
+    GENERIC_STATUS_TIMEOUT,
+)
+from ..scheduler.transport import Connection
+from .execution_context import _reset_current_context, _set_current_context
+
+HEARTBEAT_INTERVAL = 15.0  # parity: reference config.py:318
+OUTPUT_FLUSH_INTERVAL = 0.001  # batch outputs ~1 ms (beats the 20/req constant)
+OUTPUT_BATCH_MAX = 512
+
+_app_id_var: contextvars.ContextVar[str] = contextvars.ContextVar("modal_amd_app_id", default="")
+
+
+class FunctionRuntime:
+    """Worker-side state for one registered function."""
+
+    def __init__(self, runtime: "WorkerRuntime", msg: dict):
+        self.runtime = runtime
+        self.function_id = msg["function_id"]
+        self.app_id = msg.get("app_id", "")
+        self.name = msg.get("name", "")
+        self.is_generator = bool(msg.get("is_generator"))
+        self.timeout = msg.get("timeout")
+        self.max_concurrent = max(int(msg.get("max_concurrent_inputs") or 1), 1)
+        self.batch_max_size = int(msg.get("batch_max_size") or 0)
+        self.batch_linger_ms = int(msg.get("batch_linger_ms") or 0)
+        self.version = msg.get("version", 1)
+        self.definition_blob = msg["definition"]
+        self.definition_kind = msg.get("definition_kind", "serialized")
+        self.env = msg.get("env") or {}
+        self.volumes = msg.get("volumes") or {}
+        self.sem = asyncio.Semaphore(self.max_concurrent)
+        self._callable: Any = None
+        self._service: Any = None
+        self._loaded = False
+        self._load_error: Optional[BaseException] = None
+        self._enter_done = False
+        # dynamic batching state
+        self._batch: list[tuple[dict, tuple, dict]] = []
+        self._batch_flush_handle: Optional[asyncio.TimerHandle] = None
+
+    def load(self) -> Any:
+        """Deserialize/import the user function (once per definition version)."""
+        if self._loaded:
+            if self._load_error is not None:
+                raise self._load_error
+            return self._callable
+        try:
+            if self.env:
+                os.environ.update(self.env)  # secret env bundles
+            if self.volumes:
+                from .volumes import mount_volumes
+
+                mount_volumes(self.volumes, self.runtime)
+            obj = deserialize(self.definition_blob)
+            if isinstance(obj, dict) and obj.get("kind") == "cls_service":
+                # class-backed service: instantiate, run @enter hooks lazily
+                self._service = _ClsService(obj)
+                self._callable = None
+            else:
+                self._callable = obj
+            self._loaded = True
+            return self._callable
+        except BaseException as exc:
+            self._load_error = exc
+            self._loaded = True
+            raise
+
+    def get_callable(self, method_name: str) -> Any:
+        fn = self.load()
+        if self._service is not None:
+            return self._service.get_method(method_name)
+        return fn
+
+
+class _ClsService:
+    """Instantiated class service with lifecycle hooks (parity:
+    user_code_imports.py:388 ImportedClass; lifecycle hooks
+    _partial_function.py:589,617)."""
+
+    def __init__(self, spec: dict):
+        self.cls = spec["cls"]
+        self.params_args = spec.get("args", ())
+        self.params_kwargs = spec.get("kwargs", {})
+        self.instance: Any = None
+        self._entered = False
+
+    def _ensure_instance(self) -> Any:
+        if self.instance is None:
+            from ..cls import _Parameter
+
+            if "__init__" in vars(self.cls):
+                self.instance = self.cls(*self.params_args, **self.params_kwargs)
+            else:
+                inst = self.cls.__new__(self.cls)
+                for name, value in vars(self.cls).items():
+                    if isinstance(value, _Parameter) and value.default is not ...:
+                        setattr(inst, name, value.default)
+                for key, value in self.params_kwargs.items():
+                    setattr(inst, key, value)
+                self.instance = inst
+        if not self._entered:
+            self._entered = True
+            for hook_name in _lifecycle_hooks(self.cls, "enter"):
+                getattr(self.instance, hook_name)()
+        return self.instance
+
+    def get_method(self, method_name: str) -> Any:
+        inst = self._ensure_instance()
+        raw = getattr(type(inst), method_name)
+        raw_fn = getattr(raw, "raw_f", None) or raw
+        def bound(*args: Any, **kwargs: Any) -> Any:
+            return raw_fn(inst, *args, **kwargs)
+        bound.__name__ = method_name
+        if inspect.iscoroutinefunction(raw_fn):
+            async def abound(*args: Any, **kwargs: Any) -> Any:
+                return await raw_fn(inst, *args, **kwargs)
+            abound.__name__ = method_name
+            return abound
+        return bound
+
+    def exit(self) -> None:
+        if self.instance is not None and self._entered:
+            for hook_name in _lifecycle_hooks(self.cls, "exit"):
+                try:
+                    getattr(self.instance, hook_name)()
+                except Exception:
+                    traceback.print_exc()
+
+
+def _lifecycle_hooks(cls: type, kind: str) -> list[str]:
+    out = []
+    for name in dir(cls):
+        attr = inspect.getattr_static(cls, name, None)
+        if getattr(attr, "_modal_amd_lifecycle", None) == kind:
+            out.append(name)
+    return out
+
+
+class _LogForwarder:
+    """Tee stdout/stderr lines to the scheduler (parity: container log streaming)."""
+
+    def __init__(self, runtime: "WorkerRuntime", orig: Any, fd: int):
+        self.runtime = runtime
+        self.orig = orig
+        self.fd = fd
+
+    def write(self, data: str) -> int:
+        self.orig.write(data)
+        if data:
+            self.runtime.post_log(self.fd, data)
+        return len(data)
+
+    def flush(self) -> None:
+        self.orig.flush()
+
+    def isatty(self) -> bool:
+        return False
+
+    def fileno(self) -> int:
+        return self.orig.fileno()
+
+
+class WorkerRuntime:
+    def __init__(self) -> None:
+        self.worker_id = int(os.environ.get("MODAL_AMD_WORKER_ID", "-1"))
+        self.gpu_index = (
+            int(os.environ["MODAL_AMD_GPU_INDEX"]) if "MODAL_AMD_GPU_INDEX" in os.environ else None
+        )
+        self.socket_path = os.environ["MODAL_AMD_WORKER_SOCKET"]
+        self.external = os.environ.get("MODAL_AMD_EXTERNAL_WORKER") == "1"
+        self.conn: Optional[Connection] = None
+        self.task_id: str = ""
+        self.functions: dict[str, FunctionRuntime] = {}
+        self.executor = ThreadPoolExecutor(
+            max_workers=int(os.environ.get("MODAL_AMD_WORKER_THREADS", "16")),
+            thread_name_prefix="modal-amd-input",
+        )
+        run_dir = os.path.dirname(self.socket_path)
+        self.blob_store = BlobStore(os.path.join(run_dir, "blobs"))
+        self._outbox: list[dict] = []
+        self._outbox_flush_scheduled = False
+        self._running: dict[str, asyncio.Task] = {}
+        self._abandoned: set[str] = set()  # tokens whose (sync) execution timed out
+        self._shutdown = asyncio.Event()
+        self._hello_ack = asyncio.Event()
+        self._log_buffer: list[tuple[int, str]] = []
+        self._log_flush_scheduled = False
+        self.loop: Optional[asyncio.AbstractEventLoop] = None
+
+    # ---- wiring ---------------------------------------------------------
+    async def run(self) -> None:
+        self.loop = asyncio.get_running_loop()
+        try:
+            reader, writer = await asyncio.open_unix_connection(self.socket_path)
+        except (FileNotFoundError, ConnectionRefusedError):
+            return  # scheduler already gone: exit quietly
+        self.conn = Connection(reader, writer, self._handle, rpc_target=None)
+        self.conn.start()
+        await self.conn.send(
+            {
+                "t": "hello",
+                "role": "worker",
+                "worker_id": self.worker_id,
+                "gpu_index": self.gpu_index,
+                "external": self.external,
+                "pid": os.getpid(),
+            }
+        )
+        # install the process-default client so user code's handles bind here
+        from ..client import _Client
+        from .. import client as client_mod  # noqa: F401
+
+        proxy_client = _Client(client_mod.SchedulerProxy(self.conn), "container")
+        _Client.set_default(proxy_client)
+        set_client_context(proxy_client)
+
+        sys.stdout = _LogForwarder(self, sys.__stdout__, 1)  # type: ignore[assignment]
+        sys.stderr = _LogForwarder(self, sys.__stderr__, 2)  # type: ignore[assignment]
+
+        hb_task = asyncio.get_running_loop().create_task(self._heartbeat_loop())
+        closed_task = asyncio.get_running_loop().create_task(self.conn.wait_closed())
+        stop_task = asyncio.get_running_loop().create_task(self._shutdown.wait())
+        await asyncio.wait({closed_task, stop_task}, return_when=asyncio.FIRST_COMPLETED)
+        hb_task.cancel()
+        for frt in self.functions.values():
+            if frt._service is not None:
+                frt._service.exit()
+        await self._flush_outbox()
+        await self.conn.close()
+
+    async def _heartbeat_loop(self) -> None:
+        while True:
+            await asyncio.sleep(HEARTBEAT_INTERVAL)
+            try:
+                await self.conn.send({"t": "hb"})
+            except Exception:
+                return
+
+    async def _handle(self, msg: dict) -> None:
+        kind = msg.get("t")
+        if kind == "hello_ack":
+            self.task_id = msg.get("task_id", "")
+            os.environ["MODAL_AMD_TASK_ID"] = self.task_id
+            self._hello_ack.set()
+        elif kind == "def":
+            frt = FunctionRuntime(self, msg)
+            self.functions[frt.function_id] = frt
+        elif kind == "inputs":
+            frt = self.functions.get(msg["function_id"])
+            if frt is None:
+                for item in msg["items"]:
+                    self.post_output(
+                        item["token"],
+                        msg["function_id"],
+                        GENERIC_STATUS_INTERNAL_FAILURE,
+                        None,
+                        0,
+                        "worker missing function definition",
+                    )
+                return
+            for item in msg["items"]:
+                if frt.batch_max_size > 1:
+                    self._batch_add(frt, item)
+                else:
+                    task = asyncio.get_running_loop().create_task(self._run_input(frt, item))
+                    self._running[item["token"]] = task
+                    task.add_done_callback(lambda _t, tok=item["token"]: self._running.pop(tok, None))
+        elif kind == "cancel":
+            for token in msg.get("tokens", []):
+                task = self._running.get(token)
+                if task is not None:
+                    task.cancel()
+                else:
+                    self._abandoned.add(token)
+            if msg.get("terminate"):
+                self._shutdown.set()
+        elif kind == "shutdown":
+            self._shutdown.set()
+
+    # ---- execution ------------------------------------------------------
+    def _decode_args(self, item: dict) -> tuple[tuple, dict]:
+        if item.get("payload_blob"):
+            payload = self.blob_store.get(item["payload_blob"])
+        else:
+            payload = item.get("payload") or b""
+        return deserialize_payload(payload)
+
+    async def _run_input(self, frt: FunctionRuntime, item: dict) -> None:
+        token = item["token"]
+        call_id = token.rsplit(":", 2)[0]
+        async with frt.sem:
+            if token in self._abandoned:
+                self._abandoned.discard(token)
+                return
+            ctx_tokens = _set_current_context(item.get("input_id"), call_id)
+            app_tok = _app_id_var.set(frt.app_id)
+            started = time.monotonic()
+            try:
+                fn = frt.get_callable(item.get("method", ""))
+                args, kwargs = self._decode_args(item)
+                is_gen = (
+                    frt.is_generator
+                    or inspect.isgeneratorfunction(fn)
+                    or inspect.isasyncgenfunction(fn)
+                )
+                if is_gen:
+                    await self._run_generator(frt, fn, token, args, kwargs)
+                    return
+                result = await self._execute(frt, fn, args, kwargs)
+                data = serialize(result)
+                self.post_output(token, frt.function_id, GENERIC_STATUS_SUCCESS, data, DataFormat.PICKLE)
+            except asyncio.TimeoutError:
+                self.post_output(
+                    token,
+                    frt.function_id,
+                    GENERIC_STATUS_TIMEOUT,
+                    None,
+                    0,
+                    f"Function exceeded timeout of {frt.timeout}s "
+                    f"(ran {time.monotonic() - started:.1f}s)",
+                )
+            except asyncio.CancelledError:
+                self.post_output(
+                    token, frt.function_id, GENERIC_STATUS_TERMINATED, None, 0, "input cancelled"
+                )
+            except BaseException as exc:
+                self.post_output(
+                    token,
+                    frt.function_id,
+                    GENERIC_STATUS_FAILURE,
+                    self._serialize_exception(exc),
+                    DataFormat.PICKLE,
+                    "".join(traceback.format_exception_only(type(exc), exc)).strip(),
+                )
+            finally:
+                _app_id_var.reset(app_tok)
+                _reset_current_context(ctx_tokens)
+
+    async def _execute(self, frt: FunctionRuntime, fn: Any, args: tuple, kwargs: dict) -> Any:
+        if inspect.iscoroutinefunction(fn):
+            coro = fn(*args, **kwargs)
+            if frt.timeout:
+                return await asyncio.wait_for(coro, frt.timeout)
+            return await coro
+        ctx = contextvars.copy_context()
+        fut = asyncio.get_running_loop().run_in_executor(
+            self.executor, lambda: ctx.run(fn, *args, **kwargs)
+        )
+        if frt.timeout:
+            return await asyncio.wait_for(fut, frt.timeout)
+        return await fut
+
+    async def _run_generator(self, frt: FunctionRuntime, fn: Any, token: str, args: tuple, kwargs: dict) -> None:
+        """Stream generator items over the data channel, then a GeneratorDone
+        output (parity: generator_output_sender, container_io_manager.py:802)."""
+        index = 0
+        if inspect.isasyncgenfunction(fn):
+            agen = fn(*args, **kwargs)
+            async for value in agen:
+                await self._send_gen_item(token, index, value)
+                index += 1
+        else:
+            loop = asyncio.get_running_loop()
+            queue: asyncio.Queue = asyncio.Queue(maxsize=256)
+            SENTINEL = object()
+
+            ctx = contextvars.copy_context()
+
+            def pump() -> Any:
+                try:
+                    for value in ctx.run(fn, *args, **kwargs):
+                        fut = asyncio.run_coroutine_threadsafe(queue.put(value), loop)
+                        fut.result()
+                    return None
+                finally:
+                    asyncio.run_coroutine_threadsafe(queue.put(SENTINEL), loop).result()
+
+            pump_fut = loop.run_in_executor(self.executor, pump)
+            while True:
+                value = await queue.get()
+                if value is SENTINEL:
+                    break
+                await self._send_gen_item(token, index, value)
+                index += 1
+            await pump_fut  # surface exceptions
+        await self.conn.send({"t": "gen_data", "token": token, "index": index, "done": True})
+        done = GeneratorDone(items_total=index)
+        self.post_output(
+            token,
+            frt.function_id,
+            GENERIC_STATUS_SUCCESS,
+            serialize_data_format(done, DataFormat.GENERATOR_DONE),
+            DataFormat.GENERATOR_DONE,
+        )
+
+    async def _send_gen_item(self, token: str, index: int, value: Any) -> None:
+        data = serialize(value)
+        await self.conn.send(
+            {"t": "gen_data", "token": token, "index": index, "data": data, "format": int(DataFormat.PICKLE)}
+        )
+
+    # ---- dynamic batching ------------------------------------------------
+    def _batch_add(self, frt: FunctionRuntime, item: dict) -> None:
+        """Accumulate inputs; flush at batch_max_size or after batch_linger_ms
+        (parity: @modal.batched, container_io_manager.py:135-283)."""
+        frt._batch.append((item, (), {}))
+        if len(frt._batch) >= frt.batch_max_size:
+            self._batch_flush(frt)
+        elif frt._batch_flush_handle is None:
+            frt._batch_flush_handle = asyncio.get_running_loop().call_later(
+                frt.batch_linger_ms / 1000.0, self._batch_flush, frt
+            )
+
+    def _batch_flush(self, frt: FunctionRuntime) -> None:
+        if frt._batch_flush_handle is not None:
+            frt._batch_flush_handle.cancel()
+            frt._batch_flush_handle = None
+        batch = frt._batch
+        frt._batch = []
+        if batch:
+            task = asyncio.get_running_loop().create_task(self._run_batch(frt, [b[0] for b in batch]))
+            for item, _, _ in batch:
+                self._running[item["token"]] = task
+
+    async def _run_batch(self, frt: FunctionRuntime, items: list[dict]) -> None:
+        async with frt.sem:
+            try:
+                fn = frt.get_callable(items[0].get("method", ""))
+                decoded = [self._decode_args(item) for item in items]
+                # transpose: positional args and kwargs become per-arg lists
+                n_args = max((len(a) for a, _ in decoded), default=0)
+                arg_lists = [[d[0][i] if i < len(d[0]) else None for d in decoded] for i in range(n_args)]
+                kwarg_keys: set = set()
+                for _, kw in decoded:
+                    kwarg_keys.update(kw)
+                kw_lists = {k: [d[1].get(k) for d in decoded] for k in kwarg_keys}
+                results = await self._execute(frt, fn, tuple(arg_lists), kw_lists)
+                if not isinstance(results, (list, tuple)) or len(results) != len(items):
+                    raise ValueError(
+                        f"Batched function {frt.name} must return a list of {len(items)} results, "
+                        f"got {type(results).__name__}"
+                    )
+                for item, result in zip(items, results):
+                    self.post_output(
+                        item["token"], frt.function_id, GENERIC_STATUS_SUCCESS,
+                        serialize(result), DataFormat.PICKLE,
+                    )
+            except asyncio.TimeoutError:
+                for item in items:
+                    self.post_output(
+                        item["token"], frt.function_id, GENERIC_STATUS_TIMEOUT, None, 0,
+                        f"Batched call exceeded timeout of {frt.timeout}s",
+                    )
+            except BaseException as exc:
+                err = "".join(traceback.format_exception_only(type(exc), exc)).strip()
+                data = self._serialize_exception(exc)
+                for item in items:
+                    self.post_output(
+                        item["token"], frt.function_id, GENERIC_STATUS_FAILURE, data,
+                        DataFormat.PICKLE, err,
+                    )
+
+    # ---- outputs ---------------------------------------------------------
+    def _serialize_exception(self, exc: BaseException) -> bytes:
+        from ..utils.tb import clean_traceback
+
+        clean_traceback(exc)
+        try:
+            return serialize(exc)
+        except BaseException:
+            try:
+                stripped = type(exc)(*[repr(a) for a in exc.args])
+                return serialize(stripped)
+            except BaseException:
+                return serialize(RuntimeError(repr(exc)))
+
+    def post_output(
+        self,
+        token: str,
+        function_id: str,
+        status: int,
+        data: Optional[bytes],
+        data_format: int,
+        exc_repr: Optional[str] = None,
+    ) -> None:
+        item: dict[str, Any] = {
+            "token": token,
+            "function_id": function_id,
+            "status": int(status),
+            "format": int(data_format),
+        }
+        if data is not None:
+            if len(data) > INLINE_LIMIT:
+                item["data"] = None
+                item["data_blob"] = self.blob_store.put(data)
+            else:
+                item["data"] = data
+        if exc_repr:
+            item["exc"] = exc_repr
+        self._outbox.append(item)
+        if not self._outbox_flush_scheduled:
+            self._outbox_flush_scheduled = True
+            asyncio.get_running_loop().call_later(OUTPUT_FLUSH_INTERVAL, self._flush_outbox_cb)
+
+    def _flush_outbox_cb(self) -> None:
+        self._outbox_flush_scheduled = False
+        asyncio.get_running_loop().create_task(self._flush_outbox())
+
+    async def _flush_outbox(self) -> None:
+        while self._outbox:
+            chunk, self._outbox = self._outbox[:OUTPUT_BATCH_MAX], self._outbox[OUTPUT_BATCH_MAX:]
+            try:
+                await self.conn.send({"t": "outputs", "items": chunk})
+            except Exception:
+                return
+
+    def post_log(self, fd: int, data: str) -> None:
+        if self.loop is None or self.conn is None or self.conn.closed:
+            return
+        self._log_buffer.append((fd, data))
+        if not self._log_flush_scheduled:
+            self._log_flush_scheduled = True
+            try:
+                self.loop.call_soon_threadsafe(self._schedule_log_flush)
+            except RuntimeError:
+                pass
+
+    def _schedule_log_flush(self) -> None:
+        self.loop.call_later(0.02, self._flush_logs)
+
+    def _flush_logs(self) -> None:
+        self._log_flush_scheduled = False
+        buf, self._log_buffer = self._log_buffer, []
+        if not buf:
+            return
+        app_id = _app_id_var.get()
+        by_fd: dict[int, list[str]] = {}
+        for fd, data in buf:
+            by_fd.setdefault(fd, []).append(data)
+        for fd, chunks in by_fd.items():
+            asyncio.get_running_loop().create_task(
+                self._send_log(fd, "".join(chunks), app_id)
+            )
+
+    async def _send_log(self, fd: int, data: str, app_id: str) -> None:
+        try:
+            await self.conn.send({"t": "log", "fd": fd, "data": data, "app_id": app_id})
+        except Exception:
+            pass
+
+
+def main() -> None:
+    os.environ["MODAL_AMD_IS_REMOTE"] = "1"
+    runtime = WorkerRuntime()
+    asyncio.run(runtime.run())
+
+
+if __name__ == "__main__":
+    main()

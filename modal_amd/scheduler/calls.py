@@ -1,0 +1,266 @@
+"""Function-call table: the state the control plane keeps per invocation.
+
+This is the in-process re-implementation of the server-side state behind the
+reference's Function* RPCs (FunctionMap with pipelined inputs, PutInputs,
+GetOutputs with clear_on_success + lost-input signalling, RetryInputs,
+FinishInputs — /root/reference/modal_proto/api.proto, mock behavior
+/root/reference/py/test/conftest.py:2303-2485), with the statuses of
+GenericResult and the retry semantics of the invocation engine
+(/root/reference/py/modal/_functions.py:106,286-316).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import time
+from dataclasses import dataclass, field
+from typing import Any, Optional
+
+from ..utils.ids import new_id
+
+# GenericResult.status values (parity: api.proto GenericResult)
+GENERIC_STATUS_UNSPECIFIED = 0
+GENERIC_STATUS_SUCCESS = 1
+GENERIC_STATUS_FAILURE = 2
+GENERIC_STATUS_TERMINATED = 3
+GENERIC_STATUS_TIMEOUT = 4
+GENERIC_STATUS_INTERNAL_FAILURE = 5
+GENERIC_STATUS_INIT_FAILURE = 6
+
+# parity: /root/reference/py/modal/_functions.py:106
+MAX_INTERNAL_FAILURE_COUNT = 8
+
+# parity: /root/reference/py/modal/parallel_map.py:79
+MAX_INPUTS_OUTSTANDING_DEFAULT = 1000
+
+
+@dataclass
+class RetryPolicy:
+    """User-facing retry policy (parity: modal.Retries, reference retries.py:12)."""
+
+    max_retries: int = 0
+    backoff_coefficient: float = 2.0
+    initial_delay_ms: int = 1000
+    max_delay_ms: int = 60_000
+
+    def delay_ms(self, retry_count: int) -> float:
+        if retry_count <= 0:
+            return self.initial_delay_ms
+        delay = self.initial_delay_ms * (self.backoff_coefficient ** (retry_count - 1))
+        return min(delay, self.max_delay_ms)
+
+    def to_dict(self) -> dict:
+        return {
+            "max_retries": self.max_retries,
+            "backoff_coefficient": self.backoff_coefficient,
+            "initial_delay_ms": self.initial_delay_ms,
+            "max_delay_ms": self.max_delay_ms,
+        }
+
+    @classmethod
+    def from_dict(cls, d: Optional[dict]) -> "RetryPolicy":
+        return cls(**d) if d else cls()
+
+
+@dataclass
+class FunctionDef:
+    """A registered function: the scheduler's row for ``fu-`` objects.
+
+    The serialized definition travels to workers once and is cached there
+    (parity: the container entrypoint imports user code once,
+    /root/reference/py/modal/_runtime/user_code_imports.py:118).
+    """
+
+    function_id: str
+    app_id: str
+    name: str
+    definition: bytes  # cloudpickled callable or import-ref descriptor
+    definition_kind: str = "serialized"  # "serialized" | "ref"
+    is_generator: bool = False
+    needs_gpu: bool = False
+    gpu_count: int = 0
+    timeout: Optional[float] = None
+    retry_policy: RetryPolicy = field(default_factory=RetryPolicy)
+    max_concurrent_inputs: int = 1  # @modal.concurrent(max_inputs=...)
+    target_concurrent_inputs: int = 0
+    batch_max_size: int = 0  # @modal.batched
+    batch_linger_ms: int = 0
+    is_method: bool = False
+    is_class_service: bool = False
+    cluster_size: int = 0  # @modal.clustered(size=...)
+    min_containers: int = 0
+    max_containers: int = 0
+    buffer_containers: int = 0
+    scaledown_window: float = 60.0
+    definition_version: int = 1
+    metadata: dict = field(default_factory=dict)
+    web_config: Optional[dict] = None
+    secret_ids: list = field(default_factory=list)
+    volume_mounts: dict = field(default_factory=dict)  # mount path -> volume id
+    schedule: Optional[dict] = None  # {"cron": "..."} | {"period": seconds}
+
+    def placement_tag(self) -> str:
+        return "gpu" if self.needs_gpu else "any"
+
+    def public_metadata(self) -> dict:
+        return {
+            "function_name": self.name,
+            "is_generator": self.is_generator,
+            "needs_gpu": self.needs_gpu,
+            "is_method": self.is_method,
+            "cluster_size": self.cluster_size,
+            "batch_max_size": self.batch_max_size,
+            "definition_id": self.function_id,
+            **self.metadata,
+        }
+
+
+@dataclass
+class InputRecord:
+    call_id: str
+    idx: int
+    input_id: str
+    payload: bytes
+    payload_blob: Optional[str] = None  # CAS digest when payload exceeded the inline limit
+    method_name: str = ""
+    retry_count: int = 0  # user-policy retries consumed
+    internal_failures: int = 0
+    status: int = GENERIC_STATUS_UNSPECIFIED
+    final: bool = False
+    output: Optional[bytes] = None
+    output_blob: Optional[str] = None  # CAS digest for oversized outputs
+    output_format: int = 0
+    exc_repr: Optional[str] = None
+    worker_id: Optional[int] = None
+    enqueued_at: float = 0.0
+    started_at: float = 0.0
+    finished_at: float = 0.0
+    tensors: Optional[list] = None  # tensor sidecar (CUDA-IPC / pinned staging)
+    cancelled: bool = False
+
+    @property
+    def token(self) -> str:
+        """Stable retry-versioned token — plays the role of the reference's
+        per-input JWT used for lost-input detection (parallel_map.py:447-523)."""
+        return f"{self.call_id}:{self.idx}:{self.retry_count}"
+
+
+class CallRecord:
+    """One function call (``fc-``): unary, spawn, or map fan-out."""
+
+    def __init__(
+        self,
+        function_id: str,
+        kind: str,
+        return_exceptions: bool = False,
+    ):
+        self.call_id = new_id("function_call")
+        self.function_id = function_id
+        self.kind = kind  # "unary" | "spawn" | "map" | "spawn_map"
+        self.return_exceptions = return_exceptions
+        self.inputs: dict[int, InputRecord] = {}
+        self.next_idx = 0
+        self.num_inputs_final: Optional[int] = None
+        self.completed: int = 0
+        self.cancelled = False
+        self.created_at = time.time()
+        # completion-order queue of idx for streaming GetOutputs
+        self.output_ready: asyncio.Queue[int] = asyncio.Queue()
+        # per-input completion events for unary waits
+        self._waiters: dict[int, asyncio.Future] = {}
+        # generator data-out channel, per input idx
+        self.gen_queues: dict[int, asyncio.Queue] = {}
+        self.done_event = asyncio.Event()
+
+    # -- input intake ----------------------------------------------------
+    def add_input(
+        self,
+        payload: bytes,
+        method_name: str = "",
+        tensors: Optional[list] = None,
+        payload_blob: Optional[str] = None,
+    ) -> InputRecord:
+        idx = self.next_idx
+        self.next_idx += 1
+        rec = InputRecord(
+            call_id=self.call_id,
+            idx=idx,
+            input_id=new_id("input"),
+            payload=payload,
+            payload_blob=payload_blob,
+            method_name=method_name,
+            enqueued_at=time.time(),
+            tensors=tensors,
+        )
+        self.inputs[idx] = rec
+        return rec
+
+    def finish_inputs(self) -> None:
+        self.num_inputs_final = self.next_idx
+        self._check_done()
+
+    def _check_done(self) -> None:
+        if self.num_inputs_final is not None and self.completed >= self.num_inputs_final:
+            self.done_event.set()
+
+    # -- completion ------------------------------------------------------
+    def post_output(
+        self,
+        idx: int,
+        status: int,
+        output: Optional[bytes],
+        output_format: int,
+        exc_repr: Optional[str],
+        retry_count: int,
+        output_blob: Optional[str] = None,
+    ) -> bool:
+        """Record a final output for input idx. Returns False on stale/dup
+        delivery (parity: dedup by (idx, retry_count),
+        reference parallel_map.py:1416-1431)."""
+        rec = self.inputs.get(idx)
+        if rec is None or rec.final:
+            return False
+        if retry_count != rec.retry_count:
+            return False  # output from a superseded attempt
+        rec.status = status
+        rec.output = output
+        rec.output_blob = output_blob
+        rec.output_format = output_format
+        rec.exc_repr = exc_repr
+        rec.final = True
+        rec.finished_at = time.time()
+        self.completed += 1
+        self.output_ready.put_nowait(idx)
+        waiter = self._waiters.pop(idx, None)
+        if waiter is not None and not waiter.done():
+            waiter.set_result(rec)
+        self._check_done()
+        return True
+
+    async def wait_output(self, idx: int, timeout: Optional[float] = None) -> InputRecord:
+        rec = self.inputs[idx]
+        if rec.final:
+            return rec
+        fut = self._waiters.get(idx)
+        if fut is None:
+            fut = asyncio.get_running_loop().create_future()
+            self._waiters[idx] = fut
+        if timeout is None:
+            return await asyncio.shield(fut)
+        return await asyncio.wait_for(asyncio.shield(fut), timeout)
+
+    # -- generator data plane -------------------------------------------
+    def gen_queue(self, idx: int) -> asyncio.Queue:
+        q = self.gen_queues.get(idx)
+        if q is None:
+            q = asyncio.Queue()
+            self.gen_queues[idx] = q
+        return q
+
+    def stats(self) -> dict:
+        return {
+            "total": self.next_idx,
+            "completed": self.completed,
+            "pending": self.next_idx - self.completed,
+            "final": self.num_inputs_final is not None,
+        }

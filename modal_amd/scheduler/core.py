@@ -1,0 +1,582 @@
+"""The in-process scheduler: single-node re-implementation of the control plane.
+
+The reference client speaks 240 gRPC RPCs to Modal's closed cloud
+(/root/reference/modal_proto/api.proto, service at :4680). Here the control
+plane is this class: called directly (no serialization) from the client
+process, and over Unix-socket RPC (scheduler/transport.py) from workers,
+sandboxes and CLI tooling. Method names and semantics track the RPC groups
+(App*, Function*, Queue*, Dict*, Secret*, Blob*) so behavior matches the
+reference's mock servicer (/root/reference/py/test/conftest.py:701) where the
+reference documents it.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import os
+import tempfile
+import time
+from collections import deque
+from typing import Any, Optional
+
+from ..exception import InvalidError, NotFoundError
+from ..utils.ids import new_id
+from .blobs import INLINE_LIMIT, BlobStore
+from .calls import (
+    GENERIC_STATUS_FAILURE,
+    GENERIC_STATUS_INTERNAL_FAILURE,
+    GENERIC_STATUS_SUCCESS,
+    GENERIC_STATUS_TERMINATED,
+    GENERIC_STATUS_TIMEOUT,
+    MAX_INPUTS_OUTSTANDING_DEFAULT,
+    MAX_INTERNAL_FAILURE_COUNT,
+    CallRecord,
+    FunctionDef,
+    InputRecord,
+    RetryPolicy,
+)
+from .services import Services
+from .workerhost import WorkerPool
+
+OUTPUT_POLL_TIMEOUT = 55.0  # parity: function_utils.py:498 (55 s long-poll)
+
+
+class AppState:
+    def __init__(self, app_id: str, description: str, ephemeral: bool, environment: str):
+        self.app_id = app_id
+        self.description = description
+        self.ephemeral = ephemeral
+        self.environment = environment
+        self.state = "running"  # running | stopped | deployed
+        self.created_at = time.time()
+        self.last_heartbeat = time.time()
+        self.objects: dict[str, tuple[str, dict]] = {}  # tag -> (object_id, metadata)
+        self.deployment_name: Optional[str] = None
+        self.logs: deque = deque(maxlen=10_000)
+        self.log_subscribers: list[asyncio.Queue] = []
+
+
+class RPCAdapter:
+    """Whitelisted surface exposed to socket peers (workers, sandboxes, CLI)."""
+
+    _ALLOWED = {
+        "queue_get_or_create", "queue_put", "queue_get", "queue_len", "queue_clear",
+        "queue_peek", "queue_delete",
+        "dict_get_or_create", "dict_update", "dict_get", "dict_pop", "dict_contains",
+        "dict_len", "dict_items", "dict_clear", "dict_delete",
+        "secret_get_or_create", "secret_env",
+        "blob_put", "blob_get", "blob_path",
+        "function_lookup", "function_map", "function_put_inputs", "function_finish_inputs",
+        "function_get_outputs", "function_call_cancel", "function_call_info",
+        "function_get_current_stats", "generator_poll",
+        "app_lookup", "app_get_layout", "cluster_hello",
+        "volume_get_or_create", "volume_commit_files", "volume_get_file", "volume_list_files",
+        "volume_remove_file", "volume_copy_files", "volume_reload", "volume_delete", "volume_rename",
+        "sandbox_create", "sandbox_wait", "sandbox_terminate", "sandbox_poll", "sandbox_stdio_read",
+        "sandbox_stdin_write", "sandbox_exec", "sandbox_exec_wait", "sandbox_exec_poll",
+        "sandbox_list", "sandbox_tag", "sandbox_from_name", "sandbox_snapshot_fs",
+        "image_get_or_create", "image_info",
+    }
+
+    def __init__(self, scheduler: "Scheduler"):
+        self._scheduler = scheduler
+
+    def __getattr__(self, name: str) -> Any:
+        if name in RPCAdapter._ALLOWED:
+            return getattr(self._scheduler, name)
+        raise AttributeError(name)
+
+
+class Scheduler:
+    def __init__(self, run_dir: Optional[str] = None):
+        from ..config import config
+
+        self.run_dir = run_dir or config.get("run_dir") or tempfile.mkdtemp(prefix="modal-amd-")
+        os.makedirs(self.run_dir, exist_ok=True)
+        self.apps: dict[str, AppState] = {}
+        self.app_names: dict[tuple[str, str], str] = {}  # (env, name) -> app_id
+        self.functions: dict[str, FunctionDef] = {}
+        self.function_names: dict[tuple[str, str, str], str] = {}  # (env, app_name, fn) -> fu-id
+        self.calls: dict[str, CallRecord] = {}
+        self.services = Services()
+        self.blob_store = BlobStore(os.path.join(self.run_dir, "blobs"))
+        self.pool = WorkerPool(self)
+        self.rpc_adapter = RPCAdapter(self)
+        self._started = False
+        self._start_lock = asyncio.Lock()
+        self.default_environment = "main"
+        self._extra: dict[str, Any] = {}  # extension services (volumes, sandboxes, images)
+
+    # -- lifecycle -------------------------------------------------------
+    async def start(self) -> None:
+        async with self._start_lock:
+            if self._started:
+                return
+            await self.pool.start()
+            self._started = True
+
+    async def stop(self) -> None:
+        if not self._started:
+            return
+        await self.pool.stop()
+        self._started = False
+
+    def log(self, message: str) -> None:
+        import logging
+
+        logging.getLogger("modal_amd.scheduler").debug(message)
+
+    # -- apps ------------------------------------------------------------
+    async def app_create(
+        self, description: str = "", ephemeral: bool = True, environment: str = ""
+    ) -> dict:
+        app_id = new_id("app")
+        env = environment or self.default_environment
+        self.apps[app_id] = AppState(app_id, description, ephemeral, env)
+        return {"app_id": app_id}
+
+    async def app_publish(self, app_id: str, name: str) -> dict:
+        app = self._app(app_id)
+        app.deployment_name = name
+        app.state = "deployed"
+        app.ephemeral = False
+        self.app_names[(app.environment, name)] = app_id
+        for tag, (object_id, _meta) in app.objects.items():
+            if object_id.startswith("fu-"):
+                self.function_names[(app.environment, name, tag)] = object_id
+        return {"app_id": app_id, "url": f"local://{name}"}
+
+    async def app_lookup(self, name: str, environment: str = "") -> dict:
+        env = environment or self.default_environment
+        app_id = self.app_names.get((env, name))
+        if app_id is None:
+            raise NotFoundError(f"App '{name}' not found in environment '{env}'")
+        return {"app_id": app_id}
+
+    async def app_set_objects(self, app_id: str, objects: dict[str, tuple[str, dict]]) -> None:
+        self._app(app_id).objects.update(objects)
+
+    async def app_get_layout(self, app_id: str) -> dict:
+        return self.app_layout(app_id)
+
+    def app_layout(self, app_id: str) -> dict:
+        app = self.apps.get(app_id)
+        if app is None:
+            return {"app_id": app_id, "objects": {}}
+        return {
+            "app_id": app_id,
+            "objects": {tag: [oid, meta] for tag, (oid, meta) in app.objects.items()},
+        }
+
+    async def app_heartbeat(self, app_id: str) -> None:
+        self._app(app_id).last_heartbeat = time.time()
+
+    async def app_client_disconnect(self, app_id: str) -> None:
+        app = self.apps.get(app_id)
+        if app is None:
+            return
+        if app.ephemeral:
+            app.state = "stopped"
+            await self._cancel_app_calls(app_id)
+
+    async def app_stop(self, app_id: str) -> None:
+        app = self._app(app_id)
+        app.state = "stopped"
+        if app.deployment_name:
+            self.app_names.pop((app.environment, app.deployment_name), None)
+        await self._cancel_app_calls(app_id)
+
+    async def app_list(self, environment: str = "") -> list[dict]:
+        env = environment or self.default_environment
+        return [
+            {
+                "app_id": a.app_id,
+                "description": a.description,
+                "state": a.state,
+                "name": a.deployment_name,
+                "created_at": a.created_at,
+            }
+            for a in self.apps.values()
+            if a.environment == env
+        ]
+
+    async def _cancel_app_calls(self, app_id: str) -> None:
+        fids = {fid for fid, f in self.functions.items() if f.app_id == app_id}
+        for call in self.calls.values():
+            if call.function_id in fids and not call.done_event.is_set():
+                await self.function_call_cancel(call.call_id, terminate_containers=False)
+
+    def _app(self, app_id: str) -> AppState:
+        app = self.apps.get(app_id)
+        if app is None:
+            raise NotFoundError(f"App {app_id} not found")
+        return app
+
+    # -- function registry ------------------------------------------------
+    async def function_create(
+        self,
+        app_id: str,
+        name: str,
+        definition: bytes,
+        options: Optional[dict] = None,
+    ) -> dict:
+        options = options or {}
+        fid = new_id("function")
+        fdef = FunctionDef(
+            function_id=fid,
+            app_id=app_id,
+            name=name,
+            definition=definition,
+            definition_kind=options.get("definition_kind", "serialized"),
+            is_generator=bool(options.get("is_generator")),
+            needs_gpu=bool(options.get("needs_gpu")),
+            gpu_count=int(options.get("gpu_count", 1 if options.get("needs_gpu") else 0)),
+            timeout=options.get("timeout"),
+            retry_policy=RetryPolicy.from_dict(options.get("retries")),
+            max_concurrent_inputs=int(options.get("max_concurrent_inputs", 1)),
+            target_concurrent_inputs=int(options.get("target_concurrent_inputs", 0)),
+            batch_max_size=int(options.get("batch_max_size", 0)),
+            batch_linger_ms=int(options.get("batch_linger_ms", 0)),
+            is_method=bool(options.get("is_method")),
+            is_class_service=bool(options.get("is_class_service")),
+            cluster_size=int(options.get("cluster_size", 0)),
+            min_containers=int(options.get("min_containers", 0)),
+            max_containers=int(options.get("max_containers", 0)),
+            buffer_containers=int(options.get("buffer_containers", 0)),
+            scaledown_window=float(options.get("scaledown_window", 60.0)),
+            metadata=options.get("metadata") or {},
+            web_config=options.get("web_config"),
+            secret_ids=list(options.get("secret_ids") or []),
+            volume_mounts=dict(options.get("volume_mounts") or {}),
+            schedule=options.get("schedule"),
+        )
+        self.functions[fid] = fdef
+        app = self.apps.get(app_id)
+        if app is not None:
+            app.objects[name] = (fid, fdef.public_metadata())
+        return {"function_id": fid, "metadata": fdef.public_metadata()}
+
+    async def function_update(self, function_id: str, definition: bytes, options: Optional[dict] = None) -> None:
+        fdef = self.functions[function_id]
+        fdef.definition = definition
+        fdef.definition_version += 1
+        for key, value in (options or {}).items():
+            if hasattr(fdef, key):
+                setattr(fdef, key, value)
+
+    async def function_lookup(self, app_name: str, name: str, environment: str = "") -> dict:
+        env = environment or self.default_environment
+        fid = self.function_names.get((env, app_name, name))
+        if fid is None:
+            raise NotFoundError(f"Function '{app_name}/{name}' not found")
+        return {"function_id": fid, "metadata": self.functions[fid].public_metadata()}
+
+    async def function_get_current_stats(self, function_id: str) -> dict:
+        backlog = len(self.pool.pending.get(function_id, ()))
+        runners = sum(
+            1 for w in self.pool.workers.values() if w.outstanding.get(function_id, 0) > 0
+        )
+        return {"backlog": backlog, "num_total_tasks": runners}
+
+    async def function_update_autoscaler(
+        self,
+        function_id: str,
+        min_containers: Optional[int] = None,
+        max_containers: Optional[int] = None,
+        buffer_containers: Optional[int] = None,
+        scaledown_window: Optional[float] = None,
+    ) -> None:
+        fdef = self.functions[function_id]
+        if min_containers is not None:
+            fdef.min_containers = min_containers
+        if max_containers is not None:
+            fdef.max_containers = max_containers
+        if buffer_containers is not None:
+            fdef.buffer_containers = buffer_containers
+        if scaledown_window is not None:
+            fdef.scaledown_window = scaledown_window
+
+    # -- invocation -------------------------------------------------------
+    async def function_map(
+        self,
+        function_id: str,
+        kind: str = "unary",
+        pipelined_inputs: Optional[list] = None,
+        return_exceptions: bool = False,
+    ) -> dict:
+        """Create a function call; optionally enqueue the first inputs in the
+        same round-trip (parity: FunctionMap w/ pipelined_inputs,
+        reference _functions.py:163-188)."""
+        if function_id not in self.functions:
+            raise NotFoundError(f"Function {function_id} not found")
+        record = CallRecord(function_id, kind, return_exceptions)
+        self.calls[record.call_id] = record
+        if pipelined_inputs:
+            await self.function_put_inputs(record.call_id, pipelined_inputs)
+        return {
+            "function_call_id": record.call_id,
+            "retry_policy": self.functions[function_id].retry_policy.to_dict(),
+            "max_inputs_outstanding": MAX_INPUTS_OUTSTANDING_DEFAULT,
+            "sync_client_retries_enabled": True,
+        }
+
+    async def function_put_inputs(self, function_call_id: str, items: list) -> list:
+        """items: [{"payload": bytes, "method": str}] -> [{"idx", "input_id"}]."""
+        record = self._call(function_call_id)
+        out = []
+        for item in items:
+            if isinstance(item, (bytes, bytearray)):
+                item = {"payload": bytes(item)}
+            rec = record.add_input(
+                item.get("payload") or b"",
+                item.get("method", ""),
+                tensors=item.get("tensors"),
+                payload_blob=item.get("payload_blob"),
+            )
+            self.pool.enqueue(rec)
+            out.append({"idx": rec.idx, "input_id": rec.input_id})
+        return out
+
+    async def function_finish_inputs(self, function_call_id: str) -> None:
+        self._call(function_call_id).finish_inputs()
+
+    async def function_wait_output(
+        self, function_call_id: str, idx: int = 0, timeout: Optional[float] = None
+    ) -> InputRecord:
+        """In-process fast path: await one input's final output directly."""
+        return await self._call(function_call_id).wait_output(idx, timeout)
+
+    async def function_get_outputs(
+        self,
+        function_call_id: str,
+        max_values: int = 256,
+        timeout: float = OUTPUT_POLL_TIMEOUT,
+        clear_on_success: bool = True,
+    ) -> list[dict]:
+        """Streaming poll: completed outputs in completion order
+        (parity: FunctionGetOutputs long-poll, reference _functions.py:224-263)."""
+        record = self._call(function_call_id)
+        out: list[dict] = []
+        deadline = time.time() + timeout
+        while not out:
+            try:
+                remaining = deadline - time.time()
+                if remaining <= 0:
+                    break
+                idx = await asyncio.wait_for(record.output_ready.get(), remaining)
+            except asyncio.TimeoutError:
+                break
+            out.append(self._output_item(record, idx))
+            while len(out) < max_values:
+                try:
+                    idx = record.output_ready.get_nowait()
+                except asyncio.QueueEmpty:
+                    break
+                out.append(self._output_item(record, idx))
+        if clear_on_success:
+            for item in out:
+                rec = record.inputs.get(item["idx"])
+                if rec is not None:
+                    rec.payload = b""  # release memory; result already extracted
+        return out
+
+    def _output_item(self, record: CallRecord, idx: int) -> dict:
+        rec = record.inputs[idx]
+        return {
+            "idx": idx,
+            "input_id": rec.input_id,
+            "status": rec.status,
+            "data": rec.output,
+            "data_blob": rec.output_blob,
+            "format": rec.output_format,
+            "exc": rec.exc_repr,
+            "retry_count": rec.retry_count,
+        }
+
+    async def function_call_cancel(
+        self, function_call_id: str, terminate_containers: bool = False
+    ) -> None:
+        record = self._call(function_call_id)
+        record.cancelled = True
+        tokens = []
+        for rec in record.inputs.values():
+            if not rec.final:
+                rec.cancelled = True
+                tokens.append(rec.token)
+                self.finalize_input(
+                    rec, GENERIC_STATUS_TERMINATED, None, 0, "input cancelled", rec.retry_count
+                )
+        await self.pool.cancel_inputs(tokens, terminate=terminate_containers)
+
+    async def function_call_info(self, function_call_id: str) -> dict:
+        record = self._call(function_call_id)
+        return {
+            "function_call_id": record.call_id,
+            "function_id": record.function_id,
+            "kind": record.kind,
+            **record.stats(),
+        }
+
+    def _call(self, function_call_id: str) -> CallRecord:
+        record = self.calls.get(function_call_id)
+        if record is None:
+            raise NotFoundError(f"Function call {function_call_id} not found")
+        return record
+
+    # -- worker callbacks --------------------------------------------------
+    def on_worker_output(
+        self,
+        call_id: str,
+        idx: int,
+        retry_count: int,
+        status: int,
+        output: Optional[bytes],
+        output_format: int,
+        exc_repr: Optional[str],
+        output_blob: Optional[str] = None,
+    ) -> None:
+        record = self.calls.get(call_id)
+        if record is None:
+            return
+        rec = record.inputs.get(idx)
+        if rec is None or rec.final or rec.retry_count != retry_count:
+            return  # stale attempt (dedup parity: parallel_map.py:1416-1431)
+        fdef = self.functions.get(record.function_id)
+        if status == GENERIC_STATUS_FAILURE and fdef is not None:
+            policy = fdef.retry_policy
+            if rec.retry_count < policy.max_retries and not rec.cancelled:
+                rec.retry_count += 1
+                delay_s = policy.delay_ms(rec.retry_count) / 1000.0
+                if delay_s <= 0:
+                    self.pool.enqueue(rec)
+                else:
+                    self.pool.enqueue_delayed(rec, delay_s)
+                return
+        elif status == GENERIC_STATUS_INTERNAL_FAILURE:
+            rec.internal_failures += 1
+            if rec.internal_failures <= MAX_INTERNAL_FAILURE_COUNT and not rec.cancelled:
+                self.pool.enqueue(rec, front=True)
+                return
+        self.finalize_input(rec, status, output, output_format, exc_repr, retry_count, output_blob)
+
+    def finalize_input(
+        self,
+        rec: InputRecord,
+        status: int,
+        output: Optional[bytes],
+        output_format: int,
+        exc_repr: Optional[str],
+        retry_count: int,
+        output_blob: Optional[str] = None,
+    ) -> None:
+        record = self.calls.get(rec.call_id)
+        if record is None:
+            return
+        record.post_output(rec.idx, status, output, output_format, exc_repr, retry_count, output_blob)
+
+    def on_generator_data(self, msg: dict) -> None:
+        call_id, idx_s, _ = msg["token"].rsplit(":", 2)
+        record = self.calls.get(call_id)
+        if record is None:
+            return
+        record.gen_queue(int(idx_s)).put_nowait(
+            (msg.get("index", 0), msg.get("data"), msg.get("format", 0), bool(msg.get("done")))
+        )
+
+    async def generator_poll(
+        self, function_call_id: str, idx: int = 0, timeout: float = OUTPUT_POLL_TIMEOUT
+    ) -> list:
+        """RPC-friendly poll of a generator's data-out channel."""
+        record = self._call(function_call_id)
+        q = record.gen_queue(idx)
+        out = []
+        try:
+            item = await asyncio.wait_for(q.get(), timeout)
+            out.append(list(item))
+            while True:
+                out.append(list(q.get_nowait()))
+        except (asyncio.TimeoutError, asyncio.QueueEmpty):
+            pass
+        return out
+
+    def on_worker_log(self, handle: Any, msg: dict) -> None:
+        app_id = msg.get("app_id", "")
+        entry = {
+            "ts": time.time(),
+            "task_id": handle.task_id,
+            "fd": msg.get("fd", 1),
+            "data": msg.get("data", ""),
+            "app_id": app_id,
+        }
+        app = self.apps.get(app_id)
+        if app is not None:
+            app.logs.append(entry)
+            for sub in app.log_subscribers:
+                sub.put_nowait(entry)
+        else:
+            for a in self.apps.values():
+                if a.state == "running":
+                    for sub in a.log_subscribers:
+                        sub.put_nowait(entry)
+
+    def resolve_function_env(self, fdef: FunctionDef) -> dict[str, str]:
+        """Merge the env bundles of a function's secrets (later wins;
+        parity: secrets applied to container env, reference secret.py)."""
+        env: dict[str, str] = {}
+        for secret_id in fdef.secret_ids:
+            try:
+                env.update(self.services.secret_env(secret_id))
+            except Exception:
+                pass
+        return env
+
+    # -- cluster rendezvous ------------------------------------------------
+    async def cluster_hello(self, cluster_id: str, rank: int, world_size: int, addr: str = "") -> dict:
+        """In-proc TaskClusterHello: collect member addresses, release when full
+        (parity: reference _clustered_functions.py:74-86)."""
+        key = f"cluster:{cluster_id}"
+        state = self._extra.setdefault(key, {"members": {}, "event": asyncio.Event()})
+        state["members"][rank] = addr
+        if len(state["members"]) >= world_size:
+            state["event"].set()
+        await state["event"].wait()
+        return {
+            "cluster_id": cluster_id,
+            "rank": rank,
+            "world_size": world_size,
+            "addrs": [state["members"].get(r, "") for r in range(world_size)],
+        }
+
+    # -- blobs -------------------------------------------------------------
+    async def blob_put(self, data: bytes) -> dict:
+        digest = await asyncio.get_running_loop().run_in_executor(None, self.blob_store.put, data)
+        return {"blob_id": digest}
+
+    async def blob_get(self, blob_id: str) -> bytes:
+        return await asyncio.get_running_loop().run_in_executor(None, self.blob_store.get, blob_id)
+
+    async def blob_path(self, blob_id: str) -> str:
+        return self.blob_store.open_path(blob_id)
+
+    def maybe_blob(self, payload: bytes, limit: int = INLINE_LIMIT) -> dict:
+        """Inline small payloads; CAS-store large ones (parity: should_upload,
+        reference function_utils.py:586, thresholds blob_utils.py:36-39)."""
+        if len(payload) <= limit:
+            return {"payload": payload}
+        digest = self.blob_store.put(payload)
+        return {"payload_blob": digest}
+
+    # -- service passthrough (queues/dicts/secrets) -------------------------
+    def __getattr__(self, name: str) -> Any:
+        # delegate queue_/dict_/secret_ methods to Services, wrapped async
+        if name.startswith(("queue_", "dict_", "secret_")):
+            target = getattr(self.services, name, None)
+            if target is not None:
+                if asyncio.iscoroutinefunction(target):
+                    return target
+
+                async def _async_wrap(*args: Any, **kwargs: Any) -> Any:
+                    return target(*args, **kwargs)
+
+                return _async_wrap
+        raise AttributeError(name)

@@ -1,0 +1,166 @@
+"""Worker-plane transport: length-prefixed msgpack frames over Unix sockets.
+
+This replaces the reference's gRPC/HTTP2 control plane
+(/root/reference/py/modal/_utils/grpc_utils.py) with a single-node design:
+4-byte little-endian length + msgpack body, no TLS, no HTTP framing. Payload
+bytes ride inside msgpack bin values (zero re-encoding). An RPC layer on top
+gives workers access to scheduler services (queues, dicts, volumes, blobs)
+with the same semantics user code sees in the client process.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import itertools
+import struct
+from typing import Any, Awaitable, Callable, Optional
+
+import msgpack
+
+_LEN = struct.Struct("<I")
+
+MAX_FRAME = 1 << 31  # 2 GiB guard
+
+
+def pack(msg: dict) -> bytes:
+    body = msgpack.packb(msg, use_bin_type=True)
+    return _LEN.pack(len(body)) + body
+
+
+async def read_frame(reader: asyncio.StreamReader) -> Optional[dict]:
+    try:
+        header = await reader.readexactly(4)
+    except (asyncio.IncompleteReadError, ConnectionResetError):
+        return None
+    (length,) = _LEN.unpack(header)
+    if length > MAX_FRAME:
+        raise ValueError(f"Frame too large: {length}")
+    try:
+        body = await reader.readexactly(length)
+    except (asyncio.IncompleteReadError, ConnectionResetError):
+        return None
+    return msgpack.unpackb(body, raw=False, strict_map_key=False)
+
+
+class Connection:
+    """One framed connection with an RPC request/response layer.
+
+    Both directions can send one-way messages (dispatched to ``handler``) and
+    RPCs (``call``), multiplexed by message type:
+      {"t": <kind>, ...}                     one-way
+      {"t": "rpc", "i": id, "m": method, "p": params}  request
+      {"t": "rpc_r", "i": id, "r": result} / {"t": "rpc_e", "i": id, "e": msg}
+    """
+
+    def __init__(
+        self,
+        reader: asyncio.StreamReader,
+        writer: asyncio.StreamWriter,
+        handler: Callable[[dict], Awaitable[None]],
+        rpc_target: Optional[Any] = None,
+    ):
+        self.reader = reader
+        self.writer = writer
+        self.handler = handler
+        self.rpc_target = rpc_target  # object whose async methods serve inbound RPCs
+        self._rpc_seq = itertools.count(1)
+        self._pending: dict[int, asyncio.Future] = {}
+        self._send_lock = asyncio.Lock()
+        self._closed = asyncio.Event()
+        self._reader_task: Optional[asyncio.Task] = None
+
+    def start(self) -> None:
+        self._reader_task = asyncio.get_running_loop().create_task(self._read_loop())
+
+    @property
+    def closed(self) -> bool:
+        return self._closed.is_set()
+
+    async def wait_closed(self) -> None:
+        await self._closed.wait()
+
+    async def send(self, msg: dict) -> None:
+        data = pack(msg)
+        async with self._send_lock:
+            self.writer.write(data)
+            await self.writer.drain()
+
+    async def call(self, method: str, params: Any = None, timeout: Optional[float] = None) -> Any:
+        rpc_id = next(self._rpc_seq)
+        fut: asyncio.Future = asyncio.get_running_loop().create_future()
+        self._pending[rpc_id] = fut
+        try:
+            await self.send({"t": "rpc", "i": rpc_id, "m": method, "p": params})
+            if timeout is not None:
+                return await asyncio.wait_for(fut, timeout)
+            return await fut
+        finally:
+            self._pending.pop(rpc_id, None)
+
+    async def _serve_rpc(self, msg: dict) -> None:
+        rpc_id = msg["i"]
+        method = msg["m"]
+        try:
+            if self.rpc_target is None:
+                raise RuntimeError("No RPC target on this end")
+            fn = getattr(self.rpc_target, method, None)
+            if fn is None or method.startswith("_"):
+                raise RuntimeError(f"Unknown RPC method {method!r}")
+            params = msg.get("p") or {}
+            result = await fn(**params)
+            await self.send({"t": "rpc_r", "i": rpc_id, "r": result})
+        except asyncio.CancelledError:
+            raise
+        except BaseException as exc:  # report, keep connection alive
+            await self.send(
+                {"t": "rpc_e", "i": rpc_id, "e": f"{type(exc).__name__}: {exc}", "c": type(exc).__name__}
+            )
+
+    async def _read_loop(self) -> None:
+        try:
+            while True:
+                msg = await read_frame(self.reader)
+                if msg is None:
+                    break
+                kind = msg.get("t")
+                if kind == "rpc":
+                    asyncio.get_running_loop().create_task(self._serve_rpc(msg))
+                elif kind == "rpc_r":
+                    fut = self._pending.get(msg["i"])
+                    if fut is not None and not fut.done():
+                        fut.set_result(msg.get("r"))
+                elif kind == "rpc_e":
+                    fut = self._pending.get(msg["i"])
+                    if fut is not None and not fut.done():
+                        fut.set_exception(RemoteRPCError(msg.get("e", "remote error"), msg.get("c")))
+                else:
+                    await self.handler(msg)
+        except asyncio.CancelledError:
+            pass
+        except Exception:
+            pass
+        finally:
+            self._closed.set()
+            for fut in self._pending.values():
+                if not fut.done():
+                    fut.set_exception(ConnectionError("connection closed"))
+            try:
+                self.writer.close()
+            except Exception:
+                pass
+
+    async def close(self) -> None:
+        if self._reader_task is not None:
+            self._reader_task.cancel()
+        try:
+            self.writer.close()
+            await self.writer.wait_closed()
+        except Exception:
+            pass
+        self._closed.set()
+
+
+class RemoteRPCError(Exception):
+    def __init__(self, message: str, code: Optional[str] = None):
+        super().__init__(message)
+        self.code = code

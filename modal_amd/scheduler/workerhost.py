@@ -1,0 +1,472 @@
+"""Worker pool: per-GPU worker processes and the dispatch plane.
+
+MI355X-native replacement for the reference's container fleet: instead of
+cloud-scheduled containers pulling inputs over gRPC long-polls
+(/root/reference/py/modal/_runtime/container_io_manager.py:856), the scheduler
+*pushes* credit-bounded input batches to worker processes over Unix sockets —
+one worker per GPU (``HIP_VISIBLE_DEVICES``-pinned) plus CPU workers, all on
+this node. Worker death reproduces the INTERNAL_FAILURE requeue path
+(reference _functions.py:106): in-flight inputs are rescheduled up to 8 times.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import heapq
+import os
+import subprocess
+import sys
+import time
+from collections import deque
+from typing import TYPE_CHECKING, Optional
+
+from ..utils.ids import new_id
+from .calls import (
+    GENERIC_STATUS_INTERNAL_FAILURE,
+    MAX_INTERNAL_FAILURE_COUNT,
+    FunctionDef,
+    InputRecord,
+)
+from .transport import Connection
+
+if TYPE_CHECKING:
+    from .core import Scheduler
+
+# deep-enough per-worker pipeline so tiny functions amortize frame overhead;
+# the analog of the reference's 49-inputs-per-PutInputs batching
+# (/root/reference/py/modal/parallel_map.py:82) tuned for a local socket.
+DEFAULT_PIPELINE_DEPTH = 64
+
+
+class WorkerHandle:
+    def __init__(
+        self,
+        worker_id: int,
+        conn: Connection,
+        gpu_index: Optional[int],
+        proc: Optional[subprocess.Popen] = None,
+        external: bool = False,
+    ):
+        self.worker_id = worker_id
+        self.task_id = new_id("task")
+        self.conn = conn
+        self.gpu_index = gpu_index
+        self.proc = proc
+        self.external = external
+        self.alive = True
+        self.last_heartbeat = time.time()
+        # tokens of inputs currently assigned here, mapped to their records
+        self.inflight: dict[str, InputRecord] = {}
+        # per-function outstanding count (for credit computation)
+        self.outstanding: dict[str, int] = {}
+        self.functions_loaded: set[str] = set()
+        self.defs_version: dict[str, int] = {}
+        self.draining = False
+
+    @property
+    def has_gpu(self) -> bool:
+        return self.gpu_index is not None
+
+    def credit_for(self, fdef: FunctionDef) -> int:
+        cap = max(fdef.max_concurrent_inputs, 1) * 2
+        cap = max(cap, DEFAULT_PIPELINE_DEPTH if not fdef.needs_gpu else fdef.max_concurrent_inputs * 2)
+        if fdef.batch_max_size:
+            cap = max(cap, fdef.batch_max_size * 2)
+        return cap - self.outstanding.get(fdef.function_id, 0)
+
+
+class WorkerPool:
+    def __init__(self, scheduler: "Scheduler"):
+        self.scheduler = scheduler
+        self.run_dir = scheduler.run_dir
+        self.socket_path = os.path.join(self.run_dir, "scheduler.sock")
+        self.workers: dict[int, WorkerHandle] = {}
+        self._next_worker_id = 0
+        self._server: Optional[asyncio.AbstractServer] = None
+        self._dispatch_wake = asyncio.Event()
+        self._dispatch_task: Optional[asyncio.Task] = None
+        self._retry_task: Optional[asyncio.Task] = None
+        self._stopping = False
+        # pending inputs per function (FIFO) and delayed retries (timestamp heap,
+        # parity: TimestampPriorityQueue, reference async_utils.py:656)
+        self.pending: dict[str, deque[InputRecord]] = {}
+        self.delayed: list[tuple[float, int, str, InputRecord]] = []
+        self._delay_seq = 0
+        self._spawn_lock = asyncio.Lock()
+        self._pending_spawns = 0
+        self._worker_ready = asyncio.Event()
+        self._procs: list[subprocess.Popen] = []
+
+    # -- lifecycle -------------------------------------------------------
+    async def start(self) -> None:
+        os.makedirs(self.run_dir, exist_ok=True)
+        self._server = await asyncio.start_unix_server(self._on_connect, path=self.socket_path)
+        self._dispatch_task = asyncio.get_running_loop().create_task(self._dispatch_loop())
+        self._retry_task = asyncio.get_running_loop().create_task(self._retry_loop())
+
+    async def stop(self) -> None:
+        self._stopping = True
+        for task in (self._dispatch_task, self._retry_task):
+            if task is not None:
+                task.cancel()
+        for w in list(self.workers.values()):
+            try:
+                await w.conn.send({"t": "shutdown"})
+            except Exception:
+                pass
+        # give workers a moment to exit cleanly, then kill
+        deadline = time.time() + 3.0
+        for w in list(self.workers.values()):
+            if w.proc is not None:
+                try:
+                    w.proc.wait(timeout=max(0.05, deadline - time.time()))
+                except subprocess.TimeoutExpired:
+                    w.proc.kill()
+        for w in list(self.workers.values()):
+            await w.conn.close()
+        # reap any process that never connected (or is still exiting)
+        for proc in self._procs:
+            if proc.poll() is None:
+                try:
+                    proc.terminate()
+                    proc.wait(timeout=2)
+                except Exception:
+                    try:
+                        proc.kill()
+                    except Exception:
+                        pass
+        if self._server is not None:
+            self._server.close()
+            try:
+                await self._server.wait_closed()
+            except Exception:
+                pass
+        try:
+            os.unlink(self.socket_path)
+        except OSError:
+            pass
+
+    # -- worker spawning -------------------------------------------------
+    def _gpu_count(self) -> int:
+        if os.environ.get("MODAL_AMD_FAKE_GPUS"):
+            return int(os.environ["MODAL_AMD_FAKE_GPUS"])
+        try:
+            import torch
+
+            if torch.cuda.is_available():
+                return torch.cuda.device_count()
+        except Exception:
+            pass
+        return 0
+
+    async def spawn_worker(self, gpu_index: Optional[int] = None, extra_env: Optional[dict] = None) -> None:
+        """Spawn one worker process; it will connect back to our socket."""
+        worker_id = self._next_worker_id
+        self._next_worker_id += 1
+        env = dict(os.environ)
+        env["MODAL_AMD_WORKER_SOCKET"] = self.socket_path
+        env["MODAL_AMD_WORKER_ID"] = str(worker_id)
+        env["MODAL_AMD_IS_REMOTE"] = "1"
+        if gpu_index is not None:
+            env["MODAL_AMD_GPU_INDEX"] = str(gpu_index)
+            env["HIP_VISIBLE_DEVICES"] = str(gpu_index)
+            env["CUDA_VISIBLE_DEVICES"] = str(gpu_index)
+        if extra_env:
+            env.update(extra_env)
+        self._pending_spawns += 1
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "modal_amd.runtime.worker"],
+            env=env,
+            cwd=os.getcwd(),
+            start_new_session=True,
+        )
+        self._procs.append(proc)
+
+    async def ensure_workers(self, needs_gpu: bool) -> None:
+        """Lazily bring up the pool sized to the hardware (288 GB/GPU MI355X:
+        one worker per GPU; CPU functions get a small CPU pool)."""
+        have = [w for w in self.workers.values() if w.alive and (w.has_gpu or not needs_gpu)]
+        if needs_gpu:
+            have = [w for w in have if w.has_gpu]
+        if have or self._pending_spawns > 0:
+            return
+        async with self._spawn_lock:
+            have = [w for w in self.workers.values() if w.alive and ((w.has_gpu and needs_gpu) or not needs_gpu)]
+            if have or self._pending_spawns > 0:
+                return
+            from ..config import config
+
+            n_gpus = self._gpu_count()
+            if needs_gpu or n_gpus > 0:
+                count = config.get("worker_count") or max(n_gpus, 1)
+                for i in range(count):
+                    await self.spawn_worker(gpu_index=i % n_gpus if n_gpus else None)
+            else:
+                count = config.get("worker_count") or min(max((os.cpu_count() or 4) // 2, 1), 8)
+                for _ in range(count):
+                    await self.spawn_worker(gpu_index=None)
+
+    async def wait_for_worker(self, needs_gpu: bool, timeout: float = 120.0) -> None:
+        deadline = time.time() + timeout
+        while True:
+            if any(w.alive and (w.has_gpu or not needs_gpu) for w in self.workers.values()):
+                return
+            await self.ensure_workers(needs_gpu)
+            self._worker_ready.clear()
+            remaining = deadline - time.time()
+            if remaining <= 0:
+                raise TimeoutError("No worker became available")
+            try:
+                await asyncio.wait_for(self._worker_ready.wait(), min(remaining, 1.0))
+            except asyncio.TimeoutError:
+                pass
+
+    # -- connections -----------------------------------------------------
+    async def _on_connect(self, reader: asyncio.StreamReader, writer: asyncio.StreamWriter) -> None:
+        conn: Connection = None  # type: ignore[assignment]
+
+        handle_holder: dict = {}
+
+        async def handler(msg: dict) -> None:
+            kind = msg.get("t")
+            if kind == "hello":
+                role = msg.get("role", "worker")
+                if role == "worker":
+                    worker_id = msg.get("worker_id")
+                    if worker_id is None or worker_id < 0:
+                        worker_id = self._next_worker_id
+                        self._next_worker_id += 1
+                    gpu_index = msg.get("gpu_index")
+                    handle = WorkerHandle(
+                        worker_id,
+                        conn,
+                        gpu_index,
+                        external=bool(msg.get("external")),
+                    )
+                    handle_holder["h"] = handle
+                    self.workers[worker_id] = handle
+                    self._pending_spawns = max(0, self._pending_spawns - 1)
+                    self._worker_ready.set()
+                    self._dispatch_wake.set()
+                    await conn.send({"t": "hello_ack", "task_id": handle.task_id, "worker_id": worker_id})
+                    asyncio.get_running_loop().create_task(self._watch_worker(handle))
+                else:
+                    # tooling/sandbox client connection: RPC only
+                    await conn.send({"t": "hello_ack"})
+                return
+            handle = handle_holder.get("h")
+            if handle is None:
+                return
+            if kind == "outputs":
+                self._on_outputs(handle, msg)
+            elif kind == "gen_data":
+                self.scheduler.on_generator_data(msg)
+            elif kind == "hb":
+                handle.last_heartbeat = time.time()
+            elif kind == "log":
+                self.scheduler.on_worker_log(handle, msg)
+
+        conn = Connection(reader, writer, handler, rpc_target=self.scheduler.rpc_adapter)
+        conn.start()
+
+    async def _watch_worker(self, handle: WorkerHandle) -> None:
+        await handle.conn.wait_closed()
+        handle.alive = False
+        self.workers.pop(handle.worker_id, None)
+        if self._stopping:
+            return
+        # requeue in-flight inputs: the INTERNAL_FAILURE path
+        for token, rec in list(handle.inflight.items()):
+            if rec.final:
+                continue
+            rec.internal_failures += 1
+            rec.worker_id = None
+            if rec.internal_failures > MAX_INTERNAL_FAILURE_COUNT:
+                self.scheduler.finalize_input(
+                    rec,
+                    GENERIC_STATUS_INTERNAL_FAILURE,
+                    None,
+                    0,
+                    f"worker died while executing input (x{rec.internal_failures})",
+                    rec.retry_count,
+                )
+            else:
+                self.enqueue(rec, front=True)
+        handle.inflight.clear()
+
+    # -- output handling -------------------------------------------------
+    def _on_outputs(self, handle: WorkerHandle, msg: dict) -> None:
+        for item in msg["items"]:
+            token = item["token"]
+            rec = handle.inflight.pop(token, None)
+            call_id, idx_s, retry_s = token.rsplit(":", 2)
+            if rec is not None:
+                fid = rec.call_id  # noqa: F841  (token bookkeeping)
+            fdef_id = item.get("function_id")
+            if fdef_id:
+                cnt = handle.outstanding.get(fdef_id, 0)
+                if cnt > 0:
+                    handle.outstanding[fdef_id] = cnt - 1
+            self.scheduler.on_worker_output(
+                call_id=call_id,
+                idx=int(idx_s),
+                retry_count=int(retry_s),
+                status=item["status"],
+                output=item.get("data"),
+                output_format=item.get("format", 0),
+                exc_repr=item.get("exc"),
+                output_blob=item.get("data_blob"),
+            )
+        self._dispatch_wake.set()
+
+    # -- dispatch --------------------------------------------------------
+    def enqueue(self, rec: InputRecord, front: bool = False) -> None:
+        fdef = self.scheduler.functions[self._function_id_of(rec)]
+        q = self.pending.setdefault(fdef.function_id, deque())
+        if front:
+            q.appendleft(rec)
+        else:
+            q.append(rec)
+        self._dispatch_wake.set()
+
+    def enqueue_delayed(self, rec: InputRecord, delay_s: float) -> None:
+        self._delay_seq += 1
+        heapq.heappush(
+            self.delayed, (time.time() + delay_s, self._delay_seq, self._function_id_of(rec), rec)
+        )
+
+    def _function_id_of(self, rec: InputRecord) -> str:
+        return self.scheduler.calls[rec.call_id].function_id
+
+    async def _retry_loop(self) -> None:
+        while True:
+            if not self.delayed:
+                await asyncio.sleep(0.05)
+                continue
+            ready_at, _, fid, rec = self.delayed[0]
+            now = time.time()
+            if ready_at <= now:
+                heapq.heappop(self.delayed)
+                if not rec.final and not rec.cancelled:
+                    self.pending.setdefault(fid, deque()).append(rec)
+                    self._dispatch_wake.set()
+            else:
+                await asyncio.sleep(min(ready_at - now, 0.5))
+
+    async def _dispatch_loop(self) -> None:
+        while True:
+            await self._dispatch_wake.wait()
+            self._dispatch_wake.clear()
+            try:
+                await self._dispatch_once()
+            except asyncio.CancelledError:
+                raise
+            except Exception as exc:  # pragma: no cover - defensive
+                self.scheduler.log(f"dispatch error: {exc!r}")
+
+    async def _dispatch_once(self) -> None:
+        for fid, q in list(self.pending.items()):
+            if not q:
+                continue
+            fdef = self.scheduler.functions.get(fid)
+            if fdef is None:
+                q.clear()
+                continue
+            await self.ensure_workers(fdef.needs_gpu)
+            candidates = [
+                w
+                for w in self.workers.values()
+                if w.alive and not w.draining and (w.has_gpu or not fdef.needs_gpu)
+            ]
+            if fdef.needs_gpu:
+                candidates = [w for w in candidates if w.has_gpu]
+            if not candidates:
+                continue
+            # round-robin over workers by current outstanding (least-loaded first)
+            candidates.sort(key=lambda w: w.outstanding.get(fid, 0))
+            for w in candidates:
+                if not q:
+                    break
+                credit = w.credit_for(fdef)
+                if credit <= 0:
+                    continue
+                batch: list[InputRecord] = []
+                while q and len(batch) < credit:
+                    rec = q.popleft()
+                    if rec.final or rec.cancelled:
+                        continue
+                    batch.append(rec)
+                if not batch:
+                    continue
+                await self._send_batch(w, fdef, batch)
+            # anything left stays pending until credit frees up
+
+    async def _send_batch(self, w: WorkerHandle, fdef: FunctionDef, batch: list[InputRecord]) -> None:
+        try:
+            if (
+                fdef.function_id not in w.functions_loaded
+                or w.defs_version.get(fdef.function_id, 0) != fdef.definition_version
+            ):
+                await w.conn.send(
+                    {
+                        "t": "def",
+                        "function_id": fdef.function_id,
+                        "app_id": fdef.app_id,
+                        "name": fdef.name,
+                        "definition": fdef.definition,
+                        "definition_kind": fdef.definition_kind,
+                        "is_generator": fdef.is_generator,
+                        "timeout": fdef.timeout,
+                        "max_concurrent_inputs": fdef.max_concurrent_inputs,
+                        "batch_max_size": fdef.batch_max_size,
+                        "batch_linger_ms": fdef.batch_linger_ms,
+                        "version": fdef.definition_version,
+                        "app_layout": self.scheduler.app_layout(fdef.app_id),
+                        "env": self.scheduler.resolve_function_env(fdef),
+                        "volumes": fdef.volume_mounts,
+                    }
+                )
+                w.functions_loaded.add(fdef.function_id)
+                w.defs_version[fdef.function_id] = fdef.definition_version
+            items = []
+            for rec in batch:
+                rec.worker_id = w.worker_id
+                rec.started_at = time.time()
+                w.inflight[rec.token] = rec
+                item = {
+                    "token": rec.token,
+                    "input_id": rec.input_id,
+                    "payload": rec.payload,
+                    "method": rec.method_name,
+                    "retry_count": rec.retry_count,
+                }
+                if rec.payload_blob:
+                    item["payload_blob"] = rec.payload_blob
+                items.append(item)
+            w.outstanding[fdef.function_id] = w.outstanding.get(fdef.function_id, 0) + len(batch)
+            await w.conn.send({"t": "inputs", "function_id": fdef.function_id, "items": items})
+        except Exception:
+            # connection died mid-send: requeue, the watcher will clean up
+            for rec in batch:
+                if rec.token in w.inflight:
+                    del w.inflight[rec.token]
+                if not rec.final:
+                    self.enqueue(rec, front=True)
+
+    async def cancel_inputs(self, tokens: list[str], terminate: bool = False) -> None:
+        """Propagate cancellation to workers holding these inputs
+        (parity: server-pushed cancellation via heartbeats,
+        reference container_io_manager.py:645-710)."""
+        by_worker: dict[int, list[str]] = {}
+        for w in self.workers.values():
+            hit = [t for t in tokens if t in w.inflight]
+            if hit:
+                by_worker[w.worker_id] = hit
+        for worker_id, toks in by_worker.items():
+            w = self.workers.get(worker_id)
+            if w is None:
+                continue
+            try:
+                await w.conn.send({"t": "cancel", "tokens": toks, "terminate": terminate})
+            except Exception:
+                pass
