@@ -232,6 +232,10 @@ def _rebuild_handle(object_id: str, metadata: dict, client: Any) -> Any:
     """
     from ._sync import wrap
 
+    if client is None:
+        from .client import _Client
+
+        client = _Client._singleton  # same-process deserialization
     kind = id_type(object_id)
     cls = _TYPE_REGISTRY.get(kind)
     if cls is None:

@@ -193,6 +193,9 @@ class Services:
 
     def queue_delete(self, queue_id: str) -> None:
         self.queues.pop(queue_id, None)
+        self.queue_names.by_name = {
+            k: v for k, v in self.queue_names.by_name.items() if v != queue_id
+        }
 
     # ---- dicts ---------------------------------------------------------
     def dict_get_or_create(
@@ -259,6 +262,9 @@ class Services:
 
     def dict_delete(self, dict_id: str) -> None:
         self.dicts.pop(dict_id, None)
+        self.dict_names.by_name = {
+            k: v for k, v in self.dict_names.by_name.items() if v != dict_id
+        }
 
     # ---- secrets -------------------------------------------------------
     def secret_get_or_create(

@@ -68,9 +68,11 @@ class Connection:
         self._send_lock = asyncio.Lock()
         self._closed = asyncio.Event()
         self._reader_task: Optional[asyncio.Task] = None
+        self._loop: Optional[asyncio.AbstractEventLoop] = None
 
     def start(self) -> None:
-        self._reader_task = asyncio.get_running_loop().create_task(self._read_loop())
+        self._loop = asyncio.get_running_loop()
+        self._reader_task = self._loop.create_task(self._read_loop())
 
     @property
     def closed(self) -> bool:
@@ -86,6 +88,13 @@ class Connection:
             await self.writer.drain()
 
     async def call(self, method: str, params: Any = None, timeout: Optional[float] = None) -> Any:
+        # RPCs may originate on another event loop (e.g. the synchronizer loop
+        # driving user code's handles inside a worker); bridge to the loop that
+        # owns this connection's streams.
+        running = asyncio.get_running_loop()
+        if self._loop is not None and running is not self._loop:
+            fut = asyncio.run_coroutine_threadsafe(self.call(method, params, timeout), self._loop)
+            return await asyncio.wrap_future(fut)
         rpc_id = next(self._rpc_seq)
         fut: asyncio.Future = asyncio.get_running_loop().create_future()
         self._pending[rpc_id] = fut

@@ -1,0 +1,143 @@
+"""Queue / Dict / Secret semantics (parity: reference queue_test, dict_test)."""
+
+from __future__ import annotations
+
+import threading
+import time
+
+import pytest
+
+import modal_amd as modal
+from modal_amd.exception import NotFoundError, QueueEmptyError
+
+
+def test_queue_basic(client):
+    with modal.Queue.ephemeral() as q:
+        q.put(1)
+        q.put_many([2, 3])
+        assert q.len() == 3
+        assert q.get() == 1
+        assert q.get_many(2) == [2, 3]
+        assert q.get(block=False) is None
+
+
+def test_queue_partitions(client):
+    with modal.Queue.ephemeral() as q:
+        q.put(1)
+        q.put(2, partition="other")
+        assert q.len() == 1
+        assert q.len(partition="other") == 1
+        assert q.len(total=True) == 2
+        assert q.get(partition="other") == 2
+
+
+def test_queue_blocking_timeout(client):
+    with modal.Queue.ephemeral() as q:
+        t0 = time.time()
+        with pytest.raises(QueueEmptyError):
+            q.get(timeout=0.2)
+        assert 0.1 < time.time() - t0 < 5
+
+
+def test_queue_producer_consumer_threads(client):
+    with modal.Queue.ephemeral() as q:
+        got = []
+
+        def consumer():
+            for _ in range(10):
+                got.append(q.get(timeout=5))
+
+        t = threading.Thread(target=consumer)
+        t.start()
+        for i in range(10):
+            q.put(i)
+        t.join()
+        assert sorted(got) == list(range(10))
+
+
+def test_queue_named_and_delete(client):
+    q = modal.Queue.from_name("q-named", create_if_missing=True)
+    q.put("x")
+    q2 = modal.Queue.from_name("q-named")
+    assert q2.get() == "x"
+    modal.Queue.delete("q-named")
+    with pytest.raises(NotFoundError):
+        modal.Queue.from_name("q-named").hydrate()
+
+
+def test_queue_iterate(client):
+    with modal.Queue.ephemeral() as q:
+        q.put_many([1, 2, 3])
+        assert list(q.iterate()) == [1, 2, 3]
+        assert q.len() == 3  # non-destructive
+
+
+def test_dict_basic(client):
+    with modal.Dict.ephemeral() as d:
+        d.put("k", {"v": 1})
+        assert d.get("k") == {"v": 1}
+        assert d.get("missing", 5) == 5
+        assert d.contains("k")
+        assert d.len() == 1
+        d.update({"a": 1, "b": 2})
+        assert sorted([k for k in d.keys()]) == ["a", "b", "k"]
+        assert d.pop("a") == 1
+        with pytest.raises(KeyError):
+            d.pop("a")
+        d.clear()
+        assert d.len() == 0
+
+
+def test_dict_object_keys(client):
+    with modal.Dict.ephemeral() as d:
+        d.put((1, 2), "tuple-key")
+        assert d.get((1, 2)) == "tuple-key"
+
+
+def test_dict_skip_if_exists(client):
+    with modal.Dict.ephemeral() as d:
+        assert d.put("k", 1, skip_if_exists=True) is True
+        assert d.put("k", 2, skip_if_exists=True) is False
+        assert d.get("k") == 1
+
+
+def test_secret_from_dict_and_name(client):
+    s = modal.Secret.from_dict({"API_KEY": "abc"})
+    s.hydrate()
+    assert s.env() == {"API_KEY": "abc"}
+    modal.Secret.create_deployed("my-secret", {"TOK": "t1"})
+    s2 = modal.Secret.from_name("my-secret")
+    s2.hydrate()
+    assert s2.env() == {"TOK": "t1"}
+    with pytest.raises(NotFoundError):
+        modal.Secret.from_name("nope").hydrate()
+
+
+def test_secret_reaches_worker_env(client):
+    app = modal.App("test-secret-env")
+
+    @app.function(secrets=[modal.Secret.from_dict({"MY_TEST_VAR": "hello-worker"})])
+    def read_env():
+        import os
+
+        return os.environ.get("MY_TEST_VAR")
+
+    with app.run(client=client):
+        assert read_env.remote() == "hello-worker"
+
+
+def test_queue_used_inside_worker(client):
+    """Handles serialize as ids and rebind to the worker's scheduler proxy."""
+    app = modal.App("test-q-worker")
+
+    @app.function()
+    def pusher(q, n):
+        for i in range(n):
+            q.put(i * 2)
+        return q.len()
+
+    with app.run(client=client):
+        with modal.Queue.ephemeral() as q:
+            count = pusher.remote(q, 5)
+            assert count == 5
+            assert [q.get() for _ in range(5)] == [0, 2, 4, 6, 8]

@@ -1,0 +1,102 @@
+from __future__ import annotations
+
+import hashlib
+
+import pytest
+
+from modal_amd._serialization import (
+    DataFormat,
+    GeneratorDone,
+    deserialize,
+    deserialize_data_format,
+    deserialize_payload,
+    serialize,
+    serialize_data_format,
+)
+from modal_amd.utils import cbor
+
+
+def test_roundtrip_basics():
+    for obj in [1, "x", b"bytes", [1, 2], {"a": (1, 2)}, None, 3.5, {1, 2}]:
+        assert deserialize(serialize(obj)) == obj
+
+
+def test_roundtrip_closure():
+    y = 41
+
+    def f(x):
+        return x + y
+
+    g = deserialize(serialize(f))
+    assert g(1) == 42
+
+
+def test_payload_roundtrip():
+    data = serialize(("P", ((1, "a"), {"k": 2})))
+    args, kwargs = deserialize_payload(data)
+    assert args == (1, "a")
+    assert kwargs == {"k": 2}
+
+
+def test_data_formats():
+    blob = serialize_data_format({"a": 1}, DataFormat.CBOR)
+    assert deserialize_data_format(blob, DataFormat.CBOR) == {"a": 1}
+    done = serialize_data_format(GeneratorDone(5), DataFormat.GENERATOR_DONE)
+    out = deserialize_data_format(done, DataFormat.GENERATOR_DONE)
+    assert out == GeneratorDone(5)
+
+
+def test_cbor_vectors():
+    # RFC 8949 appendix A vectors (subset)
+    assert cbor.dumps(0) == bytes.fromhex("00")
+    assert cbor.dumps(1) == bytes.fromhex("01")
+    assert cbor.dumps(10) == bytes.fromhex("0a")
+    assert cbor.dumps(23) == bytes.fromhex("17")
+    assert cbor.dumps(24) == bytes.fromhex("1818")
+    assert cbor.dumps(25) == bytes.fromhex("1819")
+    assert cbor.dumps(100) == bytes.fromhex("1864")
+    assert cbor.dumps(1000) == bytes.fromhex("1903e8")
+    assert cbor.dumps(1000000) == bytes.fromhex("1a000f4240")
+    assert cbor.dumps(-1) == bytes.fromhex("20")
+    assert cbor.dumps(-10) == bytes.fromhex("29")
+    assert cbor.dumps(-100) == bytes.fromhex("3863")
+    assert cbor.dumps(False) == bytes.fromhex("f4")
+    assert cbor.dumps(True) == bytes.fromhex("f5")
+    assert cbor.dumps(None) == bytes.fromhex("f6")
+    assert cbor.dumps(1.1) == bytes.fromhex("fb3ff199999999999a")
+    assert cbor.dumps("a") == bytes.fromhex("6161")
+    assert cbor.dumps("IETF") == bytes.fromhex("6449455446")
+    assert cbor.dumps([1, 2, 3]) == bytes.fromhex("83010203")
+    assert cbor.dumps({"a": 1, "b": [2, 3]}) == bytes.fromhex("a26161016162820203")
+    assert cbor.dumps(18446744073709551616) == bytes.fromhex("c249010000000000000000")
+
+
+def test_cbor_roundtrip():
+    for obj in [0, 1, -1, 2**70, -(2**70), "héllo", b"\x00\xff", [1, [2, [3]]],
+                {"k": {"n": None}}, 3.14159, True, False, None]:
+        assert cbor.loads(cbor.dumps(obj)) == obj
+    # half-float decode
+    assert cbor.loads(bytes.fromhex("f90000")) == 0.0
+    assert cbor.loads(bytes.fromhex("f93c00")) == 1.0
+
+
+def test_handle_swap_roundtrip(client):
+    """A hydrated Queue handle pickles as an id and rehydrates on load."""
+    import modal_amd as modal
+
+    with modal.Queue.ephemeral() as q:
+        q.put(1)
+        data = serialize({"the_queue": q})
+        out = deserialize(data)
+        q2 = out["the_queue"]
+        assert q2.object_id == q.object_id
+        assert q2.get() == 1
+
+
+def test_unhydrated_handle_rejected(client):
+    import modal_amd as modal
+    from modal_amd.exception import SerializationError
+
+    q = modal.Queue.from_name("never-hydrated")
+    with pytest.raises(SerializationError):
+        serialize(q)
