@@ -1,0 +1,255 @@
+"""Flagship benchmark: Function.map items/sec across N MI355X GPUs.
+
+BASELINE.json metric: "Function.map items/sec (node) + p50 .remote() latency"
+— config 3 (100k synthetic inputs fanned across the GPUs) as the headline
+number, with config 2's p50 .remote() latency (gpu=1 torch.mm bf16) reported
+in the config block.
+
+Driver contract:
+  python bench.py --gpus N --steps K --warmup W
+For N>1 the driver launches this under torch.distributed.run with one rank
+per GPU. Rank 0 runs the client + in-process scheduler; every rank
+(including 0) contributes one worker pinned to its GPU. Steps are bracketed
+by a dist barrier + torch.cuda.synchronize on both sides; elapsed time is
+MAX-reduced over ranks; rank 0 prints one JSON line.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+# Pin each torchrun rank to its GPU *before* torch import so device 0 is the
+# rank's GPU everywhere (workers, RCCL, user payloads).
+_LOCAL_RANK = int(os.environ.get("LOCAL_RANK", "0"))
+_WORLD_SIZE = int(os.environ.get("WORLD_SIZE", "1"))
+if _WORLD_SIZE > 1 and "HIP_VISIBLE_DEVICES" not in os.environ:
+    os.environ["HIP_VISIBLE_DEVICES"] = str(_LOCAL_RANK)
+    os.environ["CUDA_VISIBLE_DEVICES"] = str(_LOCAL_RANK)
+
+ITEMS_PER_GPU = 12_500  # x8 GPUs = the 100k-input config of BASELINE.json
+
+
+def _bench_run_dir() -> str:
+    port = os.environ.get("MASTER_PORT", "0")
+    return f"/tmp/modal-amd-bench-{port}-{os.environ.get('TORCHELASTIC_RUN_ID', 'solo')}"
+
+
+def map_item_gpu(x: int) -> int:
+    import torch
+
+    cache = getattr(torch, "_ma_bench_cache", None)
+    if cache is None:
+        cache = torch.ones(4096, device="cuda", dtype=torch.bfloat16)
+        torch._ma_bench_cache = cache
+    t = cache * float(x % 7 + 1)
+    return int(t[:4].float().sum().item()) and x or x
+
+
+def map_item_cpu(x: int) -> int:
+    return x
+
+
+def p50_probe_gpu(n: int) -> float:
+    import torch
+
+    a = torch.randn(n, n, dtype=torch.bfloat16, device="cuda")
+    b = torch.randn(n, n, dtype=torch.bfloat16, device="cuda")
+    c = a @ b
+    torch.cuda.synchronize()
+    return float(c.float().mean().item())
+
+
+def run_worker_rank(rank: int, world: int) -> None:
+    """Ranks > 0: host one worker connected to rank 0's scheduler, plus
+    participate in the start/end barriers from the main thread."""
+    import threading
+
+    import torch
+    import torch.distributed as dist
+
+    socket_path = os.path.join(_bench_run_dir(), "scheduler.sock")
+    deadline = time.time() + 120
+    while not os.path.exists(socket_path):
+        if time.time() > deadline:
+            raise RuntimeError("scheduler socket never appeared")
+        time.sleep(0.05)
+
+    os.environ["MODAL_AMD_WORKER_SOCKET"] = socket_path
+    os.environ["MODAL_AMD_WORKER_ID"] = str(1000 + rank)
+    os.environ["MODAL_AMD_GPU_INDEX"] = "0" if torch.cuda.is_available() else ""
+    os.environ["MODAL_AMD_EXTERNAL_WORKER"] = "1"
+    os.environ["MODAL_AMD_IS_REMOTE"] = "1"
+    if not os.environ.get("MODAL_AMD_GPU_INDEX"):
+        del os.environ["MODAL_AMD_GPU_INDEX"]
+
+    from modal_amd.runtime.worker import WorkerRuntime
+
+    runtime = WorkerRuntime()
+    thread = threading.Thread(target=lambda: __import__("asyncio").run(runtime.run()), daemon=True)
+    thread.start()
+
+    device = torch.device("cuda:0") if torch.cuda.is_available() else None
+    # start barrier (warmup done on rank 0), timed region, end barrier
+    dist.barrier()
+    t0 = time.perf_counter()
+    dist.barrier()
+    if device is not None:
+        torch.cuda.synchronize()
+    elapsed = torch.tensor([time.perf_counter() - t0], dtype=torch.float64)
+    dist.all_reduce(elapsed, op=dist.ReduceOp.MAX)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def main() -> None:
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=8)
+    parser.add_argument("--warmup", type=int, default=2)
+    parser.add_argument("--items-per-gpu", type=int, default=ITEMS_PER_GPU)
+    args = parser.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = _WORLD_SIZE
+    n_gpus = args.gpus
+
+    import torch
+
+    has_gpu = torch.cuda.is_available()
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        backend = "nccl" if has_gpu else "gloo"
+        dist.init_process_group(backend=backend)
+        if has_gpu:
+            torch.cuda.set_device(0)
+
+    if rank != 0:
+        run_worker_rank(rank, world)
+        return
+
+    # ---- rank 0: client + scheduler -----------------------------------
+    run_dir = _bench_run_dir()
+    os.makedirs(run_dir, exist_ok=True)
+    os.environ["MODAL_AMD_RUN_DIR"] = run_dir
+
+    from modal_amd._sync import synchronizer
+    from modal_amd.client import _Client
+    from modal_amd.scheduler.core import Scheduler
+
+    async def boot():
+        scheduler = Scheduler(run_dir=run_dir)
+        await scheduler.start()
+        client = _Client(scheduler, "client")
+        _Client.set_default(client)
+        return scheduler, client
+
+    scheduler, client = synchronizer.run(boot())
+
+    # rank 0 contributes one worker for its own GPU (ranks>0 bring theirs)
+    async def spawn_local_worker():
+        await scheduler.pool.spawn_worker(gpu_index=0 if has_gpu else None)
+
+    synchronizer.run(spawn_local_worker())
+    # wait until all N workers are connected
+    want_workers = n_gpus if world > 1 else 1
+    deadline = time.time() + 180
+    while len(scheduler.pool.workers) < want_workers:
+        if time.time() > deadline:
+            raise RuntimeError(
+                f"only {len(scheduler.pool.workers)}/{want_workers} workers connected"
+            )
+        time.sleep(0.05)
+
+    import modal_amd as modal
+
+    app = modal.App("bench")
+    work_fn = map_item_gpu if has_gpu else map_item_cpu
+    item_fn = app.function(gpu=1 if has_gpu else None)(
+        modal.concurrent(max_inputs=8)(work_fn)
+    )
+    probe_fn = app.function(gpu=1 if has_gpu else None)(p50_probe_gpu) if has_gpu else None
+
+    items_per_step = args.items_per_gpu * n_gpus
+
+    ctx = app.run(client=client)
+    ctx.__enter__()
+    try:
+        # ---- warmup: also measures config-2 p50 .remote() latency ------
+        p50_ms = None
+        if probe_fn is not None:
+            lat = []
+            probe_fn.remote(256)  # first call pays worker/model init
+            for _ in range(30):
+                t = time.perf_counter()
+                probe_fn.remote(256)
+                lat.append((time.perf_counter() - t) * 1000)
+            lat.sort()
+            p50_ms = lat[len(lat) // 2]
+        for _ in range(args.warmup):
+            out = list(item_fn.map(range(items_per_step), order_outputs=False))
+            assert len(out) == items_per_step
+
+        # ---- timed region ---------------------------------------------
+        if dist is not None:
+            dist.barrier()
+        if has_gpu:
+            torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(args.steps):
+            out = list(item_fn.map(range(items_per_step), order_outputs=False))
+            assert len(out) == items_per_step
+        if dist is not None:
+            dist.barrier()
+        if has_gpu:
+            torch.cuda.synchronize()
+        elapsed = time.perf_counter() - t0
+        if dist is not None:
+            t = torch.tensor([elapsed], dtype=torch.float64)
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            elapsed = float(t.item())
+
+        total_items = args.steps * items_per_step
+        result = {
+            "metric": "map_items_per_sec",
+            "value": total_items / elapsed,
+            "unit": "items/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if has_gpu else "none",
+            "data": "synthetic",
+            "config": {
+                "model": "Function.map fan-out (BASELINE config 3)",
+                "global_batch": items_per_step,
+                "seq_len": 0,
+                "parallelism": f"map{n_gpus}",
+                "p50_remote_ms": p50_ms,
+                "per_item_gpu_op": "bf16 vector scale + readback" if has_gpu else "noop",
+                "workers": len(scheduler.pool.workers),
+            },
+        }
+        print(json.dumps(result), flush=True)
+    finally:
+        try:
+            ctx.__exit__(None, None, None)
+        except Exception:
+            pass
+        if dist is not None:
+            dist.barrier()
+            dist.destroy_process_group()
+        synchronizer.run(client.close())
+
+
+if __name__ == "__main__":
+    main()

@@ -112,6 +112,17 @@ def serialize(obj: Any) -> bytes:
     return buf.getvalue()
 
 
+def serialize_fast(obj: Any) -> bytes:
+    """Hot-path serialize: try the C pickler first (no hooks), fall back to
+    the full cloudpickle path. Safe because live handles and closures make
+    plain pickle raise (their graphs contain the scheduler/locks), which
+    routes them to the hook-aware pickler."""
+    try:
+        return pickle.dumps(obj, PICKLE_PROTOCOL)
+    except Exception:
+        return serialize(obj)
+
+
 def deserialize(data: bytes) -> Any:
     try:
         return Unpickler(io.BytesIO(data)).load()

@@ -53,7 +53,9 @@ class FunctionCallCancelledError(Error):
 
 def make_payload_item(client: Any, args: tuple, kwargs: dict) -> dict:
     """Serialize one input; offload to the CAS above the inline limit."""
-    payload = serialize(("P", (args, kwargs)))
+    from ._serialization import serialize_fast
+
+    payload = serialize_fast(("P", (args, kwargs)))
     if len(payload) > INLINE_LIMIT:
         store = client.blob_store
         if store is not None:

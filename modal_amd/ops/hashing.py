@@ -1,0 +1,110 @@
+"""Content hashing: plain SHA-256 for small payloads, GPU tree-SHA-256 above.
+
+The reference hashes every blob/Volume block with streaming SHA-256 on the
+CPU (/root/reference/py/modal/_utils/hash_utils.py:68; 8 MiB Volume blocks
+blob_utils.py:63). SHA-256 of one message is inherently serial, so the GPU
+design re-shapes the problem: a buffer is split into 64 KiB leaves, every
+leaf hashed by one CDNA4 lane (csrc/sha256.hip), and the content digest is
+SHA-256 over the concatenated leaf digests with a domain separator. The CPU
+reference implementation below produces bit-identical digests, so the CAS is
+consistent across GPU-ful and GPU-less processes, and kernel correctness is
+testable against hashlib.
+
+Small payloads bypass the GPU entirely (SURVEY.md §7 hard part 7: keep
+``.remote()`` p50 low — the crossover is config ``gpu_hash_threshold``).
+"""
+
+from __future__ import annotations
+
+import hashlib
+import struct
+from typing import Optional, Union
+
+from . import gpu_available, load_lib
+
+LEAF_SIZE = 64 * 1024  # matches the reference's 64 KiB hash streaming chunk
+TREE_DOMAIN = b"modal-amd-tree-v1"
+GPU_MIN_BYTES = 8 * 1024 * 1024  # below this, CPU wins (kernel+copy overhead)
+
+Buffer = Union[bytes, bytearray, memoryview]
+
+
+def _root_digest(total_len: int, leaf_digests: bytes) -> bytes:
+    h = hashlib.sha256()
+    h.update(TREE_DOMAIN)
+    h.update(struct.pack("<Q", total_len))
+    h.update(leaf_digests)
+    return h.digest()
+
+
+def tree_sha256_cpu(data: Buffer) -> bytes:
+    data = memoryview(data)
+    digests = bytearray()
+    for off in range(0, len(data), LEAF_SIZE):
+        digests += hashlib.sha256(data[off : off + LEAF_SIZE]).digest()
+    return _root_digest(len(data), bytes(digests))
+
+
+def _tree_sha256_gpu(data: Buffer) -> Optional[bytes]:
+    lib = load_lib(required=True)
+    import torch
+
+    n = len(data)
+    n_leaves = (n + LEAF_SIZE - 1) // LEAF_SIZE
+    src = torch.frombuffer(bytearray(data), dtype=torch.uint8).cuda()
+    offsets = torch.arange(0, n_leaves, dtype=torch.int64) * LEAF_SIZE
+    lengths = torch.full((n_leaves,), LEAF_SIZE, dtype=torch.int64)
+    if n % LEAF_SIZE:
+        lengths[-1] = n % LEAF_SIZE
+    offsets_d = offsets.cuda()
+    lengths_d = lengths.cuda()
+    out = torch.empty((n_leaves, 32), dtype=torch.uint8, device="cuda")
+    stream = torch.cuda.current_stream().cuda_stream
+    rc = lib.ma_sha256_many(
+        src.data_ptr(), offsets_d.data_ptr(), lengths_d.data_ptr(), out.data_ptr(),
+        n_leaves, stream,
+    )
+    if rc != 0:
+        raise RuntimeError(f"sha256 kernel failed: hipError {rc}")
+    torch.cuda.synchronize()
+    return _root_digest(n, out.cpu().numpy().tobytes())
+
+
+def sha256_many_gpu(buf: "object", offsets: "object", lengths: "object") -> "object":
+    """Hash arbitrary (offset, length) slices of a GPU-resident uint8 tensor.
+
+    Returns an [n, 32] uint8 CUDA tensor of standard SHA-256 digests.
+    """
+    lib = load_lib(required=True)
+    import torch
+
+    assert buf.dtype == torch.uint8 and buf.is_cuda
+    n = len(offsets)
+    offsets_d = offsets.to(device="cuda", dtype=torch.int64)
+    lengths_d = lengths.to(device="cuda", dtype=torch.int64)
+    out = torch.empty((n, 32), dtype=torch.uint8, device="cuda")
+    rc = lib.ma_sha256_many(
+        buf.data_ptr(), offsets_d.data_ptr(), lengths_d.data_ptr(), out.data_ptr(),
+        n, torch.cuda.current_stream().cuda_stream,
+    )
+    if rc != 0:
+        raise RuntimeError(f"sha256 kernel failed: hipError {rc}")
+    return out
+
+
+def tree_sha256(data: Buffer) -> bytes:
+    """GPU when it pays, CPU otherwise; identical digests either way."""
+    if len(data) >= GPU_MIN_BYTES and gpu_available():
+        return _tree_sha256_gpu(data)
+    return tree_sha256_cpu(data)
+
+
+def content_digest(data: Buffer, gpu_threshold: int = GPU_MIN_BYTES) -> str:
+    """The CAS key: plain SHA-256 below the threshold, tree digest above.
+
+    Deterministic for a given payload size, so every process computes the
+    same key regardless of whether it has a GPU.
+    """
+    if len(data) < gpu_threshold:
+        return hashlib.sha256(data).hexdigest()
+    return tree_sha256(data).hex()

@@ -303,6 +303,25 @@ class WorkerRuntime:
                         "worker missing function definition",
                     )
                 return
+            if (
+                frt.batch_max_size <= 1
+                and not frt.is_generator
+                and not frt.timeout
+                and len(msg["items"]) > 1
+            ):
+                # fast path: run the frame in max_concurrent executor chunks —
+                # the per-item thread round-trip dominates tiny map items
+                items = msg["items"]
+                n_chunks = min(frt.max_concurrent, max(1, len(items) // 4))
+                size = (len(items) + n_chunks - 1) // n_chunks
+                for start in range(0, len(items), size):
+                    chunk = items[start : start + size]
+                    task = asyncio.get_running_loop().create_task(
+                        self._run_frame_fast(frt, chunk)
+                    )
+                    for item in chunk:
+                        self._running[item["token"]] = task
+                return
             for item in msg["items"]:
                 if frt.batch_max_size > 1:
                     self._batch_add(frt, item)
@@ -380,6 +399,77 @@ class WorkerRuntime:
             finally:
                 _app_id_var.reset(app_tok)
                 _reset_current_context(ctx_tokens)
+
+    async def _run_frame_fast(self, frt: FunctionRuntime, items: list[dict]) -> None:
+        """Sequentially execute a frame of inputs in one worker thread,
+        posting all outputs in bulk (amortizes the executor hop and the
+        output-flush wakeups across the frame)."""
+        loop = asyncio.get_running_loop()
+        from .execution_context import _current_function_call_id, _current_input_id
+
+        def run_all() -> list[dict]:
+            results: list[dict] = []
+            try:
+                fn = frt.get_callable(items[0].get("method", ""))
+            except BaseException as exc:
+                err = "".join(traceback.format_exception_only(type(exc), exc)).strip()
+                data = self._serialize_exception(exc)
+                return [
+                    self._make_output(
+                        item["token"], frt.function_id, GENERIC_STATUS_FAILURE, data,
+                        DataFormat.PICKLE, err,
+                    )
+                    for item in items
+                ]
+            if inspect.iscoroutinefunction(fn) or inspect.isasyncgenfunction(fn) or inspect.isgeneratorfunction(fn):
+                return []  # signal: fall back to the general path
+            for item in items:
+                if item["token"] in self._abandoned:
+                    self._abandoned.discard(item["token"])
+                    continue
+                call_id = item["token"].rsplit(":", 2)[0]
+                tok_i = _current_input_id.set(item.get("input_id"))
+                tok_c = _current_function_call_id.set(call_id)
+                try:
+                    args, kwargs = self._decode_args(item)
+                    result = fn(*args, **kwargs)
+                    results.append(
+                        self._make_output(
+                            item["token"], frt.function_id, GENERIC_STATUS_SUCCESS,
+                            serialize(result), DataFormat.PICKLE,
+                        )
+                    )
+                except BaseException as exc:
+                    results.append(
+                        self._make_output(
+                            item["token"], frt.function_id, GENERIC_STATUS_FAILURE,
+                            self._serialize_exception(exc), DataFormat.PICKLE,
+                            "".join(traceback.format_exception_only(type(exc), exc)).strip(),
+                        )
+                    )
+                finally:
+                    _current_input_id.reset(tok_i)
+                    _current_function_call_id.reset(tok_c)
+            return results
+
+        app_tok = _app_id_var.set(frt.app_id)
+        try:
+            async with frt.sem:  # one slot per chunk-thread => <= max_concurrent items running
+                results = await loop.run_in_executor(self.executor, run_all)
+        finally:
+            _app_id_var.reset(app_tok)
+        if not results and items:
+            # not a plain sync function after all: general path per item
+            for item in items:
+                task = asyncio.get_running_loop().create_task(self._run_input(frt, item))
+                self._running[item["token"]] = task
+                task.add_done_callback(lambda _t, tok=item["token"]: self._running.pop(tok, None))
+            return
+        for item in items:
+            self._running.pop(item["token"], None)
+        self._outbox.extend(results)
+        if not self._outbox_flush_scheduled:
+            await self._flush_outbox()
 
     async def _execute(self, frt: FunctionRuntime, fn: Any, args: tuple, kwargs: dict) -> Any:
         if inspect.iscoroutinefunction(fn):
@@ -519,7 +609,7 @@ class WorkerRuntime:
             except BaseException:
                 return serialize(RuntimeError(repr(exc)))
 
-    def post_output(
+    def _make_output(
         self,
         token: str,
         function_id: str,
@@ -527,7 +617,7 @@ class WorkerRuntime:
         data: Optional[bytes],
         data_format: int,
         exc_repr: Optional[str] = None,
-    ) -> None:
+    ) -> dict:
         item: dict[str, Any] = {
             "token": token,
             "function_id": function_id,
@@ -542,6 +632,18 @@ class WorkerRuntime:
                 item["data"] = data
         if exc_repr:
             item["exc"] = exc_repr
+        return item
+
+    def post_output(
+        self,
+        token: str,
+        function_id: str,
+        status: int,
+        data: Optional[bytes],
+        data_format: int,
+        exc_repr: Optional[str] = None,
+    ) -> None:
+        item = self._make_output(token, function_id, status, data, data_format, exc_repr)
         self._outbox.append(item)
         if not self._outbox_flush_scheduled:
             self._outbox_flush_scheduled = True

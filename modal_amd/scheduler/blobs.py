@@ -24,21 +24,11 @@ BLOCK_SIZE = 8 * 1024 * 1024  # parity: volume v2 block size, blob_utils.py:63
 
 
 def _hash_bytes(data: Union[bytes, memoryview]) -> str:
-    """SHA-256 hex digest; routed to the HIP kernel for large buffers."""
-    data = bytes(data)
-    if len(data) >= 8 * 1024 * 1024:
-        try:
-            from ..ops import sha256 as gpu_sha256
+    """Content digest; routed to the HIP sha256 kernel for large buffers
+    (ops/hashing.py: plain SHA-256 below the crossover, tree digest above)."""
+    from ..ops.hashing import content_digest
 
-            digest = gpu_sha256.sha256_gpu_or_none(data)
-            if digest is not None:
-                return digest.hex()
-        except Exception:
-            pass
-    h = hashlib.sha256()
-    for off in range(0, len(data), HASH_CHUNK):
-        h.update(data[off : off + HASH_CHUNK])
-    return h.hexdigest()
+    return content_digest(bytes(data))
 
 
 class BlobStore:

@@ -35,7 +35,7 @@ if TYPE_CHECKING:
 # deep-enough per-worker pipeline so tiny functions amortize frame overhead;
 # the analog of the reference's 49-inputs-per-PutInputs batching
 # (/root/reference/py/modal/parallel_map.py:82) tuned for a local socket.
-DEFAULT_PIPELINE_DEPTH = 64
+DEFAULT_PIPELINE_DEPTH = 256
 
 
 class WorkerHandle:
