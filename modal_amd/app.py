@@ -109,7 +109,14 @@ def _build_options(
     options["_image"] = image
     options["_mounts"] = list(mounts)
     options["_schedule"] = schedule
-    for key in ("max_concurrent_inputs", "target_concurrent_inputs", "batch_max_size", "batch_linger_ms"):
+    for key in (
+        "max_concurrent_inputs",
+        "target_concurrent_inputs",
+        "batch_max_size",
+        "batch_linger_ms",
+        "cluster_size",
+        "rdma",
+    ):
         if key in flags and flags[key]:
             options[key] = flags[key]
     if flags.get("is_generator") is not None:

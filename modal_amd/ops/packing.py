@@ -57,11 +57,17 @@ def unpack_gpu(packed: "object", dst: "object", dst_offsets: "object", lengths: 
     n = len(lengths_cpu)
     if n == 0:
         return
+    # keep device copies referenced past the (async) launch: a freed temporary's
+    # block is recycled by the caching allocator before the kernel reads it
+    src_off_d = src_offsets_cpu.cuda()
+    len_d = lengths_cpu.cuda()
+    dst_off_d = dst_offsets.to(device="cuda", dtype=torch.int64)
     rc = lib.ma_unpack_segments(
-        packed.data_ptr(), src_offsets_cpu.cuda().data_ptr(), lengths_cpu.cuda().data_ptr(),
-        dst.data_ptr(), dst_offsets.to(device="cuda", dtype=torch.int64).data_ptr(),
+        packed.data_ptr(), src_off_d.data_ptr(), len_d.data_ptr(),
+        dst.data_ptr(), dst_off_d.data_ptr(),
         n, int(lengths_cpu.max().item()),
         torch.cuda.current_stream().cuda_stream,
     )
+    _ = (src_off_d, len_d, dst_off_d)  # alive past the launch; stream order covers the rest
     if rc != 0:
         raise RuntimeError(f"unpack kernel failed: hipError {rc}")

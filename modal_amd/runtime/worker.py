@@ -359,6 +359,9 @@ class WorkerRuntime:
             ctx_tokens = _set_current_context(item.get("input_id"), call_id)
             app_tok = _app_id_var.set(frt.app_id)
             started = time.monotonic()
+            cluster = item.get("cluster")
+            if cluster is not None:
+                self._setup_cluster(cluster)
             try:
                 fn = frt.get_callable(item.get("method", ""))
                 args, kwargs = self._decode_args(item)
@@ -397,8 +400,33 @@ class WorkerRuntime:
                     "".join(traceback.format_exception_only(type(exc), exc)).strip(),
                 )
             finally:
+                if cluster is not None:
+                    from ..experimental import _set_cluster_info
+
+                    _set_cluster_info(None)
                 _app_id_var.reset(app_tok)
                 _reset_current_context(ctx_tokens)
+
+    def _setup_cluster(self, cluster: dict) -> None:
+        """Rank/world bootstrap for @clustered gangs: env for
+        torch.distributed env:// rendezvous (RCCL over xGMI on GPU workers)
+        plus ClusterInfo (parity: reference _clustered_functions.py:42-94)."""
+        from ..experimental import ClusterInfo, _set_cluster_info
+
+        rank, size = cluster["rank"], cluster["size"]
+        os.environ["RANK"] = str(rank)
+        os.environ["WORLD_SIZE"] = str(size)
+        os.environ["LOCAL_RANK"] = "0"  # each worker sees exactly its own GPU
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(cluster["master_port"])
+        _set_cluster_info(
+            ClusterInfo(
+                rank=rank,
+                cluster_id=cluster["cluster_id"],
+                container_ips=["127.0.0.1"] * size,
+                fabric_ids=[0] * size,  # one xGMI hive on a single node
+            )
+        )
 
     async def _run_frame_fast(self, frt: FunctionRuntime, items: list[dict]) -> None:
         """Sequentially execute a frame of inputs in one worker thread,

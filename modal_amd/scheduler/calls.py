@@ -137,6 +137,7 @@ class InputRecord:
     finished_at: float = 0.0
     tensors: Optional[list] = None  # tensor sidecar (CUDA-IPC / pinned staging)
     cancelled: bool = False
+    cluster: Optional[dict] = None  # gang identity: {rank, size, cluster_id, master_port}
 
     @property
     def token(self) -> str:

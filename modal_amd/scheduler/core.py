@@ -323,10 +323,15 @@ class Scheduler:
     async def function_put_inputs(self, function_call_id: str, items: list) -> list:
         """items: [{"payload": bytes, "method": str}] -> [{"idx", "input_id"}]."""
         record = self._call(function_call_id)
+        fdef = self.functions.get(record.function_id)
         out = []
         for item in items:
             if isinstance(item, (bytes, bytearray)):
                 item = {"payload": bytes(item)}
+            if fdef is not None and fdef.cluster_size > 1:
+                recs = await self._put_gang_input(record, fdef, item)
+                out.append({"idx": recs[0].idx, "input_id": recs[0].input_id})
+                continue
             rec = record.add_input(
                 item.get("payload") or b"",
                 item.get("method", ""),
@@ -336,6 +341,37 @@ class Scheduler:
             self.pool.enqueue(rec)
             out.append({"idx": rec.idx, "input_id": rec.input_id})
         return out
+
+    async def _put_gang_input(self, record: CallRecord, fdef: FunctionDef, item: dict) -> list:
+        """Gang scheduling for @clustered(size=n): the same input lands on n
+        distinct workers simultaneously, each with rank/world identity; the
+        rank-0 output is the call's result (parity: reference clustered
+        semantics, _clustered_functions.py:42-94)."""
+        import socket
+
+        n = fdef.cluster_size
+        cluster_id = new_id("task")
+        # pre-allocate the rendezvous port for torch.distributed/RCCL
+        sock = socket.socket()
+        sock.bind(("127.0.0.1", 0))
+        port = sock.getsockname()[1]
+        sock.close()
+        recs = []
+        for rank in range(n):
+            rec = record.add_input(
+                item.get("payload") or b"",
+                item.get("method", ""),
+                payload_blob=item.get("payload_blob"),
+            )
+            rec.cluster = {
+                "rank": rank,
+                "size": n,
+                "cluster_id": cluster_id,
+                "master_port": port,
+            }
+            recs.append(rec)
+        await self.pool.dispatch_gang(fdef, recs)
+        return recs
 
     async def function_finish_inputs(self, function_call_id: str) -> None:
         self._call(function_call_id).finish_inputs()

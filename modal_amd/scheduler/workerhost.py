@@ -442,6 +442,8 @@ class WorkerPool:
                 }
                 if rec.payload_blob:
                     item["payload_blob"] = rec.payload_blob
+                if rec.cluster:
+                    item["cluster"] = rec.cluster
                 items.append(item)
             w.outstanding[fdef.function_id] = w.outstanding.get(fdef.function_id, 0) + len(batch)
             await w.conn.send({"t": "inputs", "function_id": fdef.function_id, "items": items})
@@ -452,6 +454,32 @@ class WorkerPool:
                     del w.inflight[rec.token]
                 if not rec.final:
                     self.enqueue(rec, front=True)
+
+    async def dispatch_gang(self, fdef: FunctionDef, recs: list) -> None:
+        """Place a gang on len(recs) distinct workers at once, bypassing the
+        credit queue (gang members block on each other, so partial placement
+        would deadlock)."""
+        n = len(recs)
+        deadline = time.time() + 120
+        while True:
+            await self.ensure_workers(fdef.needs_gpu)
+            candidates = [
+                w
+                for w in self.workers.values()
+                if w.alive and not w.draining and (w.has_gpu or not fdef.needs_gpu)
+            ]
+            if fdef.needs_gpu:
+                candidates = [w for w in candidates if w.has_gpu]
+            if len(candidates) >= n:
+                break
+            if time.time() > deadline:
+                raise TimeoutError(
+                    f"Gang of {n} workers unavailable (have {len(candidates)})"
+                )
+            await asyncio.sleep(0.05)
+        candidates.sort(key=lambda w: (sum(w.outstanding.values()), w.worker_id))
+        for rec, w in zip(recs, candidates):
+            await self._send_batch(w, fdef, [rec])
 
     async def cancel_inputs(self, tokens: list[str], terminate: bool = False) -> None:
         """Propagate cancellation to workers holding these inputs
