@@ -14,9 +14,11 @@ if sys.version_info[:2] < (3, 10):
 
 __version__ = "0.1.0"
 
-from . import exception
+from . import exception, experimental
+from ._tunnel import Tunnel, forward
 from .app import App
 from .client import Client
+from .cloud_bucket_mount import CloudBucketMount
 from .cls import Cls, parameter
 from .config import config
 from .dict import Dict
@@ -35,8 +37,16 @@ from .partial_function import (
     web_server,
     wsgi_app,
 )
+from .image import Image
+from .mount import Mount
+from .network_file_system import NetworkFileSystem
+from .proxy import Proxy
 from .queue import Queue
 from .retries import Retries
+from .sandbox import ContainerProcess, FileIO, Probe, Sandbox
+from .scheduler_placement import SchedulerPlacement
+from .snapshot import SandboxSnapshot
+from .volume import FileEntry, Volume
 from .runtime.execution_context import (
     current_function_call_id,
     current_input_id,
@@ -49,16 +59,32 @@ from .secret import Secret
 __all__ = [
     "App",
     "Client",
+    "CloudBucketMount",
     "Cls",
+    "ContainerProcess",
     "Cron",
     "Dict",
     "Error",
+    "FileEntry",
+    "FileIO",
     "Function",
     "FunctionCall",
+    "Image",
+    "Mount",
+    "NetworkFileSystem",
     "Period",
+    "Probe",
+    "Proxy",
     "Queue",
     "Retries",
+    "Sandbox",
+    "SandboxSnapshot",
+    "SchedulerPlacement",
     "Secret",
+    "Tunnel",
+    "Volume",
+    "experimental",
+    "forward",
     "asgi_app",
     "batched",
     "concurrent",

@@ -1,0 +1,41 @@
+"""NetworkFileSystem (legacy): aliased onto the Volume service.
+
+Parity: /root/reference/py/modal/network_file_system.py:361 — the reference
+keeps NFS for backward compatibility; SURVEY.md §2 row 30 allows aliasing it
+to Volume. Same API names (from_name/ephemeral/write_file/read_file/listdir/
+remove_file) over the volume backend.
+"""
+
+from __future__ import annotations
+
+from typing import Any, AsyncGenerator, BinaryIO, Optional
+
+from ._object import _Object, live_method
+from ._sync import synchronize_api
+from .volume import FileEntry, _Volume
+
+
+class _NetworkFileSystem(_Volume, type_kind="nfs"):
+    @classmethod
+    def from_name(
+        cls, name: str, *, environment_name: str = "", create_if_missing: bool = False
+    ) -> "_NetworkFileSystem":
+        async def _load(obj: "_NetworkFileSystem", resolver: Any, existing: Any) -> None:
+            resp = await resolver.client.svc.volume_get_or_create(
+                name=f"nfs/{name}",
+                environment=environment_name or "main",
+                create_if_missing=create_if_missing,
+                ephemeral=False,
+            )
+            obj._hydrate(resp["volume_id"], resolver.client, None)
+
+        return cls._from_loader(_load, rep=f"NetworkFileSystem.from_name({name!r})")
+
+    @live_method
+    async def write_file(self, remote_path: str, fp: BinaryIO) -> int:
+        data = fp.read()
+        await self._put_data(data, remote_path)
+        return len(data)
+
+
+NetworkFileSystem = synchronize_api(_NetworkFileSystem, "NetworkFileSystem")

@@ -73,6 +73,7 @@ class FunctionRuntime:
         self.definition_kind = msg.get("definition_kind", "serialized")
         self.env = msg.get("env") or {}
         self.volumes = msg.get("volumes") or {}
+        self.python_paths = msg.get("python_paths") or []
         self.sem = asyncio.Semaphore(self.max_concurrent)
         self._callable: Any = None
         self._service: Any = None
@@ -91,7 +92,10 @@ class FunctionRuntime:
             return self._callable
         try:
             if self.env:
-                os.environ.update(self.env)  # secret env bundles
+                os.environ.update(self.env)  # image + secret env bundles
+            for p in self.python_paths:
+                if p not in sys.path:
+                    sys.path.insert(0, p)  # image site-packages
             if self.volumes:
                 from .volumes import mount_volumes
 

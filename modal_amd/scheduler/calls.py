@@ -98,6 +98,7 @@ class FunctionDef:
     secret_ids: list = field(default_factory=list)
     volume_mounts: dict = field(default_factory=dict)  # mount path -> volume id
     schedule: Optional[dict] = None  # {"cron": "..."} | {"period": seconds}
+    image_id: Optional[str] = None
 
     def placement_tag(self) -> str:
         return "gpu" if self.needs_gpu else "any"

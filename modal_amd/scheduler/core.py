@@ -70,12 +70,13 @@ class RPCAdapter:
         "function_get_outputs", "function_call_cancel", "function_call_info",
         "function_get_current_stats", "generator_poll",
         "app_lookup", "app_get_layout", "cluster_hello",
-        "volume_get_or_create", "volume_commit_files", "volume_get_file", "volume_list_files",
-        "volume_remove_file", "volume_copy_files", "volume_reload", "volume_delete", "volume_rename",
+        "volume_get_or_create", "volume_put_file_blocks", "volume_get_file", "volume_list_files",
+        "volume_remove_file", "volume_copy_files", "volume_commit", "volume_reload",
+        "volume_delete", "volume_rename", "volume_dir",
         "sandbox_create", "sandbox_wait", "sandbox_terminate", "sandbox_poll", "sandbox_stdio_read",
-        "sandbox_stdin_write", "sandbox_exec", "sandbox_exec_wait", "sandbox_exec_poll",
-        "sandbox_list", "sandbox_tag", "sandbox_from_name", "sandbox_snapshot_fs",
-        "image_get_or_create", "image_info",
+        "sandbox_stdin_write", "sandbox_exec", "sandbox_list", "sandbox_set_tags",
+        "sandbox_from_name", "sandbox_snapshot_fs", "sandbox_fs_op",
+        "image_get_or_create", "image_info", "mount_get_or_create",
     }
 
     def __init__(self, scheduler: "Scheduler"):
@@ -100,6 +101,16 @@ class Scheduler:
         self.calls: dict[str, CallRecord] = {}
         self.services = Services()
         self.blob_store = BlobStore(os.path.join(self.run_dir, "blobs"))
+        from .sandboxes import SandboxService
+        from .volumes import VolumeService
+
+        self.sandbox_service = SandboxService(self.run_dir)
+        self.volume_service = VolumeService(self.run_dir, self.blob_store)
+        from .images import ImageService
+
+        self.image_service = ImageService(self.run_dir, self.blob_store)
+        self.mounts: dict[str, str] = {}  # mount_id -> materialized dir
+        self._mounts_by_hash: dict[str, str] = {}
         self.pool = WorkerPool(self)
         self.rpc_adapter = RPCAdapter(self)
         self._started = False
@@ -118,6 +129,7 @@ class Scheduler:
     async def stop(self) -> None:
         if not self._started:
             return
+        await self.sandbox_service.shutdown()
         await self.pool.stop()
         self._started = False
 
@@ -249,6 +261,7 @@ class Scheduler:
             secret_ids=list(options.get("secret_ids") or []),
             volume_mounts=dict(options.get("volume_mounts") or {}),
             schedule=options.get("schedule"),
+            image_id=options.get("image_id"),
         )
         self.functions[fid] = fdef
         app = self.apps.get(app_id)
@@ -556,15 +569,26 @@ class Scheduler:
                         sub.put_nowait(entry)
 
     def resolve_function_env(self, fdef: FunctionDef) -> dict[str, str]:
-        """Merge the env bundles of a function's secrets (later wins;
-        parity: secrets applied to container env, reference secret.py)."""
+        """Merge image env then secret env bundles (secrets win; parity:
+        image env + secrets applied to container env)."""
         env: dict[str, str] = {}
+        if fdef.image_id:
+            state = self.image_service.by_id.get(fdef.image_id)
+            if state is not None:
+                env.update(state.env)
         for secret_id in fdef.secret_ids:
             try:
                 env.update(self.services.secret_env(secret_id))
             except Exception:
                 pass
         return env
+
+    def resolve_function_pythonpaths(self, fdef: FunctionDef) -> list[str]:
+        if fdef.image_id:
+            state = self.image_service.by_id.get(fdef.image_id)
+            if state is not None:
+                return list(state.python_paths)
+        return []
 
     # -- cluster rendezvous ------------------------------------------------
     async def cluster_hello(self, cluster_id: str, rank: int, world_size: int, addr: str = "") -> dict:
@@ -601,6 +625,164 @@ class Scheduler:
             return {"payload": payload}
         digest = self.blob_store.put(payload)
         return {"payload_blob": digest}
+
+    # -- volumes -----------------------------------------------------------
+    async def volume_get_or_create(self, name=None, environment="main", create_if_missing=False, ephemeral=False) -> dict:
+        return await self.volume_service.get_or_create(name, environment, create_if_missing, ephemeral)
+
+    async def volume_put_file_blocks(self, volume_id, rel_path, block_digests, size, mode=0o644) -> dict:
+        return await self.volume_service.put_file_blocks(volume_id, rel_path, block_digests, size, mode)
+
+    async def volume_get_file(self, volume_id, rel_path, offset=0, n_bytes=-1) -> bytes:
+        return await self.volume_service.get_file(volume_id, rel_path, offset, n_bytes)
+
+    async def volume_list_files(self, volume_id, rel_path="/", recursive=True) -> list:
+        return await self.volume_service.list_files(volume_id, rel_path, recursive)
+
+    async def volume_remove_file(self, volume_id, rel_path, recursive=False) -> None:
+        return await self.volume_service.remove_file(volume_id, rel_path, recursive)
+
+    async def volume_copy_files(self, volume_id, src_paths, dst_path) -> None:
+        return await self.volume_service.copy_files(volume_id, src_paths, dst_path)
+
+    async def volume_commit(self, volume_id) -> dict:
+        return await self.volume_service.commit(volume_id)
+
+    async def volume_reload(self, volume_id) -> None:
+        return await self.volume_service.reload(volume_id)
+
+    async def volume_delete(self, volume_id) -> None:
+        return await self.volume_service.delete(volume_id)
+
+    async def volume_rename(self, volume_id, new_name, environment="main") -> None:
+        return await self.volume_service.rename(volume_id, new_name, environment)
+
+    async def volume_dir(self, volume_id) -> str:
+        return self.volume_service.volume_dir(volume_id)
+
+    # -- sandboxes ---------------------------------------------------------
+    async def sandbox_create(self, **kwargs: Any) -> dict:
+        volume_mounts = kwargs.pop("volume_mounts", None) or {}
+        volume_paths = {
+            path: self.volume_service.volume_dir(vid) for path, vid in volume_mounts.items()
+        }
+        return await self.sandbox_service.create(volume_paths=volume_paths, **kwargs)
+
+    async def sandbox_exec(self, sandbox_id, cmd, env=None, workdir=None, timeout=None, exec_id=None) -> dict:
+        return await self.sandbox_service.exec(sandbox_id, cmd, env, workdir, timeout, exec_id)
+
+    async def sandbox_stdio_read(self, target_id, fd, offset=0, max_bytes=1 << 20, timeout=55.0) -> dict:
+        return await self.sandbox_service.stdio_read(target_id, fd, offset, max_bytes, timeout)
+
+    async def sandbox_stdin_write(self, target_id, offset, data, eof=False) -> int:
+        return await self.sandbox_service.stdin_write(target_id, offset, data, eof)
+
+    async def sandbox_wait(self, target_id, timeout=None, raise_on_timeout=True) -> dict:
+        return await self.sandbox_service.wait(target_id, timeout, raise_on_timeout)
+
+    async def sandbox_poll(self, target_id) -> dict:
+        return await self.sandbox_service.poll(target_id)
+
+    async def sandbox_terminate(self, sandbox_id) -> None:
+        return await self.sandbox_service.terminate(sandbox_id)
+
+    async def sandbox_list(self, app_id=None, tags=None) -> list:
+        return await self.sandbox_service.list(app_id, tags)
+
+    async def sandbox_set_tags(self, sandbox_id, tags) -> None:
+        return await self.sandbox_service.set_tags(sandbox_id, tags)
+
+    async def sandbox_from_name(self, name, environment="main") -> dict:
+        sid = await self.sandbox_service.from_name(name, environment)
+        return {"sandbox_id": sid}
+
+    async def sandbox_snapshot_fs(self, sandbox_id) -> dict:
+        return await self.sandbox_service.snapshot_fs(sandbox_id, self.blob_store)
+
+    async def sandbox_fs_op(self, sandbox_id: str, op: str, path: str = "", **kwargs: Any) -> Any:
+        """Typed remote-FS operations inside a sandbox workdir (parity:
+        reference sandbox filesystem API, sandbox_fs.py:68, file_io.py:135)."""
+        sb = self.sandbox_service._get(sandbox_id)
+        base = sb.workdir
+        full = os.path.normpath(os.path.join(base, path.lstrip("/"))) if path else base
+        loop = asyncio.get_running_loop()
+        if op == "read":
+            def _read() -> bytes:
+                with open(full, "rb") as f:
+                    f.seek(kwargs.get("offset", 0))
+                    return f.read(kwargs.get("n", -1))
+
+            return await loop.run_in_executor(None, _read)
+        if op == "write":
+            def _write() -> int:
+                mode = "ab" if kwargs.get("append") else ("r+b" if kwargs.get("offset") else "wb")
+                if kwargs.get("offset") and not os.path.exists(full):
+                    open(full, "wb").close()
+                with open(full, mode) as f:
+                    if kwargs.get("offset"):
+                        f.seek(kwargs["offset"])
+                    return f.write(kwargs.get("data", b""))
+
+            return await loop.run_in_executor(None, _write)
+        if op == "ls":
+            return sorted(os.listdir(full))
+        if op == "stat":
+            st = os.stat(full)
+            return {"size": st.st_size, "mtime": st.st_mtime, "is_dir": os.path.isdir(full)}
+        if op == "mkdir":
+            os.makedirs(full, exist_ok=kwargs.get("parents", False) or kwargs.get("exist_ok", False))
+            return None
+        if op == "rm":
+            if os.path.isdir(full) and not os.path.islink(full):
+                if kwargs.get("recursive"):
+                    import shutil
+
+                    shutil.rmtree(full)
+                else:
+                    os.rmdir(full)
+            else:
+                os.unlink(full)
+            return None
+        if op == "exists":
+            return os.path.exists(full)
+        raise InvalidError(f"Unknown fs op {op!r}")
+
+    # -- images ------------------------------------------------------------
+    async def image_get_or_create(self, recipe: list) -> dict:
+        return await self.image_service.get_or_create(recipe)
+
+    async def image_info(self, image_id: str) -> dict:
+        return await self.image_service.info(image_id)
+
+    # -- mounts ------------------------------------------------------------
+    async def mount_get_or_create(self, manifest: list) -> dict:
+        """manifest: [[remote_path, blob_id, mode], ...] — content-addressed
+        dedup like the reference's MountGetOrCreate (api.proto:4838-4840);
+        files materialize as hard links out of the CAS."""
+        import hashlib
+        import json as _json
+
+        key = hashlib.sha256(
+            _json.dumps(sorted(manifest), sort_keys=True).encode()
+        ).hexdigest()
+        existing = self._mounts_by_hash.get(key)
+        if existing:
+            return {"mount_id": existing, "dir": self.mounts[existing]}
+        mount_id = new_id("mount")
+        root = os.path.join(self.run_dir, "mounts", mount_id)
+        for remote_path, blob_id, _mode in manifest:
+            dest = os.path.join(root, remote_path.lstrip("/"))
+            os.makedirs(os.path.dirname(dest), exist_ok=True)
+            src = self.blob_store.open_path(blob_id)
+            try:
+                os.link(src, dest)
+            except OSError:
+                import shutil
+
+                shutil.copyfile(src, dest)
+        self.mounts[mount_id] = root
+        self._mounts_by_hash[key] = mount_id
+        return {"mount_id": mount_id, "dir": root}
 
     # -- service passthrough (queues/dicts/secrets) -------------------------
     def __getattr__(self, name: str) -> Any:

@@ -1,0 +1,355 @@
+"""Sandbox service: on-demand supervised processes with exec/stdio/FS.
+
+Single-node re-implementation of the reference's Sandbox control plane +
+task command router (/root/reference/py/modal/sandbox.py:370,
+task_command_router_client.py:211). A sandbox is a process group rooted in a
+private workdir under the run dir; ``exec`` spawns additional processes in
+the same workdir/env (the container-exec analog). Stdio is captured into
+offset-addressable buffers so reads are resumable from any byte offset —
+the same contract as the reference's offset-resumable stdio streams
+(task_command_router_client.py:31,523-614).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import os
+import signal
+import time
+from typing import Any, Optional
+
+from ..exception import NotFoundError, SandboxTimeoutError
+from ..utils.ids import new_id
+
+STDIN_CHUNK = 256 * 1024  # parity: reference task_command_router_client.py:31
+
+
+class _ProcState:
+    """One supervised process: the sandbox entrypoint or an exec."""
+
+    def __init__(self, proc_id: str):
+        self.proc_id = proc_id
+        self.proc: Optional[asyncio.subprocess.Process] = None
+        self.stdout = bytearray()
+        self.stderr = bytearray()
+        self.stdout_eof = False
+        self.stderr_eof = False
+        self.cond = asyncio.Condition()
+        self.returncode: Optional[int] = None
+        self.stdin_offset = 0  # bytes accepted so far (resume dedup)
+        self._readers: list[asyncio.Task] = []
+
+    async def attach(self, proc: asyncio.subprocess.Process) -> None:
+        self.proc = proc
+        loop = asyncio.get_running_loop()
+        if proc.stdout is not None:
+            self._readers.append(loop.create_task(self._pump(proc.stdout, 1)))
+        if proc.stderr is not None:
+            self._readers.append(loop.create_task(self._pump(proc.stderr, 2)))
+        self._readers.append(loop.create_task(self._wait()))
+
+    async def _pump(self, stream: asyncio.StreamReader, fd: int) -> None:
+        buf = self.stdout if fd == 1 else self.stderr
+        while True:
+            chunk = await stream.read(64 * 1024)
+            async with self.cond:
+                if not chunk:
+                    if fd == 1:
+                        self.stdout_eof = True
+                    else:
+                        self.stderr_eof = True
+                    self.cond.notify_all()
+                    return
+                buf.extend(chunk)
+                self.cond.notify_all()
+
+    async def _wait(self) -> None:
+        rc = await self.proc.wait()
+        async with self.cond:
+            self.returncode = rc
+            self.stdout_eof = True
+            self.stderr_eof = True
+            self.cond.notify_all()
+
+    async def read(self, fd: int, offset: int, max_bytes: int, timeout: float) -> dict:
+        deadline = time.monotonic() + timeout
+        async with self.cond:
+            while True:
+                buf = self.stdout if fd == 1 else self.stderr
+                eof = self.stdout_eof if fd == 1 else self.stderr_eof
+                if offset < len(buf):
+                    data = bytes(buf[offset : offset + max_bytes])
+                    return {"data": data, "eof": eof and offset + len(data) >= len(buf),
+                            "next_offset": offset + len(data)}
+                if eof:
+                    return {"data": b"", "eof": True, "next_offset": offset}
+                remaining = deadline - time.monotonic()
+                if remaining <= 0:
+                    return {"data": b"", "eof": False, "next_offset": offset}
+                try:
+                    await asyncio.wait_for(self.cond.wait(), remaining)
+                except asyncio.TimeoutError:
+                    return {"data": b"", "eof": False, "next_offset": offset}
+
+    async def write_stdin(self, offset: int, data: bytes, eof: bool) -> int:
+        """Offset-resumable stdin: bytes before stdin_offset are dedup'd
+        (parity: resumable stdin stream, reference :523-614)."""
+        if self.proc is None or self.proc.stdin is None:
+            raise NotFoundError("stdin not available")
+        if offset < self.stdin_offset:
+            data = data[self.stdin_offset - offset :]
+        if data:
+            self.proc.stdin.write(data)
+            self.stdin_offset += len(data)
+            await self.proc.stdin.drain()
+        if eof:
+            self.proc.stdin.close()
+        return self.stdin_offset
+
+    async def wait(self, timeout: Optional[float] = None) -> int:
+        async with self.cond:
+            deadline = None if timeout is None else time.monotonic() + timeout
+            while self.returncode is None:
+                remaining = None if deadline is None else deadline - time.monotonic()
+                if remaining is not None and remaining <= 0:
+                    raise asyncio.TimeoutError
+                await asyncio.wait_for(self.cond.wait(), remaining)
+            return self.returncode
+
+    def kill(self, sig: int = signal.SIGKILL) -> None:
+        if self.proc is not None and self.returncode is None:
+            try:
+                os.killpg(self.proc.pid, sig)
+            except (ProcessLookupError, PermissionError):
+                try:
+                    self.proc.kill()
+                except ProcessLookupError:
+                    pass
+
+
+class SandboxState:
+    def __init__(self, sandbox_id: str, workdir: str, env: dict, app_id: str):
+        self.sandbox_id = sandbox_id
+        self.task_id = new_id("task")
+        self.workdir = workdir
+        self.env = env
+        self.app_id = app_id
+        self.main = _ProcState(sandbox_id)
+        self.execs: dict[str, _ProcState] = {}
+        self.created_at = time.time()
+        self.timeout: Optional[float] = None
+        self.timeout_task: Optional[asyncio.Task] = None
+        self.timed_out = False
+        self.name: Optional[str] = None
+        self.tags: dict[str, str] = {}
+        self.gpu_index: Optional[int] = None
+
+
+class SandboxService:
+    def __init__(self, run_dir: str):
+        self.run_dir = run_dir
+        self.root = os.path.join(run_dir, "sandboxes")
+        os.makedirs(self.root, exist_ok=True)
+        self.sandboxes: dict[str, SandboxState] = {}
+        self.by_name: dict[tuple[str, str], str] = {}
+
+    def _get(self, sandbox_id: str) -> SandboxState:
+        sb = self.sandboxes.get(sandbox_id)
+        if sb is None:
+            raise NotFoundError(f"Sandbox {sandbox_id} not found")
+        return sb
+
+    def _proc(self, target_id: str) -> _ProcState:
+        if target_id in self.sandboxes:
+            return self.sandboxes[target_id].main
+        for sb in self.sandboxes.values():
+            if target_id in sb.execs:
+                return sb.execs[target_id]
+        raise NotFoundError(f"No process {target_id}")
+
+    async def create(
+        self,
+        entrypoint_args: list[str],
+        env: Optional[dict] = None,
+        workdir: Optional[str] = None,
+        timeout: Optional[float] = None,
+        gpu: Optional[int] = None,
+        app_id: str = "",
+        name: Optional[str] = None,
+        environment: str = "main",
+        volume_paths: Optional[dict] = None,
+        cpu: Optional[float] = None,
+        memory: Optional[int] = None,
+    ) -> dict:
+        sandbox_id = new_id("sandbox")
+        sb_dir = os.path.join(self.root, sandbox_id)
+        os.makedirs(sb_dir, exist_ok=True)
+        full_env = dict(os.environ)
+        full_env.update(env or {})
+        full_env["MODAL_AMD_SANDBOX_ID"] = sandbox_id
+        if gpu is not None:
+            full_env["HIP_VISIBLE_DEVICES"] = str(gpu)
+            full_env["CUDA_VISIBLE_DEVICES"] = str(gpu)
+        # volume mounts: symlink shared trees into the sandbox workdir space
+        for mount_path, vol_dir in (volume_paths or {}).items():
+            os.makedirs(vol_dir, exist_ok=True)
+            link = mount_path if os.path.isabs(mount_path) else os.path.join(sb_dir, mount_path)
+            try:
+                os.makedirs(os.path.dirname(link), exist_ok=True)
+                if not os.path.exists(link):
+                    os.symlink(vol_dir, link)
+            except OSError:
+                pass
+        state = SandboxState(sandbox_id, workdir or sb_dir, full_env, app_id)
+        state.gpu_index = gpu
+        if name:
+            self.by_name[(environment, name)] = sandbox_id
+            state.name = name
+        self.sandboxes[sandbox_id] = state
+        args = entrypoint_args or ["sleep", "infinity"]
+        proc = await asyncio.create_subprocess_exec(
+            *args,
+            cwd=state.workdir,
+            env=full_env,
+            stdin=asyncio.subprocess.PIPE,
+            stdout=asyncio.subprocess.PIPE,
+            stderr=asyncio.subprocess.PIPE,
+            start_new_session=True,
+        )
+        await state.main.attach(proc)
+        if timeout:
+            state.timeout = timeout
+
+            async def _timeout_kill() -> None:
+                await asyncio.sleep(timeout)
+                state.timed_out = True
+                state.main.kill()
+
+            state.timeout_task = asyncio.get_running_loop().create_task(_timeout_kill())
+        return {"sandbox_id": sandbox_id, "task_id": state.task_id}
+
+    async def exec(
+        self,
+        sandbox_id: str,
+        cmd: list[str],
+        env: Optional[dict] = None,
+        workdir: Optional[str] = None,
+        timeout: Optional[float] = None,
+        exec_id: Optional[str] = None,
+    ) -> dict:
+        sb = self._get(sandbox_id)
+        # exec_id idempotency (parity: command-router exec_id semantics)
+        if exec_id and exec_id in sb.execs:
+            return {"exec_id": exec_id}
+        exec_id = exec_id or new_id("task")
+        state = _ProcState(exec_id)
+        sb.execs[exec_id] = state
+        full_env = dict(sb.env)
+        full_env.update(env or {})
+        proc = await asyncio.create_subprocess_exec(
+            *cmd,
+            cwd=workdir or sb.workdir,
+            env=full_env,
+            stdin=asyncio.subprocess.PIPE,
+            stdout=asyncio.subprocess.PIPE,
+            stderr=asyncio.subprocess.PIPE,
+            start_new_session=True,
+        )
+        await state.attach(proc)
+        if timeout:
+
+            async def _timeout_kill() -> None:
+                await asyncio.sleep(timeout)
+                state.kill()
+
+            asyncio.get_running_loop().create_task(_timeout_kill())
+        return {"exec_id": exec_id}
+
+    async def stdio_read(
+        self, target_id: str, fd: int, offset: int, max_bytes: int = 1 << 20, timeout: float = 55.0
+    ) -> dict:
+        return await self._proc(target_id).read(fd, offset, max_bytes, timeout)
+
+    async def stdin_write(self, target_id: str, offset: int, data: bytes, eof: bool = False) -> int:
+        return await self._proc(target_id).write_stdin(offset, data, eof)
+
+    async def wait(self, target_id: str, timeout: Optional[float] = None, raise_on_timeout: bool = True) -> dict:
+        state = self._proc(target_id)
+        sb = self.sandboxes.get(target_id)
+        try:
+            rc = await state.wait(timeout)
+        except asyncio.TimeoutError:
+            if raise_on_timeout:
+                raise SandboxTimeoutError(f"{target_id} still running after {timeout}s") from None
+            return {"returncode": None, "running": True}
+        timed_out = bool(sb.timed_out) if sb is not None else False
+        return {"returncode": rc, "running": False, "timed_out": timed_out}
+
+    async def poll(self, target_id: str) -> dict:
+        state = self._proc(target_id)
+        sb = self.sandboxes.get(target_id)
+        return {
+            "returncode": state.returncode,
+            "running": state.returncode is None,
+            "timed_out": bool(sb.timed_out) if sb is not None else False,
+        }
+
+    async def terminate(self, sandbox_id: str) -> None:
+        sb = self._get(sandbox_id)
+        if sb.timeout_task is not None:
+            sb.timeout_task.cancel()
+        for ex in sb.execs.values():
+            ex.kill()
+        sb.main.kill()
+
+    async def list(self, app_id: Optional[str] = None, tags: Optional[dict] = None) -> list[dict]:
+        out = []
+        for sb in self.sandboxes.values():
+            if app_id and sb.app_id != app_id:
+                continue
+            if tags and any(sb.tags.get(k) != v for k, v in tags.items()):
+                continue
+            out.append(
+                {
+                    "sandbox_id": sb.sandbox_id,
+                    "task_id": sb.task_id,
+                    "created_at": sb.created_at,
+                    "returncode": sb.main.returncode,
+                    "name": sb.name,
+                    "tags": sb.tags,
+                }
+            )
+        return out
+
+    async def set_tags(self, sandbox_id: str, tags: dict) -> None:
+        self._get(sandbox_id).tags.update(tags)
+
+    async def from_name(self, name: str, environment: str = "main") -> str:
+        sid = self.by_name.get((environment, name))
+        if sid is None:
+            raise NotFoundError(f"Sandbox '{name}' not found")
+        return sid
+
+    async def snapshot_fs(self, sandbox_id: str, blob_store: Any) -> dict:
+        """Tar the sandbox workdir into the CAS (parity: SandboxSnapshotFs)."""
+        sb = self._get(sandbox_id)
+        import io
+        import tarfile
+
+        buf = io.BytesIO()
+        with tarfile.open(fileobj=buf, mode="w:gz") as tar:
+            tar.add(sb.workdir, arcname=".")
+        digest = blob_store.put(buf.getvalue())
+        return {"image_id": new_id("image"), "blob_id": digest}
+
+    async def shutdown(self) -> None:
+        for sb in list(self.sandboxes.values()):
+            try:
+                await self.terminate(sb.sandbox_id)
+            except Exception:
+                pass
+
+    def cleanup_dirs(self) -> None:
+        import shutil
+
+        shutil.rmtree(self.root, ignore_errors=True)

@@ -424,6 +424,7 @@ class WorkerPool:
                         "app_layout": self.scheduler.app_layout(fdef.app_id),
                         "env": self.scheduler.resolve_function_env(fdef),
                         "volumes": fdef.volume_mounts,
+                        "python_paths": self.scheduler.resolve_function_pythonpaths(fdef),
                     }
                 )
                 w.functions_loaded.add(fdef.function_id)
