@@ -154,18 +154,13 @@ class _ClsService:
         return self.instance
 
     def get_method(self, method_name: str) -> Any:
+        import functools
+
         inst = self._ensure_instance()
-        raw = getattr(type(inst), method_name)
+        raw = inspect.getattr_static(type(inst), method_name)
         raw_fn = getattr(raw, "raw_f", None) or raw
-        def bound(*args: Any, **kwargs: Any) -> Any:
-            return raw_fn(inst, *args, **kwargs)
-        bound.__name__ = method_name
-        if inspect.iscoroutinefunction(raw_fn):
-            async def abound(*args: Any, **kwargs: Any) -> Any:
-                return await raw_fn(inst, *args, **kwargs)
-            abound.__name__ = method_name
-            return abound
-        return bound
+        # functools.partial preserves generator/coroutine detection in inspect
+        return functools.partial(raw_fn, inst)
 
     def exit(self) -> None:
         if self.instance is not None and self._entered:
@@ -342,6 +337,14 @@ class WorkerRuntime:
                     self._abandoned.add(token)
             if msg.get("terminate"):
                 self._shutdown.set()
+        elif kind == "app_stop":
+            app_id = msg.get("app_id")
+            for fid in [f for f, frt in self.functions.items() if frt.app_id == app_id]:
+                frt = self.functions.pop(fid)
+                if frt._service is not None:
+                    await asyncio.get_running_loop().run_in_executor(
+                        self.executor, frt._service.exit
+                    )
         elif kind == "shutdown":
             self._shutdown.set()
 

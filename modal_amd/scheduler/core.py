@@ -190,6 +190,16 @@ class Scheduler:
         if app.ephemeral:
             app.state = "stopped"
             await self._cancel_app_calls(app_id)
+            await self._notify_app_stop(app_id)
+
+    async def _notify_app_stop(self, app_id: str) -> None:
+        """Tell workers to tear down this app's services (@exit hooks run;
+        parity: container shutdown lifecycle, task_lifecycle_manager.py:78)."""
+        for w in list(self.pool.workers.values()):
+            try:
+                await w.conn.send({"t": "app_stop", "app_id": app_id})
+            except Exception:
+                pass
 
     async def app_stop(self, app_id: str) -> None:
         app = self._app(app_id)
@@ -197,6 +207,7 @@ class Scheduler:
         if app.deployment_name:
             self.app_names.pop((app.environment, app.deployment_name), None)
         await self._cancel_app_calls(app_id)
+        await self._notify_app_stop(app_id)
 
     async def app_list(self, environment: str = "") -> list[dict]:
         env = environment or self.default_environment

@@ -1,0 +1,155 @@
+"""Class services, @batched dynamic batching, @concurrent input slots."""
+
+from __future__ import annotations
+
+import time
+
+import pytest
+
+import modal_amd as modal
+
+
+def test_cls_method_remote(client):
+    app = modal.App("cls-app")
+
+    @app.cls()
+    class Counter:
+        base = modal.parameter(default=100)
+
+        @modal.enter()
+        def setup(self):
+            self.offset = 7
+
+        @modal.method()
+        def add(self, x):
+            return self.base + self.offset + x
+
+    with app.run(client=client):
+        c = Counter()
+        assert c.add.remote(1) == 108
+
+
+def test_cls_parametrized(client):
+    app = modal.App("cls-param")
+
+    @app.cls()
+    class Scaler:
+        factor = modal.parameter(default=2)
+
+        @modal.method()
+        def scale(self, x):
+            return x * self.factor
+
+    with app.run(client=client):
+        assert Scaler().scale.remote(10) == 20
+        assert Scaler(factor=5).scale.remote(10) == 50
+
+
+def test_cls_lifecycle_exit_hook(client, run_dir):
+    app = modal.App("cls-exit")
+    marker = f"{run_dir}/exit-marker"
+
+    @app.cls()
+    class Svc:
+        @modal.enter()
+        def up(self):
+            self.path = None
+
+        @modal.method()
+        def ping(self, path):
+            self.path = path
+            type(self)._path = path
+            return "pong"
+
+        @modal.exit()
+        def down(self):
+            with open(type(self)._path, "w") as f:
+                f.write("exited")
+
+    with app.run(client=client):
+        assert Svc().ping.remote(marker) == "pong"
+    # worker shutdown runs @exit hooks; give the pool a moment
+    deadline = time.time() + 10
+    while time.time() < deadline:
+        try:
+            assert open(marker).read() == "exited"
+            break
+        except FileNotFoundError:
+            time.sleep(0.2)
+    else:
+        pytest.fail("exit hook never ran")
+
+
+def test_cls_generator_method(client):
+    app = modal.App("cls-gen")
+
+    @app.cls()
+    class Gen:
+        @modal.method()
+        def items(self, n):
+            for i in range(n):
+                yield i * 2
+
+    with app.run(client=client):
+        assert list(Gen().items.remote_gen(3)) == [0, 2, 4]
+
+
+def test_batched_function(client):
+    app = modal.App("batched-app")
+
+    @app.function()
+    @modal.batched(max_batch_size=4, wait_ms=300)
+    def batch_double(xs):
+        # xs arrives as a list; return a list of the same length
+        assert isinstance(xs, list)
+        return [x * 2 for x in xs]
+
+    with app.run(client=client):
+        results = list(batch_double.map(range(10)))
+        assert results == [x * 2 for x in range(10)]
+
+
+def test_batched_error_propagates_per_item(client):
+    app = modal.App("batched-err")
+
+    @app.function()
+    @modal.batched(max_batch_size=8, wait_ms=100)
+    def bad_batch(xs):
+        raise ValueError("batch failed")
+
+    with app.run(client=client):
+        out = list(bad_batch.map(range(4), return_exceptions=True))
+        assert all(isinstance(o, ValueError) for o in out)
+
+
+def test_concurrent_overlap(client):
+    app = modal.App("concurrent-app")
+
+    @app.function()
+    @modal.concurrent(max_inputs=8)
+    def sleeper(x):
+        time.sleep(0.3)
+        return x
+
+    with app.run(client=client):
+        t0 = time.time()
+        out = sorted(sleeper.map(range(8), order_outputs=False))
+        elapsed = time.time() - t0
+        assert out == list(range(8))
+        # 8 x 0.3 s sequential would be 2.4 s+; concurrency should crush that
+        assert elapsed < 1.8, f"no overlap: {elapsed:.2f}s"
+
+
+def test_cls_with_options_and_from_name(client):
+    app = modal.App("cls-deploy")
+
+    @app.cls()
+    class Echo:
+        @modal.method()
+        def say(self, v):
+            return f"echo:{v}"
+
+    app.deploy(name="cls-deployed", client=client)
+    remote_cls = modal.Cls.from_name("cls-deployed", "Echo")
+    obj = remote_cls()
+    assert obj.say.remote("hi") == "echo:hi"
