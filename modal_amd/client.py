@@ -121,6 +121,15 @@ class _Client:
         # worker processes install their client explicitly before user code runs;
         # reaching here in a worker means env-based attach to the host scheduler.
         socket_path = os.environ.get("MODAL_AMD_ATTACH_SOCKET")
+        if not socket_path:
+            # a configured run_dir with a live daemon socket means "attach"
+            from .config import config
+
+            run_dir = config.get("run_dir")
+            if run_dir:
+                candidate = os.path.join(run_dir, "scheduler.sock")
+                if os.path.exists(candidate):
+                    socket_path = candidate
         if socket_path:
             client = await cls.connect(socket_path)
         else:

@@ -1,0 +1,110 @@
+"""CLI surface tests (click runner; parity with reference cli commands)."""
+
+from __future__ import annotations
+
+import os
+import textwrap
+
+import pytest
+from click.testing import CliRunner
+
+from modal_amd.cli.entry_point import entrypoint_cli
+
+
+@pytest.fixture()
+def runner():
+    return CliRunner()
+
+
+@pytest.fixture()
+def app_file(tmp_path):
+    path = tmp_path / "myapp.py"
+    path.write_text(
+        textwrap.dedent(
+            """
+            import modal_amd as modal
+
+            app = modal.App("cli-test-app")
+
+            @app.function()
+            def double(x: int):
+                return x * 2
+
+            @app.local_entrypoint()
+            def main(n: int = 3):
+                print("entrypoint says", double.remote(n))
+            """
+        )
+    )
+    return str(path)
+
+
+def test_cli_help(runner):
+    result = runner.invoke(entrypoint_cli, ["--help"])
+    assert result.exit_code == 0
+    for cmd in ["run", "deploy", "serve", "app", "volume", "queue", "dict", "secret", "config"]:
+        assert cmd in result.output
+
+
+def test_cli_run_function(runner, app_file, client):
+    result = runner.invoke(entrypoint_cli, ["run", f"{app_file}::app.double", "21"])
+    assert result.exit_code == 0, result.output
+    assert "42" in result.output
+
+
+def test_cli_run_entrypoint(runner, app_file, client):
+    result = runner.invoke(entrypoint_cli, ["run", f"{app_file}::app.main", "--n=5"])
+    assert result.exit_code == 0, result.output
+    assert "entrypoint says 10" in result.output
+
+
+def test_cli_deploy_and_app_list(runner, app_file, client):
+    result = runner.invoke(entrypoint_cli, ["deploy", app_file, "--name", "cli-deployed"])
+    assert result.exit_code == 0, result.output
+    result = runner.invoke(entrypoint_cli, ["app", "list"])
+    assert "cli-deployed" in result.output
+
+
+def test_cli_queue_roundtrip(runner, client):
+    assert runner.invoke(entrypoint_cli, ["queue", "create", "cliq"]).exit_code == 0
+    import modal_amd as modal
+
+    modal.Queue.from_name("cliq").put("hello")
+    result = runner.invoke(entrypoint_cli, ["queue", "len", "cliq"])
+    assert result.output.strip() == "1"
+    result = runner.invoke(entrypoint_cli, ["queue", "peek", "cliq"])
+    assert "hello" in result.output
+    assert runner.invoke(entrypoint_cli, ["queue", "delete", "cliq", "--yes"]).exit_code == 0
+
+
+def test_cli_volume_roundtrip(runner, client, tmp_path):
+    src = tmp_path / "data.txt"
+    src.write_text("cli-volume-data")
+    assert runner.invoke(entrypoint_cli, ["volume", "create", "cliv"]).exit_code == 0
+    res = runner.invoke(entrypoint_cli, ["volume", "put", "cliv", str(src), "/"])
+    assert res.exit_code == 0, res.output
+    res = runner.invoke(entrypoint_cli, ["volume", "ls", "cliv"])
+    assert "data.txt" in res.output
+    dest = tmp_path / "out.txt"
+    res = runner.invoke(entrypoint_cli, ["volume", "get", "cliv", "data.txt", str(dest)])
+    assert res.exit_code == 0, res.output
+    assert dest.read_text() == "cli-volume-data"
+
+
+def test_cli_secret_and_config(runner, client):
+    res = runner.invoke(entrypoint_cli, ["secret", "create", "clis", "KEY=val"])
+    assert res.exit_code == 0, res.output
+    res = runner.invoke(entrypoint_cli, ["secret", "list"])
+    assert "clis" in res.output
+    res = runner.invoke(entrypoint_cli, ["config", "show"])
+    assert res.exit_code == 0
+    assert "heartbeat_interval" in res.output
+
+
+def test_cli_dict_roundtrip(runner, client):
+    assert runner.invoke(entrypoint_cli, ["dict", "create", "clid"]).exit_code == 0
+    import modal_amd as modal
+
+    modal.Dict.from_name("clid").put("k", [1, 2])
+    res = runner.invoke(entrypoint_cli, ["dict", "get", "clid", "k"])
+    assert "[1, 2]" in res.output
