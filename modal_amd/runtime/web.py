@@ -1,0 +1,167 @@
+"""Worker-side web wrappers: FastAPI / ASGI / WSGI / web_server bridges.
+
+Parity: /root/reference/py/modal/_runtime/asgi.py — ``asgi_app_wrapper``
+(:99), ``wsgi_app_wrapper`` (:233), web_server proxy (:280-470),
+``magic_fastapi_app`` (:240). A request arrives as a dict (method, path,
+query_string, headers, body) and the wrapper returns
+{status, headers, body}.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import inspect
+from typing import Any, Callable, Optional
+
+
+def magic_fastapi_app(fn: Callable, method: str, docs: bool) -> Any:
+    from fastapi import FastAPI
+
+    app = FastAPI(docs_url="/docs" if docs else None)
+    app.add_api_route("/", fn, methods=[method.upper()])
+    return app
+
+
+async def run_asgi(app: Any, req: dict) -> dict:
+    scope = {
+        "type": "http",
+        "asgi": {"version": "3.0", "spec_version": "2.3"},
+        "http_version": "1.1",
+        "method": req["method"],
+        "scheme": "http",
+        "path": req.get("path") or "/",
+        "raw_path": (req.get("path") or "/").encode(),
+        "query_string": (req.get("query_string") or "").encode(),
+        "root_path": "",
+        "headers": [(k.lower().encode(), v.encode()) for k, v in req.get("headers", [])],
+        "server": ("127.0.0.1", 80),
+        "client": ("127.0.0.1", 0),
+    }
+    incoming = [{"type": "http.request", "body": req.get("body") or b"", "more_body": False}]
+    sent: list[dict] = []
+    done = asyncio.Event()
+
+    async def receive() -> dict:
+        if incoming:
+            return incoming.pop(0)
+        await done.wait()
+        return {"type": "http.disconnect"}
+
+    async def send(message: dict) -> None:
+        sent.append(message)
+        if message.get("type") == "http.response.body" and not message.get("more_body"):
+            done.set()
+
+    await app(scope, receive, send)
+    status = 500
+    headers: list = []
+    body = b""
+    for message in sent:
+        if message["type"] == "http.response.start":
+            status = message["status"]
+            headers = [
+                [k.decode(), v.decode()] for k, v in message.get("headers", [])
+            ]
+        elif message["type"] == "http.response.body":
+            body += message.get("body", b"")
+    return {"status": status, "headers": headers, "body": body}
+
+
+def run_wsgi(app: Any, req: dict) -> dict:
+    import io
+
+    environ = {
+        "REQUEST_METHOD": req["method"],
+        "SCRIPT_NAME": "",
+        "PATH_INFO": req.get("path") or "/",
+        "QUERY_STRING": req.get("query_string") or "",
+        "SERVER_NAME": "127.0.0.1",
+        "SERVER_PORT": "80",
+        "SERVER_PROTOCOL": "HTTP/1.1",
+        "wsgi.version": (1, 0),
+        "wsgi.url_scheme": "http",
+        "wsgi.input": io.BytesIO(req.get("body") or b""),
+        "wsgi.errors": io.StringIO(),
+        "wsgi.multithread": True,
+        "wsgi.multiprocess": False,
+        "wsgi.run_once": False,
+    }
+    for key, value in req.get("headers", []):
+        cgi_key = "HTTP_" + key.upper().replace("-", "_")
+        if key.lower() == "content-type":
+            environ["CONTENT_TYPE"] = value
+        elif key.lower() == "content-length":
+            environ["CONTENT_LENGTH"] = value
+        else:
+            environ[cgi_key] = value
+
+    captured: dict = {"status": 500, "headers": []}
+
+    def start_response(status: str, headers: list, exc_info: Any = None) -> Any:
+        captured["status"] = int(status.split()[0])
+        captured["headers"] = [[k, v] for k, v in headers]
+
+    chunks = app(environ, start_response)
+    body = b"".join(chunks)
+    if hasattr(chunks, "close"):
+        chunks.close()
+    return {"status": captured["status"], "headers": captured["headers"], "body": body}
+
+
+class WebEndpointRuntime:
+    """Built once per web function in the worker; callable per request."""
+
+    def __init__(self, web_config: dict, raw_fn: Callable):
+        self.config = web_config
+        self.kind = web_config["type"]
+        self.raw_fn = raw_fn
+        self._app: Any = None
+        self._started = False
+
+    def _ensure_app(self) -> Any:
+        if self._app is None:
+            if self.kind == "fastapi":
+                self._app = magic_fastapi_app(
+                    self.raw_fn, self.config.get("method", "GET"), self.config.get("docs", False)
+                )
+            elif self.kind == "asgi":
+                self._app = self.raw_fn()
+            elif self.kind == "wsgi":
+                self._app = self.raw_fn()
+        return self._app
+
+    async def handle(self, req: dict) -> dict:
+        if self.kind in ("fastapi", "asgi"):
+            return await run_asgi(self._ensure_app(), req)
+        if self.kind == "wsgi":
+            loop = asyncio.get_running_loop()
+            return await loop.run_in_executor(None, run_wsgi, self._ensure_app(), req)
+        if self.kind == "web_server":
+            if not self._started:
+                self._started = True
+                result = self.raw_fn()
+                if inspect.iscoroutine(result):
+                    await result
+                # give the server a beat to bind (parity: startup_timeout)
+                await asyncio.sleep(min(self.config.get("startup_timeout", 5.0), 0.5))
+            return await self._proxy(req)
+        raise ValueError(f"Unknown web endpoint type {self.kind}")
+
+    async def _proxy(self, req: dict) -> dict:
+        import aiohttp
+
+        port = self.config["port"]
+        url = f"http://127.0.0.1:{port}{req.get('path') or '/'}"
+        if req.get("query_string"):
+            url += f"?{req['query_string']}"
+        async with aiohttp.ClientSession() as session:
+            async with session.request(
+                req["method"], url, data=req.get("body") or None,
+                headers={k: v for k, v in req.get("headers", [])},
+            ) as resp:
+                body = await resp.read()
+                return {
+                    "status": resp.status,
+                    "headers": [[k, v] for k, v in resp.headers.items()],
+                    "body": body,
+                }

@@ -74,6 +74,8 @@ class FunctionRuntime:
         self.env = msg.get("env") or {}
         self.volumes = msg.get("volumes") or {}
         self.python_paths = msg.get("python_paths") or []
+        self.web_config = msg.get("web_config")
+        self._web_runtime: Any = None
         self.sem = asyncio.Semaphore(self.max_concurrent)
         self._callable: Any = None
         self._service: Any = None
@@ -370,6 +372,18 @@ class WorkerRuntime:
             if cluster is not None:
                 self._setup_cluster(cluster)
             try:
+                if frt.web_config and item.get("method") == "__web__":
+                    if frt._web_runtime is None:
+                        from .web import WebEndpointRuntime
+
+                        frt._web_runtime = WebEndpointRuntime(frt.web_config, frt.load())
+                    args, kwargs = self._decode_args(item)
+                    result = await frt._web_runtime.handle(args[0])
+                    self.post_output(
+                        token, frt.function_id, GENERIC_STATUS_SUCCESS,
+                        serialize(result), DataFormat.PICKLE,
+                    )
+                    return
                 fn = frt.get_callable(item.get("method", ""))
                 args, kwargs = self._decode_args(item)
                 is_gen = (

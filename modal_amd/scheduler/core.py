@@ -111,6 +111,9 @@ class Scheduler:
         self.image_service = ImageService(self.run_dir, self.blob_store)
         self.mounts: dict[str, str] = {}  # mount_id -> materialized dir
         self._mounts_by_hash: dict[str, str] = {}
+        from .web import WebGateway
+
+        self.web_gateway = WebGateway(self)
         self.pool = WorkerPool(self)
         self.rpc_adapter = RPCAdapter(self)
         self._started = False
@@ -130,6 +133,7 @@ class Scheduler:
         if not self._started:
             return
         await self.sandbox_service.shutdown()
+        await self.web_gateway.stop()
         await self.pool.stop()
         self._started = False
 
@@ -275,6 +279,10 @@ class Scheduler:
             image_id=options.get("image_id"),
         )
         self.functions[fid] = fdef
+        if fdef.web_config:
+            await self.web_gateway.ensure_started()
+            label = fdef.web_config.get("label") or name
+            fdef.metadata["web_url"] = self.web_gateway.register(label, fid)
         app = self.apps.get(app_id)
         if app is not None:
             app.objects[name] = (fid, fdef.public_metadata())

@@ -425,6 +425,7 @@ class WorkerPool:
                         "env": self.scheduler.resolve_function_env(fdef),
                         "volumes": fdef.volume_mounts,
                         "python_paths": self.scheduler.resolve_function_pythonpaths(fdef),
+                        "web_config": fdef.web_config,
                     }
                 )
                 w.functions_loaded.add(fdef.function_id)
