@@ -127,11 +127,17 @@ class Scheduler:
             if self._started:
                 return
             await self.pool.start()
+            from .cron import ScheduleRunner
+
+            self.schedule_runner = ScheduleRunner(self)
+            self.schedule_runner.start()
             self._started = True
 
     async def stop(self) -> None:
         if not self._started:
             return
+        if getattr(self, "schedule_runner", None) is not None:
+            self.schedule_runner.stop()
         await self.sandbox_service.shutdown()
         await self.web_gateway.stop()
         await self.pool.stop()
@@ -279,6 +285,8 @@ class Scheduler:
             image_id=options.get("image_id"),
         )
         self.functions[fid] = fdef
+        if fdef.min_containers or fdef.buffer_containers:
+            asyncio.get_running_loop().create_task(self.pool.ensure_min(fdef))
         if fdef.web_config:
             await self.web_gateway.ensure_started()
             label = fdef.web_config.get("label") or name

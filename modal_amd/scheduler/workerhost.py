@@ -399,7 +399,52 @@ class WorkerPool:
                 if not batch:
                     continue
                 await self._send_batch(w, fdef, batch)
-            # anything left stays pending until credit frees up
+            if q:
+                # backlog remains with no free credit: autoscale
+                # (parity: min/max_containers autoscaler settings,
+                # reference _functions.py:1195-1292)
+                await self._maybe_scale_up(fdef, backlog=len(q), active=len(candidates))
+
+    async def _maybe_scale_up(self, fdef: FunctionDef, backlog: int, active: int) -> None:
+        if self._pending_spawns > 0:
+            return
+        limit = self._pool_limit(fdef)
+        if fdef.max_containers:
+            limit = min(limit, fdef.max_containers)
+        current = sum(
+            1 for w in self.workers.values() if w.alive and (w.has_gpu or not fdef.needs_gpu)
+        )
+        if current >= limit:
+            return
+        want = min(limit - current, max(1, backlog // DEFAULT_PIPELINE_DEPTH))
+        n_gpus = self._gpu_count()
+        for i in range(want):
+            gpu_index = (current + i) % n_gpus if (fdef.needs_gpu and n_gpus) else (
+                (current + i) % n_gpus if n_gpus else None
+            )
+            await self.spawn_worker(gpu_index=gpu_index)
+
+    async def ensure_min(self, fdef: FunctionDef) -> None:
+        """Warm pool: keep min_containers workers alive for this function."""
+        want = min(fdef.min_containers + fdef.buffer_containers, self._pool_limit(fdef))
+        current = sum(
+            1 for w in self.workers.values() if w.alive and (w.has_gpu or not fdef.needs_gpu)
+        ) + self._pending_spawns
+        n_gpus = self._gpu_count()
+        for i in range(max(0, want - current)):
+            gpu_index = (current + i) % n_gpus if n_gpus else None
+            await self.spawn_worker(gpu_index=gpu_index if fdef.needs_gpu or n_gpus else None)
+
+    def _pool_limit(self, fdef: FunctionDef) -> int:
+        from ..config import config
+
+        n_gpus = self._gpu_count()
+        if fdef.needs_gpu:
+            return max(n_gpus, 1)
+        configured = config.get("worker_count")
+        if configured:
+            return configured
+        return min(max((os.cpu_count() or 4) // 2, 1), 8)
 
     async def _send_batch(self, w: WorkerHandle, fdef: FunctionDef, batch: list[InputRecord]) -> None:
         try:

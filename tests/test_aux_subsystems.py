@@ -1,0 +1,181 @@
+"""Schedules, autoscaler logic, telemetry, snapshot state machine, flash."""
+
+from __future__ import annotations
+
+import json
+import os
+import socket
+import struct
+import threading
+import time
+from datetime import datetime, timezone
+
+import pytest
+
+import modal_amd as modal
+
+
+def test_cron_matcher():
+    from modal_amd.scheduler.cron import cron_matches
+
+    dt = datetime(2026, 9, 13, 14, 30, tzinfo=timezone.utc)  # Sunday
+    assert cron_matches("30 14 * * *", dt)
+    assert cron_matches("*/15 * * * *", dt)
+    assert not cron_matches("31 14 * * *", dt)
+    assert cron_matches("30 14 13 9 *", dt)
+    assert not cron_matches("30 14 14 9 *", dt)
+    assert cron_matches("* * * * 0", dt)  # sunday == 0
+    assert not cron_matches("* * * * 1", dt)
+
+
+def test_period_schedule_fires(client, run_dir):
+    app = modal.App("sched-app")
+    marker = f"{run_dir}/fired"
+
+    @app.function(schedule=modal.Period(seconds=0.5))
+    def tick():
+        with open(marker, "a") as f:
+            f.write("x")
+
+    app.deploy(name="sched-deployed", client=client)
+    deadline = time.time() + 15
+    while time.time() < deadline:
+        if os.path.exists(marker) and len(open(marker).read()) >= 1:
+            break
+        time.sleep(0.2)
+    else:
+        pytest.fail("schedule never fired")
+
+
+def test_flash_autoscaler_decisions():
+    from modal_amd.flash import FlashAutoscaler
+
+    metric = {"v": 100.0}
+    scaler = FlashAutoscaler(
+        lambda: metric["v"], target_value=50.0, min_replicas=1, max_replicas=8,
+        scale_up_stabilization=0.0, scale_down_stabilization=10.0,
+    )
+    # metric 2x target -> double replicas, immediately (no up window)
+    assert scaler.decide(2, now=100.0) == 4
+    # metric at target -> hold
+    metric["v"] = 50.0
+    assert scaler.decide(4, now=101.0) == 4
+    # metric far below target -> scale down only after the window
+    metric["v"] = 10.0
+    assert scaler.decide(4, now=102.0) == 4  # pending
+    assert scaler.decide(4, now=105.0) == 4  # still inside window
+    assert scaler.decide(4, now=113.0) == 1  # window elapsed
+
+
+def test_flash_manager_registry():
+    from modal_amd.flash import FlashManager
+
+    mgr = FlashManager()
+    mgr.register("svc", "http://127.0.0.1:9000")
+    assert mgr.list()[0].name == "svc"
+    mgr.deregister("svc")
+    assert mgr.list() == []
+
+
+def test_import_telemetry_socket(tmp_path):
+    from modal_amd.runtime.telemetry import instrument_imports, uninstrument_imports
+
+    sock_path = str(tmp_path / "telemetry.sock")
+    received = []
+    server = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+    server.bind(sock_path)
+    server.listen(1)
+
+    def acceptor():
+        conn, _ = server.accept()
+        buf = b""
+        while True:
+            try:
+                data = conn.recv(4096)
+            except OSError:
+                return
+            if not data:
+                return
+            buf += data
+            while len(buf) >= 4:
+                (n,) = struct.unpack("<I", buf[:4])
+                if len(buf) < 4 + n:
+                    break
+                received.append(json.loads(buf[4 : 4 + n]))
+                buf = buf[4 + n :]
+
+    t = threading.Thread(target=acceptor, daemon=True)
+    t.start()
+    try:
+        instrument_imports(sock_path)
+        import wsgiref.handlers  # noqa: F401  - something not yet imported
+
+        deadline = time.time() + 5
+        while time.time() < deadline and not any(
+            "wsgiref" in m.get("name", "") for m in received
+        ):
+            time.sleep(0.05)
+        assert any("wsgiref" in m.get("name", "") for m in received)
+        assert all(m["event"] == "module_load" for m in received)
+    finally:
+        uninstrument_imports()
+        server.close()
+
+
+def test_snapshot_state_machine_cpu():
+    from modal_amd.runtime.gpu_snapshot import CudaCheckpointState, GPUMemorySnapshot
+
+    snap = GPUMemorySnapshot()
+    assert snap.state is CudaCheckpointState.RUNNING
+    snap.checkpoint()  # no GPU here: transitions straight to CHECKPOINTED
+    assert snap.state is CudaCheckpointState.CHECKPOINTED
+    snap.restore()
+    assert snap.state is CudaCheckpointState.RUNNING
+    with pytest.raises(RuntimeError):
+        snap.restore()
+
+
+@pytest.mark.gpu
+def test_snapshot_roundtrip_gpu():
+    torch = pytest.importorskip("torch")
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from modal_amd.runtime.gpu_snapshot import GPUMemorySnapshot
+
+    t1 = torch.arange(1000, device="cuda", dtype=torch.float32)
+    t2 = torch.full((64, 64), 3.0, device="cuda")
+    expect1, expect2 = t1.cpu().clone(), t2.cpu().clone()
+    snap = GPUMemorySnapshot()
+    snap.checkpoint()
+    assert t1.device.type == "cpu" and t1.numel() == 0  # paged out
+    snap.restore()
+    assert t1.is_cuda and t2.is_cuda
+    assert torch.equal(t1.cpu(), expect1)
+    assert torch.equal(t2.cpu(), expect2)
+
+
+def test_autoscaler_spawns_for_backlog(client):
+    """Backlogged CPU function grows the pool beyond one worker."""
+    app = modal.App("scale-app")
+
+    @app.function()
+    def slow(x):
+        time.sleep(0.4)
+        return x
+
+    with app.run(client=client):
+        out = sorted(slow.map(range(24), order_outputs=False))
+        assert out == list(range(24))
+        svc = client.svc
+        assert len(svc.pool.workers) >= 2, "autoscaler never scaled up"
+
+
+def test_environments_and_workspace():
+    import modal_amd.environments as envs
+    from modal_amd.workspace import Workspace
+
+    envs.create_environment("staging")
+    names = [e.name for e in envs.list_environments()]
+    assert "staging" in names and "main" in names
+    envs.delete_environment("staging")
+    assert Workspace.current().name
