@@ -22,7 +22,11 @@ from typing import Optional, Union
 
 from . import gpu_available, load_lib
 
-LEAF_SIZE = 64 * 1024  # matches the reference's 64 KiB hash streaming chunk
+# Leaf size trades tree overhead against GPU parallelism: SHA-256 is serial
+# within a leaf, so leaves are the unit of parallelism. 16 KiB gives 4x the
+# workgroups of the reference's 64 KiB streaming chunk (hash_utils.py:11) —
+# measured latency-bound at 64 KiB (profiles/README.md).
+LEAF_SIZE = 16 * 1024
 TREE_DOMAIN = b"modal-amd-tree-v1"
 GPU_MIN_BYTES = 8 * 1024 * 1024  # below this, CPU wins (kernel+copy overhead)
 
@@ -97,6 +101,66 @@ def tree_sha256(data: Buffer) -> bytes:
     if len(data) >= GPU_MIN_BYTES and gpu_available():
         return _tree_sha256_gpu(data)
     return tree_sha256_cpu(data)
+
+
+def content_digests_batch(buffers: list, gpu_threshold: int = GPU_MIN_BYTES) -> list[str]:
+    """CAS keys for many buffers with ONE kernel launch over all leaves.
+
+    This is the volume-upload hot path: a multi-GiB file's 8 MiB blocks hash
+    in a single batched dispatch (thousands of leaves -> full chip), instead
+    of one under-occupied launch per block.
+    """
+    total = sum(len(b) for b in buffers)
+    if total < gpu_threshold or not gpu_available():
+        return [content_digest(b, gpu_threshold) for b in buffers]
+    lib = load_lib(required=True)
+    import torch
+
+    blob = bytearray()
+    spans: list[tuple[int, int]] = []  # (first_leaf_index, n_leaves) per buffer
+    offsets: list[int] = []
+    lengths: list[int] = []
+    small_idx: dict[int, str] = {}
+    for i, data in enumerate(buffers):
+        if len(data) < gpu_threshold:
+            # digest form must depend only on the buffer's own size so every
+            # path (single, batch, CPU, GPU) computes the same CAS key
+            import hashlib as _hl
+
+            small_idx[i] = _hl.sha256(bytes(data)).hexdigest()
+            spans.append((len(offsets), 0))
+            continue
+        base = len(blob)
+        blob += bytes(data)
+        n_leaves = (len(data) + LEAF_SIZE - 1) // LEAF_SIZE
+        spans.append((len(offsets), n_leaves))
+        for leaf in range(n_leaves):
+            offsets.append(base + leaf * LEAF_SIZE)
+            lengths.append(min(LEAF_SIZE, len(data) - leaf * LEAF_SIZE))
+    if not offsets:
+        return [small_idx[i] for i in range(len(buffers))]
+    src = torch.frombuffer(blob, dtype=torch.uint8).cuda()
+    out = torch.empty((len(offsets), 32), dtype=torch.uint8, device="cuda")
+    off_d = torch.tensor(offsets, dtype=torch.int64).cuda()
+    len_d = torch.tensor(lengths, dtype=torch.int64).cuda()
+    rc = lib.ma_sha256_many(
+        src.data_ptr(), off_d.data_ptr(), len_d.data_ptr(), out.data_ptr(),
+        len(offsets), torch.cuda.current_stream().cuda_stream,
+    )
+    if rc != 0:
+        raise RuntimeError(f"sha256 kernel failed: hipError {rc}")
+    torch.cuda.synchronize()
+    leaf_digests = out.cpu().numpy().tobytes()
+    results: list[str] = []
+    for i, data in enumerate(buffers):
+        if i in small_idx:
+            results.append(small_idx[i])
+        else:
+            first, n_leaves = spans[i]
+            results.append(
+                _root_digest(len(data), leaf_digests[first * 32 : (first + n_leaves) * 32]).hex()
+            )
+    return results
 
 
 def content_digest(data: Buffer, gpu_threshold: int = GPU_MIN_BYTES) -> str:

@@ -41,11 +41,10 @@ class BlobStore:
     def _path(self, digest: str) -> str:
         return os.path.join(self.root, digest[:2], digest)
 
-    def put(self, data: Union[bytes, memoryview]) -> str:
-        digest = _hash_bytes(data)
+    def _store(self, digest: str, data: Union[bytes, memoryview]) -> None:
         path = self._path(digest)
         if os.path.exists(path):
-            return digest
+            return
         os.makedirs(os.path.dirname(path), exist_ok=True)
         fd, tmp = tempfile.mkstemp(dir=os.path.dirname(path))
         try:
@@ -58,7 +57,21 @@ class BlobStore:
             except OSError:
                 pass
             raise
+
+    def put(self, data: Union[bytes, memoryview]) -> str:
+        digest = _hash_bytes(data)
+        self._store(digest, data)
         return digest
+
+    def put_many(self, buffers: list) -> list[str]:
+        """Batched put: one GPU hash dispatch over all buffers' leaves
+        (the volume-upload hot path; see ops/hashing.content_digests_batch)."""
+        from ..ops.hashing import content_digests_batch
+
+        digests = content_digests_batch(buffers)
+        for digest, data in zip(digests, buffers):
+            self._store(digest, data)
+        return digests
 
     def put_file(self, src_path: str) -> str:
         h = hashlib.sha256()

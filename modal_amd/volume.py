@@ -161,10 +161,10 @@ class _Volume(_Object, type_kind="volume"):
         """Block-wise upload through the CAS: hash each 8 MiB block (HIP
         kernel above the crossover), store, then commit the manifest."""
         store = self._client.blob_store
-        digests = []
-        for off in range(0, max(len(data), 1), BLOCK_SIZE):
-            block = data[off : off + BLOCK_SIZE]
-            digests.append(store.put(block))
+        blocks = [
+            bytes(data[off : off + BLOCK_SIZE]) for off in range(0, max(len(data), 1), BLOCK_SIZE)
+        ]
+        digests = store.put_many(blocks)  # one batched GPU hash dispatch
         resp = await self._client.svc.volume_put_file_blocks(
             volume_id=self.object_id,
             rel_path=remote_path,

@@ -17,7 +17,7 @@ assert torch.cuda.is_available()
 
 def bench_sha(total_mb: int = 512) -> None:
     n = total_mb * 1024 * 1024
-    leaf = 64 * 1024
+    from modal_amd.ops.hashing import LEAF_SIZE as leaf
     n_leaves = n // leaf
     buf = torch.randint(0, 256, (n,), dtype=torch.uint8, device="cuda")
     offsets = (torch.arange(n_leaves, dtype=torch.int64) * leaf)

@@ -138,3 +138,15 @@ def test_blobstore_uses_gpu_hash(tmp_path):
     digest = store.put(data)
     assert digest == tree_sha256_cpu(data).hex()
     assert store.get(digest) == data
+
+
+def test_content_digests_batch_gpu_matches_cpu():
+    _require_gpu()
+    from modal_amd.ops.hashing import content_digest, content_digests_batch
+
+    import os as _os
+
+    buffers = [_os.urandom(9 * 1024 * 1024), b"small", _os.urandom(8 * 1024 * 1024), b""]
+    got = content_digests_batch(buffers)
+    expect = [content_digest(b) for b in buffers]
+    assert got == expect
