@@ -14,7 +14,7 @@ if sys.version_info[:2] < (3, 10):
 
 __version__ = "0.1.0"
 
-from . import exception, experimental
+from . import billing, environments, exception, experimental
 from ._tunnel import Tunnel, forward
 from .app import App
 from .client import Client
@@ -45,6 +45,7 @@ from .queue import Queue
 from .retries import Retries
 from .sandbox import ContainerProcess, FileIO, Probe, Sandbox
 from .scheduler_placement import SchedulerPlacement
+from .server import Server
 from .snapshot import SandboxSnapshot
 from .volume import FileEntry, Volume
 from .runtime.execution_context import (
@@ -81,6 +82,9 @@ __all__ = [
     "SandboxSnapshot",
     "SchedulerPlacement",
     "Secret",
+    "Server",
+    "billing",
+    "environments",
     "Tunnel",
     "Volume",
     "experimental",
