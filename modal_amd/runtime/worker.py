@@ -287,6 +287,9 @@ class WorkerRuntime:
         if kind == "hello_ack":
             self.task_id = msg.get("task_id", "")
             os.environ["MODAL_AMD_TASK_ID"] = self.task_id
+            if msg.get("ring_in"):
+                # worker's out-ring is the scheduler's in-ring and vice versa
+                self.conn.attach_rings(msg.get("ring_out"), msg.get("ring_in"), create=False)
             self._hello_ack.set()
         elif kind == "def":
             frt = FunctionRuntime(self, msg)

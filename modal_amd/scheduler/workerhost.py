@@ -248,7 +248,19 @@ class WorkerPool:
                     self._pending_spawns = max(0, self._pending_spawns - 1)
                     self._worker_ready.set()
                     self._dispatch_wake.set()
-                    await conn.send({"t": "hello_ack", "task_id": handle.task_id, "worker_id": worker_id})
+                    # shm ring pair: bulk frames bypass the socket
+                    to_worker = os.path.join(self.run_dir, f"ring.{handle.task_id}.in")
+                    from_worker = os.path.join(self.run_dir, f"ring.{handle.task_id}.out")
+                    rings_ok = conn.attach_rings(to_worker, from_worker, create=True)
+                    await conn.send(
+                        {
+                            "t": "hello_ack",
+                            "task_id": handle.task_id,
+                            "worker_id": worker_id,
+                            "ring_in": to_worker if rings_ok else None,
+                            "ring_out": from_worker if rings_ok else None,
+                        }
+                    )
                     asyncio.get_running_loop().create_task(self._watch_worker(handle))
                 else:
                     # tooling/sandbox client connection: RPC only
