@@ -81,6 +81,13 @@ def get_client_context() -> Any:
     return _client_context.get()
 
 
+def _torch_tensor_cls() -> Any:
+    import sys as _sys
+
+    torch = _sys.modules.get("torch")
+    return torch.Tensor if torch is not None else None
+
+
 class Pickler(cloudpickle.CloudPickler):
     def persistent_id(self, obj: Any) -> Any:
         impl = getattr(obj, "_impl", None)
@@ -93,17 +100,86 @@ class Pickler(cloudpickle.CloudPickler):
                     "(use it inside a running app, or call .hydrate())"
                 )
             return ("modal-amd-object", obj.object_id, obj._get_metadata())
+        tensor_cls = _torch_tensor_cls()
+        if tensor_cls is not None and type(obj) is tensor_cls:
+            import os as _os
+            import sys as _sys
+
+            worker_mod = _sys.modules.get("modal_amd.runtime.worker")
+            runtime = getattr(worker_mod, "RUNTIME", None) if worker_mod else None
+            if runtime is not None and (
+                obj.is_cuda or _os.environ.get("MODAL_AMD_MESH_ALL_TENSORS")
+            ):
+                # worker-side export: the tensor stays on its GPU; consumers
+                # fetch it over RCCL/xGMI (runtime/devicemesh.py)
+                from .runtime.devicemesh import meta_of
+
+                token = runtime.tensor_table.register(obj)
+                return ("modal-amd-devtensor", runtime.task_id, token, meta_of(obj))
+            if runtime is None and obj.is_cuda:
+                # client-side CUDA tensor: host-stage (worker devices differ)
+                return ("modal-amd-staged-cuda", obj.detach().cpu(), None)
         return None
 
 
 class Unpickler(pickle.Unpickler):
     def persistent_load(self, pid: Any) -> Any:
-        tag, object_id, metadata = pid
-        if tag != "modal-amd-object":
-            raise DeserializationError(f"Unknown persistent id tag {tag!r}")
-        if _handle_factory is None:
-            raise DeserializationError("Object layer not initialized; cannot rebuild handles")
-        return _handle_factory(object_id, metadata, _client_context.get())
+        tag = pid[0]
+        if tag == "modal-amd-object":
+            _tag, object_id, metadata = pid
+            if _handle_factory is None:
+                raise DeserializationError("Object layer not initialized; cannot rebuild handles")
+            return _handle_factory(object_id, metadata, _client_context.get())
+        if tag == "modal-amd-devtensor":
+            _tag, owner_task, token, meta = pid
+            import sys as _sys
+
+            worker_mod = _sys.modules.get("modal_amd.runtime.worker")
+            runtime = getattr(worker_mod, "RUNTIME", None) if worker_mod else None
+            if runtime is not None:
+                return runtime.fetch_device_tensor(owner_task, token, meta)
+            return _client_pull_tensor(owner_task, token, meta)
+        if tag == "modal-amd-staged-cuda":
+            tensor = pid[1]
+            try:
+                import torch
+
+                if torch.cuda.is_available():
+                    return tensor.cuda()
+            except Exception:
+                pass
+            return tensor
+        raise DeserializationError(f"Unknown persistent id tag {tag!r}")
+
+
+def _client_pull_tensor(owner_task: str, token: str, meta: dict) -> Any:
+    """Non-worker consumer (the client process): host-staged pull.
+
+    Must run OFF the framework event loop (callers scan payloads for the
+    devtensor marker and deserialize in a thread when present)."""
+    import pickle as _pickle
+
+    from ._sync import synchronizer
+    from .client import _Client
+
+    client = _client_context.get() or _Client._singleton
+    if client is None:
+        raise DeserializationError("No client available to pull device tensor")
+    raw = synchronizer.run_future(
+        client.svc.tensor_pull_relay(owner_task=owner_task, token=token)
+    ).result(timeout=120)
+    if raw is None:
+        raise DeserializationError(f"device tensor {token} expired on {owner_task}")
+    tensor = _pickle.loads(raw)
+    if meta.get("device") == "cuda":
+        try:
+            import torch
+
+            if torch.cuda.is_available():
+                tensor = tensor.cuda()
+        except Exception:
+            pass
+    return tensor
 
 
 def serialize(obj: Any) -> bytes:

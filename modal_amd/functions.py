@@ -63,12 +63,20 @@ def make_payload_item(client: Any, args: tuple, kwargs: dict) -> dict:
     return {"payload": payload}
 
 
-def process_output_item(item: dict, client: Any) -> Any:
+async def process_output_item(item: dict, client: Any) -> Any:
     """Decode one completed output: return the value or raise the remote error
     (parity: _process_result, reference function_utils.py:527-583)."""
     data = item.get("data")
     if item.get("data_blob"):
         data = client.blob_store.get(item["data_blob"])
+    if (
+        data is not None
+        and item.get("status") == GENERIC_STATUS_SUCCESS
+        and b"modal-amd-devtensor" in data
+    ):
+        # device-tensor pulls block on a cross-process transfer: never
+        # deserialize those on the event loop
+        return await asyncio.get_running_loop().run_in_executor(None, deserialize, data)
     status = item.get("status")
     if status == GENERIC_STATUS_SUCCESS:
         if item.get("format") == DataFormat.GENERATOR_DONE:
@@ -155,7 +163,7 @@ class _Invocation:
 
     async def run_function(self, timeout: Optional[float] = None) -> Any:
         item = await await_output_item(self.client, self.call_id, 0, timeout)
-        return process_output_item(item, self.client)
+        return await process_output_item(item, self.client)
 
     async def run_generator(self) -> AsyncGenerator[Any, None]:
         """Consume the generator data channel until the done marker
@@ -173,7 +181,7 @@ class _Invocation:
                 info = await svc.function_call_info(function_call_id=self.call_id)
                 if info.get("completed", 0) >= 1:
                     item = await await_output_item(self.client, self.call_id, 0, 1.0)
-                    process_output_item(item, self.client)  # raises on failure
+                    await process_output_item(item, self.client)  # raises on failure
                     return
                 continue
             for index, data, fmt, is_done in entries:
@@ -186,7 +194,7 @@ class _Invocation:
                     next_index += 1
         # drain the final output so exceptions surface
         item = await await_output_item(self.client, self.call_id, 0, None)
-        result = process_output_item(item, self.client)
+        result = await process_output_item(item, self.client)
         if isinstance(result, GeneratorDone):
             return
 
@@ -476,7 +484,7 @@ class _FunctionCall(_Object, type_kind="function_call"):
         if self._has_cached:
             return self._cached
         item = await await_output_item(self._client, self.object_id, 0, timeout)
-        value = process_output_item(item, self._client)
+        value = await process_output_item(item, self._client)
         self._cached = value
         self._has_cached = True
         return value

@@ -94,11 +94,11 @@ async def map_invocation(
     next_output_idx = 0
     ordering_buffer: dict[int, Any] = {}
 
-    def decode(out: dict) -> Any:
+    async def decode(out: dict) -> Any:
         if out["status"] == GENERIC_STATUS_SUCCESS and not return_exceptions:
-            return process_output_item(out, client)
+            return await process_output_item(out, client)
         try:
-            return process_output_item(out, client)
+            return await process_output_item(out, client)
         except BaseException as exc:
             if return_exceptions:
                 return exc
@@ -116,7 +116,7 @@ async def map_invocation(
             for out in outs:
                 received += 1
                 sem.release()
-                value = decode(out)
+                value = await decode(out)
                 if order_outputs:
                     ordering_buffer[out["idx"]] = value
                     while next_output_idx in ordering_buffer:

@@ -54,6 +54,32 @@ OUTPUT_BATCH_MAX = 512
 
 _app_id_var: contextvars.ContextVar[str] = contextvars.ContextVar("modal_amd_app_id", default="")
 
+#: the live worker runtime of this process (None in client processes);
+#: serialization hooks use it to export/fetch device tensors
+RUNTIME: Optional["WorkerRuntime"] = None
+
+
+class WorkerRPCTarget:
+    """RPCs the scheduler can invoke ON a worker (reverse direction)."""
+
+    def __init__(self, runtime: "WorkerRuntime"):
+        self._runtime = runtime
+
+    async def tensor_pull(self, token: str) -> Optional[bytes]:
+        """Host-staged export of a registered tensor (for non-mesh consumers)."""
+        tensor = self._runtime.tensor_table.get(token)
+        if tensor is None:
+            return None
+        import pickle as _pickle
+
+        loop = asyncio.get_running_loop()
+        return await loop.run_in_executor(
+            None, lambda: _pickle.dumps(tensor.detach().cpu(), 4)
+        )
+
+    async def ping(self) -> str:
+        return self._runtime.task_id
+
 
 class FunctionRuntime:
     """Worker-side state for one registered function."""
@@ -223,6 +249,10 @@ class WorkerRuntime:
         )
         run_dir = os.path.dirname(self.socket_path)
         self.blob_store = BlobStore(os.path.join(run_dir, "blobs"))
+        from .devicemesh import DeviceMesh, TensorTable
+
+        self.mesh = DeviceMesh()
+        self.tensor_table = TensorTable()
         self._outbox: list[dict] = []
         self._outbox_flush_scheduled = False
         self._running: dict[str, asyncio.Task] = {}
@@ -235,12 +265,14 @@ class WorkerRuntime:
 
     # ---- wiring ---------------------------------------------------------
     async def run(self) -> None:
+        global RUNTIME
+        RUNTIME = self
         self.loop = asyncio.get_running_loop()
         try:
             reader, writer = await asyncio.open_unix_connection(self.socket_path)
         except (FileNotFoundError, ConnectionRefusedError):
             return  # scheduler already gone: exit quietly
-        self.conn = Connection(reader, writer, self._handle, rpc_target=None)
+        self.conn = Connection(reader, writer, self._handle, rpc_target=WorkerRPCTarget(self))
         self.conn.start()
         await self.conn.send(
             {
@@ -342,6 +374,20 @@ class WorkerRuntime:
                     self._abandoned.add(token)
             if msg.get("terminate"):
                 self._shutdown.set()
+        elif kind == "mesh_init":
+            self.mesh.init(msg["rank"], msg["world"], msg["port"], msg["backend"])
+
+            def wait_and_ack() -> bool:
+                return self.mesh.wait_ready()
+
+            ok = await asyncio.get_running_loop().run_in_executor(self.executor, wait_and_ack)
+            await self.conn.send({"t": "mesh_ready", "ok": ok, "rank": msg["rank"]})
+        elif kind == "dev_send":
+            tensor = self.tensor_table.get(msg["token"])
+            if tensor is not None:
+                self.mesh.submit_send(tensor, msg["dst_rank"])
+        elif kind == "dev_recv":
+            self.mesh.submit_recv(msg["xfer_id"], msg["meta"], msg["src_rank"])
         elif kind == "app_stop":
             app_id = msg.get("app_id")
             for fid in [f for f, frt in self.functions.items() if frt.app_id == app_id]:
@@ -353,12 +399,56 @@ class WorkerRuntime:
         elif kind == "shutdown":
             self._shutdown.set()
 
+    # ---- device tensors --------------------------------------------------
+    def fetch_device_tensor(self, owner_task: str, token: str, meta: dict) -> Any:
+        """Materialize another worker's exported tensor (executor threads
+        only: blocks on the mesh transfer or the host-staged pull)."""
+        if owner_task == self.task_id:
+            tensor = self.tensor_table.get(token)
+            if tensor is not None:
+                return tensor
+        from .._sync import synchronizer
+        from ..client import _Client
+
+        svc = _Client._singleton.svc
+        resp = synchronizer.run(
+            svc.device_transfer(
+                owner_task=owner_task, token=token, meta=meta, dest_task=self.task_id
+            )
+        )
+        if resp and resp.get("xfer_id"):
+            return self.mesh.wait_result(resp["xfer_id"])
+        raw = synchronizer.run(svc.tensor_pull_relay(owner_task=owner_task, token=token))
+        if raw is None:
+            raise RuntimeError(f"device tensor {token} no longer available on {owner_task}")
+        import pickle as _pickle
+
+        tensor = _pickle.loads(raw)
+        if meta.get("device") == "cuda":
+            import torch
+
+            if torch.cuda.is_available():
+                tensor = tensor.cuda()
+        return tensor
+
     # ---- execution ------------------------------------------------------
     def _decode_args(self, item: dict) -> tuple[tuple, dict]:
         if item.get("payload_blob"):
             payload = self.blob_store.get(item["payload_blob"])
         else:
             payload = item.get("payload") or b""
+        return deserialize_payload(payload)
+
+    async def _decode_args_async(self, item: dict) -> tuple[tuple, dict]:
+        """Device-tensor markers block on transfers: decode those off-loop."""
+        if item.get("payload_blob"):
+            payload = self.blob_store.get(item["payload_blob"])
+        else:
+            payload = item.get("payload") or b""
+        if b"modal-amd-devtensor" in payload:
+            return await asyncio.get_running_loop().run_in_executor(
+                self.executor, deserialize_payload, payload
+            )
         return deserialize_payload(payload)
 
     async def _run_input(self, frt: FunctionRuntime, item: dict) -> None:
@@ -380,7 +470,7 @@ class WorkerRuntime:
                         from .web import WebEndpointRuntime
 
                         frt._web_runtime = WebEndpointRuntime(frt.web_config, frt.load())
-                    args, kwargs = self._decode_args(item)
+                    args, kwargs = await self._decode_args_async(item)
                     result = await frt._web_runtime.handle(args[0])
                     self.post_output(
                         token, frt.function_id, GENERIC_STATUS_SUCCESS,
@@ -388,7 +478,7 @@ class WorkerRuntime:
                     )
                     return
                 fn = frt.get_callable(item.get("method", ""))
-                args, kwargs = self._decode_args(item)
+                args, kwargs = await self._decode_args_async(item)
                 is_gen = (
                     frt.is_generator
                     or inspect.isgeneratorfunction(fn)
@@ -613,7 +703,7 @@ class WorkerRuntime:
         async with frt.sem:
             try:
                 fn = frt.get_callable(items[0].get("method", ""))
-                decoded = [self._decode_args(item) for item in items]
+                decoded = [await self._decode_args_async(item) for item in items]
                 # transpose: positional args and kwargs become per-arg lists
                 n_args = max((len(a) for a, _ in decoded), default=0)
                 arg_lists = [[d[0][i] if i < len(d[0]) else None for d in decoded] for i in range(n_args)]

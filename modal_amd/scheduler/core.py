@@ -77,6 +77,7 @@ class RPCAdapter:
         "sandbox_stdin_write", "sandbox_exec", "sandbox_list", "sandbox_set_tags",
         "sandbox_from_name", "sandbox_snapshot_fs", "sandbox_fs_op",
         "image_get_or_create", "image_info", "mount_get_or_create",
+        "device_transfer", "tensor_pull_relay",
     }
 
     def __init__(self, scheduler: "Scheduler"):
@@ -616,6 +617,99 @@ class Scheduler:
             if state is not None:
                 return list(state.python_paths)
         return []
+
+    # -- device mesh (RCCL/xGMI tensor plane) ------------------------------
+    async def ensure_mesh(self) -> dict:
+        """Form the worker collective plane once: rank assignment + readiness
+        (the tensor-transfer substrate of runtime/devicemesh.py)."""
+        state = self._extra.setdefault(
+            "mesh", {"formed": False, "ranks": {}, "lock": asyncio.Lock(), "ready": {}}
+        )
+        async with state["lock"]:
+            if state["formed"]:
+                return state
+            members = [w for w in self.pool.workers.values() if w.alive]
+            gpu_members = [w for w in members if w.has_gpu]
+            if gpu_members and self.pool._gpu_count() > 0:
+                members = gpu_members
+                backend = "nccl"
+            else:
+                backend = "gloo"
+            if len(members) < 2:
+                raise NotFoundError("device mesh needs >= 2 workers")
+            members.sort(key=lambda w: w.worker_id)
+            import socket as socket_mod
+
+            sock = socket_mod.socket()
+            sock.bind(("127.0.0.1", 0))
+            port = sock.getsockname()[1]
+            sock.close()
+            ranks = {w.task_id: i for i, w in enumerate(members)}
+            events = {w.task_id: asyncio.Event() for w in members}
+            state["ready"] = events
+            for w in members:
+                await w.conn.send(
+                    {
+                        "t": "mesh_init",
+                        "rank": ranks[w.task_id],
+                        "world": len(members),
+                        "port": port,
+                        "backend": backend,
+                    }
+                )
+            await asyncio.wait_for(
+                asyncio.gather(*(e.wait() for e in events.values())), timeout=120
+            )
+            state["ranks"] = ranks
+            state["backend"] = backend
+            state["formed"] = True
+            return state
+
+    def on_mesh_ready(self, handle: Any, ok: bool) -> None:
+        state = self._extra.get("mesh")
+        if state:
+            event = state["ready"].get(handle.task_id)
+            if event is not None and ok:
+                event.set()
+
+    def _worker_by_task(self, task_id: str) -> Any:
+        for w in self.pool.workers.values():
+            if w.task_id == task_id:
+                return w
+        return None
+
+    async def device_transfer(
+        self, owner_task: str, token: str, meta: dict, dest_task: str
+    ) -> dict:
+        """Coordinate a p2p tensor move: owner sends, dest receives
+        (RCCL over xGMI on GPU pools; gloo on CPU pools)."""
+        try:
+            state = await self.ensure_mesh()
+        except Exception:
+            return {"fallback": True}
+        ranks = state["ranks"]
+        if owner_task not in ranks or dest_task not in ranks or owner_task == dest_task:
+            return {"fallback": True}
+        owner = self._worker_by_task(owner_task)
+        dest = self._worker_by_task(dest_task)
+        if owner is None or dest is None:
+            return {"fallback": True}
+        xfer_id = new_id("blob")
+        # dest's dev_recv is ordered before this RPC's reply on its socket
+        await dest.conn.send(
+            {"t": "dev_recv", "xfer_id": xfer_id, "meta": meta, "src_rank": ranks[owner_task]}
+        )
+        await owner.conn.send(
+            {"t": "dev_send", "token": token, "dst_rank": ranks[dest_task], "xfer_id": xfer_id}
+        )
+        return {"xfer_id": xfer_id, "src_rank": ranks[owner_task]}
+
+    async def tensor_pull_relay(self, owner_task: str, token: str) -> Optional[bytes]:
+        """Host-staged fallback for non-mesh consumers."""
+        owner = self._worker_by_task(owner_task)
+        if owner is None:
+            return None
+        return await owner.conn.call("tensor_pull", {"token": token}, timeout=120)
 
     # -- cluster rendezvous ------------------------------------------------
     async def cluster_hello(self, cluster_id: str, rank: int, world_size: int, addr: str = "") -> dict:

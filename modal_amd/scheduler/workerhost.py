@@ -275,6 +275,8 @@ class WorkerPool:
                 self.scheduler.on_generator_data(msg)
             elif kind == "hb":
                 handle.last_heartbeat = time.time()
+            elif kind == "mesh_ready":
+                self.scheduler.on_mesh_ready(handle, bool(msg.get("ok")))
             elif kind == "log":
                 self.scheduler.on_worker_log(handle, msg)
 
