@@ -1,0 +1,128 @@
+"""Failure detection / recovery: worker death requeue, cancellation, caps.
+
+Parity targets: INTERNAL_FAILURE requeue up to 8x without consuming user
+retries (reference _functions.py:106), heartbeat/cancellation propagation
+(container_io_manager.py:645-710), kill-switch semantics.
+"""
+
+from __future__ import annotations
+
+import os
+import time
+
+import pytest
+
+import modal_amd as modal
+from modal_amd.functions import FunctionCallCancelledError
+
+
+def test_worker_death_requeues_input(client, run_dir):
+    """An input whose worker dies mid-flight reruns on another worker."""
+    app = modal.App("death-app")
+    marker = os.path.join(run_dir, "died-once")
+
+    @app.function()
+    def maybe_die(x, path):
+        import os as _os
+
+        if x == 3 and not _os.path.exists(path):
+            with open(path, "w") as f:
+                f.write("dying")
+            _os._exit(1)  # hard-kill the whole worker process
+        return x * 10
+
+    with app.run(client=client):
+        out = sorted(maybe_die.map(range(6), kwargs={"path": marker}, order_outputs=False))
+        assert out == [0, 10, 20, 30, 40, 50]
+        assert os.path.exists(marker), "the poisoned input never ran"
+
+
+def test_repeated_internal_failure_finalizes(client, run_dir):
+    """An input that always kills its worker eventually fails with
+    InternalFailure after the 8-requeue cap (reference _functions.py:106)."""
+    app = modal.App("death-cap-app")
+
+    @app.function()
+    def always_die():
+        import os as _os
+
+        _os._exit(1)
+
+    from modal_amd.exception import InternalFailure
+    import modal_amd.scheduler.calls as calls_mod
+
+    # shrink the cap so the test is fast
+    old_cap = calls_mod.MAX_INTERNAL_FAILURE_COUNT
+    import modal_amd.scheduler.workerhost as wh
+    import modal_amd.scheduler.core as core_mod
+
+    wh.MAX_INTERNAL_FAILURE_COUNT = 2
+    core_mod.MAX_INTERNAL_FAILURE_COUNT = 2
+    try:
+        with app.run(client=client):
+            with pytest.raises(InternalFailure):
+                always_die.remote()
+    finally:
+        wh.MAX_INTERNAL_FAILURE_COUNT = old_cap
+        core_mod.MAX_INTERNAL_FAILURE_COUNT = old_cap
+
+
+def test_cancel_function_call(client):
+    app = modal.App("cancel-app")
+
+    @app.function()
+    def slow():
+        time.sleep(30)
+        return "done"
+
+    with app.run(client=client):
+        fc = slow.spawn()
+        time.sleep(0.3)
+        fc.cancel()
+        with pytest.raises(FunctionCallCancelledError):
+            fc.get(timeout=10)
+
+
+def test_exception_has_clean_traceback(client):
+    app = modal.App("tb-app")
+
+    @app.function()
+    def inner_fail():
+        def deep():
+            raise ValueError("deep failure")
+
+        deep()
+
+    with app.run(client=client):
+        try:
+            inner_fail.remote()
+            pytest.fail("should have raised")
+        except ValueError as exc:
+            import traceback
+
+            frames = traceback.extract_tb(exc.__traceback__)
+            filenames = [f.filename for f in frames]
+            # user frames survive the wire; asyncio/framework frames are cut
+            assert any("test_failure_paths" in f for f in filenames)
+            assert not any("asyncio" in f for f in filenames)
+
+
+def test_map_with_failures_and_retries(client, run_dir):
+    """User retry policy applies per input inside a map."""
+    app = modal.App("map-retry-app")
+
+    @app.function(retries=modal.Retries(max_retries=2, initial_delay=1.0))
+    def flaky(x, base):
+        import os as _os
+
+        path = f"{base}/attempt-{x}"
+        n = int(open(path).read()) if _os.path.exists(path) else 0
+        with open(path, "w") as f:
+            f.write(str(n + 1))
+        if x % 3 == 0 and n == 0:
+            raise RuntimeError(f"transient {x}")
+        return x
+
+    with app.run(client=client):
+        out = sorted(flaky.map(range(6), kwargs={"base": run_dir}, order_outputs=False))
+        assert out == list(range(6))
