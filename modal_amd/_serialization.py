@@ -278,26 +278,33 @@ def _join_tensors(obj: Any, tensors: list) -> Any:
     return obj
 
 
+def _walk_for_tensors(obj: Any, tensor_cls: type) -> bool:
+    t = type(obj)
+    if t is tensor_cls:
+        return True
+    if t is tuple or t is list:
+        for x in obj:
+            if _walk_for_tensors(x, tensor_cls):
+                return True
+        return False
+    if t is dict:
+        for v in obj.values():
+            if _walk_for_tensors(v, tensor_cls):
+                return True
+        return False
+    return isinstance(obj, tensor_cls)
+
+
 def contains_tensors(args: tuple, kwargs: dict) -> bool:
-    try:
-        import sys
+    import sys
 
-        torch = sys.modules.get("torch")
-        if torch is None:
-            return False
-    except Exception:  # pragma: no cover
+    torch = sys.modules.get("torch")
+    if torch is None:
         return False
-
-    def walk(obj: Any) -> bool:
-        if isinstance(obj, torch.Tensor):
-            return True
-        if type(obj) in (tuple, list):
-            return any(walk(x) for x in obj)
-        if type(obj) is dict:
-            return any(walk(v) for v in obj.values())
-        return False
-
-    return walk(args) or walk(kwargs)
+    tensor_cls = torch.Tensor
+    return _walk_for_tensors(args, tensor_cls) or (
+        bool(kwargs) and _walk_for_tensors(kwargs, tensor_cls)
+    )
 
 
 def serialize_payload(args: tuple, kwargs: dict) -> tuple[bytes, list]:

@@ -26,6 +26,10 @@ from ..scheduler.calls import GENERIC_STATUS_SUCCESS
 
 PUT_BATCH_SIZE = 512  # parity: spawn_map batch, reference parallel_map.py:83
 OUTPUT_FETCH_MAX = 1024
+# items per shared pickle chunk: one C-pickler pass serves ~64 inputs on the
+# client AND one unpickle serves them on the worker (SURVEY §2 row 6's
+# "tensor-aware fast path" generalized to all small map payloads)
+CHUNK_ITEMS = 64
 
 
 async def _iterate_maybe_async(it: Any) -> AsyncGenerator[Any, None]:
@@ -66,21 +70,58 @@ async def map_invocation(
 
     async def pump() -> None:
         nonlocal total_inputs
-        batch: list[dict] = []
+        from .._serialization import serialize_fast
+
+        chunk_buf: list = []
+        chunk_seq = 0
+        items: list[dict] = []
+        chunks: dict[str, bytes] = {}
+
+        import sys as _sys
+
+        def _chunk_serialize(buf: list) -> bytes:
+            # tensors need the hook-aware pickler (device staging / mesh export)
+            if "torch" in _sys.modules:
+                from .._serialization import _walk_for_tensors, serialize
+
+                tensor_cls = _sys.modules["torch"].Tensor
+                if _walk_for_tensors(buf, tensor_cls):
+                    return serialize(("C", buf))
+            return serialize_fast(("C", buf))
+
+        async def flush_chunk() -> None:
+            nonlocal chunk_buf, chunk_seq, items, chunks
+            if not chunk_buf:
+                return
+            payload = _chunk_serialize(chunk_buf)
+            chunk_id = f"{call_id}.c{chunk_seq}"
+            chunk_seq += 1
+            chunks[chunk_id] = payload
+            for ci in range(len(chunk_buf)):
+                item = {"chunk": chunk_id, "ci": ci}
+                if fn._method_name:
+                    item["method"] = fn._method_name
+                items.append(item)
+            chunk_buf = []
+            if len(items) >= PUT_BATCH_SIZE:
+                await svc.function_put_inputs(
+                    function_call_id=call_id, items=items, chunks=chunks
+                )
+                items, chunks = [], {}
+
         try:
             async for args, extra_kwargs in _iterate_maybe_async(input_iter):
                 kw = {**kwargs_common, **extra_kwargs} if extra_kwargs else kwargs_common
-                item = make_payload_item(client, args, kw)
-                if fn._method_name:
-                    item["method"] = fn._method_name
-                batch.append(item)
+                chunk_buf.append((args, kw))
                 total_inputs += 1
                 await sem.acquire()
-                if len(batch) >= PUT_BATCH_SIZE:
-                    await svc.function_put_inputs(function_call_id=call_id, items=batch)
-                    batch = []
-            if batch:
-                await svc.function_put_inputs(function_call_id=call_id, items=batch)
+                if len(chunk_buf) >= CHUNK_ITEMS:
+                    await flush_chunk()
+            await flush_chunk()
+            if items:
+                await svc.function_put_inputs(
+                    function_call_id=call_id, items=items, chunks=chunks
+                )
             await svc.function_finish_inputs(function_call_id=call_id)
         except BaseException as exc:
             pump_error.append(exc)

@@ -253,6 +253,9 @@ class WorkerRuntime:
 
         self.mesh = DeviceMesh()
         self.tensor_table = TensorTable()
+        # shared chunk payloads (map fast path): chunk_id -> entry
+        self._chunk_cache: dict[str, dict] = {}
+        self._chunk_order: list[str] = []
         self._outbox: list[dict] = []
         self._outbox_flush_scheduled = False
         self._running: dict[str, asyncio.Task] = {}
@@ -327,6 +330,12 @@ class WorkerRuntime:
             frt = FunctionRuntime(self, msg)
             self.functions[frt.function_id] = frt
         elif kind == "inputs":
+            for cid, data in (msg.get("chunks") or {}).items():
+                self._chunk_cache[cid] = {"raw": data, "decoded": None}
+                self._chunk_order.append(cid)
+                while len(self._chunk_order) > 512:
+                    old = self._chunk_order.pop(0)
+                    self._chunk_cache.pop(old, None)
             frt = self.functions.get(msg["function_id"])
             if frt is None:
                 for item in msg["items"]:
@@ -432,6 +441,24 @@ class WorkerRuntime:
         return tensor
 
     # ---- execution ------------------------------------------------------
+    def _resolve_item_args(self, item: dict) -> tuple[tuple, dict]:
+        """Thread-context arg resolution: chunk payloads decode once and
+        serve every item of the chunk."""
+        cid = item.get("chunk")
+        if cid is not None:
+            entry = self._chunk_cache.get(cid)
+            if entry is None:
+                raise RuntimeError(f"chunk {cid} not delivered to this worker")
+            decoded = entry["decoded"]
+            if decoded is None:
+                kind, decoded = deserialize(entry["raw"])
+                assert kind == "C"
+                entry["decoded"] = decoded
+                entry["raw"] = None
+            args, kwargs = decoded[item.get("ci", 0)]
+            return args, kwargs
+        return self._decode_args(item)
+
     def _decode_args(self, item: dict) -> tuple[tuple, dict]:
         if item.get("payload_blob"):
             payload = self.blob_store.get(item["payload_blob"])
@@ -441,6 +468,10 @@ class WorkerRuntime:
 
     async def _decode_args_async(self, item: dict) -> tuple[tuple, dict]:
         """Device-tensor markers block on transfers: decode those off-loop."""
+        if item.get("chunk") is not None:
+            return await asyncio.get_running_loop().run_in_executor(
+                self.executor, self._resolve_item_args, item
+            )
         if item.get("payload_blob"):
             payload = self.blob_store.get(item["payload_blob"])
         else:
@@ -573,7 +604,7 @@ class WorkerRuntime:
                 tok_i = _current_input_id.set(item.get("input_id"))
                 tok_c = _current_function_call_id.set(call_id)
                 try:
-                    args, kwargs = self._decode_args(item)
+                    args, kwargs = self._resolve_item_args(item)
                     result = fn(*args, **kwargs)
                     results.append(
                         self._make_output(

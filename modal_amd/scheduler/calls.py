@@ -139,6 +139,8 @@ class InputRecord:
     tensors: Optional[list] = None  # tensor sidecar (CUDA-IPC / pinned staging)
     cancelled: bool = False
     cluster: Optional[dict] = None  # gang identity: {rank, size, cluster_id, master_port}
+    chunk_id: Optional[str] = None  # shared chunk payload (map fan-out fast path)
+    chunk_index: int = 0
 
     @property
     def token(self) -> str:
@@ -161,6 +163,9 @@ class CallRecord:
         self.kind = kind  # "unary" | "spawn" | "map" | "spawn_map"
         self.return_exceptions = return_exceptions
         self.inputs: dict[int, InputRecord] = {}
+        # shared chunk payloads: chunk_id -> {"data": bytes, "refs": int}
+        # (one pickled list serves many inputs; freed when all are final)
+        self.chunks: dict[str, dict] = {}
         self.next_idx = 0
         self.num_inputs_final: Optional[int] = None
         self.completed: int = 0
@@ -231,6 +236,12 @@ class CallRecord:
         rec.exc_repr = exc_repr
         rec.final = True
         rec.finished_at = time.time()
+        if rec.chunk_id:
+            chunk = self.chunks.get(rec.chunk_id)
+            if chunk is not None:
+                chunk["refs"] -= 1
+                if chunk["refs"] <= 0:
+                    del self.chunks[rec.chunk_id]
         self.completed += 1
         self.output_ready.put_nowait(idx)
         waiter = self._waiters.pop(idx, None)

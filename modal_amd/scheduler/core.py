@@ -361,11 +361,19 @@ class Scheduler:
             "sync_client_retries_enabled": True,
         }
 
-    async def function_put_inputs(self, function_call_id: str, items: list) -> list:
-        """items: [{"payload": bytes, "method": str}] -> [{"idx", "input_id"}]."""
+    async def function_put_inputs(
+        self, function_call_id: str, items: list, chunks: Optional[dict] = None
+    ) -> list:
+        """items: [{"payload": bytes, "method": str} | {"chunk": id, "ci": i}]
+        -> [{"idx", "input_id"}]. ``chunks`` carries shared pickled arg-lists
+        (the map fast path: one pickle per ~64 inputs)."""
         record = self._call(function_call_id)
         fdef = self.functions.get(record.function_id)
+        if chunks:
+            for chunk_id, data in chunks.items():
+                record.chunks[chunk_id] = {"data": data, "refs": 0}
         out = []
+        batch_recs = []
         for item in items:
             if isinstance(item, (bytes, bytearray)):
                 item = {"payload": bytes(item)}
@@ -379,8 +387,17 @@ class Scheduler:
                 tensors=item.get("tensors"),
                 payload_blob=item.get("payload_blob"),
             )
-            self.pool.enqueue(rec)
+            chunk_id = item.get("chunk")
+            if chunk_id:
+                rec.chunk_id = chunk_id
+                rec.chunk_index = item.get("ci", 0)
+                chunk = record.chunks.get(chunk_id)
+                if chunk is not None:
+                    chunk["refs"] += 1
+            batch_recs.append(rec)
             out.append({"idx": rec.idx, "input_id": rec.input_id})
+        if batch_recs:
+            self.pool.enqueue_many(record.function_id, batch_recs)
         return out
 
     async def _put_gang_input(self, record: CallRecord, fdef: FunctionDef, item: dict) -> list:

@@ -61,6 +61,7 @@ class WorkerHandle:
         self.outstanding: dict[str, int] = {}
         self.functions_loaded: set[str] = set()
         self.defs_version: dict[str, int] = {}
+        self.chunks_sent: set[str] = set()
         self.draining = False
 
     @property
@@ -334,13 +335,18 @@ class WorkerPool:
         self._dispatch_wake.set()
 
     # -- dispatch --------------------------------------------------------
-    def enqueue(self, rec: InputRecord, front: bool = False) -> None:
-        fdef = self.scheduler.functions[self._function_id_of(rec)]
-        q = self.pending.setdefault(fdef.function_id, deque())
+    def enqueue(self, rec: InputRecord, front: bool = False, function_id: Optional[str] = None) -> None:
+        fid = function_id or self._function_id_of(rec)
+        q = self.pending.setdefault(fid, deque())
         if front:
             q.appendleft(rec)
         else:
             q.append(rec)
+        self._dispatch_wake.set()
+
+    def enqueue_many(self, function_id: str, recs: list) -> None:
+        q = self.pending.setdefault(function_id, deque())
+        q.extend(recs)
         self._dispatch_wake.set()
 
     def enqueue_delayed(self, rec: InputRecord, delay_s: float) -> None:
@@ -490,6 +496,8 @@ class WorkerPool:
                 w.functions_loaded.add(fdef.function_id)
                 w.defs_version[fdef.function_id] = fdef.definition_version
             items = []
+            chunks_needed: dict[str, bytes] = {}
+            record = None
             for rec in batch:
                 rec.worker_id = w.worker_id
                 rec.started_at = time.time()
@@ -497,17 +505,31 @@ class WorkerPool:
                 item = {
                     "token": rec.token,
                     "input_id": rec.input_id,
-                    "payload": rec.payload,
                     "method": rec.method_name,
                     "retry_count": rec.retry_count,
                 }
-                if rec.payload_blob:
-                    item["payload_blob"] = rec.payload_blob
+                if rec.chunk_id:
+                    item["chunk"] = rec.chunk_id
+                    item["ci"] = rec.chunk_index
+                    if rec.chunk_id not in w.chunks_sent:
+                        if record is None or record.call_id != rec.call_id:
+                            record = self.scheduler.calls.get(rec.call_id)
+                        chunk = record.chunks.get(rec.chunk_id) if record else None
+                        if chunk is not None:
+                            chunks_needed[rec.chunk_id] = chunk["data"]
+                            w.chunks_sent.add(rec.chunk_id)
+                else:
+                    item["payload"] = rec.payload
+                    if rec.payload_blob:
+                        item["payload_blob"] = rec.payload_blob
                 if rec.cluster:
                     item["cluster"] = rec.cluster
                 items.append(item)
             w.outstanding[fdef.function_id] = w.outstanding.get(fdef.function_id, 0) + len(batch)
-            await w.conn.send({"t": "inputs", "function_id": fdef.function_id, "items": items})
+            frame = {"t": "inputs", "function_id": fdef.function_id, "items": items}
+            if chunks_needed:
+                frame["chunks"] = chunks_needed
+            await w.conn.send(frame)
         except Exception:
             # connection died mid-send: requeue, the watcher will clean up
             for rec in batch:
