@@ -69,6 +69,14 @@ async def process_output_item(item: dict, client: Any) -> Any:
     data = item.get("data")
     if item.get("data_blob"):
         data = client.blob_store.get(item["data_blob"])
+    if item.get("out_chunk") and item.get("status") == GENERIC_STATUS_SUCCESS:
+        # value shared inside a frame-level output chunk (worker fast path)
+        import pickle as _pickle
+
+        chunk_data = item.get("chunk_data")
+        if chunk_data is None:
+            raise ExecutionError("output chunk bytes missing from response")
+        return _pickle.loads(chunk_data)[item.get("out_ci", 0)]
     if (
         data is not None
         and item.get("status") == GENERIC_STATUS_SUCCESS
@@ -118,7 +126,7 @@ async def await_output_item(
     svc = client.svc
     if _is_inproc(svc):
         rec = await svc.function_wait_output(call_id, idx, timeout)
-        return {
+        item = {
             "idx": idx,
             "status": rec.status,
             "data": rec.output,
@@ -126,6 +134,12 @@ async def await_output_item(
             "format": rec.output_format,
             "exc": rec.exc_repr,
         }
+        if rec.out_chunk:
+            chunk = svc.out_chunks.get(rec.out_chunk)
+            item["out_chunk"] = rec.out_chunk
+            item["out_ci"] = rec.out_ci
+            item["chunk_data"] = chunk["data"] if chunk else None
+        return item
     deadline = None if timeout is None else time.monotonic() + timeout
     while True:
         remaining = 55.0 if deadline is None else min(55.0, deadline - time.monotonic())

@@ -135,9 +135,23 @@ async def map_invocation(
     next_output_idx = 0
     ordering_buffer: dict[int, Any] = {}
 
+    out_chunk_cache: dict[str, list] = {}
+
     async def decode(out: dict) -> Any:
-        if out["status"] == GENERIC_STATUS_SUCCESS and not return_exceptions:
-            return await process_output_item(out, client)
+        cid = out.get("out_chunk")
+        if cid is not None and out["status"] == GENERIC_STATUS_SUCCESS:
+            values = out_chunk_cache.get(cid)
+            if values is None:
+                import pickle as _pickle
+
+                data = out.get("chunk_data")
+                if data is None:
+                    return await process_output_item(out, client)  # raises missing-chunk
+                values = _pickle.loads(data)
+                out_chunk_cache[cid] = values
+                if len(out_chunk_cache) > 64:
+                    out_chunk_cache.pop(next(iter(out_chunk_cache)))
+            return values[out.get("out_ci", 0)]
         try:
             return await process_output_item(out, client)
         except BaseException as exc:

@@ -580,7 +580,7 @@ class WorkerRuntime:
         loop = asyncio.get_running_loop()
         from .execution_context import _current_function_call_id, _current_input_id
 
-        def run_all() -> list[dict]:
+        def run_all() -> Any:
             results: list[dict] = []
             try:
                 fn = frt.get_callable(items[0].get("method", ""))
@@ -595,7 +595,11 @@ class WorkerRuntime:
                     for item in items
                 ]
             if inspect.iscoroutinefunction(fn) or inspect.isasyncgenfunction(fn) or inspect.isgeneratorfunction(fn):
-                return []  # signal: fall back to the general path
+                return None  # signal: fall back to the general path
+            # happy path: collect raw values and pickle the whole frame once
+            # (mirrors the input chunking; one serialize per ~64 outputs)
+            values: list = []
+            value_tokens: list[str] = []
             for item in items:
                 if item["token"] in self._abandoned:
                     self._abandoned.discard(item["token"])
@@ -605,13 +609,8 @@ class WorkerRuntime:
                 tok_c = _current_function_call_id.set(call_id)
                 try:
                     args, kwargs = self._resolve_item_args(item)
-                    result = fn(*args, **kwargs)
-                    results.append(
-                        self._make_output(
-                            item["token"], frt.function_id, GENERIC_STATUS_SUCCESS,
-                            serialize(result), DataFormat.PICKLE,
-                        )
-                    )
+                    values.append(fn(*args, **kwargs))
+                    value_tokens.append(item["token"])
                 except BaseException as exc:
                     results.append(
                         self._make_output(
@@ -623,15 +622,30 @@ class WorkerRuntime:
                 finally:
                     _current_input_id.reset(tok_i)
                     _current_function_call_id.reset(tok_c)
+            if value_tokens:
+                from ._serialize_chunk import serialize_value_chunk
+
+                chunk_data, inline = serialize_value_chunk(values)
+                if inline is not None:
+                    # tiny frames or unpicklable-in-bulk: per-item fallback
+                    for token, data in zip(value_tokens, inline):
+                        results.append(
+                            self._make_output(
+                                token, frt.function_id, GENERIC_STATUS_SUCCESS,
+                                data, DataFormat.PICKLE,
+                            )
+                        )
+                else:
+                    return (results, value_tokens, chunk_data)
             return results
 
         app_tok = _app_id_var.set(frt.app_id)
         try:
             async with frt.sem:  # one slot per chunk-thread => <= max_concurrent items running
-                results = await loop.run_in_executor(self.executor, run_all)
+                ret = await loop.run_in_executor(self.executor, run_all)
         finally:
             _app_id_var.reset(app_tok)
-        if not results and items:
+        if ret is None:
             # not a plain sync function after all: general path per item
             for item in items:
                 task = asyncio.get_running_loop().create_task(self._run_input(frt, item))
@@ -640,7 +654,25 @@ class WorkerRuntime:
             return
         for item in items:
             self._running.pop(item["token"], None)
-        self._outbox.extend(results)
+        if isinstance(ret, tuple):
+            results, value_tokens, chunk_data = ret
+            if results:
+                self._outbox.extend(results)
+            try:
+                await self.conn.send(
+                    {
+                        "t": "outputs_chunk",
+                        "function_id": frt.function_id,
+                        "tokens": value_tokens,
+                        "data": chunk_data,
+                    }
+                )
+            except Exception:
+                return
+            if self._outbox and not self._outbox_flush_scheduled:
+                await self._flush_outbox()
+            return
+        self._outbox.extend(ret)
         if not self._outbox_flush_scheduled:
             await self._flush_outbox()
 

@@ -141,6 +141,8 @@ class InputRecord:
     cluster: Optional[dict] = None  # gang identity: {rank, size, cluster_id, master_port}
     chunk_id: Optional[str] = None  # shared chunk payload (map fan-out fast path)
     chunk_index: int = 0
+    out_chunk: Optional[str] = None  # shared output chunk (worker fast path)
+    out_ci: int = 0
 
     @property
     def token(self) -> str:
@@ -220,6 +222,8 @@ class CallRecord:
         exc_repr: Optional[str],
         retry_count: int,
         output_blob: Optional[str] = None,
+        out_chunk: Optional[str] = None,
+        out_ci: int = 0,
     ) -> bool:
         """Record a final output for input idx. Returns False on stale/dup
         delivery (parity: dedup by (idx, retry_count),
@@ -232,6 +236,8 @@ class CallRecord:
         rec.status = status
         rec.output = output
         rec.output_blob = output_blob
+        rec.out_chunk = out_chunk
+        rec.out_ci = out_ci
         rec.output_format = output_format
         rec.exc_repr = exc_repr
         rec.final = True

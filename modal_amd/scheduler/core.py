@@ -100,6 +100,10 @@ class Scheduler:
         self.functions: dict[str, FunctionDef] = {}
         self.function_names: dict[tuple[str, str, str], str] = {}  # (env, app_name, fn) -> fu-id
         self.calls: dict[str, CallRecord] = {}
+        # shared output chunks: chunk_id -> {"data": bytes, "refs": int}
+        self.out_chunks: dict[str, dict] = {}
+        self._out_chunk_order: list[str] = []
+        self._out_chunk_seq = 0
         self.services = Services()
         self.blob_store = BlobStore(os.path.join(self.run_dir, "blobs"))
         from .sandboxes import SandboxService
@@ -467,16 +471,32 @@ class Scheduler:
                 except asyncio.QueueEmpty:
                     break
                 out.append(self._output_item(record, idx))
+        # attach shared output-chunk bytes once per chunk per response
+        chunks_seen: set[str] = set()
+        for item in out:
+            cid = item.get("out_chunk")
+            if cid and cid not in chunks_seen:
+                chunks_seen.add(cid)
+                chunk = self.out_chunks.get(cid)
+                if chunk is not None:
+                    item["chunk_data"] = chunk["data"]
         if clear_on_success:
             for item in out:
                 rec = record.inputs.get(item["idx"])
                 if rec is not None:
                     rec.payload = b""  # release memory; result already extracted
+                cid = item.get("out_chunk")
+                if cid:
+                    chunk = self.out_chunks.get(cid)
+                    if chunk is not None:
+                        chunk["refs"] -= 1
+                        if chunk["refs"] <= 0:
+                            self.out_chunks.pop(cid, None)
         return out
 
     def _output_item(self, record: CallRecord, idx: int) -> dict:
         rec = record.inputs[idx]
-        return {
+        item = {
             "idx": idx,
             "input_id": rec.input_id,
             "status": rec.status,
@@ -486,6 +506,10 @@ class Scheduler:
             "exc": rec.exc_repr,
             "retry_count": rec.retry_count,
         }
+        if rec.out_chunk:
+            item["out_chunk"] = rec.out_chunk
+            item["out_ci"] = rec.out_ci
+        return item
 
     async def function_call_cancel(
         self, function_call_id: str, terminate_containers: bool = False
@@ -518,6 +542,16 @@ class Scheduler:
         return record
 
     # -- worker callbacks --------------------------------------------------
+    def register_out_chunk(self, data: bytes, refs: int) -> str:
+        self._out_chunk_seq += 1
+        chunk_id = f"oc{self._out_chunk_seq}"
+        self.out_chunks[chunk_id] = {"data": data, "refs": refs}
+        self._out_chunk_order.append(chunk_id)
+        while len(self._out_chunk_order) > 20_000:  # stale-delivery leak bound
+            old = self._out_chunk_order.pop(0)
+            self.out_chunks.pop(old, None)
+        return chunk_id
+
     def on_worker_output(
         self,
         call_id: str,
@@ -528,6 +562,8 @@ class Scheduler:
         output_format: int,
         exc_repr: Optional[str],
         output_blob: Optional[str] = None,
+        out_chunk: Optional[str] = None,
+        out_ci: int = 0,
     ) -> None:
         record = self.calls.get(call_id)
         if record is None:
@@ -551,7 +587,10 @@ class Scheduler:
             if rec.internal_failures <= MAX_INTERNAL_FAILURE_COUNT and not rec.cancelled:
                 self.pool.enqueue(rec, front=True)
                 return
-        self.finalize_input(rec, status, output, output_format, exc_repr, retry_count, output_blob)
+        self.finalize_input(
+            rec, status, output, output_format, exc_repr, retry_count, output_blob,
+            out_chunk, out_ci,
+        )
 
     def finalize_input(
         self,
@@ -562,11 +601,16 @@ class Scheduler:
         exc_repr: Optional[str],
         retry_count: int,
         output_blob: Optional[str] = None,
+        out_chunk: Optional[str] = None,
+        out_ci: int = 0,
     ) -> None:
         record = self.calls.get(rec.call_id)
         if record is None:
             return
-        record.post_output(rec.idx, status, output, output_format, exc_repr, retry_count, output_blob)
+        record.post_output(
+            rec.idx, status, output, output_format, exc_repr, retry_count, output_blob,
+            out_chunk, out_ci,
+        )
 
     def on_generator_data(self, msg: dict) -> None:
         call_id, idx_s, _ = msg["token"].rsplit(":", 2)

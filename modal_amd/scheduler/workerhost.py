@@ -272,6 +272,8 @@ class WorkerPool:
                 return
             if kind == "outputs":
                 self._on_outputs(handle, msg)
+            elif kind == "outputs_chunk":
+                self._on_outputs_chunk(handle, msg)
             elif kind == "gen_data":
                 self.scheduler.on_generator_data(msg)
             elif kind == "hb":
@@ -331,6 +333,33 @@ class WorkerPool:
                 output_format=item.get("format", 0),
                 exc_repr=item.get("exc"),
                 output_blob=item.get("data_blob"),
+            )
+        self._dispatch_wake.set()
+
+    def _on_outputs_chunk(self, handle: WorkerHandle, msg: dict) -> None:
+        """One pickled value-list covering many outputs (worker fast path)."""
+        tokens = msg["tokens"]
+        chunk_id = self.scheduler.register_out_chunk(msg["data"], len(tokens))
+        fdef_id = msg.get("function_id")
+        from .calls import GENERIC_STATUS_SUCCESS
+
+        for ci, token in enumerate(tokens):
+            handle.inflight.pop(token, None)
+            call_id, idx_s, retry_s = token.rsplit(":", 2)
+            if fdef_id:
+                cnt = handle.outstanding.get(fdef_id, 0)
+                if cnt > 0:
+                    handle.outstanding[fdef_id] = cnt - 1
+            self.scheduler.on_worker_output(
+                call_id=call_id,
+                idx=int(idx_s),
+                retry_count=int(retry_s),
+                status=GENERIC_STATUS_SUCCESS,
+                output=None,
+                output_format=0,
+                exc_repr=None,
+                out_chunk=chunk_id,
+                out_ci=ci,
             )
         self._dispatch_wake.set()
 
