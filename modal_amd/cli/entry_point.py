@@ -190,6 +190,40 @@ def app_logs(app_id: str) -> None:
         click.echo(entry.get("data", ""), nl=False)
 
 
+@app_cli.command(name="history")
+@click.argument("name")
+def app_history(name: str) -> None:
+    client = _get_client()
+    for row in synchronizer.run(client.svc.app_history(name=name)):
+        click.echo(f"v{row['version']}  {row['app_id']}  {time.ctime(row['deployed_at'])}")
+
+
+@app_cli.command(name="rollback")
+@click.argument("name")
+@click.option("--version", default=-1, type=int)
+def app_rollback(name: str, version: int) -> None:
+    client = _get_client()
+    out = synchronizer.run(client.svc.app_rollback(name=name, version=version))
+    click.echo(f"'{name}' now serves v{out['version']} ({out['app_id']})")
+
+
+@entrypoint_cli.group(name="billing")
+def billing_cli() -> None:
+    """Usage accounting (local GPU-seconds)."""
+
+
+@billing_cli.command(name="summary")
+def billing_summary() -> None:
+    from ..billing import usage_summary
+
+    for row in usage_summary():
+        gpu = "gpu" if row["gpu"] else "cpu"
+        click.echo(
+            f"{row['function']:30s} {gpu}  inputs={row['inputs']:<8d} "
+            f"runtime={row['runtime_seconds']:.2f}s"
+        )
+
+
 # ---- container (= worker) ----------------------------------------------
 
 

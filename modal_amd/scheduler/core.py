@@ -167,11 +167,36 @@ class Scheduler:
         app.deployment_name = name
         app.state = "deployed"
         app.ephemeral = False
+        previous = self.app_names.get((app.environment, name))
         self.app_names[(app.environment, name)] = app_id
+        history = self._extra.setdefault("deploy_history", {}).setdefault(
+            (app.environment, name), []
+        )
+        history.append({"app_id": app_id, "deployed_at": time.time(), "version": len(history) + 1})
         for tag, (object_id, _meta) in app.objects.items():
             if object_id.startswith("fu-"):
                 self.function_names[(app.environment, name, tag)] = object_id
-        return {"app_id": app_id, "url": f"local://{name}"}
+        return {"app_id": app_id, "url": f"local://{name}", "version": len(history)}
+
+    async def app_history(self, name: str, environment: str = "") -> list:
+        env = environment or self.default_environment
+        return list(self._extra.get("deploy_history", {}).get((env, name), []))
+
+    async def app_rollback(self, name: str, environment: str = "", version: int = -1) -> dict:
+        """Re-point the deployed name at an earlier version (parity:
+        modal app rollback / deployment history RPCs, reference runner.py:590)."""
+        env = environment or self.default_environment
+        history = self._extra.get("deploy_history", {}).get((env, name), [])
+        if not history:
+            raise NotFoundError(f"No deployment history for '{name}'")
+        entry = history[version if version != -1 else -2 if len(history) > 1 else -1]
+        app_id = entry["app_id"]
+        app = self._app(app_id)
+        self.app_names[(env, name)] = app_id
+        for tag, (object_id, _meta) in app.objects.items():
+            if object_id.startswith("fu-"):
+                self.function_names[(env, name, tag)] = object_id
+        return {"app_id": app_id, "version": entry["version"]}
 
     async def app_lookup(self, name: str, environment: str = "") -> dict:
         env = environment or self.default_environment

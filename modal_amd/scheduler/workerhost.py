@@ -104,10 +104,34 @@ class WorkerPool:
         self._server = await asyncio.start_unix_server(self._on_connect, path=self.socket_path)
         self._dispatch_task = asyncio.get_running_loop().create_task(self._dispatch_loop())
         self._retry_task = asyncio.get_running_loop().create_task(self._retry_loop())
+        self._health_task = asyncio.get_running_loop().create_task(self._health_loop())
+
+    async def _health_loop(self) -> None:
+        """Worker supervision beyond process liveness: a worker that stops
+        heartbeating (hung GPU kernel, wedged loop) is declared dead and its
+        in-flight inputs requeue (the INTERNAL_FAILURE path). The MI355X
+        analog of the reference's server-side container health tracking."""
+        while True:
+            await asyncio.sleep(15.0)
+            now = time.time()
+            for w in list(self.workers.values()):
+                if not w.alive:
+                    continue
+                stale = now - w.last_heartbeat
+                if stale > 60.0 and w.inflight:
+                    self.scheduler.log(
+                        f"worker {w.worker_id} missed heartbeats for {stale:.0f}s; recycling"
+                    )
+                    try:
+                        if w.proc is not None:
+                            w.proc.kill()
+                    except Exception:
+                        pass
+                    await w.conn.close()  # triggers _watch_worker requeue
 
     async def stop(self) -> None:
         self._stopping = True
-        for task in (self._dispatch_task, self._retry_task):
+        for task in (self._dispatch_task, self._retry_task, getattr(self, "_health_task", None)):
             if task is not None:
                 task.cancel()
         for w in list(self.workers.values()):

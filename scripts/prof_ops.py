@@ -57,7 +57,30 @@ def verify() -> None:
     print("sha256 verify ok")
 
 
+
+
+
+def bench_sha_leaf_sweep(total_mb: int = 512) -> None:
+    """Leaf-size sweep: occupancy/latency tradeoff for the tree hash."""
+    n = total_mb * 1024 * 1024
+    buf = torch.randint(0, 256, (n,), dtype=torch.uint8, device="cuda")
+    for leaf_kb in (4, 8, 16, 32):
+        leaf = leaf_kb * 1024
+        n_leaves = n // leaf
+        offsets = torch.arange(n_leaves, dtype=torch.int64) * leaf
+        lengths = torch.full((n_leaves,), leaf, dtype=torch.int64)
+        sha256_many_gpu(buf, offsets, lengths)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(5):
+            sha256_many_gpu(buf, offsets, lengths)
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / 5
+        print(f"leaf={leaf_kb:3d}KiB: {total_mb} MiB in {dt*1000:.2f} ms -> {total_mb/1024/dt:.2f} GiB/s")
+
+
 if __name__ == "__main__":
     verify()
+    bench_sha_leaf_sweep()
     bench_sha()
     bench_pack()
