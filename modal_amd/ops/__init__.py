@@ -40,6 +40,8 @@ def load_lib(required: bool = False) -> Optional[ctypes.CDLL]:
             lib = ctypes.CDLL(path)
             lib.ma_sha256_many.restype = ctypes.c_int
             lib.ma_sha256_many.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_int, ctypes.c_void_p]
+            lib.ma_sha256_many2.restype = ctypes.c_int
+            lib.ma_sha256_many2.argtypes = lib.ma_sha256_many.argtypes
             lib.ma_pack_segments.restype = ctypes.c_int
             lib.ma_pack_segments.argtypes = (
                 [ctypes.c_void_p] * 5 + [ctypes.c_int, ctypes.c_long, ctypes.c_void_p]

@@ -74,10 +74,14 @@ def _tree_sha256_gpu(data: Buffer) -> Optional[bytes]:
     return _root_digest(n, out.cpu().numpy().tobytes())
 
 
-def sha256_many_gpu(buf: "object", offsets: "object", lengths: "object") -> "object":
+def sha256_many_gpu(
+    buf: "object", offsets: "object", lengths: "object", ilp: int = 1
+) -> "object":
     """Hash arbitrary (offset, length) slices of a GPU-resident uint8 tensor.
 
     Returns an [n, 32] uint8 CUDA tensor of standard SHA-256 digests.
+    ilp=2 uses the dual-chain kernel (two messages per lane, interleaved
+    rounds) — wins when occupancy is low (few messages).
     """
     lib = load_lib(required=True)
     import torch
@@ -87,7 +91,8 @@ def sha256_many_gpu(buf: "object", offsets: "object", lengths: "object") -> "obj
     offsets_d = offsets.to(device="cuda", dtype=torch.int64)
     lengths_d = lengths.to(device="cuda", dtype=torch.int64)
     out = torch.empty((n, 32), dtype=torch.uint8, device="cuda")
-    rc = lib.ma_sha256_many(
+    fn = lib.ma_sha256_many2 if ilp == 2 else lib.ma_sha256_many
+    rc = fn(
         buf.data_ptr(), offsets_d.data_ptr(), lengths_d.data_ptr(), out.data_ptr(),
         n, torch.cuda.current_stream().cuda_stream,
     )

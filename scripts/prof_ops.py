@@ -69,14 +69,15 @@ def bench_sha_leaf_sweep(total_mb: int = 512) -> None:
         n_leaves = n // leaf
         offsets = torch.arange(n_leaves, dtype=torch.int64) * leaf
         lengths = torch.full((n_leaves,), leaf, dtype=torch.int64)
-        sha256_many_gpu(buf, offsets, lengths)
-        torch.cuda.synchronize()
-        t0 = time.perf_counter()
-        for _ in range(5):
-            sha256_many_gpu(buf, offsets, lengths)
-        torch.cuda.synchronize()
-        dt = (time.perf_counter() - t0) / 5
-        print(f"leaf={leaf_kb:3d}KiB: {total_mb} MiB in {dt*1000:.2f} ms -> {total_mb/1024/dt:.2f} GiB/s")
+        for ilp in (1, 2):
+            sha256_many_gpu(buf, offsets, lengths, ilp=ilp)
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(5):
+                sha256_many_gpu(buf, offsets, lengths, ilp=ilp)
+            torch.cuda.synchronize()
+            dt = (time.perf_counter() - t0) / 5
+            print(f"leaf={leaf_kb:3d}KiB ilp={ilp}: {total_mb} MiB in {dt*1000:.2f} ms -> {total_mb/1024/dt:.2f} GiB/s")
 
 
 if __name__ == "__main__":

@@ -150,3 +150,27 @@ def test_content_digests_batch_gpu_matches_cpu():
     got = content_digests_batch(buffers)
     expect = [content_digest(b) for b in buffers]
     assert got == expect
+
+
+def test_sha256_ilp2_matches_hashlib():
+    torch = _require_gpu()
+    from modal_amd.ops.hashing import sha256_many_gpu
+
+    rng = random.Random(13)
+    lengths = [0, 1, 63, 64, 65, 4096, 16384, 16385] + [rng.randrange(0, 40_000) for _ in range(21)]
+    blob = bytes(rng.randrange(256) for _ in range(sum(lengths)))
+    offsets, off = [], 0
+    for ln in lengths:
+        offsets.append(off)
+        off += ln
+    buf = torch.frombuffer(bytearray(blob), dtype=torch.uint8).cuda()
+    out = sha256_many_gpu(
+        buf,
+        torch.tensor(offsets, dtype=torch.int64),
+        torch.tensor(lengths, dtype=torch.int64),
+        ilp=2,
+    )
+    torch.cuda.synchronize()
+    got = out.cpu().numpy().tobytes()
+    for i, (o, ln) in enumerate(zip(offsets, lengths)):
+        assert got[i * 32 : (i + 1) * 32] == hashlib.sha256(blob[o : o + ln]).digest(), i
