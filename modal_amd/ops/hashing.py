@@ -23,10 +23,10 @@ from typing import Optional, Union
 from . import gpu_available, load_lib
 
 # Leaf size trades tree overhead against GPU parallelism: SHA-256 is serial
-# within a leaf, so leaves are the unit of parallelism. 16 KiB gives 4x the
-# workgroups of the reference's 64 KiB streaming chunk (hash_utils.py:11) —
-# measured latency-bound at 64 KiB (profiles/README.md).
-LEAF_SIZE = 16 * 1024
+# within a leaf, so leaves are the unit of parallelism. Swept 4/8/16/32 KiB
+# on MI355X (profiles/README.md): 8 KiB wins (911 GiB/s at 512 MiB) — enough
+# waves to hide the round-dependency chain, padding overhead still small.
+LEAF_SIZE = 8 * 1024
 TREE_DOMAIN = b"modal-amd-tree-v1"
 GPU_MIN_BYTES = 8 * 1024 * 1024  # below this, CPU wins (kernel+copy overhead)
 
