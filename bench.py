@@ -136,6 +136,9 @@ def main() -> None:
 
     # ---- rank 0: client + scheduler -----------------------------------
     run_dir = _bench_run_dir()
+    import shutil
+
+    shutil.rmtree(run_dir, ignore_errors=True)  # stale sockets from prior N
     os.makedirs(run_dir, exist_ok=True)
     os.environ["MODAL_AMD_RUN_DIR"] = run_dir
 

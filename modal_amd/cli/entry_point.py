@@ -533,6 +533,52 @@ def launch(template: Optional[str]) -> None:
     click.echo("Templates available locally: jupyter (modal-amd launch jupyter)")
 
 
+@entrypoint_cli.command(name="curl")
+@click.argument("url")
+@click.option("-X", "--request", "method", default="GET")
+@click.option("-d", "--data", default=None)
+def curl(url: str, method: str, data: Optional[str]) -> None:
+    """Call a deployed web endpoint (parity: modal curl)."""
+    import urllib.request
+
+    req = urllib.request.Request(
+        url, data=data.encode() if data else None, method=method.upper()
+    )
+    try:
+        with urllib.request.urlopen(req, timeout=60) as resp:
+            click.echo(resp.read().decode("utf-8", errors="replace"))
+    except Exception as exc:
+        click.echo(f"request failed: {exc}", err=True)
+        sys.exit(1)
+
+
+@entrypoint_cli.command(name="setup")
+def setup() -> None:
+    """First-time setup (parity: modal setup). Local runtime needs no auth."""
+    click.echo("modal-amd runs entirely on this node; no account setup needed.")
+    click.echo("Optional: set MODAL_AMD_RUN_DIR and run `modal-amd daemon` for a")
+    click.echo("persistent scheduler that other processes attach to.")
+
+
+@entrypoint_cli.group(name="cluster")
+def cluster_cli() -> None:
+    """Inspect gang-scheduled (clustered) runs."""
+
+
+@cluster_cli.command(name="info")
+def cluster_info() -> None:
+    client = _get_client()
+    svc = client.svc
+    if hasattr(svc, "pool"):
+        gpus = [w for w in svc.pool.workers.values() if w.has_gpu]
+        click.echo(f"workers: {len(svc.pool.workers)} ({len(gpus)} GPU-pinned)")
+        mesh = svc._extra.get("mesh") or {}
+        if mesh.get("formed"):
+            click.echo(f"device mesh: formed, backend={mesh.get('backend')}, ranks={mesh.get('ranks')}")
+        else:
+            click.echo("device mesh: not formed")
+
+
 def main() -> None:
     try:
         entrypoint_cli(standalone_mode=True)

@@ -271,10 +271,17 @@ class WorkerRuntime:
         global RUNTIME
         RUNTIME = self
         self.loop = asyncio.get_running_loop()
-        try:
-            reader, writer = await asyncio.open_unix_connection(self.socket_path)
-        except (FileNotFoundError, ConnectionRefusedError):
-            return  # scheduler already gone: exit quietly
+        deadline = time.time() + (60.0 if self.external else 0.0)
+        while True:
+            try:
+                reader, writer = await asyncio.open_unix_connection(self.socket_path)
+                break
+            except (FileNotFoundError, ConnectionRefusedError):
+                # external workers (torchrun ranks) may race the scheduler's
+                # socket (re)creation; spawned workers exit quietly
+                if time.time() >= deadline:
+                    return
+                await asyncio.sleep(0.05)
         self.conn = Connection(reader, writer, self._handle, rpc_target=WorkerRPCTarget(self))
         self.conn.start()
         await self.conn.send(

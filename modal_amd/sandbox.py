@@ -387,6 +387,42 @@ class _Sandbox(_Object, type_kind="sandbox"):
             sandbox_id=self.object_id, op="exists", path=path
         )
 
+    async def watch(self, path: str, poll_interval: float = 0.5) -> AsyncGenerator[dict, None]:
+        """Poll-based file watch (parity: sandbox_fs watch API): yields
+        {"path", "event"} dicts for created/modified/deleted entries."""
+        if not self._is_hydrated:
+            await self.hydrate()
+        known: dict[str, float] = {}
+        first = True
+        while True:
+            try:
+                names = await self._client.svc.sandbox_fs_op(
+                    sandbox_id=self.object_id, op="ls", path=path
+                )
+            except Exception:
+                return
+            current: dict[str, float] = {}
+            for name in names:
+                try:
+                    st = await self._client.svc.sandbox_fs_op(
+                        sandbox_id=self.object_id, op="stat", path=f"{path}/{name}"
+                    )
+                    current[name] = st["mtime"]
+                except Exception:
+                    continue
+            if not first:
+                for name, mtime in current.items():
+                    if name not in known:
+                        yield {"path": f"{path}/{name}", "event": "created"}
+                    elif known[name] != mtime:
+                        yield {"path": f"{path}/{name}", "event": "modified"}
+                for name in known:
+                    if name not in current:
+                        yield {"path": f"{path}/{name}", "event": "deleted"}
+            known = current
+            first = False
+            await asyncio.sleep(poll_interval)
+
     # -- snapshots / tunnels ----------------------------------------------
     @live_method
     async def snapshot_filesystem(self, timeout: float = 55.0) -> Any:
