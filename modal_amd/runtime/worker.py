@@ -858,8 +858,11 @@ class WorkerRuntime:
         item = self._make_output(token, function_id, status, data, data_format, exc_repr)
         self._outbox.append(item)
         if not self._outbox_flush_scheduled:
+            # flush NOW (latency path: a lone unary result should not wait on
+            # a coalescing timer); the flag suppresses re-entrant sends so
+            # bursts posted in the same tick still batch into one frame
             self._outbox_flush_scheduled = True
-            asyncio.get_running_loop().call_later(OUTPUT_FLUSH_INTERVAL, self._flush_outbox_cb)
+            asyncio.get_running_loop().call_soon(self._flush_outbox_cb)
 
     def _flush_outbox_cb(self) -> None:
         self._outbox_flush_scheduled = False

@@ -512,9 +512,11 @@ class WorkerPool:
         from ..config import config
 
         n_gpus = self._gpu_count()
-        if fdef.needs_gpu:
-            return max(n_gpus, 1)
         configured = config.get("worker_count")
+        if fdef.needs_gpu:
+            # >1 worker per GPU is a legitimate MI355X shape (288 GB HBM per
+            # device easily hosts several concurrent payload processes)
+            return max(configured or 0, n_gpus, 1)
         if configured:
             return configured
         return min(max((os.cpu_count() or 4) // 2, 1), 8)
