@@ -31,6 +31,9 @@ if _WORLD_SIZE > 1 and "HIP_VISIBLE_DEVICES" not in os.environ:
     os.environ["CUDA_VISIBLE_DEVICES"] = str(_LOCAL_RANK)
 
 ITEMS_PER_GPU = 12_500  # x8 GPUs = the 100k-input config of BASELINE.json
+# 288 GB HBM3E per MI355X easily hosts several payload processes; 4 workers
+# per GPU measured 2x items/s vs 1 (profiles/README.md)
+WORKERS_PER_GPU = 4
 
 
 def _bench_run_dir() -> str:
@@ -91,6 +94,20 @@ def run_worker_rank(rank: int, world: int) -> None:
     runtime = WorkerRuntime()
     thread = threading.Thread(target=lambda: __import__("asyncio").run(runtime.run()), daemon=True)
     thread.start()
+    # extra workers for this rank's GPU (they inherit HIP_VISIBLE_DEVICES)
+    import subprocess
+    import sys as _sys
+
+    extra = []
+    for k in range(1, WORKERS_PER_GPU if torch.cuda.is_available() else 1):
+        env = dict(os.environ)
+        env["MODAL_AMD_WORKER_ID"] = str(1000 + rank + 100 * k)
+        extra.append(
+            subprocess.Popen(
+                [_sys.executable, "-m", "modal_amd.runtime.worker"], env=env,
+                start_new_session=True,
+            )
+        )
 
     device = torch.device("cuda:0") if torch.cuda.is_available() else None
     # start barrier (warmup done on rank 0), timed region, end barrier
@@ -103,6 +120,11 @@ def run_worker_rank(rank: int, world: int) -> None:
     dist.all_reduce(elapsed, op=dist.ReduceOp.MAX)
     dist.barrier()
     dist.destroy_process_group()
+    for proc in extra:
+        try:
+            proc.terminate()
+        except Exception:
+            pass
 
 
 def main() -> None:
@@ -157,11 +179,13 @@ def main() -> None:
 
     # rank 0 contributes one worker for its own GPU (ranks>0 bring theirs)
     async def spawn_local_worker():
-        await scheduler.pool.spawn_worker(gpu_index=0 if has_gpu else None)
+        for _ in range(WORKERS_PER_GPU if has_gpu else 1):
+            await scheduler.pool.spawn_worker(gpu_index=0 if has_gpu else None)
 
     synchronizer.run(spawn_local_worker())
-    # wait until all N workers are connected
-    want_workers = n_gpus if world > 1 else 1
+    # wait until every rank's workers are connected
+    per_rank = WORKERS_PER_GPU if has_gpu else 1
+    want_workers = per_rank * (n_gpus if world > 1 else 1)
     deadline = time.time() + 180
     while len(scheduler.pool.workers) < want_workers:
         if time.time() > deadline:
