@@ -116,7 +116,7 @@ class FunctionDef:
         }
 
 
-@dataclass
+@dataclass(slots=True)
 class InputRecord:
     call_id: str
     idx: int
@@ -194,7 +194,8 @@ class CallRecord:
         rec = InputRecord(
             call_id=self.call_id,
             idx=idx,
-            input_id=new_id("input"),
+            # derived id: unique per runtime without a per-input entropy draw
+            input_id=f"in-{self.call_id[3:]}-{idx}",
             payload=payload,
             payload_blob=payload_blob,
             method_name=method_name,
@@ -255,6 +256,37 @@ class CallRecord:
             waiter.set_result(rec)
         self._check_done()
         return True
+
+    def post_outputs_bulk(self, triples: list, out_chunk: str) -> int:
+        """Bulk success delivery for a shared output chunk:
+        triples = [(idx, retry_count, chunk_index), ...]. One queue entry
+        covers the whole group. Returns the number actually recorded."""
+        ready: list[int] = []
+        now = time.time()
+        for idx, retry_count, ci in triples:
+            rec = self.inputs.get(idx)
+            if rec is None or rec.final or retry_count != rec.retry_count:
+                continue
+            rec.status = GENERIC_STATUS_SUCCESS
+            rec.out_chunk = out_chunk
+            rec.out_ci = ci
+            rec.final = True
+            rec.finished_at = now
+            if rec.chunk_id:
+                chunk = self.chunks.get(rec.chunk_id)
+                if chunk is not None:
+                    chunk["refs"] -= 1
+                    if chunk["refs"] <= 0:
+                        del self.chunks[rec.chunk_id]
+            ready.append(idx)
+            waiter = self._waiters.pop(idx, None)
+            if waiter is not None and not waiter.done():
+                waiter.set_result(rec)
+        if ready:
+            self.completed += len(ready)
+            self.output_ready.put_nowait(ready)
+            self._check_done()
+        return len(ready)
 
     async def wait_output(self, idx: int, timeout: Optional[float] = None) -> InputRecord:
         rec = self.inputs[idx]

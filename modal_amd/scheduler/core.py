@@ -486,16 +486,18 @@ class Scheduler:
                 remaining = deadline - time.time()
                 if remaining <= 0:
                     break
-                idx = await asyncio.wait_for(record.output_ready.get(), remaining)
+                entry = await asyncio.wait_for(record.output_ready.get(), remaining)
             except asyncio.TimeoutError:
                 break
-            out.append(self._output_item(record, idx))
+            for idx in entry if isinstance(entry, list) else (entry,):
+                out.append(self._output_item(record, idx))
             while len(out) < max_values:
                 try:
-                    idx = record.output_ready.get_nowait()
+                    entry = record.output_ready.get_nowait()
                 except asyncio.QueueEmpty:
                     break
-                out.append(self._output_item(record, idx))
+                for idx in entry if isinstance(entry, list) else (entry,):
+                    out.append(self._output_item(record, idx))
         # attach shared output-chunk bytes once per chunk per response
         chunks_seen: set[str] = set()
         for item in out:

@@ -361,30 +361,24 @@ class WorkerPool:
         self._dispatch_wake.set()
 
     def _on_outputs_chunk(self, handle: WorkerHandle, msg: dict) -> None:
-        """One pickled value-list covering many outputs (worker fast path)."""
+        """One pickled value-list covering many outputs (worker fast path);
+        bulk bookkeeping grouped per call."""
         tokens = msg["tokens"]
         chunk_id = self.scheduler.register_out_chunk(msg["data"], len(tokens))
         fdef_id = msg.get("function_id")
-        from .calls import GENERIC_STATUS_SUCCESS
-
+        if fdef_id:
+            cnt = handle.outstanding.get(fdef_id, 0)
+            handle.outstanding[fdef_id] = max(cnt - len(tokens), 0)
+        inflight_pop = handle.inflight.pop
+        by_call: dict[str, list] = {}
         for ci, token in enumerate(tokens):
-            handle.inflight.pop(token, None)
+            inflight_pop(token, None)
             call_id, idx_s, retry_s = token.rsplit(":", 2)
-            if fdef_id:
-                cnt = handle.outstanding.get(fdef_id, 0)
-                if cnt > 0:
-                    handle.outstanding[fdef_id] = cnt - 1
-            self.scheduler.on_worker_output(
-                call_id=call_id,
-                idx=int(idx_s),
-                retry_count=int(retry_s),
-                status=GENERIC_STATUS_SUCCESS,
-                output=None,
-                output_format=0,
-                exc_repr=None,
-                out_chunk=chunk_id,
-                out_ci=ci,
-            )
+            by_call.setdefault(call_id, []).append((int(idx_s), int(retry_s), ci))
+        for call_id, triples in by_call.items():
+            record = self.scheduler.calls.get(call_id)
+            if record is not None:
+                record.post_outputs_bulk(triples, chunk_id)
         self._dispatch_wake.set()
 
     # -- dispatch --------------------------------------------------------
