@@ -174,3 +174,31 @@ def test_sha256_ilp2_matches_hashlib():
     got = out.cpu().numpy().tobytes()
     for i, (o, ln) in enumerate(zip(offsets, lengths)):
         assert got[i * 32 : (i + 1) * 32] == hashlib.sha256(blob[o : o + ln]).digest(), i
+
+
+def test_sandbox_gpu_visible(client):
+    _require_gpu()
+    import modal_amd as modal
+
+    sb = modal.Sandbox.create(
+        "python3", "-c", "import torch; print(torch.cuda.is_available())", gpu=1
+    )
+    rc = sb.wait(raise_on_termination=False)
+    assert rc == 0, sb.stderr.read()
+    assert sb.stdout.read().strip() == "True"
+
+
+def test_volume_gpu_hash_upload(client):
+    """Config 4: volume batch_upload of a large file rides the GPU hash."""
+    _require_gpu()
+    import io
+    import os as _os
+
+    import modal_amd as modal
+
+    payload = _os.urandom(1 << 20) * 24  # 24 MiB -> 3 blocks, GPU-batched hash
+    with modal.Volume.ephemeral() as vol:
+        with vol.batch_upload() as batch:
+            batch.put_file(io.BytesIO(payload), "model.bin")
+        data = b"".join(chunk for chunk in vol.read_file("model.bin"))
+        assert data == payload
