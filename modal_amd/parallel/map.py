@@ -96,6 +96,13 @@ async def map_invocation(
             payload = _chunk_serialize(chunk_buf)
             chunk_id = f"{call_id}.c{chunk_seq}"
             chunk_seq += 1
+            if len(payload) > 2 * 1024 * 1024:
+                # big-item chunks spill to the CAS (parity: the 2 MiB inline
+                # payload limit, blob_utils.py:36) — workers read from the
+                # shared store instead of the wire
+                store = client.blob_store
+                if store is not None:
+                    payload = {"blob": store.put(payload)}
             chunks[chunk_id] = payload
             for ci in range(len(chunk_buf)):
                 item = {"chunk": chunk_id, "ci": ci}
@@ -109,13 +116,18 @@ async def map_invocation(
                 )
                 items, chunks = [], {}
 
+        approx_bytes = 0
         try:
             async for args, extra_kwargs in _iterate_maybe_async(input_iter):
                 kw = {**kwargs_common, **extra_kwargs} if extra_kwargs else kwargs_common
                 chunk_buf.append((args, kw))
+                for a in args:
+                    if type(a) in (bytes, bytearray, str):
+                        approx_bytes += len(a)
                 total_inputs += 1
                 await sem.acquire()
-                if len(chunk_buf) >= CHUNK_ITEMS:
+                if len(chunk_buf) >= CHUNK_ITEMS or approx_bytes > 4 * 1024 * 1024:
+                    approx_bytes = 0
                     await flush_chunk()
             await flush_chunk()
             if items:

@@ -23,8 +23,12 @@ def serialize_value_chunk(values: list) -> tuple[Optional[bytes], Optional[list]
 
             return None, [serialize(v) for v in values]
     try:
-        return pickle.dumps(values, 4), None
+        data = pickle.dumps(values, 4)
     except Exception:
         from .._serialization import serialize
 
         return None, [serialize(v) for v in values]
+    if len(data) > 2 * 1024 * 1024:
+        # oversized frames go per-item so each output can blob-offload
+        return None, [pickle.dumps(v, 4) for v in values]
+    return data, None

@@ -458,7 +458,10 @@ class WorkerRuntime:
                 raise RuntimeError(f"chunk {cid} not delivered to this worker")
             decoded = entry["decoded"]
             if decoded is None:
-                kind, decoded = deserialize(entry["raw"])
+                raw = entry["raw"]
+                if isinstance(raw, dict):  # CAS-spilled chunk (>2 MiB)
+                    raw = self.blob_store.get(raw["blob"])
+                kind, decoded = deserialize(raw)
                 assert kind == "C"
                 entry["decoded"] = decoded
                 entry["raw"] = None

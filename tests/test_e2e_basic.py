@@ -187,3 +187,36 @@ def test_retries_with_state_file(client, run_dir):
     with app.run(client=client):
         assert flaky.remote(marker) == "ok"
         assert int(open(marker).read()) == 3
+
+
+def test_map_large_payloads(client):
+    """Per-item payloads in the MiB range spill chunks to the CAS."""
+    import os as _os
+
+    app = modal.App("test-big-map")
+
+    @app.function()
+    def digest(blob):
+        import hashlib
+
+        return hashlib.sha256(blob).hexdigest()
+
+    blobs = [_os.urandom(1 << 20) for _ in range(6)]  # 6 x 1 MiB
+    with app.run(client=client):
+        import hashlib
+
+        out = list(digest.map(blobs))
+        assert out == [hashlib.sha256(b).hexdigest() for b in blobs]
+
+
+def test_map_large_outputs(client):
+    app = modal.App("test-big-out")
+
+    @app.function()
+    def inflate(n):
+        return bytes([n % 256]) * (3 * 1024 * 1024)  # 3 MiB result
+
+    with app.run(client=client):
+        out = list(inflate.map(range(3)))
+        for i, blob in enumerate(out):
+            assert blob == bytes([i % 256]) * (3 * 1024 * 1024)
