@@ -687,15 +687,22 @@ class WorkerRuntime:
             await self._flush_outbox()
 
     async def _execute(self, frt: FunctionRuntime, fn: Any, args: tuple, kwargs: dict) -> Any:
+        from ..utils.tracing import enabled as trace_enabled, trace_range
+
         if inspect.iscoroutinefunction(fn):
             coro = fn(*args, **kwargs)
             if frt.timeout:
                 return await asyncio.wait_for(coro, frt.timeout)
             return await coro
         ctx = contextvars.copy_context()
-        fut = asyncio.get_running_loop().run_in_executor(
-            self.executor, lambda: ctx.run(fn, *args, **kwargs)
-        )
+
+        def invoke() -> Any:
+            if trace_enabled():
+                with trace_range(f"modal_amd::{frt.name}"):
+                    return ctx.run(fn, *args, **kwargs)
+            return ctx.run(fn, *args, **kwargs)
+
+        fut = asyncio.get_running_loop().run_in_executor(self.executor, invoke)
         if frt.timeout:
             return await asyncio.wait_for(fut, frt.timeout)
         return await fut
