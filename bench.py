@@ -205,6 +205,17 @@ def main() -> None:
 
     items_per_step = args.items_per_gpu * n_gpus
 
+    def run_map_step() -> None:
+        """One map step driven on the framework loop (no per-item sync bridge)."""
+
+        async def _consume() -> None:
+            n = 0
+            async for _ in item_fn.map.aio(range(items_per_step), order_outputs=False):
+                n += 1
+            assert n == items_per_step
+
+        synchronizer.run(_consume())
+
     ctx = app.run(client=client)
     ctx.__enter__()
     try:
@@ -220,8 +231,7 @@ def main() -> None:
             lat.sort()
             p50_ms = lat[len(lat) // 2]
         for _ in range(args.warmup):
-            out = list(item_fn.map(range(items_per_step), order_outputs=False))
-            assert len(out) == items_per_step
+            run_map_step()
 
         # ---- timed region ---------------------------------------------
         if dist is not None:
@@ -230,8 +240,7 @@ def main() -> None:
             torch.cuda.synchronize()
         t0 = time.perf_counter()
         for _ in range(args.steps):
-            out = list(item_fn.map(range(items_per_step), order_outputs=False))
-            assert len(out) == items_per_step
+            run_map_step()
         if dist is not None:
             dist.barrier()
         if has_gpu:

@@ -60,6 +60,9 @@ class RPCAdapter:
     """Whitelisted surface exposed to socket peers (workers, sandboxes, CLI)."""
 
     _ALLOWED = {
+        "app_create", "app_publish", "app_heartbeat", "app_client_disconnect", "app_stop",
+        "app_list", "app_history", "app_rollback", "app_set_objects",
+        "function_create", "function_update", "function_update_autoscaler",
         "queue_get_or_create", "queue_put", "queue_get", "queue_len", "queue_clear",
         "queue_peek", "queue_delete",
         "dict_get_or_create", "dict_update", "dict_get", "dict_pop", "dict_contains",
