@@ -18,7 +18,7 @@ async def usage_summary(client: Any = None) -> list[dict]:
 
     client = client or await _Client.from_env()
     svc = client.svc
-    if not hasattr(svc, "calls"):
+    if getattr(svc, "is_proxy", False):
         return []
     per_function: dict[str, dict] = {}
     for record in svc.calls.values():

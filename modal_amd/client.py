@@ -54,6 +54,10 @@ def map_remote_error(exc: RemoteRPCError) -> Exception:
 class SchedulerProxy:
     """Socket-backed scheduler: async methods forwarded as RPC frames."""
 
+    #: class attribute (normal lookup beats __getattr__): lets callers
+    #: distinguish the proxy from the in-process Scheduler
+    is_proxy = True
+
     def __init__(self, conn: Connection):
         self._conn = conn
 

@@ -111,7 +111,7 @@ async def process_output_item(item: dict, client: Any) -> Any:
 
 
 def _is_inproc(svc: Any) -> bool:
-    return hasattr(svc, "pool")
+    return not getattr(svc, "is_proxy", False)
 
 
 async def await_output_item(

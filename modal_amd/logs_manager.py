@@ -20,7 +20,7 @@ def fetch_app_logs(app_id: str, client: Any = None) -> list[dict]:
     async def fetch() -> list[dict]:
         c = client or await _Client.from_env()
         svc = c.svc
-        if hasattr(svc, "apps"):
+        if not getattr(svc, "is_proxy", False):
             app = svc.apps.get(app_id)
             return list(app.logs) if app else []
         return []

@@ -176,7 +176,7 @@ class AppRunContext:
 
     def _start_log_stream(self) -> None:
         svc = self.client.svc
-        if not hasattr(svc, "apps"):
+        if getattr(svc, "is_proxy", False):
             return  # remote attach: log streaming handled by logs manager
         from .output import get_output_manager
 

@@ -55,6 +55,7 @@ class WorkerHandle:
         self.external = external
         self.alive = True
         self.last_heartbeat = time.time()
+        self.last_active = time.time()  # last input assignment (scaledown clock)
         # tokens of inputs currently assigned here, mapped to their records
         self.inflight: dict[str, InputRecord] = {}
         # per-function outstanding count (for credit computation)
@@ -106,6 +107,36 @@ class WorkerPool:
         self._retry_task = asyncio.get_running_loop().create_task(self._retry_loop())
         self._health_task = asyncio.get_running_loop().create_task(self._health_loop())
 
+    async def _scaledown_once(self) -> None:
+        """Reap idle workers beyond the warm floor (parity: scaledown_window
+        autoscaler setting, reference _functions.py:1195-1292)."""
+        window = min(
+            (f.scaledown_window for f in self.scheduler.functions.values() if f.scaledown_window),
+            default=60.0,
+        )
+        floor = max(
+            (f.min_containers + f.buffer_containers for f in self.scheduler.functions.values()),
+            default=0,
+        )
+        now = time.time()
+        idle = [
+            w
+            for w in self.workers.values()
+            if w.alive
+            and not w.draining
+            and not w.inflight
+            and not w.external  # torchrun-owned workers are not ours to reap
+            and now - w.last_active > window
+        ]
+        alive = sum(1 for w in self.workers.values() if w.alive)
+        excess = alive - max(floor, 1)
+        for w in idle[: max(excess, 0)]:
+            w.draining = True
+            try:
+                await w.conn.send({"t": "shutdown"})
+            except Exception:
+                pass
+
     async def _health_loop(self) -> None:
         """Worker supervision beyond process liveness: a worker that stops
         heartbeating (hung GPU kernel, wedged loop) is declared dead and its
@@ -113,6 +144,10 @@ class WorkerPool:
         analog of the reference's server-side container health tracking."""
         while True:
             await asyncio.sleep(15.0)
+            try:
+                await self._scaledown_once()
+            except Exception:
+                pass
             now = time.time()
             for w in list(self.workers.values()):
                 if not w.alive:
@@ -544,6 +579,7 @@ class WorkerPool:
                 )
                 w.functions_loaded.add(fdef.function_id)
                 w.defs_version[fdef.function_id] = fdef.definition_version
+            w.last_active = time.time()
             items = []
             chunks_needed: dict[str, bytes] = {}
             record = None

@@ -1,0 +1,103 @@
+"""Cross-process attach mode (daemon scheduler) + native core unit tests."""
+
+from __future__ import annotations
+
+import os
+import subprocess
+import sys
+import tempfile
+import time
+
+import pytest
+
+
+def test_shm_ring_roundtrip(tmp_path):
+    core = pytest.importorskip("modal_amd._core")
+    path = str(tmp_path / "ring")
+    writer = core.ShmRing(path, 1 << 20, True)
+    reader = core.ShmRing(path, 0, False)
+    assert writer.push(b"alpha")
+    assert writer.push(b"B" * 100_000)
+    assert reader.pop_all() == [b"alpha", b"B" * 100_000]
+    # wraparound across the 1 MiB boundary
+    blob = os.urandom(300_000)
+    for _ in range(24):
+        assert writer.push(blob)
+        assert reader.pop_all() == [blob]
+    # full-ring refusal, then drain
+    count = 0
+    while writer.push(b"x" * 100_000):
+        count += 1
+    assert count >= 9
+    assert len(reader.pop_all()) == count
+    assert writer.pending_bytes() == 0
+
+
+def test_pack_payloads_roundtrip():
+    core = pytest.importorskip("modal_amd._core")
+    items = [b"", b"a", os.urandom(1000), b"zz" * 5000]
+    blob = core.pack_payloads(items)
+    assert core.unpack_payloads(blob) == items
+
+
+def test_attach_to_daemon_scheduler(tmp_path):
+    """A second process's scheduler serves this client over the socket
+    (the cross-process wire contract: app create, function create, map,
+    queue ops)."""
+    run_dir = str(tmp_path / "daemon")
+    daemon = subprocess.Popen(
+        [sys.executable, "-m", "modal_amd.cli.entry_point", "daemon", "--run-dir", run_dir],
+        stdout=subprocess.PIPE,
+        stderr=subprocess.STDOUT,
+        start_new_session=True,
+    )
+    sock = os.path.join(run_dir, "scheduler.sock")
+    try:
+        deadline = time.time() + 30
+        while not os.path.exists(sock):
+            assert daemon.poll() is None, daemon.stdout.read().decode()
+            assert time.time() < deadline, "daemon socket never appeared"
+            time.sleep(0.05)
+        env = dict(os.environ)
+        env["MODAL_AMD_ATTACH_SOCKET"] = sock
+        script = """
+import modal_amd as modal
+
+app = modal.App("attach-test")
+
+@app.function()
+def double(x):
+    return x * 2
+
+with app.run():
+    assert double.remote(21) == 42
+    out = sorted(double.map(range(20), order_outputs=False))
+    assert out == [2 * x for x in range(20)]
+    with modal.Queue.ephemeral() as q:
+        q.put("cross-process")
+        assert q.get() == "cross-process"
+print("ATTACH_OK")
+"""
+        proc = subprocess.run(
+            [sys.executable, "-c", script], env=env, capture_output=True, text=True, timeout=120
+        )
+        assert proc.returncode == 0, proc.stdout + proc.stderr
+        assert "ATTACH_OK" in proc.stdout
+    finally:
+        daemon.terminate()
+        try:
+            daemon.wait(timeout=5)
+        except subprocess.TimeoutExpired:
+            daemon.kill()
+
+
+def test_file_pattern_matcher():
+    from modal_amd.file_pattern_matcher import FilePatternMatcher
+
+    m = FilePatternMatcher("*.pyc", "build/**", "!build/keep.txt")
+    assert m("foo.pyc")
+    assert m("build/a/b.o")
+    assert not m("build/keep.txt")
+    assert not m("src/main.py")
+    m2 = FilePatternMatcher("**/__pycache__")
+    assert m2("a/b/__pycache__")
