@@ -172,7 +172,30 @@ class ImageService:
                     elif stripped.upper().startswith("WORKDIR "):
                         state.workdir = stripped.split(None, 1)[1]
             elif kind == "run_function":
-                log.append("run_function layer recorded (executed by runner at deploy)")
+                # build-time function execution in a child process rooted in
+                # the image dir (parity: reference Image.run_function builds)
+                payload_hex = layer.get("payload_hex", "")
+                script = (
+                    "import cloudpickle, sys\n"
+                    "fn, args, kwargs = cloudpickle.loads(bytes.fromhex(sys.argv[1]))\n"
+                    "fn(*args, **kwargs)\n"
+                )
+
+                def run_fn_layer() -> None:
+                    full_env = dict(os.environ)
+                    full_env.update(state.env)
+                    proc = subprocess.run(
+                        [sys.executable, "-c", script, payload_hex],
+                        cwd=state.root, env=full_env, capture_output=True, text=True,
+                        timeout=600,
+                    )
+                    log.append(f"run_function {layer.get('name')}:\n{proc.stdout}{proc.stderr}")
+                    if proc.returncode != 0:
+                        raise ExecutionError(
+                            f"Image.run_function({layer.get('name')}) failed:\n{proc.stderr[-2000:]}"
+                        )
+
+                await loop.run_in_executor(None, run_fn_layer)
             else:
                 log.append(f"unknown layer kind {kind!r} (ignored)")
         state.build_log = "\n".join(log)

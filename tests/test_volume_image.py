@@ -122,3 +122,26 @@ def test_mount_python_packages(client):
     m = modal.Mount.from_local_python_packages("modal_amd.utils")
     m.hydrate()
     assert m.object_id.startswith("mo-")
+
+
+def test_image_run_function_executes(client, tmp_path):
+    marker = str(tmp_path / "built-by-fn")
+
+    def build_step(path):
+        with open(path, "w") as f:
+            f.write("ran at build time")
+
+    img = modal.Image.debian_slim().run_function(build_step, marker)
+    img.hydrate()
+    assert open(marker).read() == "ran at build time"
+
+
+def test_image_run_function_failure_surfaces(client):
+    def bad_step():
+        raise RuntimeError("build exploded")
+
+    img = modal.Image.debian_slim().run_function(bad_step)
+    from modal_amd.exception import ExecutionError
+
+    with pytest.raises(ExecutionError, match="build exploded|run_function"):
+        img.hydrate()
