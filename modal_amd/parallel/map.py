@@ -32,6 +32,26 @@ OUTPUT_FETCH_MAX = 1024
 CHUNK_ITEMS = 64
 
 
+class BulkSemaphore:
+    """Counting semaphore with O(1) bulk acquire/release (the 1000-outstanding
+    backpressure cap acquires per chunk, not per item)."""
+
+    def __init__(self, value: int):
+        self._value = value
+        self._event = asyncio.Event()
+        self._event.set()
+
+    async def acquire(self, n: int = 1) -> None:
+        while self._value < n:
+            self._event.clear()
+            await self._event.wait()
+        self._value -= n
+
+    def release(self, n: int = 1) -> None:
+        self._value += n
+        self._event.set()
+
+
 async def _iterate_maybe_async(it: Any) -> AsyncGenerator[Any, None]:
     if hasattr(it, "__aiter__"):
         async for item in it:
@@ -63,7 +83,7 @@ async def map_invocation(
     resp = await svc.function_map(function_id=fn.object_id, kind="map")
     call_id = resp["function_call_id"]
     max_outstanding = resp.get("max_inputs_outstanding") or 1000
-    sem = asyncio.Semaphore(max_outstanding)
+    sem = BulkSemaphore(max_outstanding)
     pump_done = asyncio.Event()
     total_inputs = 0
     pump_error: list[BaseException] = []
@@ -74,8 +94,6 @@ async def map_invocation(
 
         chunk_buf: list = []
         chunk_seq = 0
-        items: list[dict] = []
-        chunks: dict[str, bytes] = {}
 
         import sys as _sys
 
@@ -90,12 +108,14 @@ async def map_invocation(
             return serialize_fast(("C", buf))
 
         async def flush_chunk() -> None:
-            nonlocal chunk_buf, chunk_seq, items, chunks
+            nonlocal chunk_buf, chunk_seq
             if not chunk_buf:
                 return
             payload = _chunk_serialize(chunk_buf)
             chunk_id = f"{call_id}.c{chunk_seq}"
             chunk_seq += 1
+            count = len(chunk_buf)
+            chunk_buf = []
             if len(payload) > 2 * 1024 * 1024:
                 # big-item chunks spill to the CAS (parity: the 2 MiB inline
                 # payload limit, blob_utils.py:36) — workers read from the
@@ -103,18 +123,13 @@ async def map_invocation(
                 store = client.blob_store
                 if store is not None:
                     payload = {"blob": store.put(payload)}
-            chunks[chunk_id] = payload
-            for ci in range(len(chunk_buf)):
-                item = {"chunk": chunk_id, "ci": ci}
-                if fn._method_name:
-                    item["method"] = fn._method_name
-                items.append(item)
-            chunk_buf = []
-            if len(items) >= PUT_BATCH_SIZE:
-                await svc.function_put_inputs(
-                    function_call_id=call_id, items=items, chunks=chunks
-                )
-                items, chunks = [], {}
+            await svc.function_put_chunk(
+                function_call_id=call_id,
+                chunk_id=chunk_id,
+                payload=payload,
+                count=count,
+                method=fn._method_name or "",
+            )
 
         approx_bytes = 0
         try:
@@ -125,15 +140,13 @@ async def map_invocation(
                     if type(a) in (bytes, bytearray, str):
                         approx_bytes += len(a)
                 total_inputs += 1
-                await sem.acquire()
                 if len(chunk_buf) >= CHUNK_ITEMS or approx_bytes > 4 * 1024 * 1024:
                     approx_bytes = 0
+                    await sem.acquire(len(chunk_buf))
                     await flush_chunk()
+            if chunk_buf:
+                await sem.acquire(len(chunk_buf))
             await flush_chunk()
-            if items:
-                await svc.function_put_inputs(
-                    function_call_id=call_id, items=items, chunks=chunks
-                )
             await svc.function_finish_inputs(function_call_id=call_id)
         except BaseException as exc:
             pump_error.append(exc)
@@ -181,6 +194,28 @@ async def map_invocation(
             if pump_error:
                 raise pump_error[0]
             for out in outs:
+                if out.get("group"):
+                    # range-protocol group: one pickled value list for ~64 idxs
+                    import pickle as _pickle
+
+                    values = _pickle.loads(out["chunk_data"])
+                    cis = out["cis"]
+                    base = out["idx_base"]
+                    pairs = (
+                        enumerate(values) if cis is None else zip(cis, values)
+                    )
+                    received += len(values)
+                    sem.release(len(values))
+                    if order_outputs:
+                        for ci, value in pairs:
+                            ordering_buffer[base + ci] = value
+                        while next_output_idx in ordering_buffer:
+                            yield ordering_buffer.pop(next_output_idx)
+                            next_output_idx += 1
+                    else:
+                        for _ci, value in pairs:
+                            yield value
+                    continue
                 received += 1
                 sem.release()
                 value = await decode(out)

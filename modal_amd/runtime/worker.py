@@ -381,6 +381,26 @@ class WorkerRuntime:
                     task = asyncio.get_running_loop().create_task(self._run_input(frt, item))
                     self._running[item["token"]] = task
                     task.add_done_callback(lambda _t, tok=item["token"]: self._running.pop(tok, None))
+        elif kind == "inputs_chunk":
+            frt = self.functions.get(msg["function_id"])
+            if frt is None:
+                await self.conn.send(
+                    {
+                        "t": "chunk_done",
+                        "token": msg["token"],
+                        "call_id": msg["call_id"],
+                        "chunk_id": msg["chunk_id"],
+                        "function_id": msg["function_id"],
+                        "count": msg["count"],
+                        "failed": True,
+                    }
+                )
+                return
+            task = asyncio.get_running_loop().create_task(self._run_chunk(frt, msg))
+            self._running[msg["token"]] = task
+            task.add_done_callback(
+                lambda _t, tok=msg["token"]: self._running.pop(tok, None)
+            )
         elif kind == "cancel":
             for token in msg.get("tokens", []):
                 task = self._running.get(token)
@@ -582,6 +602,124 @@ class WorkerRuntime:
                 fabric_ids=[0] * size,  # one xGMI hive on a single node
             )
         )
+
+    async def _run_chunk(self, frt: FunctionRuntime, msg: dict) -> None:
+        """Range-protocol execution: one frame = one chunk of ~64 inputs.
+        Splits into max_concurrent executor sub-ranges; one serialized
+        value-chunk goes back (per-item bytes only for failures/tensors)."""
+        count = msg["count"]
+        token = msg["token"]
+        call_id = msg["call_id"]
+        self._chunk_cache[msg["chunk_id"]] = {"raw": msg.get("payload"), "decoded": None}
+        self._chunk_order.append(msg["chunk_id"])
+        loop = asyncio.get_running_loop()
+
+        values: dict[int, Any] = {}
+        errors: dict[int, tuple] = {}
+
+        def run_range(start: int, end: int) -> None:
+            try:
+                fn = frt.get_callable(msg.get("method", ""))
+            except BaseException as exc:
+                data = self._serialize_exception(exc)
+                err = "".join(traceback.format_exception_only(type(exc), exc)).strip()
+                for ci in range(start, end):
+                    errors[ci] = (data, err)
+                return
+            item = {"chunk": msg["chunk_id"], "ci": 0}
+            tok_c = _current_function_call_id.set(call_id)
+            try:
+                for ci in range(start, end):
+                    item["ci"] = ci
+                    tok_i = _current_input_id.set(f"in-{call_id[3:]}-{ci}")
+                    try:
+                        args, kwargs = self._resolve_item_args(item)
+                        values[ci] = fn(*args, **kwargs)
+                    except BaseException as exc:
+                        errors[ci] = (
+                            self._serialize_exception(exc),
+                            "".join(traceback.format_exception_only(type(exc), exc)).strip(),
+                        )
+                    finally:
+                        _current_input_id.reset(tok_i)
+            finally:
+                _current_function_call_id.reset(tok_c)
+
+        from .execution_context import _current_function_call_id, _current_input_id
+
+        try:
+            fn0 = await loop.run_in_executor(self.executor, frt.load)
+            if frt._service is not None:
+                fn0 = frt.get_callable(msg.get("method", ""))
+        except BaseException as exc:
+            data = self._serialize_exception(exc)
+            err = "".join(traceback.format_exception_only(type(exc), exc)).strip()
+            for ci in range(count):
+                errors[ci] = (data, err)
+            fn0 = None
+
+        app_tok = _app_id_var.set(frt.app_id)
+        try:
+            if fn0 is not None and (
+                inspect.iscoroutinefunction(fn0) or inspect.isasyncgenfunction(fn0)
+            ):
+                # async user function: per-item awaits on this loop
+                item = {"chunk": msg["chunk_id"], "ci": 0}
+                for ci in range(count):
+                    item["ci"] = ci
+                    try:
+                        args, kwargs = await loop.run_in_executor(
+                            self.executor, self._resolve_item_args, dict(item)
+                        )
+                        values[ci] = await self._execute(frt, fn0, args, kwargs)
+                    except BaseException as exc:
+                        errors[ci] = (
+                            self._serialize_exception(exc),
+                            "".join(traceback.format_exception_only(type(exc), exc)).strip(),
+                        )
+            elif fn0 is not None:
+                n_ranges = min(frt.max_concurrent, count)
+                size = (count + n_ranges - 1) // n_ranges
+
+                async def run_one(start: int, end: int) -> None:
+                    async with frt.sem:
+                        await loop.run_in_executor(self.executor, run_range, start, end)
+
+                await asyncio.gather(
+                    *(run_one(s, min(s + size, count)) for s in range(0, count, size))
+                )
+        except asyncio.CancelledError:
+            return
+        finally:
+            _app_id_var.reset(app_tok)
+
+        reply: dict[str, Any] = {
+            "t": "chunk_done",
+            "token": token,
+            "call_id": call_id,
+            "chunk_id": msg["chunk_id"],
+            "function_id": frt.function_id,
+            "count": count,
+        }
+        if errors:
+            reply["exceptions"] = {str(ci): [d, r] for ci, (d, r) in errors.items()}
+        if values:
+            ordered_cis = sorted(values)
+            from ._serialize_chunk import serialize_value_chunk
+
+            def pack() -> tuple:
+                return serialize_value_chunk([values[ci] for ci in ordered_cis])
+
+            data, per_item = await loop.run_in_executor(self.executor, pack)
+            if data is not None:
+                reply["data"] = data
+                reply["cis"] = None if len(ordered_cis) == count else ordered_cis
+            else:
+                reply["items"] = {str(ci): b for ci, b in zip(ordered_cis, per_item)}
+        try:
+            await self.conn.send(reply)
+        except Exception:
+            pass
 
     async def _run_frame_fast(self, frt: FunctionRuntime, items: list[dict]) -> None:
         """Sequentially execute a frame of inputs in one worker thread,

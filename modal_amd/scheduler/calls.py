@@ -151,6 +151,36 @@ class InputRecord:
         return f"{self.call_id}:{self.idx}:{self.retry_count}"
 
 
+class ChunkGroup:
+    """Scheduling state for one map chunk dispatched as a unit.
+
+    The map fast path never materializes per-item InputRecords while things
+    succeed: a chunk of ~64 inputs is one pending descriptor, one wire frame,
+    one completion. Items that fail (user exception, worker death past the
+    chunk level) materialize real InputRecords and re-enter the per-item FSM,
+    so retry/dedup semantics stay per input (parity: parallel_map.py FSM).
+    """
+
+    __slots__ = (
+        "chunk_id", "base_idx", "count", "method", "state", "worker_id", "done_cis",
+        "internal_failures",
+    )
+
+    def __init__(self, chunk_id: str, base_idx: int, count: int, method: str = ""):
+        self.chunk_id = chunk_id
+        self.base_idx = base_idx
+        self.count = count
+        self.method = method
+        self.state = "pending"  # pending | inflight | done
+        self.worker_id: Optional[int] = None
+        self.done_cis: Optional[set] = None  # populated only on partial success
+        self.internal_failures = 0
+
+    @property
+    def token(self) -> str:
+        return f"g:{self.chunk_id}"
+
+
 class CallRecord:
     """One function call (``fc-``): unary, spawn, or map fan-out."""
 
@@ -168,6 +198,8 @@ class CallRecord:
         # shared chunk payloads: chunk_id -> {"data": bytes, "refs": int}
         # (one pickled list serves many inputs; freed when all are final)
         self.chunks: dict[str, dict] = {}
+        # fast-path chunk scheduling state (range protocol)
+        self.chunk_groups: dict[str, ChunkGroup] = {}
         self.next_idx = 0
         self.num_inputs_final: Optional[int] = None
         self.completed: int = 0
@@ -204,6 +236,54 @@ class CallRecord:
         )
         self.inputs[idx] = rec
         return rec
+
+    def add_chunk(self, chunk_id: str, payload: Any, count: int, method: str = "") -> ChunkGroup:
+        """Register a whole chunk without materializing per-item records."""
+        group = ChunkGroup(chunk_id, self.next_idx, count, method)
+        self.next_idx += count
+        self.chunks[chunk_id] = {"data": payload, "refs": count}
+        self.chunk_groups[chunk_id] = group
+        return group
+
+    def materialize_chunk_item(self, group: ChunkGroup, ci: int) -> InputRecord:
+        """Create the real per-item record for a chunk member (failure path)."""
+        idx = group.base_idx + ci
+        rec = self.inputs.get(idx)
+        if rec is not None:
+            return rec
+        rec = InputRecord(
+            call_id=self.call_id,
+            idx=idx,
+            input_id=f"in-{self.call_id[3:]}-{idx}",
+            payload=b"",
+            enqueued_at=time.time(),
+            method_name=group.method,
+        )
+        rec.chunk_id = group.chunk_id
+        rec.chunk_index = ci
+        self.inputs[idx] = rec
+        return rec
+
+    def complete_chunk_success(self, group: ChunkGroup, cis: Optional[list], out_chunk: str) -> None:
+        """Bulk completion of a chunk's successful members.
+        cis=None means the whole chunk succeeded."""
+        if group.state == "done":
+            return
+        n = group.count if cis is None else len(cis)
+        chunk = self.chunks.get(group.chunk_id)
+        if chunk is not None:
+            chunk["refs"] -= n
+            if chunk["refs"] <= 0:
+                del self.chunks[group.chunk_id]
+        self.completed += n
+        if cis is None:
+            group.state = "done"
+            self.output_ready.put_nowait(("g", group.base_idx, group.count, out_chunk, None))
+        else:
+            group.done_cis = set(cis)
+            group.state = "done"
+            self.output_ready.put_nowait(("g", group.base_idx, group.count, out_chunk, list(cis)))
+        self._check_done()
 
     def finish_inputs(self) -> None:
         self.num_inputs_final = self.next_idx

@@ -69,7 +69,8 @@ class RPCAdapter:
         "dict_len", "dict_items", "dict_clear", "dict_delete",
         "secret_get_or_create", "secret_env",
         "blob_put", "blob_get", "blob_path",
-        "function_lookup", "function_map", "function_put_inputs", "function_finish_inputs",
+        "function_lookup", "function_map", "function_put_inputs", "function_put_chunk",
+        "function_finish_inputs",
         "function_get_outputs", "function_call_cancel", "function_call_info",
         "function_get_current_stats", "generator_poll",
         "app_lookup", "app_get_layout", "cluster_hello",
@@ -463,6 +464,39 @@ class Scheduler:
         await self.pool.dispatch_gang(fdef, recs)
         return recs
 
+    async def function_put_chunk(
+        self,
+        function_call_id: str,
+        chunk_id: str,
+        payload: Any,
+        count: int,
+        method: str = "",
+    ) -> dict:
+        """Range-protocol intake: one call registers a whole chunk of inputs
+        (~64) with NO per-item records; items materialize only on failure.
+        Falls back to per-item intake for functions needing per-item
+        treatment (gangs, batching, generators, timeouts)."""
+        record = self._call(function_call_id)
+        fdef = self.functions.get(record.function_id)
+        fast = (
+            fdef is not None
+            and fdef.cluster_size <= 1
+            and fdef.batch_max_size <= 1
+            and not fdef.is_generator
+            and not fdef.timeout
+            and not fdef.web_config
+        )
+        if not fast:
+            items = [{"chunk": chunk_id, "ci": ci, "method": method} for ci in range(count)]
+            return {
+                "items": await self.function_put_inputs(
+                    function_call_id, items, chunks={chunk_id: payload}
+                )
+            }
+        group = record.add_chunk(chunk_id, payload, count, method)
+        self.pool.enqueue_chunk(record, group, record.function_id)
+        return {"idx_base": group.base_idx, "count": count}
+
     async def function_finish_inputs(self, function_call_id: str) -> None:
         self._call(function_call_id).finish_inputs()
 
@@ -483,6 +517,26 @@ class Scheduler:
         (parity: FunctionGetOutputs long-poll, reference _functions.py:224-263)."""
         record = self._call(function_call_id)
         out: list[dict] = []
+
+        def expand(entry: Any) -> None:
+            if type(entry) is tuple and entry and entry[0] == "g":
+                _tag, base, count, out_chunk, cis = entry
+                out.append(
+                    {
+                        "group": True,
+                        "idx_base": base,
+                        "count": count,
+                        "cis": cis,
+                        "out_chunk": out_chunk,
+                        "status": 1,
+                    }
+                )
+            elif isinstance(entry, list):
+                for idx in entry:
+                    out.append(self._output_item(record, idx))
+            else:
+                out.append(self._output_item(record, entry))
+
         deadline = time.time() + timeout
         while not out:
             try:
@@ -492,15 +546,13 @@ class Scheduler:
                 entry = await asyncio.wait_for(record.output_ready.get(), remaining)
             except asyncio.TimeoutError:
                 break
-            for idx in entry if isinstance(entry, list) else (entry,):
-                out.append(self._output_item(record, idx))
+            expand(entry)
             while len(out) < max_values:
                 try:
                     entry = record.output_ready.get_nowait()
                 except asyncio.QueueEmpty:
                     break
-                for idx in entry if isinstance(entry, list) else (entry,):
-                    out.append(self._output_item(record, idx))
+                expand(entry)
         # attach shared output-chunk bytes once per chunk per response
         chunks_seen: set[str] = set()
         for item in out:
@@ -512,14 +564,18 @@ class Scheduler:
                     item["chunk_data"] = chunk["data"]
         if clear_on_success:
             for item in out:
-                rec = record.inputs.get(item["idx"])
-                if rec is not None:
-                    rec.payload = b""  # release memory; result already extracted
                 cid = item.get("out_chunk")
+                if item.get("group"):
+                    n = item["count"] if item["cis"] is None else len(item["cis"])
+                else:
+                    n = 1
+                    rec = record.inputs.get(item["idx"])
+                    if rec is not None:
+                        rec.payload = b""  # release memory; result extracted
                 if cid:
                     chunk = self.out_chunks.get(cid)
                     if chunk is not None:
-                        chunk["refs"] -= 1
+                        chunk["refs"] -= n
                         if chunk["refs"] <= 0:
                             self.out_chunks.pop(cid, None)
         return out
@@ -554,6 +610,18 @@ class Scheduler:
                 self.finalize_input(
                     rec, GENERIC_STATUS_TERMINATED, None, 0, "input cancelled", rec.retry_count
                 )
+        for group in record.chunk_groups.values():
+            if group.state != "done":
+                group.state = "done"
+                tokens.append(group.token)
+                for ci in range(group.count):
+                    if record.inputs.get(group.base_idx + ci) is not None:
+                        continue  # already materialized + finalized above
+                    rec = record.materialize_chunk_item(group, ci)
+                    rec.cancelled = True
+                    self.finalize_input(
+                        rec, GENERIC_STATUS_TERMINATED, None, 0, "input cancelled", rec.retry_count
+                    )
         await self.pool.cancel_inputs(tokens, terminate=terminate_containers)
 
     async def function_call_info(self, function_call_id: str) -> dict:

@@ -333,6 +333,8 @@ class WorkerPool:
                 self._on_outputs(handle, msg)
             elif kind == "outputs_chunk":
                 self._on_outputs_chunk(handle, msg)
+            elif kind == "chunk_done":
+                self._on_chunk_done(handle, msg)
             elif kind == "gen_data":
                 self.scheduler.on_generator_data(msg)
             elif kind == "hb":
@@ -352,7 +354,30 @@ class WorkerPool:
         if self._stopping:
             return
         # requeue in-flight inputs: the INTERNAL_FAILURE path
-        for token, rec in list(handle.inflight.items()):
+        for token, entry in list(handle.inflight.items()):
+            if type(entry) is tuple:  # chunk group
+                _tag, record, group = entry
+                if group.state == "done" or record.cancelled:
+                    continue
+                group.internal_failures += 1
+                if group.internal_failures > MAX_INTERNAL_FAILURE_COUNT:
+                    for ci in range(group.count):
+                        rec = record.materialize_chunk_item(group, ci)
+                        self.scheduler.finalize_input(
+                            rec,
+                            GENERIC_STATUS_INTERNAL_FAILURE,
+                            None,
+                            0,
+                            f"worker died while executing chunk (x{group.internal_failures})",
+                            rec.retry_count,
+                        )
+                    group.state = "done"
+                else:
+                    group.state = "pending"
+                    group.worker_id = None
+                    self.enqueue_chunk(record, group, record.function_id, front=True)
+                continue
+            rec = entry
             if rec.final:
                 continue
             rec.internal_failures += 1
@@ -431,6 +456,16 @@ class WorkerPool:
         q.extend(recs)
         self._dispatch_wake.set()
 
+    def enqueue_chunk(self, record: Any, group: Any, function_id: str, front: bool = False) -> None:
+        """Queue a whole chunk group as one pending descriptor."""
+        q = self.pending.setdefault(function_id, deque())
+        entry = ("g", record, group)
+        if front:
+            q.appendleft(entry)
+        else:
+            q.append(entry)
+        self._dispatch_wake.set()
+
     def enqueue_delayed(self, rec: InputRecord, delay_s: float) -> None:
         self._delay_seq += 1
         heapq.heappush(
@@ -493,14 +528,22 @@ class WorkerPool:
                 if credit <= 0:
                     continue
                 batch: list[InputRecord] = []
-                while q and len(batch) < credit:
-                    rec = q.popleft()
+                while q and credit > 0:
+                    entry = q.popleft()
+                    if type(entry) is tuple:  # ("g", record, group) chunk descriptor
+                        _tag, record, group = entry
+                        if group.state != "pending" or record.cancelled:
+                            continue
+                        await self._send_group(w, record, group, fdef)
+                        credit -= group.count
+                        continue
+                    rec = entry
                     if rec.final or rec.cancelled:
                         continue
                     batch.append(rec)
-                if not batch:
-                    continue
-                await self._send_batch(w, fdef, batch)
+                    credit -= 1
+                if batch:
+                    await self._send_batch(w, fdef, batch)
             if q:
                 # backlog remains with no free credit: autoscale
                 # (parity: min/max_containers autoscaler settings,
@@ -550,35 +593,38 @@ class WorkerPool:
             return configured
         return min(max((os.cpu_count() or 4) // 2, 1), 8)
 
+    async def _send_def(self, w: WorkerHandle, fdef: FunctionDef) -> None:
+        await w.conn.send(
+            {
+                "t": "def",
+                "function_id": fdef.function_id,
+                "app_id": fdef.app_id,
+                "name": fdef.name,
+                "definition": fdef.definition,
+                "definition_kind": fdef.definition_kind,
+                "is_generator": fdef.is_generator,
+                "timeout": fdef.timeout,
+                "max_concurrent_inputs": fdef.max_concurrent_inputs,
+                "batch_max_size": fdef.batch_max_size,
+                "batch_linger_ms": fdef.batch_linger_ms,
+                "version": fdef.definition_version,
+                "app_layout": self.scheduler.app_layout(fdef.app_id),
+                "env": self.scheduler.resolve_function_env(fdef),
+                "volumes": fdef.volume_mounts,
+                "python_paths": self.scheduler.resolve_function_pythonpaths(fdef),
+                "web_config": fdef.web_config,
+            }
+        )
+        w.functions_loaded.add(fdef.function_id)
+        w.defs_version[fdef.function_id] = fdef.definition_version
+
     async def _send_batch(self, w: WorkerHandle, fdef: FunctionDef, batch: list[InputRecord]) -> None:
         try:
             if (
                 fdef.function_id not in w.functions_loaded
                 or w.defs_version.get(fdef.function_id, 0) != fdef.definition_version
             ):
-                await w.conn.send(
-                    {
-                        "t": "def",
-                        "function_id": fdef.function_id,
-                        "app_id": fdef.app_id,
-                        "name": fdef.name,
-                        "definition": fdef.definition,
-                        "definition_kind": fdef.definition_kind,
-                        "is_generator": fdef.is_generator,
-                        "timeout": fdef.timeout,
-                        "max_concurrent_inputs": fdef.max_concurrent_inputs,
-                        "batch_max_size": fdef.batch_max_size,
-                        "batch_linger_ms": fdef.batch_linger_ms,
-                        "version": fdef.definition_version,
-                        "app_layout": self.scheduler.app_layout(fdef.app_id),
-                        "env": self.scheduler.resolve_function_env(fdef),
-                        "volumes": fdef.volume_mounts,
-                        "python_paths": self.scheduler.resolve_function_pythonpaths(fdef),
-                        "web_config": fdef.web_config,
-                    }
-                )
-                w.functions_loaded.add(fdef.function_id)
-                w.defs_version[fdef.function_id] = fdef.definition_version
+                await self._send_def(w, fdef)
             w.last_active = time.time()
             items = []
             chunks_needed: dict[str, bytes] = {}
@@ -622,6 +668,109 @@ class WorkerPool:
                     del w.inflight[rec.token]
                 if not rec.final:
                     self.enqueue(rec, front=True)
+
+    async def _send_group(self, w: WorkerHandle, record: Any, group: Any, fdef: FunctionDef) -> None:
+        """One frame carries a whole chunk (range protocol)."""
+        try:
+            if (
+                fdef.function_id not in w.functions_loaded
+                or w.defs_version.get(fdef.function_id, 0) != fdef.definition_version
+            ):
+                await self._send_def(w, fdef)
+            group.state = "inflight"
+            group.worker_id = w.worker_id
+            w.last_active = time.time()
+            w.inflight[group.token] = ("g", record, group)
+            w.outstanding[fdef.function_id] = (
+                w.outstanding.get(fdef.function_id, 0) + group.count
+            )
+            chunk = record.chunks.get(group.chunk_id) or {}
+            await w.conn.send(
+                {
+                    "t": "inputs_chunk",
+                    "function_id": fdef.function_id,
+                    "call_id": record.call_id,
+                    "chunk_id": group.chunk_id,
+                    "token": group.token,
+                    "count": group.count,
+                    "payload": chunk.get("data"),
+                    "method": group.method,
+                    "max_concurrent": fdef.max_concurrent_inputs,
+                }
+            )
+        except Exception:
+            w.inflight.pop(group.token, None)
+            group.state = "pending"
+            if not record.cancelled:
+                self.enqueue_chunk(record, group, fdef.function_id, front=True)
+
+    def _on_chunk_done(self, handle: WorkerHandle, msg: dict) -> None:
+        """Completion of a range-protocol chunk."""
+        from .calls import GENERIC_STATUS_FAILURE, GENERIC_STATUS_SUCCESS
+
+        handle.inflight.pop(msg["token"], None)
+        fid = msg.get("function_id")
+        count = msg.get("count", 0)
+        if fid:
+            handle.outstanding[fid] = max(handle.outstanding.get(fid, 0) - count, 0)
+        record = self.scheduler.calls.get(msg["call_id"])
+        if record is None:
+            return
+        group = record.chunk_groups.get(msg["chunk_id"])
+        if group is None or group.state == "done":
+            return
+        if msg.get("failed"):
+            # worker-side setup failure (e.g. missing def): internal requeue
+            group.internal_failures += 1
+            if group.internal_failures <= MAX_INTERNAL_FAILURE_COUNT and not record.cancelled:
+                group.state = "pending"
+                self.enqueue_chunk(record, group, record.function_id, front=True)
+            else:
+                from .calls import GENERIC_STATUS_INTERNAL_FAILURE as _GIF
+
+                for ci in range(group.count):
+                    rec = record.materialize_chunk_item(group, ci)
+                    self.scheduler.finalize_input(
+                        rec, _GIF, None, 0, "chunk repeatedly failed to start", rec.retry_count
+                    )
+                group.state = "done"
+            return
+        # per-item exceptions materialize real records (the per-item FSM owns
+        # retries from here)
+        for ci_s, (data, repr_s) in (msg.get("exceptions") or {}).items():
+            ci = int(ci_s)
+            rec = record.materialize_chunk_item(group, ci)
+            self.scheduler.on_worker_output(
+                call_id=record.call_id,
+                idx=rec.idx,
+                retry_count=rec.retry_count,
+                status=GENERIC_STATUS_FAILURE,
+                output=data,
+                output_format=1,
+                exc_repr=repr_s,
+            )
+        # per-item success payloads (tensor/oversized fallback)
+        for ci_s, data in (msg.get("items") or {}).items():
+            ci = int(ci_s)
+            rec = record.materialize_chunk_item(group, ci)
+            self.scheduler.on_worker_output(
+                call_id=record.call_id,
+                idx=rec.idx,
+                retry_count=rec.retry_count,
+                status=GENERIC_STATUS_SUCCESS,
+                output=data,
+                output_format=1,
+                exc_repr=None,
+            )
+        cis = msg.get("cis")  # None => all count items succeeded in order
+        data = msg.get("data")
+        if data is not None:
+            slot = self.scheduler.register_out_chunk(data, count if cis is None else len(cis))
+            record.complete_chunk_success(group, cis, slot)
+        else:
+            group.state = "done"
+            record._check_done()
+        self._dispatch_wake.set()
 
     async def dispatch_gang(self, fdef: FunctionDef, recs: list) -> None:
         """Place a gang on len(recs) distinct workers at once, bypassing the
