@@ -70,8 +70,9 @@ class WorkerHandle:
         return self.gpu_index is not None
 
     def credit_for(self, fdef: FunctionDef) -> int:
-        cap = max(fdef.max_concurrent_inputs, 1) * 2
-        cap = max(cap, DEFAULT_PIPELINE_DEPTH if not fdef.needs_gpu else fdef.max_concurrent_inputs * 2)
+        # deep enough for >=3 chunk groups in flight so a worker never idles
+        # between group completions (map chunks are ~64 items each)
+        cap = max(fdef.max_concurrent_inputs * 2, DEFAULT_PIPELINE_DEPTH)
         if fdef.batch_max_size:
             cap = max(cap, fdef.batch_max_size * 2)
         return cap - self.outstanding.get(fdef.function_id, 0)
