@@ -478,13 +478,18 @@ class WorkerRuntime:
                 raise RuntimeError(f"chunk {cid} not delivered to this worker")
             decoded = entry["decoded"]
             if decoded is None:
-                raw = entry["raw"]
-                if isinstance(raw, dict):  # CAS-spilled chunk (>2 MiB)
-                    raw = self.blob_store.get(raw["blob"])
-                kind, decoded = deserialize(raw)
-                assert kind == "C"
-                entry["decoded"] = decoded
-                entry["raw"] = None
+                # one decode serves every range-thread of the chunk
+                lock = entry.setdefault("lock", __import__("threading").Lock())
+                with lock:
+                    decoded = entry["decoded"]
+                    if decoded is None:
+                        raw = entry["raw"]
+                        if isinstance(raw, dict):  # CAS-spilled chunk (>2 MiB)
+                            raw = self.blob_store.get(raw["blob"])
+                        kind, decoded = deserialize(raw)
+                        assert kind == "C"
+                        entry["decoded"] = decoded
+                        entry["raw"] = None
             args, kwargs = decoded[item.get("ci", 0)]
             return args, kwargs
         return self._decode_args(item)

@@ -485,6 +485,7 @@ class Scheduler:
             and not fdef.is_generator
             and not fdef.timeout
             and not fdef.web_config
+            and not os.environ.get("MODAL_AMD_NO_RANGE")  # A/B escape hatch
         )
         if not fast:
             items = [{"chunk": chunk_id, "ci": ci, "method": method} for ci in range(count)]
