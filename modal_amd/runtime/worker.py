@@ -683,7 +683,11 @@ class WorkerRuntime:
                             "".join(traceback.format_exception_only(type(exc), exc)).strip(),
                         )
             elif fn0 is not None:
-                n_ranges = min(frt.max_concurrent, count)
+                if count <= frt.max_concurrent:
+                    n_ranges = count  # small chunks: full per-item overlap
+                else:
+                    # >=16 items per executor hop, up to max_concurrent ranges
+                    n_ranges = min(frt.max_concurrent, max(1, count // 16))
                 size = (count + n_ranges - 1) // n_ranges
 
                 async def run_one(start: int, end: int) -> None:
