@@ -14,8 +14,11 @@ if sys.version_info[:2] < (3, 10):
 
 __version__ = "0.1.0"
 
-from . import billing, environments, exception, experimental
+from . import billing, environments, exception, experimental, types
 from ._tunnel import Tunnel, forward
+from .environments import Environment
+from .file_pattern_matcher import FilePatternMatcher
+from .workspace import Workspace
 from .app import App
 from .client import Client
 from .cloud_bucket_mount import CloudBucketMount
@@ -65,9 +68,11 @@ __all__ = [
     "ContainerProcess",
     "Cron",
     "Dict",
+    "Environment",
     "Error",
     "FileEntry",
     "FileIO",
+    "FilePatternMatcher",
     "Function",
     "FunctionCall",
     "Image",
@@ -87,8 +92,10 @@ __all__ = [
     "environments",
     "Tunnel",
     "Volume",
+    "Workspace",
     "experimental",
     "forward",
+    "types",
     "asgi_app",
     "batched",
     "concurrent",
