@@ -48,6 +48,16 @@ def load_lib(required: bool = False) -> Optional[ctypes.CDLL]:
             )
             lib.ma_unpack_segments.restype = ctypes.c_int
             lib.ma_unpack_segments.argtypes = lib.ma_pack_segments.argtypes
+            lib.ma_lz4_compress.restype = ctypes.c_int
+            lib.ma_lz4_compress.argtypes = [
+                ctypes.c_void_p, ctypes.c_long, ctypes.c_void_p, ctypes.c_void_p,
+                ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
+            ]
+            lib.ma_lz4_decompress.restype = ctypes.c_int
+            lib.ma_lz4_decompress.argtypes = [
+                ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                ctypes.c_long, ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+            ]
             _lib = lib
             return lib
         except OSError as exc:

@@ -1053,14 +1053,7 @@ class Scheduler:
         root = os.path.join(self.run_dir, "mounts", mount_id)
         for remote_path, blob_id, _mode in manifest:
             dest = os.path.join(root, remote_path.lstrip("/"))
-            os.makedirs(os.path.dirname(dest), exist_ok=True)
-            src = self.blob_store.open_path(blob_id)
-            try:
-                os.link(src, dest)
-            except OSError:
-                import shutil
-
-                shutil.copyfile(src, dest)
+            self.blob_store.materialize(blob_id, dest)
         self.mounts[mount_id] = root
         self._mounts_by_hash[key] = mount_id
         return {"mount_id": mount_id, "dir": root}

@@ -103,18 +103,13 @@ class VolumeService:
         dest = self._safe_path(vol, rel_path)
         os.makedirs(os.path.dirname(dest), exist_ok=True)
         if len(block_digests) == 1:
-            # hard-link single-block files straight out of the CAS
-            try:
-                if os.path.exists(dest):
-                    os.unlink(dest)
-                os.link(self.blob_store.open_path(block_digests[0]), dest)
-            except OSError:
-                shutil.copyfile(self.blob_store.open_path(block_digests[0]), dest)
+            # single-block files come straight out of the CAS (hard link when
+            # stored raw, decompressed write when LZ4-compressed)
+            self.blob_store.materialize(block_digests[0], dest)
         else:
             with open(dest, "wb") as f:
                 for digest in block_digests:
-                    with open(self.blob_store.open_path(digest), "rb") as bf:
-                        shutil.copyfileobj(bf, f)
+                    f.write(self.blob_store.get(digest))
         vol.manifest[rel_path.lstrip("/")] = {
             "size": size,
             "blocks": block_digests,
