@@ -47,6 +47,7 @@ __device__ int lz4_compress_segment(const uint8_t* __restrict__ src, int src_len
   int ip = 0;       // current position
   int anchor = 0;   // start of pending literals
   int op = 0;       // output position
+  int misses = 0;   // LZ4-style skip acceleration through matchless regions
   const int mflimit = src_len - MFLIMIT;
 
   if (src_len >= MIN_MATCH + LASTLITERALS) {
@@ -57,6 +58,7 @@ __device__ int lz4_compress_segment(const uint8_t* __restrict__ src, int src_len
       table[h] = (uint16_t)ip;
       if (cand != 0xFFFF && cand < ip && (ip - cand) <= 0xFFFF &&
           load32(src + cand) == seq) {
+        misses = 0;
         // extend the match
         int mlen = MIN_MATCH;
         const int maxm = src_len - LASTLITERALS - ip;
@@ -75,7 +77,16 @@ __device__ int lz4_compress_segment(const uint8_t* __restrict__ src, int src_len
           while (rest >= 255) { dst[op++] = 255; rest -= 255; }
           dst[op++] = (uint8_t)rest;
         }
-        for (int i = 0; i < lit; ++i) dst[op++] = src[anchor + i];
+        {
+          int i = 0;
+          for (; i + 4 <= lit; i += 4) {  // word-wise literal copy
+            uint32_t v;
+            __builtin_memcpy(&v, src + anchor + i, 4);
+            __builtin_memcpy(dst + op + i, &v, 4);
+          }
+          for (; i < lit; ++i) dst[op + i] = src[anchor + i];
+          op += lit;
+        }
         uint16_t off = (uint16_t)(ip - cand);
         dst[op++] = (uint8_t)(off & 0xFF);
         dst[op++] = (uint8_t)(off >> 8);
@@ -87,7 +98,8 @@ __device__ int lz4_compress_segment(const uint8_t* __restrict__ src, int src_len
         ip += mlen;
         anchor = ip;
       } else {
-        ++ip;
+        ip += 1 + (misses >> 6);  // accelerate through incompressible runs
+        ++misses;
       }
     }
   }
@@ -138,7 +150,15 @@ __device__ int lz4_decompress_segment(const uint8_t* __restrict__ src, int src_l
       } while (b == 255);
     }
     if (ip + lit > src_len || op + lit > dst_len) return -1;
-    for (int i = 0; i < lit; ++i) dst[op + i] = src[ip + i];
+    {
+      int i = 0;
+      for (; i + 4 <= lit; i += 4) {
+        uint32_t v;
+        __builtin_memcpy(&v, src + ip + i, 4);
+        __builtin_memcpy(dst + op + i, &v, 4);
+      }
+      for (; i < lit; ++i) dst[op + i] = src[ip + i];
+    }
     ip += lit;
     op += lit;
     if (ip >= src_len) break;  // trailing-literal sequence
@@ -158,7 +178,17 @@ __device__ int lz4_decompress_segment(const uint8_t* __restrict__ src, int src_l
     mlen += MIN_MATCH;
     if (op + mlen > dst_len) return -1;
     const uint8_t* match = dst + op - off;
-    for (int i = 0; i < mlen; ++i) dst[op + i] = match[i];  // overlap-safe fwd copy
+    if (off >= 8) {  // wide copy is overlap-safe at this distance
+      int i = 0;
+      for (; i + 8 <= mlen; i += 8) {
+        uint64_t v;
+        __builtin_memcpy(&v, match + i, 8);
+        __builtin_memcpy(dst + op + i, &v, 8);
+      }
+      for (; i < mlen; ++i) dst[op + i] = match[i];
+    } else {
+      for (int i = 0; i < mlen; ++i) dst[op + i] = match[i];  // byte fwd copy
+    }
     op += mlen;
   }
   return op;
