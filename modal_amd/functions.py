@@ -597,3 +597,16 @@ def _install_sync_map_methods() -> None:
 
 
 _install_sync_map_methods()
+
+
+async def _gather_impl(*function_calls: Any) -> list:
+    """Module-level gather (parity: modal.functions.gather)."""
+    from ._sync import unwrap
+
+    impls = [unwrap(fc) for fc in function_calls]
+    return list(await asyncio.gather(*(fc.get() for fc in impls)))
+
+
+from ._sync import dual_function as _dual_function  # noqa: E402
+
+gather = _dual_function(_gather_impl)

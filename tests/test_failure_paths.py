@@ -126,3 +126,67 @@ def test_map_with_failures_and_retries(client, run_dir):
     with app.run(client=client):
         out = sorted(flaky.map(range(6), kwargs={"base": run_dir}, order_outputs=False))
         assert out == list(range(6))
+
+
+def test_cancel_with_terminate_recycles_worker(client):
+    """terminate_containers=True kills the executing worker; the pool
+    recovers for subsequent calls (parity: FunctionCall.cancel(
+    terminate_containers=True))."""
+    app = modal.App("cancel-term-app")
+
+    @app.function()
+    def hang():
+        time.sleep(60)
+
+    @app.function()
+    def ping():
+        return "alive"
+
+    with app.run(client=client):
+        fc = hang.spawn()
+        time.sleep(0.5)
+        fc.cancel(terminate_containers=True)
+        with pytest.raises(FunctionCallCancelledError):
+            fc.get(timeout=10)
+        # pool recovers: a fresh worker serves the next call
+        assert ping.remote() == "alive"
+
+
+def test_function_call_from_id_inside_worker(client):
+    """A FunctionCall handle passed between functions stays usable."""
+    app = modal.App("fc-handoff")
+
+    @app.function()
+    def slow_value():
+        time.sleep(0.3)
+        return 99
+
+    @app.function()
+    def waiter(call_id: str):
+        import modal_amd as modal
+
+        fc = modal.FunctionCall.from_id(call_id)
+        return fc.get(timeout=30) + 1
+
+    with app.run(client=client):
+        fc = slow_value.spawn()
+        assert waiter.remote(fc.object_id) == 100
+
+
+def test_map_early_break_cleans_up(client):
+    """Breaking out of a map iterator mid-stream cancels the pump."""
+    app = modal.App("early-break")
+
+    @app.function()
+    def ident(x):
+        return x
+
+    with app.run(client=client):
+        seen = 0
+        for _value in ident.map(range(10_000), order_outputs=False):
+            seen += 1
+            if seen >= 50:
+                break
+        assert seen == 50
+        # the runtime stays healthy for the next call
+        assert ident.remote(7) == 7
