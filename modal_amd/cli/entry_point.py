@@ -232,6 +232,23 @@ def container_cli() -> None:
     """Manage workers (the container analog)."""
 
 
+@container_cli.command(name="stats")
+def container_stats() -> None:
+    client = _get_client()
+    stats = synchronizer.run(client.svc.node_stats())
+    click.echo(f"pending inputs: {stats['pending_inputs']}  active calls: {stats['active_calls']}")
+    for w in stats["workers"]:
+        gpu = w.get("gpu")
+        mem = ""
+        if gpu:
+            used = (gpu["hbm_total"] - gpu["hbm_free"]) / 1e9
+            mem = f"  hbm {used:.1f}/{gpu['hbm_total']/1e9:.0f} GB"
+        click.echo(
+            f"{w['task_id']}  worker-{w['worker_id']}  gpu={w['gpu_index']}  "
+            f"inflight={w['inflight']}  hb_age={w['last_heartbeat_age']:.0f}s{mem}"
+        )
+
+
 @container_cli.command(name="list")
 def container_list() -> None:
     client = _get_client()

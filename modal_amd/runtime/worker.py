@@ -316,11 +316,30 @@ class WorkerRuntime:
         await self._flush_outbox()
         await self.conn.close()
 
+    def _gpu_stats(self) -> Optional[dict]:
+        """HBM gauges for the scheduler's per-GPU view."""
+        torch = sys.modules.get("torch")
+        if torch is None or not torch.cuda.is_initialized():
+            return None
+        try:
+            free, total = torch.cuda.mem_get_info()
+            return {
+                "hbm_total": total,
+                "hbm_free": free,
+                "hbm_reserved": torch.cuda.memory_reserved(),
+            }
+        except Exception:
+            return None
+
     async def _heartbeat_loop(self) -> None:
         while True:
             await asyncio.sleep(HEARTBEAT_INTERVAL)
             try:
-                await self.conn.send({"t": "hb"})
+                msg: dict = {"t": "hb"}
+                stats = self._gpu_stats()
+                if stats:
+                    msg["gpu"] = stats
+                await self.conn.send(msg)
             except Exception:
                 return
 

@@ -72,7 +72,7 @@ class RPCAdapter:
         "function_lookup", "function_map", "function_put_inputs", "function_put_chunk",
         "function_finish_inputs",
         "function_get_outputs", "function_call_cancel", "function_call_info",
-        "function_get_current_stats", "generator_poll",
+        "function_get_current_stats", "generator_poll", "node_stats",
         "app_lookup", "app_get_layout", "cluster_hello",
         "volume_get_or_create", "volume_put_file_blocks", "volume_get_file", "volume_list_files",
         "volume_remove_file", "volume_copy_files", "volume_commit", "volume_reload",
@@ -344,6 +344,30 @@ class Scheduler:
         if fid is None:
             raise NotFoundError(f"Function '{app_name}/{name}' not found")
         return {"function_id": fid, "metadata": self.functions[fid].public_metadata()}
+
+    async def node_stats(self) -> dict:
+        """Node-level gauges: workers, in-flight inputs, per-GPU HBM
+        (SURVEY §5.5's MI355X metrics obligation)."""
+        workers = []
+        for w in self.pool.workers.values():
+            row = {
+                "worker_id": w.worker_id,
+                "task_id": w.task_id,
+                "gpu_index": w.gpu_index,
+                "alive": w.alive,
+                "inflight": len(w.inflight),
+                "last_heartbeat_age": max(0.0, time.time() - w.last_heartbeat),
+            }
+            if w.gpu_stats:
+                row["gpu"] = w.gpu_stats
+            workers.append(row)
+        return {
+            "workers": workers,
+            "pending_inputs": sum(len(q) for q in self.pool.pending.values()),
+            "active_calls": sum(
+                1 for c in self.calls.values() if not c.done_event.is_set()
+            ),
+        }
 
     async def function_get_current_stats(self, function_id: str) -> dict:
         backlog = len(self.pool.pending.get(function_id, ()))

@@ -56,6 +56,7 @@ class WorkerHandle:
         self.alive = True
         self.last_heartbeat = time.time()
         self.last_active = time.time()  # last input assignment (scaledown clock)
+        self.gpu_stats: Optional[dict] = None  # HBM gauges from heartbeats
         # tokens of inputs currently assigned here, mapped to their records
         self.inflight: dict[str, InputRecord] = {}
         # per-function outstanding count (for credit computation)
@@ -340,6 +341,8 @@ class WorkerPool:
                 self.scheduler.on_generator_data(msg)
             elif kind == "hb":
                 handle.last_heartbeat = time.time()
+                if msg.get("gpu"):
+                    handle.gpu_stats = msg["gpu"]
             elif kind == "mesh_ready":
                 self.scheduler.on_mesh_ready(handle, bool(msg.get("ok")))
             elif kind == "log":
