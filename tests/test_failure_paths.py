@@ -190,3 +190,71 @@ def test_map_early_break_cleans_up(client):
         assert seen == 50
         # the runtime stays healthy for the next call
         assert ident.remote(7) == 7
+
+
+def test_chunked_map_mixed_failures_with_retries(client, run_dir):
+    """Range-protocol stress: chunks with partial failures materialize
+    per-item records that retry with the chunk payload still available."""
+    app = modal.App("chunk-stress")
+
+    @app.function(retries=modal.Retries(max_retries=2, initial_delay=1.0))
+    def sometimes(x, base):
+        import os as _os
+
+        if x % 7 == 0:
+            marker = f"{base}/m-{x}"
+            if not _os.path.exists(marker):
+                open(marker, "w").write("1")
+                raise ValueError(f"first-attempt failure {x}")
+        return x * 2
+
+    with app.run(client=client):
+        out = sorted(
+            sometimes.map(range(200), kwargs={"base": run_dir}, order_outputs=False)
+        )
+        assert out == [x * 2 for x in range(200)]
+
+
+def test_concurrent_maps_same_function(client):
+    """Two maps over one function interleave without cross-talk."""
+    import threading
+
+    app = modal.App("concurrent-maps")
+
+    @app.function()
+    def tag(x, label):
+        return f"{label}:{x}"
+
+    results = {}
+
+    with app.run(client=client):
+        def run_map(label):
+            results[label] = sorted(
+                tag.map(range(300), kwargs={"label": label}, order_outputs=False)
+            )
+
+        t1 = threading.Thread(target=run_map, args=("a",))
+        t2 = threading.Thread(target=run_map, args=("b",))
+        t1.start(); t2.start(); t1.join(); t2.join()
+        assert results["a"] == sorted(f"a:{x}" for x in range(300))
+        assert results["b"] == sorted(f"b:{x}" for x in range(300))
+
+
+def test_chunked_map_worker_death_requeues_chunk(client, run_dir):
+    """A worker dying mid-chunk requeues the whole chunk elsewhere."""
+    app = modal.App("chunk-death")
+    marker = os.path.join(run_dir, "chunk-killed")
+
+    @app.function()
+    def kill_once(x, path):
+        import os as _os
+
+        if x == 70 and not _os.path.exists(path):
+            open(path, "w").write("x")
+            _os._exit(1)
+        return x
+
+    with app.run(client=client):
+        out = sorted(kill_once.map(range(150), kwargs={"path": marker}, order_outputs=False))
+        assert out == list(range(150))
+        assert os.path.exists(marker)
