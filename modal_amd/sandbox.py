@@ -229,8 +229,18 @@ class _Sandbox(_Object, type_kind="sandbox"):
         app_id = ""
         if app is not None and getattr(app, "_app_id", None):
             app_id = app._app_id
+        restore_image_id = None
+        if image is not None:
+            image_impl = unwrap(image)
+            from ._object import Resolver
+
+            if not image_impl.is_hydrated:
+                await Resolver(client).load(image_impl)
+            restore_image_id = image_impl.object_id
+            env_dict = {**getattr(image_impl, "_env_cache", {}), **env_dict}
         resp = await client.svc.sandbox_create(
             entrypoint_args=list(entrypoint_args),
+            restore_image_id=restore_image_id,
             env=env_dict,
             workdir=workdir,
             timeout=timeout,

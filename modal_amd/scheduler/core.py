@@ -971,7 +971,17 @@ class Scheduler:
         volume_paths = {
             path: self.volume_service.volume_dir(vid) for path, vid in volume_mounts.items()
         }
-        return await self.sandbox_service.create(volume_paths=volume_paths, **kwargs)
+        restore_image = kwargs.pop("restore_image_id", None)
+        restore_blob = None
+        if restore_image:
+            # image ids minted by sandbox_snapshot_fs carry their tar blob
+            restore_blob = self._extra.get("snapshot_blobs", {}).get(restore_image)
+        return await self.sandbox_service.create(
+            volume_paths=volume_paths,
+            restore_blob=restore_blob,
+            blob_store=self.blob_store,
+            **kwargs,
+        )
 
     async def sandbox_exec(self, sandbox_id, cmd, env=None, workdir=None, timeout=None, exec_id=None) -> dict:
         return await self.sandbox_service.exec(sandbox_id, cmd, env, workdir, timeout, exec_id)
@@ -1002,7 +1012,9 @@ class Scheduler:
         return {"sandbox_id": sid}
 
     async def sandbox_snapshot_fs(self, sandbox_id) -> dict:
-        return await self.sandbox_service.snapshot_fs(sandbox_id, self.blob_store)
+        resp = await self.sandbox_service.snapshot_fs(sandbox_id, self.blob_store)
+        self._extra.setdefault("snapshot_blobs", {})[resp["image_id"]] = resp["blob_id"]
+        return resp
 
     async def sandbox_fs_op(self, sandbox_id: str, op: str, path: str = "", **kwargs: Any) -> Any:
         """Typed remote-FS operations inside a sandbox workdir (parity:

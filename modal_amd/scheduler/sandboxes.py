@@ -180,10 +180,14 @@ class SandboxService:
         volume_paths: Optional[dict] = None,
         cpu: Optional[float] = None,
         memory: Optional[int] = None,
+        restore_blob: Optional[str] = None,
+        blob_store: Any = None,
     ) -> dict:
         sandbox_id = new_id("sandbox")
         sb_dir = os.path.join(self.root, sandbox_id)
         os.makedirs(sb_dir, exist_ok=True)
+        if restore_blob and blob_store is not None:
+            await self.restore_fs(sb_dir, blob_store, restore_blob)
         full_env = dict(os.environ)
         full_env.update(env or {})
         full_env["MODAL_AMD_SANDBOX_ID"] = sandbox_id
@@ -329,6 +333,17 @@ class SandboxService:
         if sid is None:
             raise NotFoundError(f"Sandbox '{name}' not found")
         return sid
+
+    async def restore_fs(self, workdir: str, blob_store: Any, blob_id: str) -> None:
+        """Materialize a filesystem snapshot (tar.gz in the CAS) into a fresh
+        sandbox workdir (parity: restore via _experimental_from_snapshot,
+        reference sandbox.py:2210-2337)."""
+        import io
+        import tarfile
+
+        data = blob_store.get(blob_id)
+        with tarfile.open(fileobj=io.BytesIO(data), mode="r:gz") as tar:
+            tar.extractall(workdir)
 
     async def snapshot_fs(self, sandbox_id: str, blob_store: Any) -> dict:
         """Tar the sandbox workdir into the CAS (parity: SandboxSnapshotFs)."""

@@ -115,3 +115,17 @@ def test_sandbox_snapshot_fs(client):
         assert img.object_id.startswith("im-")
     finally:
         sb.terminate()
+
+
+def test_sandbox_snapshot_and_restore(client):
+    """fs snapshot -> new sandbox restored from it (parity: sandbox
+    _experimental_snapshot / from_snapshot flow)."""
+    sb = modal.Sandbox.create("bash", "-c", "echo snapshot-state > state.txt; sleep 60")
+    try:
+        time.sleep(0.4)
+        img = sb.snapshot_filesystem()
+    finally:
+        sb.terminate()
+    sb2 = modal.Sandbox.create("bash", "-c", "cat state.txt", image=img)
+    sb2.wait(raise_on_termination=False)
+    assert sb2.stdout.read().strip() == "snapshot-state"
