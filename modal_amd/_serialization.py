@@ -11,10 +11,10 @@ Parity with the reference (/root/reference/py/modal/_serialization.py):
   ``DataFormat`` enum {PICKLE, CBOR, ASGI, GENERATOR_DONE}
   (reference :365,393; api.proto:115).
 
-MI355X-native addition: ``serialize_payload`` has a tensor-aware fast path —
-torch tensors are extracted out of the pickle stream and carried as raw
-buffers, so worker transports can move them with pinned hipMemcpyAsync or
-CUDA-IPC instead of copying through pickle.
+MI355X-native addition: torch tensors get their own persistent-id forms —
+worker-side CUDA tensors export as device-mesh markers (fetched over
+RCCL/xGMI, runtime/devicemesh.py) and client-side CUDA tensors host-stage —
+so device payloads never round-trip through pickle copies.
 """
 
 from __future__ import annotations
@@ -238,46 +238,6 @@ def deserialize_data_format(data: bytes, data_format: int) -> Any:
 # ---------------------------------------------------------------------------
 
 
-def _split_tensors(obj: Any, out: list) -> Any:
-    """Replace torch tensors with index markers, collecting them in `out`.
-
-    Only walks plain containers (tuple/list/dict) — tensors nested inside
-    arbitrary user objects still go through pickle, which is correct but not
-    zero-copy.
-    """
-    import torch  # local import: serialization must not force torch at import time
-
-    if isinstance(obj, torch.Tensor):
-        out.append(obj)
-        return _TensorRef(len(out) - 1)
-    if type(obj) is tuple:
-        return tuple(_split_tensors(x, out) for x in obj)
-    if type(obj) is list:
-        return [_split_tensors(x, out) for x in obj]
-    if type(obj) is dict:
-        return {k: _split_tensors(v, out) for k, v in obj.items()}
-    return obj
-
-
-class _TensorRef:
-    __slots__ = ("index",)
-
-    def __init__(self, index: int):
-        self.index = index
-
-
-def _join_tensors(obj: Any, tensors: list) -> Any:
-    if isinstance(obj, _TensorRef):
-        return tensors[obj.index]
-    if type(obj) is tuple:
-        return tuple(_join_tensors(x, tensors) for x in obj)
-    if type(obj) is list:
-        return [_join_tensors(x, tensors) for x in obj]
-    if type(obj) is dict:
-        return {k: _join_tensors(v, tensors) for k, v in obj.items()}
-    return obj
-
-
 def _walk_for_tensors(obj: Any, tensor_cls: type) -> bool:
     t = type(obj)
     if t is tensor_cls:
@@ -307,22 +267,13 @@ def contains_tensors(args: tuple, kwargs: dict) -> bool:
     )
 
 
-def serialize_payload(args: tuple, kwargs: dict) -> tuple[bytes, list]:
-    """Serialize an (args, kwargs) payload, extracting top-level torch tensors.
-
-    Returns (pickled_structure, tensors). Transports decide how tensors move
-    (inline copy, pinned staging, CUDA-IPC, RCCL P2P).
-    """
-    if contains_tensors(args, kwargs):
-        tensors: list = []
-        stripped = _split_tensors((args, kwargs), tensors)
-        return serialize(("T", stripped)), tensors
-    return serialize(("P", (args, kwargs))), []
+def serialize_payload(args: tuple, kwargs: dict) -> bytes:
+    """Serialize an (args, kwargs) payload. Tensors ride the persistent-id
+    hooks (device-mesh export / host staging), so no sidecar is needed."""
+    return serialize(("P", (args, kwargs)))
 
 
 def deserialize_payload(data: bytes, tensors: Optional[list] = None) -> tuple[tuple, dict]:
     kind, payload = deserialize(data)
-    if kind == "T":
-        payload = _join_tensors(payload, tensors or [])
     args, kwargs = payload
     return args, kwargs

@@ -53,8 +53,20 @@ class FunctionCallCancelledError(Error):
 
 def make_payload_item(client: Any, args: tuple, kwargs: dict) -> dict:
     """Serialize one input; offload to the CAS above the inline limit."""
+    import sys as _sys
+
     from ._serialization import serialize_fast
 
+    if "torch" in _sys.modules:
+        # tensors must go through the hook-aware pickler (mesh export /
+        # host staging) — plain pickle would bake in the client's device index
+        from ._serialization import _walk_for_tensors, contains_tensors
+
+        if contains_tensors(args, kwargs):
+            payload = serialize(("P", (args, kwargs)))
+            if len(payload) > INLINE_LIMIT and client.blob_store is not None:
+                return {"payload": b"", "payload_blob": client.blob_store.put(payload)}
+            return {"payload": payload}
     payload = serialize_fast(("P", (args, kwargs)))
     if len(payload) > INLINE_LIMIT:
         store = client.blob_store
