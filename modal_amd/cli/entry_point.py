@@ -477,6 +477,31 @@ def secret_list() -> None:
             click.echo(f"{sid}  {env}/{name}")
 
 
+@entrypoint_cli.group(name="image")
+def image_cli() -> None:
+    """Inspect built images."""
+
+
+@image_cli.command(name="list")
+def image_list() -> None:
+    client = _get_client()
+    svc = client.svc
+    if hasattr(svc, "image_service"):
+        for state in svc.image_service.by_id.values():
+            click.echo(f"{state.image_id}  built={state.built}  {state.recipe_hash[:12]}")
+
+
+@image_cli.command(name="info")
+@click.argument("image_id")
+def image_info(image_id: str) -> None:
+    client = _get_client()
+    info = synchronizer.run(client.svc.image_info(image_id=image_id))
+    click.echo(json.dumps({k: v for k, v in info.items() if k != "build_log"}, indent=2))
+    if info.get("build_log"):
+        click.echo("--- build log (tail) ---")
+        click.echo(info["build_log"][-1500:])
+
+
 # ---- config / profile / token / environment ------------------------------
 
 

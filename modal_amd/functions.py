@@ -60,7 +60,7 @@ def make_payload_item(client: Any, args: tuple, kwargs: dict) -> dict:
     if "torch" in _sys.modules:
         # tensors must go through the hook-aware pickler (mesh export /
         # host staging) — plain pickle would bake in the client's device index
-        from ._serialization import _walk_for_tensors, contains_tensors
+        from ._serialization import contains_tensors
 
         if contains_tensors(args, kwargs):
             payload = serialize(("P", (args, kwargs)))

@@ -141,3 +141,21 @@ def test_queue_used_inside_worker(client):
             count = pusher.remote(q, 5)
             assert count == 5
             assert [q.get() for _ in range(5)] == [0, 2, 4, 6, 8]
+
+
+def test_aio_variants_for_resources(client):
+    """Every resource method carries a working .aio twin."""
+    import asyncio
+
+    async def main():
+        q = await modal.Queue.from_name("aio-q", create_if_missing=True).hydrate.aio()
+        await q.put.aio("via-aio")
+        assert await q.get.aio() == "via-aio"
+        d = await modal.Dict.from_name("aio-d", create_if_missing=True).hydrate.aio()
+        await d.put.aio("k", 1)
+        assert await d.get.aio("k") == 1
+        vol = await modal.Volume.from_name("aio-v", create_if_missing=True).hydrate.aio()
+        entries = await vol.listdir.aio("/")
+        assert entries == []
+
+    asyncio.run(main())
