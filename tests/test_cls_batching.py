@@ -153,3 +153,18 @@ def test_cls_with_options_and_from_name(client):
     remote_cls = modal.Cls.from_name("cls-deployed", "Echo")
     obj = remote_cls()
     assert obj.say.remote("hi") == "echo:hi"
+
+
+def test_batched_actually_groups(client):
+    """The runtime accumulates multiple logical calls into one execution."""
+    app = modal.App("batch-grouping")
+
+    @app.function()
+    @modal.batched(max_batch_size=32, wait_ms=150)
+    def sizes(xs):
+        return [len(xs)] * len(xs)
+
+    with app.run(client=client):
+        out = list(sizes.map(range(64), order_outputs=False))
+        assert len(out) == 64
+        assert max(out) > 1, "no batching happened"
