@@ -84,6 +84,8 @@ class Connection:
         # shm ring pair: our producer side and our consumer side
         self.ring_out: Any = None
         self.ring_in: Any = None
+        self.ring_frames_sent = 0
+        self.ring_frames_received = 0
 
     def attach_rings(self, out_path: Optional[str], in_path: Optional[str], create: bool) -> bool:
         """Attach the shared-memory bulk channel (both sides call this)."""
@@ -115,6 +117,7 @@ class Connection:
             # bulk path: payload through the shm ring, doorbell on the socket
             pushed = self.ring_out.push(body)
             if pushed:
+                self.ring_frames_sent += 1
                 async with self._send_lock:
                     self.writer.write(pack({"t": "rb"}))
                     await self.writer.drain()
@@ -188,6 +191,7 @@ class Connection:
                     # doorbell: drain the inbound ring and dispatch its frames
                     if self.ring_in is not None:
                         for body in self.ring_in.pop_all():
+                            self.ring_frames_received += 1
                             await self._dispatch(
                                 msgpack.unpackb(body, raw=False, strict_map_key=False)
                             )

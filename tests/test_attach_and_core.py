@@ -101,3 +101,27 @@ def test_file_pattern_matcher():
     assert not m("src/main.py")
     m2 = FilePatternMatcher("**/__pycache__")
     assert m2("a/b/__pycache__")
+
+
+def test_ring_carries_bulk_frames(client):
+    """Frames >= RING_MIN_FRAME ride the C++ shm ring, not the socket."""
+    import os as _os
+
+    import modal_amd as modal
+
+    app = modal.App("ring-app")
+
+    @app.function()
+    def digest_one(blob):
+        import hashlib
+
+        return hashlib.sha256(blob).hexdigest()
+
+    with app.run(client=client):
+        payload = _os.urandom(300_000)  # 300 KB unary arg -> bulk frame
+        import hashlib
+
+        assert digest_one.remote(payload) == hashlib.sha256(payload).hexdigest()
+        svc = client.svc
+        sent = sum(w.conn.ring_frames_sent for w in svc.pool.workers.values())
+        assert sent >= 1, "bulk frame never used the shm ring"
