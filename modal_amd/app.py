@@ -125,6 +125,9 @@ def _build_options(
         options["web_config"] = flags["web"]
     if extra:
         options.update(extra)
+    placement = options.get("placement")
+    if placement is not None and hasattr(placement, "to_dict"):
+        options["placement"] = placement.to_dict()
     return options
 
 

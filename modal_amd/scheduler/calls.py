@@ -99,6 +99,7 @@ class FunctionDef:
     volume_mounts: dict = field(default_factory=dict)  # mount path -> volume id
     schedule: Optional[dict] = None  # {"cron": "..."} | {"period": seconds}
     image_id: Optional[str] = None
+    placement: Optional[dict] = None  # {"gpu_index": i, "gpu_set": [..]} hints
 
     def placement_tag(self) -> str:
         return "gpu" if self.needs_gpu else "any"

@@ -317,6 +317,7 @@ class Scheduler:
             volume_mounts=dict(options.get("volume_mounts") or {}),
             schedule=options.get("schedule"),
             image_id=options.get("image_id"),
+            placement=options.get("placement"),
         )
         self.functions[fid] = fdef
         if fdef.min_containers or fdef.buffer_containers:

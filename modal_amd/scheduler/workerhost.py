@@ -521,6 +521,7 @@ class WorkerPool:
             ]
             if fdef.needs_gpu:
                 candidates = [w for w in candidates if w.has_gpu]
+            candidates = self._apply_placement(fdef, candidates)
             if not candidates:
                 continue
             # round-robin over workers by current outstanding (least-loaded first)
@@ -553,6 +554,23 @@ class WorkerPool:
                 # (parity: min/max_containers autoscaler settings,
                 # reference _functions.py:1195-1292)
                 await self._maybe_scale_up(fdef, backlog=len(q), active=len(candidates))
+
+    def _apply_placement(self, fdef: FunctionDef, candidates: list) -> list:
+        """GPU-affinity steering (SchedulerPlacement gpu_index/gpu_set —
+        the single-node analog of region/zone constraints)."""
+        placement = fdef.placement
+        if not placement:
+            return candidates
+        gpu_index = placement.get("gpu_index")
+        if gpu_index is not None:
+            # strict pin: wait for that GPU's worker rather than mis-placing
+            return [w for w in candidates if w.gpu_index == gpu_index]
+        gpu_set = placement.get("gpu_set")
+        if gpu_set:
+            subset = [w for w in candidates if w.gpu_index in gpu_set]
+            if subset:
+                return subset
+        return candidates
 
     async def _maybe_scale_up(self, fdef: FunctionDef, backlog: int, active: int) -> None:
         if self._pending_spawns > 0:

@@ -179,3 +179,36 @@ def test_environments_and_workspace():
     assert "staging" in names and "main" in names
     envs.delete_environment("staging")
     assert Workspace.current().name
+
+
+def test_scheduler_placement_steers_dispatch(run_dir, monkeypatch):
+    """SchedulerPlacement(gpu_index=N) pins work to that GPU's worker."""
+    monkeypatch.setenv("MODAL_AMD_FAKE_GPUS", "3")
+    monkeypatch.setenv("MODAL_AMD_WORKER_COUNT", "3")
+    from modal_amd._sync import synchronizer
+    from modal_amd.client import _Client
+    from modal_amd.scheduler.core import Scheduler
+
+    async def make():
+        s = Scheduler(run_dir=run_dir)
+        await s.start()
+        c = _Client(s, "client")
+        _Client.set_default(c)
+        return s, c
+
+    s, c = synchronizer.run(make())
+    try:
+        app = modal.App("placement-app")
+
+        @app.function(placement=modal.SchedulerPlacement(gpu_index=2))
+        def where():
+            import os as _os
+
+            return _os.environ.get("HIP_VISIBLE_DEVICES")
+
+        with app.run(client=c):
+            results = {where.remote() for _ in range(4)}
+            assert results == {"2"}, results
+    finally:
+        synchronizer.run(c.close())
+        _Client._singleton = None
