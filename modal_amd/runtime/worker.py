@@ -447,6 +447,12 @@ class WorkerRuntime:
             app_id = msg.get("app_id")
             for fid in [f for f, frt in self.functions.items() if frt.app_id == app_id]:
                 frt = self.functions.pop(fid)
+                # volume commit on exit (parity: task_lifecycle_manager.py:117)
+                for volume_id in set(frt.volumes.values()):
+                    try:
+                        await self.conn.call("volume_commit", {"volume_id": volume_id}, timeout=10)
+                    except Exception:
+                        pass
                 if frt._service is not None:
                     await asyncio.get_running_loop().run_in_executor(
                         self.executor, frt._service.exit
