@@ -131,12 +131,59 @@ def serve(ref: str, timeout: Optional[float]) -> None:
 @click.argument("ref", required=False)
 @click.option("--cmd", default="/bin/bash")
 def shell(ref: Optional[str], cmd: str) -> None:
-    """Interactive shell in a sandbox (parity: modal shell)."""
+    """Interactive shell in a sandbox (parity: modal shell; line-mode pipes)."""
+    import threading
+
     import modal_amd as modal
 
     sb = modal.Sandbox.create(cmd, "-i")
-    click.echo(f"Started sandbox {sb.object_id} running {cmd!r} (non-tty pipe mode)")
-    sb.terminate()
+    click.echo(f"[sandbox {sb.object_id}] {cmd!r} — line mode, 'exit' or Ctrl-D to leave")
+
+    stop = threading.Event()
+
+    def stream(reader: Any, out: Any) -> None:
+        try:
+            while not stop.is_set():
+                data, eof = reader.read_chunk(timeout=0.5)
+                if data:
+                    out.write(data.decode("utf-8", errors="replace") if isinstance(data, bytes) else data)
+                    out.flush()
+                if eof:
+                    stop.set()
+                    return
+        except Exception:
+            stop.set()
+
+    threads = [
+        threading.Thread(target=stream, args=(sb.stdout, sys.stdout), daemon=True),
+        threading.Thread(target=stream, args=(sb.stderr, sys.stderr), daemon=True),
+    ]
+    for t in threads:
+        t.start()
+    try:
+        if sys.stdin.isatty():
+            while not stop.is_set():
+                try:
+                    line = input()
+                except EOFError:
+                    break
+                if line.strip() == "exit":
+                    break
+                sb.stdin.write(line + "\n")
+                sb.stdin.drain()
+        else:
+            # non-interactive stdin: pipe it through and wait
+            data = sys.stdin.read()
+            if data:
+                sb.stdin.write(data)
+            sb.stdin.write_eof()
+            sb.stdin.drain()
+            sb.wait(raise_on_termination=False)
+    except KeyboardInterrupt:
+        pass
+    finally:
+        stop.set()
+        sb.terminate()
 
 
 @entrypoint_cli.command()

@@ -108,3 +108,11 @@ def test_cli_dict_roundtrip(runner, client):
     modal.Dict.from_name("clid").put("k", [1, 2])
     res = runner.invoke(entrypoint_cli, ["dict", "get", "clid", "k"])
     assert "[1, 2]" in res.output
+
+
+def test_cli_shell_piped(runner, client):
+    result = runner.invoke(
+        entrypoint_cli, ["shell", "--cmd", "/bin/bash"], input="echo piped-$((2+3))\n"
+    )
+    assert result.exit_code == 0, result.output
+    assert "piped-5" in result.output
