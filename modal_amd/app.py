@@ -8,7 +8,6 @@ Parity: /root/reference/py/modal/app.py — ``_App`` (:140), decorators
 
 from __future__ import annotations
 
-import inspect
 import typing
 from typing import Any, Callable, Optional, Sequence, Union
 

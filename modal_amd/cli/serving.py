@@ -6,7 +6,6 @@ the reference uses watchfiles; offline we poll mtimes (0.5 s cadence).
 
 from __future__ import annotations
 
-import importlib
 import os
 import sys
 import time

@@ -22,8 +22,6 @@ from typing import Any, Optional
 from ._sync import synchronize_api
 from .exception import (
     AlreadyExistsError,
-    ClientClosed,
-    Error,
     ExecutionError,
     InvalidError,
     NotFoundError,

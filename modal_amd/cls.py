@@ -10,11 +10,11 @@ methods dispatch by name; parametrized instances bind to derived rows.
 from __future__ import annotations
 
 import threading
-from typing import Any, Callable, Optional
+from typing import Any, Optional
 
 from ._serialization import serialize
 from ._sync import synchronizer, unwrap, wrap
-from .exception import InvalidError, NotFoundError
+from .exception import InvalidError
 from .functions import _Function
 from .partial_function import PartialFunction
 

@@ -13,7 +13,6 @@ from typing import Any, AsyncGenerator, Optional
 from ._object import _Object, live_method
 from ._serialization import deserialize, serialize
 from ._sync import synchronize_api, synchronizer, wrap
-from .exception import NotFoundError
 
 
 class _Dict(_Object, type_kind="dict"):

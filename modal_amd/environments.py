@@ -8,7 +8,7 @@ in the scheduler's name tables.
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Any, Optional
+from typing import Optional
 
 from ._sync import dual_function
 

@@ -34,7 +34,6 @@ from .exception import (
     FunctionTimeoutError,
     InternalFailure,
     InvalidError,
-    NotFoundError,
     RemoteError,
 )
 from .scheduler.blobs import INLINE_LIMIT

@@ -13,7 +13,6 @@ like the reference's layer dedup).
 from __future__ import annotations
 
 import os
-import sys
 from typing import Any, Callable, Optional, Sequence, Union
 
 from ._object import _Object, live_method

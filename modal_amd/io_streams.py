@@ -8,7 +8,7 @@ contract (task_command_router_client.py:431-614).
 from __future__ import annotations
 
 import enum
-from typing import Any, AsyncGenerator, Optional
+from typing import Any, AsyncGenerator
 
 from ._sync import synchronize_api
 

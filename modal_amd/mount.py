@@ -10,7 +10,7 @@ materialize once per content set; python-source mounts land on workers'
 from __future__ import annotations
 
 import os
-from typing import Any, Callable, Optional, Sequence, Union
+from typing import Any, Callable, Optional, Union
 
 from ._object import _Object
 from ._sync import synchronize_api

@@ -8,11 +8,11 @@ remove_file) over the volume backend.
 
 from __future__ import annotations
 
-from typing import Any, AsyncGenerator, BinaryIO, Optional
+from typing import Any, BinaryIO
 
-from ._object import _Object, live_method
+from ._object import live_method
 from ._sync import synchronize_api
-from .volume import FileEntry, _Volume
+from .volume import _Volume
 
 
 class _NetworkFileSystem(_Volume, type_kind="nfs"):

@@ -7,8 +7,6 @@ at HBM copy speed, replacing per-item host-side bytes joins.
 
 from __future__ import annotations
 
-from typing import Optional
-
 from . import load_lib
 
 

@@ -19,7 +19,7 @@ client-side retry manager + input_jwts lost-input protocol.
 from __future__ import annotations
 
 import asyncio
-from typing import Any, AsyncGenerator, Optional
+from typing import Any, AsyncGenerator
 
 from ..functions import make_payload_item, process_output_item
 from ..scheduler.calls import GENERIC_STATUS_SUCCESS

@@ -10,14 +10,12 @@ AppClientDisconnect + final log drain on exit (:326-366).
 from __future__ import annotations
 
 import asyncio
-import sys
 from typing import Any, Optional
 
 from ._object import Resolver
 from ._serialization import serialize
 from ._sync import synchronizer, unwrap
 from .client import HEARTBEAT_INTERVAL, _Client
-from .exception import InvalidError
 
 
 def _sanitize_options(options: dict) -> dict:

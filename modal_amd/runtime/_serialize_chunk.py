@@ -9,7 +9,7 @@ from __future__ import annotations
 
 import pickle
 import sys
-from typing import Any, Optional
+from typing import Optional
 
 
 def serialize_value_chunk(values: list) -> tuple[Optional[bytes], Optional[list]]:

@@ -16,7 +16,7 @@ from __future__ import annotations
 import enum
 import gc
 import sys
-from typing import Any, Optional
+from typing import Any
 
 # parity: the runtime retries without snapshot when it sees this exit code
 CUDA_CHECKPOINT_SENTINEL_EXIT = 222

@@ -11,7 +11,7 @@ from __future__ import annotations
 
 import asyncio
 import inspect
-from typing import Any, Callable, Optional
+from typing import Any, Callable
 
 
 def magic_fastapi_app(fn: Callable, method: str, docs: bool) -> Any:

@@ -14,11 +14,11 @@ from __future__ import annotations
 
 import asyncio
 from dataclasses import dataclass
-from typing import Any, AsyncGenerator, Optional, Sequence, Union
+from typing import Any, AsyncGenerator, Optional, Sequence
 
 from ._object import _Object, live_method
 from ._sync import synchronize_api, unwrap, wrap
-from .exception import InvalidError, SandboxTimeoutError
+from .exception import SandboxTimeoutError
 from .io_streams import _StreamReader, _StreamWriter
 
 

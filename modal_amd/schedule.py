@@ -7,9 +7,6 @@ evaluator (see modal_amd/scheduler/cron.py).
 
 from __future__ import annotations
 
-from datetime import timedelta
-from typing import Optional
-
 from .exception import InvalidError
 
 

@@ -14,7 +14,7 @@ from __future__ import annotations
 import hashlib
 import os
 import tempfile
-from typing import Optional, Union
+from typing import Union
 
 INLINE_LIMIT = 2 * 1024 * 1024  # parity: blob_utils.py:36
 SPAWN_INLINE_LIMIT = 8 * 1024  # parity: blob_utils.py:39  (async spawn payloads)

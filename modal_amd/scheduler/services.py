@@ -15,7 +15,7 @@ from __future__ import annotations
 import asyncio
 import time
 from collections import deque
-from typing import Any, Optional
+from typing import Optional
 
 from ..exception import AlreadyExistsError, NotFoundError, QueueFullError
 from ..utils.ids import new_id

@@ -16,11 +16,11 @@ from __future__ import annotations
 import os
 import shutil
 import time
-from typing import Any, Optional
+from typing import Optional
 
 from ..exception import AlreadyExistsError, InvalidError, NotFoundError
 from ..utils.ids import new_id
-from .blobs import BLOCK_SIZE, BlobStore
+from .blobs import BlobStore
 
 
 class VolumeState:

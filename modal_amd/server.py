@@ -7,8 +7,6 @@ it maps onto the class-service machinery with @enter-started serving.
 
 from __future__ import annotations
 
-from typing import Any
-
 from .cls import Cls
 
 

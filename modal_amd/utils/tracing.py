@@ -13,7 +13,7 @@ from __future__ import annotations
 import contextlib
 import ctypes
 import os
-from typing import Any, Iterator, Optional
+from typing import Iterator, Optional
 
 _lib: Optional[ctypes.CDLL] = None
 _checked = False

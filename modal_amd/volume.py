@@ -15,7 +15,7 @@ from typing import Any, AsyncGenerator, BinaryIO, Optional, Union
 
 from ._object import _Object, live_method
 from ._sync import synchronize_api, synchronizer, wrap
-from .exception import InvalidError, NotFoundError
+from .exception import InvalidError
 from .scheduler.blobs import BLOCK_SIZE
 
 
