@@ -54,6 +54,35 @@ OUTPUT_BATCH_MAX = 512
 
 _app_id_var: contextvars.ContextVar[str] = contextvars.ContextVar("modal_amd_app_id", default="")
 
+import threading as _threading
+
+_slot_tls = _threading.local()
+
+
+def _invoke_on_slot_stream(fn: Any, args: tuple, kwargs: dict) -> Any:
+    """Run a user callable on this executor thread's own HIP stream.
+
+    Concurrency slots each get a side stream ("per-GPU worker
+    runs N HIP streams"), so one item's device sync (``.item()``, ``.cpu()``)
+    waits only its own work instead of convoying every slot's kernels on the
+    default stream. Disable with MODAL_AMD_SLOT_STREAMS=0.
+    """
+    torch = sys.modules.get("torch")
+    if (
+        torch is None
+        or os.environ.get("MODAL_AMD_SLOT_STREAMS", "1") in ("0", "false")
+        or not torch.cuda.is_initialized()
+    ):
+        return fn(*args, **kwargs)
+    stream = getattr(_slot_tls, "stream", None)
+    if stream is None:
+        stream = torch.cuda.Stream()
+        _slot_tls.stream = stream
+    with torch.cuda.stream(stream):
+        result = fn(*args, **kwargs)
+    stream.synchronize()  # results must be safe to read/serialize anywhere
+    return result
+
 #: the live worker runtime of this process (None in client processes);
 #: serialization hooks use it to export/fetch device tensors
 RUNTIME: Optional["WorkerRuntime"] = None
@@ -664,7 +693,7 @@ class WorkerRuntime:
                     tok_i = _current_input_id.set(f"in-{call_id[3:]}-{ci}")
                     try:
                         args, kwargs = self._resolve_item_args(item)
-                        values[ci] = fn(*args, **kwargs)
+                        values[ci] = _invoke_on_slot_stream(fn, args, kwargs)
                     except BaseException as exc:
                         errors[ci] = (
                             self._serialize_exception(exc),
@@ -871,8 +900,8 @@ class WorkerRuntime:
         def invoke() -> Any:
             if trace_enabled():
                 with trace_range(f"modal_amd::{frt.name}"):
-                    return ctx.run(fn, *args, **kwargs)
-            return ctx.run(fn, *args, **kwargs)
+                    return ctx.run(_invoke_on_slot_stream, fn, args, kwargs)
+            return ctx.run(_invoke_on_slot_stream, fn, args, kwargs)
 
         fut = asyncio.get_running_loop().run_in_executor(self.executor, invoke)
         if frt.timeout:
