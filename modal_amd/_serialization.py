@@ -85,7 +85,9 @@ def _torch_tensor_cls() -> Any:
     import sys as _sys
 
     torch = _sys.modules.get("torch")
-    return torch.Tensor if torch is not None else None
+    # torch may be *partially initialized* (another thread mid-import):
+    # getattr can raise/return None until the module body finishes.
+    return getattr(torch, "Tensor", None) if torch is not None else None
 
 
 class Pickler(cloudpickle.CloudPickler):

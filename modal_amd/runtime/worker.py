@@ -68,11 +68,17 @@ def _invoke_on_slot_stream(fn: Any, args: tuple, kwargs: dict) -> Any:
     default stream. Disable with MODAL_AMD_SLOT_STREAMS=0.
     """
     torch = sys.modules.get("torch")
-    if (
-        torch is None
-        or os.environ.get("MODAL_AMD_SLOT_STREAMS", "1") in ("0", "false")
-        or not torch.cuda.is_initialized()
-    ):
+    # sys.modules can expose a *partially initialized* torch while another
+    # thread is mid-import; fall back to a plain call until it is whole.
+    cuda = getattr(torch, "cuda", None) if torch is not None else None
+    try:
+        if (
+            cuda is None
+            or os.environ.get("MODAL_AMD_SLOT_STREAMS", "1") in ("0", "false")
+            or not cuda.is_initialized()
+        ):
+            return fn(*args, **kwargs)
+    except AttributeError:
         return fn(*args, **kwargs)
     stream = getattr(_slot_tls, "stream", None)
     if stream is None:
