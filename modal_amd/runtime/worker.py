@@ -62,10 +62,13 @@ _slot_tls = _threading.local()
 def _invoke_on_slot_stream(fn: Any, args: tuple, kwargs: dict) -> Any:
     """Run a user callable on this executor thread's own HIP stream.
 
-    Concurrency slots each get a side stream ("per-GPU worker
-    runs N HIP streams"), so one item's device sync (``.item()``, ``.cpu()``)
-    waits only its own work instead of convoying every slot's kernels on the
-    default stream. Disable with MODAL_AMD_SLOT_STREAMS=0.
+    Concurrency slots each get a side stream, so one item's device sync
+    (``.item()``, ``.cpu()``) waits only its own work instead of convoying
+    every slot's kernels on the default stream. OFF by default: measured
+    2x SLOWER on the map bench (21.5k vs 44.3k items/s, same box) — per-item
+    ``stream.synchronize()`` busy-spins cost more than the convoy for short
+    per-item ops. Opt in with MODAL_AMD_SLOT_STREAMS=1 for workloads with
+    long per-item kernels.
     """
     torch = sys.modules.get("torch")
     # sys.modules can expose a *partially initialized* torch while another
@@ -74,7 +77,7 @@ def _invoke_on_slot_stream(fn: Any, args: tuple, kwargs: dict) -> Any:
     try:
         if (
             cuda is None
-            or os.environ.get("MODAL_AMD_SLOT_STREAMS", "1") in ("0", "false")
+            or os.environ.get("MODAL_AMD_SLOT_STREAMS", "0") in ("0", "false")
             or not cuda.is_initialized()
         ):
             return fn(*args, **kwargs)
