@@ -22,8 +22,8 @@ def magic_fastapi_app(fn: Callable, method: str, docs: bool) -> Any:
     return app
 
 
-async def run_asgi(app: Any, req: dict) -> dict:
-    scope = {
+def _build_scope(req: dict) -> dict:
+    return {
         "type": "http",
         "asgi": {"version": "3.0", "spec_version": "2.3"},
         "http_version": "1.1",
@@ -37,6 +37,46 @@ async def run_asgi(app: Any, req: dict) -> dict:
         "server": ("127.0.0.1", 80),
         "client": ("127.0.0.1", 0),
     }
+
+
+async def run_asgi_streaming(app: Any, req: dict, emit: Callable) -> None:
+    """Run the ASGI app, forwarding the response head then each body chunk
+    through ``emit`` as they are produced (parity: the reference streams
+    bodies via data_out chunks, asgi.py:140)."""
+    scope = _build_scope(req)
+    incoming = [{"type": "http.request", "body": req.get("body") or b"", "more_body": False}]
+    done = asyncio.Event()
+    state = {"status": 500, "headers": [], "head_sent": False}
+
+    async def receive() -> dict:
+        if incoming:
+            return incoming.pop(0)
+        await done.wait()
+        return {"type": "http.disconnect"}
+
+    async def send_head_once() -> None:
+        if not state["head_sent"]:
+            state["head_sent"] = True
+            await emit({"status": state["status"], "headers": state["headers"]})
+
+    async def send(message: dict) -> None:
+        if message.get("type") == "http.response.start":
+            state["status"] = message["status"]
+            state["headers"] = [[k.decode(), v.decode()] for k, v in message.get("headers", [])]
+        elif message.get("type") == "http.response.body":
+            await send_head_once()
+            body = message.get("body", b"")
+            if body:
+                await emit(bytes(body))
+            if not message.get("more_body"):
+                done.set()
+
+    await app(scope, receive, send)
+    await send_head_once()  # head even for empty-bodied responses
+
+
+async def run_asgi(app: Any, req: dict) -> dict:
+    scope = _build_scope(req)
     incoming = [{"type": "http.request", "body": req.get("body") or b"", "more_body": False}]
     sent: list[dict] = []
     done = asyncio.Event()
@@ -68,6 +108,21 @@ async def run_asgi(app: Any, req: dict) -> dict:
 
 
 def run_wsgi(app: Any, req: dict) -> dict:
+    environ = _wsgi_environ(req)
+    captured: dict = {"status": 500, "headers": []}
+
+    def start_response(status: str, headers: list, exc_info: Any = None) -> Any:
+        captured["status"] = int(status.split()[0])
+        captured["headers"] = [[k, v] for k, v in headers]
+
+    chunks = app(environ, start_response)
+    body = b"".join(chunks)
+    if hasattr(chunks, "close"):
+        chunks.close()
+    return {"status": captured["status"], "headers": captured["headers"], "body": body}
+
+
+def _wsgi_environ(req: dict) -> dict:
     import io
 
     environ = {
@@ -87,25 +142,40 @@ def run_wsgi(app: Any, req: dict) -> dict:
         "wsgi.run_once": False,
     }
     for key, value in req.get("headers", []):
-        cgi_key = "HTTP_" + key.upper().replace("-", "_")
         if key.lower() == "content-type":
             environ["CONTENT_TYPE"] = value
         elif key.lower() == "content-length":
             environ["CONTENT_LENGTH"] = value
         else:
-            environ[cgi_key] = value
+            environ["HTTP_" + key.upper().replace("-", "_")] = value
+    return environ
 
-    captured: dict = {"status": 500, "headers": []}
+
+def run_wsgi_streaming(app: Any, req: dict, emit_threadsafe: Callable) -> None:
+    """Iterate the WSGI body on this (executor) thread, emitting the head
+    before the first chunk (WSGI apps may call start_response lazily)."""
+    environ = _wsgi_environ(req)
+    captured: dict = {"status": 500, "headers": [], "head_sent": False}
 
     def start_response(status: str, headers: list, exc_info: Any = None) -> Any:
         captured["status"] = int(status.split()[0])
         captured["headers"] = [[k, v] for k, v in headers]
 
+    def head_once() -> None:
+        if not captured["head_sent"]:
+            captured["head_sent"] = True
+            emit_threadsafe({"status": captured["status"], "headers": captured["headers"]})
+
     chunks = app(environ, start_response)
-    body = b"".join(chunks)
-    if hasattr(chunks, "close"):
-        chunks.close()
-    return {"status": captured["status"], "headers": captured["headers"], "body": body}
+    try:
+        for chunk in chunks:
+            head_once()
+            if chunk:
+                emit_threadsafe(bytes(chunk))
+    finally:
+        head_once()
+        if hasattr(chunks, "close"):
+            chunks.close()
 
 
 class WebEndpointRuntime:
@@ -129,6 +199,54 @@ class WebEndpointRuntime:
             elif self.kind == "wsgi":
                 self._app = self.raw_fn()
         return self._app
+
+    async def handle_streaming(self, req: dict, emit: Callable) -> None:
+        """Streaming counterpart of :meth:`handle`: calls ``await emit(head)``
+        with ``{"status", "headers"}`` then ``await emit(chunk_bytes)`` per
+        body chunk, returning when the response is complete."""
+        if self.kind in ("fastapi", "asgi"):
+            await run_asgi_streaming(self._ensure_app(), req, emit)
+            return
+        if self.kind == "wsgi":
+            loop = asyncio.get_running_loop()
+
+            def emit_threadsafe(obj: Any) -> None:
+                asyncio.run_coroutine_threadsafe(emit(obj), loop).result()
+
+            await loop.run_in_executor(
+                None, run_wsgi_streaming, self._ensure_app(), req, emit_threadsafe
+            )
+            return
+        if self.kind == "web_server":
+            if not self._started:
+                self._started = True
+                result = self.raw_fn()
+                if inspect.iscoroutine(result):
+                    await result
+                await asyncio.sleep(min(self.config.get("startup_timeout", 5.0), 0.5))
+            await self._proxy_streaming(req, emit)
+            return
+        raise ValueError(f"Unknown web endpoint type {self.kind}")
+
+    async def _proxy_streaming(self, req: dict, emit: Callable) -> None:
+        import aiohttp
+
+        port = self.config["port"]
+        url = f"http://127.0.0.1:{port}{req.get('path') or '/'}"
+        if req.get("query_string"):
+            url += f"?{req['query_string']}"
+        async with aiohttp.ClientSession() as session:
+            async with session.request(
+                req["method"], url, data=req.get("body") or None,
+                headers={k: v for k, v in req.get("headers", [])},
+            ) as resp:
+                await emit({
+                    "status": resp.status,
+                    "headers": [[k, v] for k, v in resp.headers.items()],
+                })
+                async for chunk in resp.content.iter_chunked(1 << 16):
+                    if chunk:
+                        await emit(chunk)
 
     async def handle(self, req: dict) -> dict:
         if self.kind in ("fastapi", "asgi"):

@@ -600,10 +600,21 @@ class WorkerRuntime:
 
                         frt._web_runtime = WebEndpointRuntime(frt.web_config, frt.load())
                     args, kwargs = await self._decode_args_async(item)
-                    result = await frt._web_runtime.handle(args[0])
+                    # stream head + body chunks over the generator data
+                    # channel (parity: asgi.py:140 streams via data_out)
+                    index = 0
+
+                    async def emit(obj: Any) -> None:
+                        nonlocal index
+                        await self._send_gen_item(token, index, obj)
+                        index += 1
+
+                    await frt._web_runtime.handle_streaming(args[0], emit)
+                    await self.conn.send({"t": "gen_data", "token": token, "index": index, "done": True})
                     self.post_output(
                         token, frt.function_id, GENERIC_STATUS_SUCCESS,
-                        serialize(result), DataFormat.PICKLE,
+                        serialize_data_format(GeneratorDone(items_total=index), DataFormat.GENERATOR_DONE),
+                        DataFormat.GENERATOR_DONE,
                     )
                     return
                 fn = frt.get_callable(item.get("method", ""))

@@ -82,15 +82,59 @@ class WebGateway:
         )
         call_id = resp["function_call_id"]
         await self.scheduler.function_finish_inputs(function_call_id=call_id)
-        rec = await self.scheduler.function_wait_output(call_id, 0, timeout=300)
-        if rec.status != GENERIC_STATUS_SUCCESS:
-            return web.Response(status=500, text=rec.exc_repr or "function failed")
+        # The worker streams {status, headers} then body chunks over the
+        # generator data channel; forward them as they arrive. The final
+        # output (GeneratorDone or a failure) settles the call.
         from .._serialization import deserialize
 
-        out = deserialize(rec.output)
-        headers = {k: v for k, v in out.get("headers", [])}
-        headers.pop("Content-Length", None)
-        headers.pop("content-length", None)
-        return web.Response(
-            status=out.get("status", 200), body=out.get("body", b""), headers=headers
+        record = self.scheduler._call(call_id)
+        q = record.gen_queue(0)
+        output_task = asyncio.ensure_future(
+            self.scheduler.function_wait_output(call_id, 0, timeout=300)
         )
+        stream: Any = None
+
+        async def forward(entry: tuple) -> bool:
+            """Apply one gen-channel entry; True when the stream is done."""
+            nonlocal stream
+            _index, data, _fmt, done_flag = entry
+            if done_flag:
+                return True
+            value = deserialize(data)
+            if stream is None:
+                headers = {k: v for k, v in value.get("headers", [])}
+                headers.pop("Content-Length", None)
+                headers.pop("content-length", None)
+                stream = web.StreamResponse(status=value.get("status", 200), headers=headers)
+                await stream.prepare(request)
+            else:
+                await stream.write(value)
+            return False
+
+        try:
+            finished = False
+            while not finished:
+                get_task = asyncio.ensure_future(q.get())
+                await asyncio.wait({get_task, output_task}, return_when=asyncio.FIRST_COMPLETED)
+                if get_task.done():
+                    finished = await forward(get_task.result())
+                    continue
+                get_task.cancel()
+                rec = output_task.result()
+                if rec.status != GENERIC_STATUS_SUCCESS:
+                    if stream is None:
+                        return web.Response(status=500, text=rec.exc_repr or "function failed")
+                    break  # mid-stream failure: truncate the response
+                # success settled first: drain whatever is already queued
+                while not finished:
+                    try:
+                        finished = await forward(q.get_nowait())
+                    except asyncio.QueueEmpty:
+                        finished = True
+        finally:
+            if not output_task.done():
+                output_task.cancel()
+        if stream is None:
+            return web.Response(status=500, text="web function produced no response")
+        await stream.write_eof()
+        return stream

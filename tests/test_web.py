@@ -91,3 +91,54 @@ def test_fastapi_post_body(client):
         assert status in (200, 422), body
         if status == 200:
             assert json.loads(body) == {"got": {"a": 1}}
+
+
+def test_streaming_response_arrives_incrementally(client):
+    """Body chunks reach the HTTP client BEFORE the handler finishes
+    (true streaming through the generator data channel, not buffering)."""
+    import time
+
+    app = modal.App("stream-app")
+
+    @app.function()
+    @modal.asgi_app()
+    def streamer():
+        import asyncio
+
+        async def app_impl(scope, receive, send):
+            await send({"type": "http.response.start", "status": 200, "headers": []})
+            await send({"type": "http.response.body", "body": b"early|", "more_body": True})
+            await asyncio.sleep(1.5)
+            await send({"type": "http.response.body", "body": b"late", "more_body": False})
+
+        return app_impl
+
+    with app.run(client=client):
+        t0 = time.monotonic()
+        with urllib.request.urlopen(streamer.web_url, timeout=30) as resp:
+            first = resp.read(6)
+            first_latency = time.monotonic() - t0
+            rest = resp.read()
+        assert first == b"early|"
+        assert rest == b"late"
+        # the first chunk must not have waited for the 1.5 s sleep
+        assert first_latency < 1.2, f"first chunk took {first_latency:.2f}s (buffered?)"
+
+
+def test_streaming_wsgi_chunks(client):
+    app = modal.App("wsgi-stream-app")
+
+    @app.function()
+    @modal.wsgi_app()
+    def wsgi_streamer():
+        def app_impl(environ, start_response):
+            start_response("200 OK", [("Content-Type", "text/plain")])
+            for i in range(5):
+                yield f"chunk{i};".encode()
+
+        return app_impl
+
+    with app.run(client=client):
+        status, body = _get(wsgi_streamer.web_url)
+        assert status == 200
+        assert body == b"".join(f"chunk{i};".encode() for i in range(5))
