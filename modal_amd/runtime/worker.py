@@ -118,6 +118,12 @@ class WorkerRPCTarget:
     async def ping(self) -> str:
         return self._runtime.task_id
 
+    async def app_stop(self, app_id: str) -> bool:
+        """Synchronous teardown: the scheduler awaits @exit hooks and the
+        exit-time volume commit before app.run() returns."""
+        await self._runtime._app_teardown(app_id)
+        return True
+
 
 class FunctionRuntime:
     """Worker-side state for one registered function."""
@@ -482,19 +488,7 @@ class WorkerRuntime:
         elif kind == "dev_recv":
             self.mesh.submit_recv(msg["xfer_id"], msg["meta"], msg["src_rank"])
         elif kind == "app_stop":
-            app_id = msg.get("app_id")
-            for fid in [f for f, frt in self.functions.items() if frt.app_id == app_id]:
-                frt = self.functions.pop(fid)
-                # volume commit on exit (parity: task_lifecycle_manager.py:117)
-                for volume_id in set(frt.volumes.values()):
-                    try:
-                        await self.conn.call("volume_commit", {"volume_id": volume_id}, timeout=10)
-                    except Exception:
-                        pass
-                if frt._service is not None:
-                    await asyncio.get_running_loop().run_in_executor(
-                        self.executor, frt._service.exit
-                    )
+            await self._app_teardown(msg.get("app_id"))
         elif kind == "shutdown":
             self._shutdown.set()
 
@@ -660,6 +654,26 @@ class WorkerRuntime:
                     _set_cluster_info(None)
                 _app_id_var.reset(app_tok)
                 _reset_current_context(ctx_tokens)
+
+    async def _app_teardown(self, app_id: str) -> None:
+        """Drop the app's services: @exit hooks run BEFORE the exit-time
+        volume commit so files they write are included (parity: lifecycle
+        finalization then task_lifecycle_manager.py:117-120)."""
+        volume_ids: set = set()
+        for fid in [f for f, frt in self.functions.items() if frt.app_id == app_id]:
+            frt = self.functions.pop(fid)
+            volume_ids.update(frt.volumes.values())
+            if frt._service is not None:
+                await asyncio.get_running_loop().run_in_executor(
+                    self.executor, frt._service.exit
+                )
+        if volume_ids:
+            await asyncio.get_running_loop().run_in_executor(None, os.sync)
+            for volume_id in volume_ids:
+                try:
+                    await self.conn.call("volume_commit", {"volume_id": volume_id}, timeout=10)
+                except Exception:
+                    pass
 
     def _setup_cluster(self, cluster: dict) -> None:
         """Rank/world bootstrap for @clustered gangs: env for

@@ -237,13 +237,18 @@ class Scheduler:
             await self._notify_app_stop(app_id)
 
     async def _notify_app_stop(self, app_id: str) -> None:
-        """Tell workers to tear down this app's services (@exit hooks run;
-        parity: container shutdown lifecycle, task_lifecycle_manager.py:78)."""
-        for w in list(self.pool.workers.values()):
+        """Tell workers to tear down this app's services and WAIT for them:
+        @exit hooks + exit-time volume commit finish before app.run()
+        returns (parity: container shutdown lifecycle,
+        task_lifecycle_manager.py:78,117)."""
+
+        async def one(w: Any) -> None:
             try:
-                await w.conn.send({"t": "app_stop", "app_id": app_id})
+                await w.conn.call("app_stop", {"app_id": app_id}, timeout=30)
             except Exception:
                 pass
+
+        await asyncio.gather(*(one(w) for w in list(self.pool.workers.values())))
 
     async def app_stop(self, app_id: str) -> None:
         app = self._app(app_id)

@@ -145,3 +145,32 @@ def test_image_run_function_failure_surfaces(client):
 
     with pytest.raises(ExecutionError, match="build exploded|run_function"):
         img.hydrate()
+
+
+def test_exit_hook_writes_land_before_volume_commit(client):
+    """@exit hooks run before the exit-time volume commit (parity:
+    lifecycle finalization then task_lifecycle_manager.py:117-120), so
+    files written in @exit are visible in the committed volume."""
+    app = modal.App("exit-vol-app")
+    vol = modal.Volume.from_name("exit-vol", create_if_missing=True)
+
+    @app.cls(volumes={"/exit-vol": vol})
+    class Svc:
+        @modal.enter()
+        def up(self):
+            self.n = 0
+
+        @modal.method()
+        def bump(self):
+            self.n += 1
+            return self.n
+
+        @modal.exit()
+        def down(self):
+            with open("/exit-vol/final.txt", "w") as f:
+                f.write(f"count={self.n}")
+
+    with app.run(client=client):
+        assert Svc().bump.remote() == 1
+    v2 = modal.Volume.from_name("exit-vol")
+    assert b"".join(v2.read_file("final.txt")) == b"count=1"
