@@ -131,10 +131,18 @@ def serve(ref: str, timeout: Optional[float]) -> None:
 @click.argument("ref", required=False)
 @click.option("--cmd", default="/bin/bash")
 def shell(ref: Optional[str], cmd: str) -> None:
-    """Interactive shell in a sandbox (parity: modal shell; line-mode pipes)."""
+    """Interactive shell in a sandbox (parity: modal shell).
+
+    On a TTY: full PTY mode — raw terminal relay with resize propagation
+    (parity: the reference's PTY shell, _pty.py + shell.py). Piped stdin
+    falls back to line mode."""
     import threading
 
     import modal_amd as modal
+
+    if sys.stdin.isatty():
+        _pty_shell(modal, cmd)
+        return
 
     sb = modal.Sandbox.create(cmd, "-i")
     click.echo(f"[sandbox {sb.object_id}] {cmd!r} — line mode, 'exit' or Ctrl-D to leave")
@@ -184,6 +192,72 @@ def shell(ref: Optional[str], cmd: str) -> None:
     finally:
         stop.set()
         sb.terminate()
+
+
+def _pty_shell(modal: Any, cmd: str) -> None:
+    """Raw-terminal PTY relay: local tty <-> sandbox PTY exec."""
+    import os
+    import shutil
+    import signal
+    import termios
+    import threading
+    import tty
+
+    size = shutil.get_terminal_size()
+    sb = modal.Sandbox.create("sleep", "86400")
+    p = sb.exec(cmd, pty_info={"rows": size.lines, "cols": size.columns}, text=False)
+    click.echo(f"[sandbox {sb.object_id}] {cmd!r} — PTY mode, exit the shell to leave")
+
+    stop = threading.Event()
+    resize_needed = threading.Event()
+    signal.signal(signal.SIGWINCH, lambda *_a: resize_needed.set())
+
+    def pump_in() -> None:
+        try:
+            while not stop.is_set():
+                data = os.read(0, 4096)
+                if not data:
+                    return
+                p.stdin.write(data)
+                p.stdin.drain()
+        except Exception:
+            pass
+
+    def pump_out() -> None:
+        try:
+            while not stop.is_set():
+                if resize_needed.is_set():
+                    resize_needed.clear()
+                    s = shutil.get_terminal_size()
+                    p.resize(s.lines, s.columns)
+                data, eof = p.stdout.read_chunk(timeout=0.5)
+                if data:
+                    out = data if isinstance(data, bytes) else data.encode()
+                    sys.stdout.buffer.write(out)
+                    sys.stdout.buffer.flush()
+                if eof:
+                    stop.set()
+                    return
+        except Exception:
+            stop.set()
+
+    saved = termios.tcgetattr(0)
+    tty.setraw(0)
+    threads = [
+        threading.Thread(target=pump_in, daemon=True),
+        threading.Thread(target=pump_out, daemon=True),
+    ]
+    for t in threads:
+        t.start()
+    try:
+        p.wait()
+    except KeyboardInterrupt:
+        pass
+    finally:
+        stop.set()
+        termios.tcsetattr(0, termios.TCSADRAIN, saved)
+        sb.terminate()
+        sys.stdout.write("\n")
 
 
 @entrypoint_cli.command()

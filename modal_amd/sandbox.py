@@ -58,6 +58,10 @@ class _ContainerProcess:
         self._stderr: Optional[_StreamReader] = None
         self._stdin: Optional[_StreamWriter] = None
 
+    async def resize(self, rows: int, cols: int) -> None:
+        """Resize a PTY-backed exec's terminal (parity: TaskExecResize)."""
+        await self._client.svc.sandbox_resize(target_id=self._exec_id, rows=rows, cols=cols)
+
     @property
     def stdout(self) -> _StreamReader:
         if self._stdout is None:
@@ -353,7 +357,11 @@ class _Sandbox(_Object, type_kind="sandbox"):
         bufsize: int = -1,
         stdout: Any = None,
         stderr: Any = None,
+        pty_info: Optional[dict] = None,
     ) -> _ContainerProcess:
+        """``pty_info={"rows": R, "cols": C}`` runs the command on a
+        pseudo-terminal (stdout+stderr merged; parity: reference
+        sandbox.py exec pty_info)."""
         env_dict = dict(env or {})
         if secrets:
             env_dict.update(await _resolve_env(secrets, self._client))
@@ -363,6 +371,9 @@ class _Sandbox(_Object, type_kind="sandbox"):
             env=env_dict or None,
             workdir=workdir,
             timeout=timeout,
+            pty=pty_info is not None,
+            rows=(pty_info or {}).get("rows", 24),
+            cols=(pty_info or {}).get("cols", 80),
         )
         return _ContainerProcess(resp["exec_id"], self._client, text=text)
 

@@ -79,7 +79,7 @@ class RPCAdapter:
         "volume_delete", "volume_rename", "volume_dir",
         "sandbox_create", "sandbox_wait", "sandbox_terminate", "sandbox_poll", "sandbox_stdio_read",
         "sandbox_stdin_write", "sandbox_exec", "sandbox_list", "sandbox_set_tags",
-        "sandbox_from_name", "sandbox_snapshot_fs", "sandbox_fs_op",
+        "sandbox_from_name", "sandbox_snapshot_fs", "sandbox_fs_op", "sandbox_resize",
         "image_get_or_create", "image_info", "mount_get_or_create",
         "device_transfer", "tensor_pull_relay",
     }
@@ -989,8 +989,16 @@ class Scheduler:
             **kwargs,
         )
 
-    async def sandbox_exec(self, sandbox_id, cmd, env=None, workdir=None, timeout=None, exec_id=None) -> dict:
-        return await self.sandbox_service.exec(sandbox_id, cmd, env, workdir, timeout, exec_id)
+    async def sandbox_exec(
+        self, sandbox_id, cmd, env=None, workdir=None, timeout=None, exec_id=None,
+        pty=False, rows=24, cols=80,
+    ) -> dict:
+        return await self.sandbox_service.exec(
+            sandbox_id, cmd, env, workdir, timeout, exec_id, pty=pty, rows=rows, cols=cols
+        )
+
+    async def sandbox_resize(self, target_id, rows, cols) -> None:
+        await self.sandbox_service.resize(target_id, rows, cols)
 
     async def sandbox_stdio_read(self, target_id, fd, offset=0, max_bytes=1 << 20, timeout=55.0) -> dict:
         return await self.sandbox_service.stdio_read(target_id, fd, offset, max_bytes, timeout)

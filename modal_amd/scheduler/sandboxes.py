@@ -38,6 +38,7 @@ class _ProcState:
         self.returncode: Optional[int] = None
         self.stdin_offset = 0  # bytes accepted so far (resume dedup)
         self._readers: list[asyncio.Task] = []
+        self.pty_master: Optional[int] = None  # PTY-backed exec (shell)
 
     async def attach(self, proc: asyncio.subprocess.Process) -> None:
         self.proc = proc
@@ -47,6 +48,45 @@ class _ProcState:
         if proc.stderr is not None:
             self._readers.append(loop.create_task(self._pump(proc.stderr, 2)))
         self._readers.append(loop.create_task(self._wait()))
+
+    async def attach_pty(self, proc: asyncio.subprocess.Process, master: int) -> None:
+        """PTY-backed process: output (stdout+stderr merged, PTY semantics)
+        comes from the master fd; stdin writes go to it."""
+        self.proc = proc
+        self.pty_master = master
+        loop = asyncio.get_running_loop()
+        self._readers.append(loop.create_task(self._pump_pty()))
+        self._readers.append(loop.create_task(self._wait()))
+
+    async def _pump_pty(self) -> None:
+        loop = asyncio.get_running_loop()
+        while True:
+            try:
+                chunk = await loop.run_in_executor(None, os.read, self.pty_master, 65536)
+            except OSError:
+                chunk = b""  # EIO when the last slave fd closes = EOF
+            async with self.cond:
+                if not chunk:
+                    self.stdout_eof = True
+                    self.stderr_eof = True
+                    self.cond.notify_all()
+                    try:
+                        os.close(self.pty_master)
+                    except OSError:
+                        pass
+                    self.pty_master = None
+                    return
+                self.stdout.extend(chunk)
+                self.cond.notify_all()
+
+    def resize(self, rows: int, cols: int) -> None:
+        if self.pty_master is None:
+            raise NotFoundError(f"{self.proc_id} has no PTY")
+        import fcntl
+        import struct
+        import termios
+
+        fcntl.ioctl(self.pty_master, termios.TIOCSWINSZ, struct.pack("HHHH", rows, cols, 0, 0))
 
     async def _pump(self, stream: asyncio.StreamReader, fd: int) -> None:
         buf = self.stdout if fd == 1 else self.stderr
@@ -94,6 +134,15 @@ class _ProcState:
     async def write_stdin(self, offset: int, data: bytes, eof: bool) -> int:
         """Offset-resumable stdin: bytes before stdin_offset are dedup'd
         (parity: resumable stdin stream, reference :523-614)."""
+        if self.pty_master is not None:
+            if offset < self.stdin_offset:
+                data = data[self.stdin_offset - offset :]
+            if data:
+                loop = asyncio.get_running_loop()
+                await loop.run_in_executor(None, os.write, self.pty_master, bytes(data))
+                self.stdin_offset += len(data)
+            # PTY has no half-close; EOF is signalled in-band (^D) by clients
+            return self.stdin_offset
         if self.proc is None or self.proc.stdin is None:
             raise NotFoundError("stdin not available")
         if offset < self.stdin_offset:
@@ -240,6 +289,9 @@ class SandboxService:
         workdir: Optional[str] = None,
         timeout: Optional[float] = None,
         exec_id: Optional[str] = None,
+        pty: bool = False,
+        rows: int = 24,
+        cols: int = 80,
     ) -> dict:
         sb = self._get(sandbox_id)
         # exec_id idempotency (parity: command-router exec_id semantics)
@@ -250,16 +302,49 @@ class SandboxService:
         sb.execs[exec_id] = state
         full_env = dict(sb.env)
         full_env.update(env or {})
-        proc = await asyncio.create_subprocess_exec(
-            *cmd,
-            cwd=workdir or sb.workdir,
-            env=full_env,
-            stdin=asyncio.subprocess.PIPE,
-            stdout=asyncio.subprocess.PIPE,
-            stderr=asyncio.subprocess.PIPE,
-            start_new_session=True,
-        )
-        await state.attach(proc)
+        if pty:
+            # interactive exec (modal-amd shell): run on a pseudo-terminal,
+            # stdout+stderr merged, child gets it as controlling tty
+            import fcntl
+            import pty as _pty
+            import struct
+            import termios
+
+            master, slave = _pty.openpty()
+            fcntl.ioctl(slave, termios.TIOCSWINSZ, struct.pack("HHHH", rows, cols, 0, 0))
+            full_env.setdefault("TERM", "xterm-256color")
+
+            def _make_ctty() -> None:
+                os.setsid()
+                fcntl.ioctl(0, termios.TIOCSCTTY, 0)
+
+            try:
+                proc = await asyncio.create_subprocess_exec(
+                    *cmd,
+                    cwd=workdir or sb.workdir,
+                    env=full_env,
+                    stdin=slave,
+                    stdout=slave,
+                    stderr=slave,
+                    preexec_fn=_make_ctty,
+                )
+            except BaseException:
+                os.close(master)
+                raise
+            finally:
+                os.close(slave)
+            await state.attach_pty(proc, master)
+        else:
+            proc = await asyncio.create_subprocess_exec(
+                *cmd,
+                cwd=workdir or sb.workdir,
+                env=full_env,
+                stdin=asyncio.subprocess.PIPE,
+                stdout=asyncio.subprocess.PIPE,
+                stderr=asyncio.subprocess.PIPE,
+                start_new_session=True,
+            )
+            await state.attach(proc)
         if timeout:
 
             async def _timeout_kill() -> None:
@@ -276,6 +361,10 @@ class SandboxService:
 
     async def stdin_write(self, target_id: str, offset: int, data: bytes, eof: bool = False) -> int:
         return await self._proc(target_id).write_stdin(offset, data, eof)
+
+    async def resize(self, target_id: str, rows: int, cols: int) -> None:
+        """Propagate a terminal resize to a PTY-backed exec (SIGWINCH)."""
+        self._proc(target_id).resize(rows, cols)
 
     async def wait(self, target_id: str, timeout: Optional[float] = None, raise_on_timeout: bool = True) -> dict:
         state = self._proc(target_id)

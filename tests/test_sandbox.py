@@ -129,3 +129,27 @@ def test_sandbox_snapshot_and_restore(client):
     sb2 = modal.Sandbox.create("bash", "-c", "cat state.txt", image=img)
     sb2.wait(raise_on_termination=False)
     assert sb2.stdout.read().strip() == "snapshot-state"
+
+
+def test_sandbox_pty_exec(client):
+    """PTY-backed exec: the command sees a real controlling terminal,
+    stdout+stderr are merged, and resize propagates TIOCSWINSZ."""
+    sb = modal.Sandbox.create("sleep", "60")
+    try:
+        p = sb.exec(
+            "bash", "-c", "tty; echo err-msg >&2; stty size; read x; echo got:$x",
+            pty_info={"rows": 31, "cols": 97},
+        )
+        p.resize(40, 120)
+        p.stdin.write("ping\n")
+        p.stdin.drain()
+        assert p.wait() == 0
+        out = p.stdout.read()
+        assert "/dev/pts/" in out          # a real PTY, controlling tty works
+        assert "err-msg" in out            # stderr merged into the PTY stream
+        # window size came through TIOCSWINSZ (initial 31x97, or 40x120 if
+        # the resize() landed before the shell ran stty)
+        assert "31 97" in out or "40 120" in out
+        assert "got:ping" in out           # interactive stdin through the master
+    finally:
+        sb.terminate()
