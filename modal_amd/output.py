@@ -35,6 +35,20 @@ class OutputManager:
             sys.stdout.write(message + "\n")
             sys.stdout.flush()
 
+    def print_step(self, message: str) -> None:
+        """A completed step line (parity: OutputManager.step_completed —
+        the reference's green-check progress lines)."""
+        if self._console is not None:
+            self._console.print(f"[green]\N{HEAVY CHECK MARK}[/green] {message}")
+        else:
+            sys.stdout.write(f"+ {message}\n")
+            sys.stdout.flush()
+
+    def make_map_progress(self, description: str = "Running map") -> "MapProgress":
+        """Live completed/submitted progress for Function.map fan-outs
+        (parity: the reference's function-call progress display)."""
+        return MapProgress(self._console, description)
+
     def print_log(self, entry: dict) -> None:
         data = entry.get("data", "")
         stream = sys.stderr if entry.get("fd") == 2 else sys.stdout
@@ -45,6 +59,50 @@ class OutputManager:
             data = "".join(f"{ts} {line}\n" for line in data.splitlines())
         stream.write(data)
         stream.flush()
+
+
+class MapProgress:
+    """Progress bar for a map fan-out: rich on a TTY, silent otherwise.
+    ``update`` is cheap enough to call per output batch."""
+
+    def __init__(self, console: Any, description: str):
+        self._progress = None
+        self._task_id = None
+        self.completed = 0
+        self.submitted = 0
+        if console is not None:
+            try:
+                from rich.progress import (
+                    BarColumn, MofNCompleteColumn, Progress, SpinnerColumn,
+                    TaskProgressColumn, TimeElapsedColumn,
+                )
+
+                self._progress = Progress(
+                    SpinnerColumn(), "[progress.description]{task.description}",
+                    BarColumn(), MofNCompleteColumn(), TaskProgressColumn(),
+                    TimeElapsedColumn(), console=console, transient=True,
+                )
+                self._progress.start()
+                self._task_id = self._progress.add_task(description, total=None)
+            except Exception:
+                self._progress = None
+
+    def update(self, completed: int, submitted: int, done_submitting: bool = False) -> None:
+        self.completed = completed
+        self.submitted = submitted
+        if self._progress is not None:
+            self._progress.update(
+                self._task_id, completed=completed,
+                total=submitted if done_submitting else None,
+            )
+
+    def close(self) -> None:
+        if self._progress is not None:
+            try:
+                self._progress.stop()
+            except Exception:
+                pass
+            self._progress = None
 
 
 def get_output_manager() -> Optional[OutputManager]:

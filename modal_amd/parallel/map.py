@@ -160,6 +160,14 @@ async def map_invocation(
     next_output_idx = 0
     ordering_buffer: dict[int, Any] = {}
 
+    from ..output import get_output_manager
+
+    _mgr = get_output_manager()
+    progress = None
+    if _mgr is not None:
+        label = getattr(fn, "_info_name", None) or getattr(fn, "_method_name", "") or "map"
+        progress = _mgr.make_map_progress(f"Running {label}")
+
     out_chunk_cache: dict[str, list] = {}
 
     async def decode(out: dict) -> Any:
@@ -193,6 +201,8 @@ async def map_invocation(
             )
             if pump_error:
                 raise pump_error[0]
+            if progress is not None:
+                progress.update(received, total_inputs, pump_done.is_set())
             for out in outs:
                 if out.get("group"):
                     # range-protocol group: one pickled value list for ~64 idxs
@@ -227,6 +237,8 @@ async def map_invocation(
                 else:
                     yield value
     finally:
+        if progress is not None:
+            progress.close()
         if not pump_task.done():
             pump_task.cancel()
         try:

@@ -148,6 +148,14 @@ class AppRunContext:
         app._app_id = resp["app_id"]
         app._running_client = client
         await _register_functions(app, client, app._app_id)
+        from .output import get_output_manager
+
+        manager = get_output_manager()
+        if manager is not None:  # parity: OutputManager step lines
+            manager.print_step(f"Initialized app {app.description or app._app_id}.")
+            names = sorted(app._functions.keys()) if getattr(app, "_functions", None) else []
+            if names:
+                manager.print_step(f"Created functions: {', '.join(names)}.")
         self._heartbeat_task = asyncio.get_running_loop().create_task(self._heartbeat_loop())
         self._start_log_stream()
         return app

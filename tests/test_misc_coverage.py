@@ -111,3 +111,30 @@ def test_cloud_bucket_mount_local_dir(tmp_path):
     cbm = modal.CloudBucketMount("my-bucket", key_prefix="data/")
     path = cbm.local_dir(root=str(tmp_path))
     assert "my-bucket" in path and path.endswith("data/")
+
+
+def test_output_steps_and_map_progress(client, capsys):
+    """enable_output prints step lines and MapProgress tracks counters
+    (non-TTY: plain '+' lines, no rich bar)."""
+    import modal_amd as modal
+    from modal_amd.output import MapProgress, get_output_manager
+
+    app = modal.App("out-app")
+
+    @app.function()
+    def double(x):
+        return x * 2
+
+    with modal.enable_output():
+        mgr = get_output_manager()
+        assert mgr is not None
+        with app.run(client=client):
+            assert sorted(double.map(range(10))) == [x * 2 for x in range(10)]
+    captured = capsys.readouterr()
+    assert "Initialized app out-app." in captured.out
+    assert "double" in captured.out  # created-functions step line
+
+    p = MapProgress(None, "test")  # no console: silent counters
+    p.update(3, 10)
+    assert (p.completed, p.submitted) == (3, 10)
+    p.close()
