@@ -422,6 +422,8 @@ def synchronize_api(impl_cls: type, name: Optional[str] = None) -> type:
                 "__aexit__",
                 "__call__",
                 "__getitem__",
+                "__setitem__",
+                "__delitem__",
                 "__len__",
                 "__contains__",
                 "__iter__",

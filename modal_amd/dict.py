@@ -88,6 +88,26 @@ class _Dict(_Object, type_kind="dict"):
     async def contains(self, key: Any) -> bool:
         return await self._client.svc.dict_contains(dict_id=self.object_id, key=serialize(key))
 
+    # dict-style sugar (parity: reference dict.py __getitem__/__setitem__)
+    async def __getitem__(self, key: Any) -> Any:
+        sentinel = object()
+        value = await self.get(key, sentinel)
+        if value is sentinel:
+            raise KeyError(key)
+        return value
+
+    async def __setitem__(self, key: Any, value: Any) -> None:
+        await self.put(key, value)
+
+    async def __delitem__(self, key: Any) -> None:
+        await self.pop(key)
+
+    async def __contains__(self, key: Any) -> bool:
+        return await self.contains(key)
+
+    async def __len__(self) -> int:
+        return await self.len()
+
     @live_method
     async def len(self) -> int:
         return await self._client.svc.dict_len(dict_id=self.object_id)

@@ -129,6 +129,16 @@ class Scheduler:
         self._start_lock = asyncio.Lock()
         self.default_environment = "main"
         self._extra: dict[str, Any] = {}  # extension services (volumes, sandboxes, images)
+        self._persist_task: Optional[asyncio.Task] = None
+        self._persist_digest = b""
+        from . import persist
+
+        try:  # same run_dir => same deployments (durable control plane)
+            persist.load(self)
+        except Exception:
+            logging.getLogger("modal_amd.scheduler").warning(
+                "state restore failed", exc_info=True
+            )
 
     # -- lifecycle -------------------------------------------------------
     async def start(self) -> None:
@@ -140,6 +150,7 @@ class Scheduler:
 
             self.schedule_runner = ScheduleRunner(self)
             self.schedule_runner.start()
+            self._persist_task = asyncio.get_running_loop().create_task(self._persist_loop())
             self._started = True
 
     async def stop(self) -> None:
@@ -147,10 +158,35 @@ class Scheduler:
             return
         if getattr(self, "schedule_runner", None) is not None:
             self.schedule_runner.stop()
+        if self._persist_task is not None:
+            self._persist_task.cancel()
+            self._persist_task = None
+        from . import persist
+
+        try:  # final snapshot so a clean stop never loses deployments
+            self._persist_digest = persist.save_if_changed(self, self._persist_digest)
+        except Exception:
+            pass
         await self.sandbox_service.shutdown()
         await self.web_gateway.stop()
         await self.pool.stop()
         self._started = False
+
+    async def _persist_loop(self) -> None:
+        """Snapshot named/deployed state every 2 s when it changed
+        (scheduler/persist.py; writes are atomic replaces)."""
+        from . import persist
+
+        while True:
+            await asyncio.sleep(2.0)
+            try:
+                # on the loop: snapshot iterates live dicts that only the
+                # loop mutates (state is small; the write is one file)
+                self._persist_digest = persist.save_if_changed(self, self._persist_digest)
+            except Exception:
+                logging.getLogger("modal_amd.scheduler").warning(
+                    "state snapshot failed", exc_info=True
+                )
 
     def log(self, message: str) -> None:
         import logging
