@@ -370,6 +370,14 @@ def container_stats() -> None:
         )
 
 
+@entrypoint_cli.command()
+def metrics() -> None:
+    """Node metrics in Prometheus exposition format (also served at the
+    web gateway's /_metrics)."""
+    client = _get_client()
+    click.echo(synchronizer.run(client.svc.node_metrics()), nl=False)
+
+
 @container_cli.command(name="list")
 def container_list() -> None:
     client = _get_client()

@@ -72,7 +72,7 @@ class RPCAdapter:
         "function_lookup", "function_map", "function_put_inputs", "function_put_chunk",
         "function_finish_inputs",
         "function_get_outputs", "function_call_cancel", "function_call_info",
-        "function_get_current_stats", "generator_poll", "node_stats",
+        "function_get_current_stats", "generator_poll", "node_stats", "node_metrics",
         "app_lookup", "app_get_layout", "cluster_hello",
         "volume_get_or_create", "volume_put_file_blocks", "volume_get_file", "volume_list_files",
         "volume_remove_file", "volume_copy_files", "volume_commit", "volume_reload",
@@ -131,6 +131,13 @@ class Scheduler:
         self._extra: dict[str, Any] = {}  # extension services (volumes, sandboxes, images)
         self._persist_task: Optional[asyncio.Task] = None
         self._persist_digest = b""
+        # observability counters (SURVEY §5.5 MI355X line: items/sec +
+        # p50/p99 in the scheduler itself; exposed via node_metrics)
+        from collections import deque as _deque
+
+        self.metrics_counters = {"inputs_total": 0, "outputs_total": 0, "failures_total": 0}
+        self._unary_latencies: Any = _deque(maxlen=4096)
+        self._rate_window: Any = _deque(maxlen=64)
         from . import persist
 
         try:  # same run_dir => same deployments (durable control plane)
@@ -411,6 +418,57 @@ class Scheduler:
             ),
         }
 
+    async def node_metrics(self) -> str:
+        """Prometheus-exposition snapshot of the gauges/counters SURVEY §5.5
+        obligates: per-GPU HBM, worker liveness, items/sec, p50/p99 latency."""
+        now = time.time()
+        counters = self.metrics_counters
+        self._rate_window.append((now, counters["outputs_total"]))
+        lines = [
+            "# TYPE modal_amd_inputs_total counter",
+            f"modal_amd_inputs_total {counters['inputs_total']}",
+            "# TYPE modal_amd_outputs_total counter",
+            f"modal_amd_outputs_total {counters['outputs_total']}",
+            "# TYPE modal_amd_failures_total counter",
+            f"modal_amd_failures_total {counters['failures_total']}",
+        ]
+        if len(self._rate_window) >= 2:
+            (t0, c0), (t1, c1) = self._rate_window[0], self._rate_window[-1]
+            if t1 > t0:
+                lines += [
+                    "# TYPE modal_amd_items_per_sec gauge",
+                    f"modal_amd_items_per_sec {(c1 - c0) / (t1 - t0):.3f}",
+                ]
+        if self._unary_latencies:
+            lat = sorted(self._unary_latencies)
+            lines += [
+                "# TYPE modal_amd_unary_latency_seconds summary",
+                f'modal_amd_unary_latency_seconds{{quantile="0.5"}} {lat[len(lat) // 2]:.6f}',
+                f'modal_amd_unary_latency_seconds{{quantile="0.99"}} {lat[min(len(lat) - 1, int(len(lat) * 0.99))]:.6f}',
+            ]
+        lines += [
+            "# TYPE modal_amd_workers gauge",
+            f"modal_amd_workers {len(self.pool.workers)}",
+            "# TYPE modal_amd_pending_inputs gauge",
+            f"modal_amd_pending_inputs {sum(len(q) for q in self.pool.pending.values())}",
+            "# TYPE modal_amd_active_calls gauge",
+            f"modal_amd_active_calls {sum(1 for c in self.calls.values() if not c.done_event.is_set())}",
+        ]
+        gpu_lines = []
+        for w in self.pool.workers.values():
+            if w.gpu_stats and w.gpu_index is not None:
+                g = w.gpu_stats
+                gpu_lines.append(
+                    f'modal_amd_gpu_hbm_free_bytes{{gpu="{w.gpu_index}",worker="{w.worker_id}"}} {g.get("hbm_free", 0)}'
+                )
+                gpu_lines.append(
+                    f'modal_amd_gpu_hbm_total_bytes{{gpu="{w.gpu_index}",worker="{w.worker_id}"}} {g.get("hbm_total", 0)}'
+                )
+        if gpu_lines:
+            lines.append("# TYPE modal_amd_gpu_hbm_free_bytes gauge")
+            lines.extend(gpu_lines)
+        return "\n".join(lines) + "\n"
+
     async def function_get_current_stats(self, function_id: str) -> dict:
         backlog = len(self.pool.pending.get(function_id, ()))
         runners = sum(
@@ -467,6 +525,7 @@ class Scheduler:
         -> [{"idx", "input_id"}]. ``chunks`` carries shared pickled arg-lists
         (the map fast path: one pickle per ~64 inputs)."""
         record = self._call(function_call_id)
+        self.metrics_counters["inputs_total"] += len(items)
         fdef = self.functions.get(record.function_id)
         if chunks:
             for chunk_id, data in chunks.items():
@@ -561,6 +620,7 @@ class Scheduler:
                 )
             }
         group = record.add_chunk(chunk_id, payload, count, method)
+        self.metrics_counters["inputs_total"] += count
         self.pool.enqueue_chunk(record, group, record.function_id)
         return {"idx_base": group.base_idx, "count": count}
 
@@ -772,6 +832,11 @@ class Scheduler:
         record = self.calls.get(rec.call_id)
         if record is None:
             return
+        self.metrics_counters["outputs_total"] += 1
+        if status != GENERIC_STATUS_SUCCESS:
+            self.metrics_counters["failures_total"] += 1
+        if record.kind == "unary":
+            self._unary_latencies.append(time.time() - record.created_at)
         record.post_output(
             rec.idx, status, output, output_format, exc_repr, retry_count, output_blob,
             out_chunk, out_ci,

@@ -65,6 +65,10 @@ class WebGateway:
 
         label = request.match_info["label"]
         tail = request.match_info.get("tail") or "/"
+        if label == "_metrics":  # Prometheus scrape target
+            return web.Response(
+                text=await self.scheduler.node_metrics(), content_type="text/plain"
+            )
         function_id = self.routes.get(label)
         if function_id is None:
             return web.Response(status=404, text=f"No web function '{label}'")

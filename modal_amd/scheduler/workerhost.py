@@ -789,6 +789,9 @@ class WorkerPool:
         if data is not None:
             slot = self.scheduler.register_out_chunk(data, count if cis is None else len(cis))
             record.complete_chunk_success(group, cis, slot)
+            self.scheduler.metrics_counters["outputs_total"] += (
+                count if cis is None else len(cis)
+            )
         else:
             group.state = "done"
             record._check_done()

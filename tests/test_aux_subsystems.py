@@ -212,3 +212,43 @@ def test_scheduler_placement_steers_dispatch(run_dir, monkeypatch):
     finally:
         synchronizer.run(c.close())
         _Client._singleton = None
+
+
+def test_node_metrics_exposition(client):
+    """scheduler-level items/sec + p50/p99 + worker gauges in
+    Prometheus text format, over RPC and over the web gateway."""
+    import urllib.request
+
+    import modal_amd as modal
+    from modal_amd._sync import synchronizer
+
+    app = modal.App("metrics-app")
+
+    @app.function()
+    def ident(x):
+        return x
+
+    @app.function()
+    @modal.fastapi_endpoint()
+    def tiny():
+        return {"ok": True}
+
+    with app.run(client=client):
+        assert ident.remote(5) == 5
+        assert sorted(ident.map(range(40))) == list(range(40))
+        text = synchronizer.run(client.svc.node_metrics())
+        assert "modal_amd_inputs_total" in text
+        counters = {
+            line.split()[0]: float(line.split()[1])
+            for line in text.splitlines()
+            if line and not line.startswith("#") and "{" not in line
+        }
+        assert counters["modal_amd_inputs_total"] >= 41
+        assert counters["modal_amd_outputs_total"] >= 41
+        assert counters["modal_amd_failures_total"] == 0
+        assert "modal_amd_unary_latency_seconds" in text  # p50/p99 present
+        # gateway scrape target
+        base = tiny.web_url.rsplit("/", 1)[0]
+        with urllib.request.urlopen(base + "/_metrics", timeout=10) as resp:
+            scraped = resp.read().decode()
+        assert "modal_amd_workers" in scraped
