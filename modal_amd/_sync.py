@@ -42,6 +42,30 @@ class Synchronizer:
         self._thread: Optional[threading.Thread] = None
         self._pid = os.getpid()
         self._stopped = False
+        #: worker processes set this: blocking calls then run on a
+        #: per-thread event loop IN the calling thread instead of hopping
+        #: to the shared loop — a cross-thread wakeup costs ~1-2 ms on
+        #: contended hosts, dominating per-item handle RPCs (Queue.get).
+        #: Requires loop-affine transports (client.py UserCodeProxy keys
+        #: its connections by running loop).
+        self.thread_local_blocking = False
+        self._tls = threading.local()
+
+    def _thread_loop(self) -> asyncio.AbstractEventLoop:
+        loop = getattr(self._tls, "loop", None)
+        if loop is None or loop.is_closed():
+            loop = asyncio.new_event_loop()
+            self._tls.loop = loop
+        return loop
+
+    def _use_thread_local(self) -> bool:
+        if not self.thread_local_blocking or self.in_loop_thread():
+            return False
+        try:
+            asyncio.get_running_loop()
+            return False  # caller is inside some loop: keep bridge semantics
+        except RuntimeError:
+            return True
 
     # -- loop lifecycle -------------------------------------------------
     def _ensure_loop(self) -> asyncio.AbstractEventLoop:
@@ -100,6 +124,8 @@ class Synchronizer:
             raise RuntimeError(
                 "Blocking API called from within the framework event loop; use the .aio variant"
             )
+        if self._use_thread_local():
+            return self._thread_loop().run_until_complete(coro)
         fut = asyncio.run_coroutine_threadsafe(coro, self._ensure_loop())
         try:
             return fut.result()
@@ -125,6 +151,14 @@ class Synchronizer:
         (a per-item run_coroutine_threadsafe round-trip costs ~150 us and
         would dominate map() throughput).
         """
+        if self._use_thread_local():
+            # worker processes: drive the agen in THIS thread, no hops
+            loop = self._thread_loop()
+            while True:
+                try:
+                    yield loop.run_until_complete(agen.__anext__())
+                except StopAsyncIteration:
+                    return
         import queue as _queue
 
         loop = self._ensure_loop()

