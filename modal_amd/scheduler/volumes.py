@@ -107,9 +107,29 @@ class VolumeService:
             # stored raw, decompressed write when LZ4-compressed)
             self.blob_store.materialize(block_digests[0], dest)
         else:
-            with open(dest, "wb") as f:
-                for digest in block_digests:
-                    f.write(self.blob_store.get(digest))
+            # multi-block concat OFF the event loop, zero-copy for raw
+            # blocks (this was 85% of upload wall time when done with
+            # read()+write() on the loop)
+            import asyncio
+
+            def _concat() -> None:
+                with open(dest, "wb") as f:
+                    for digest in block_digests:
+                        src = self.blob_store.open_path(digest)
+                        with open(src, "rb") as bf:
+                            if bf.read(6) == b"MALZ41":
+                                f.write(self.blob_store.get(digest))  # GPU decompress
+                                continue
+                            bf.seek(0)
+                            n = os.fstat(bf.fileno()).st_size
+                            off = 0
+                            while off < n:
+                                sent = os.sendfile(f.fileno(), bf.fileno(), off, n - off)
+                                if sent == 0:
+                                    break
+                                off += sent
+
+            await asyncio.get_running_loop().run_in_executor(None, _concat)
         vol.manifest[rel_path.lstrip("/")] = {
             "size": size,
             "blocks": block_digests,
@@ -167,9 +187,14 @@ class VolumeService:
         path = self._safe_path(vol, rel_path)
         if not os.path.isfile(path):
             raise NotFoundError(f"{rel_path} not in volume")
-        with open(path, "rb") as f:
-            f.seek(offset)
-            return f.read(n_bytes if n_bytes >= 0 else -1)
+        import asyncio
+
+        def _read() -> bytes:
+            with open(path, "rb") as f:
+                f.seek(offset)
+                return f.read(n_bytes if n_bytes >= 0 else -1)
+
+        return await asyncio.get_running_loop().run_in_executor(None, _read)
 
     async def list_files(self, volume_id: str, rel_path: str = "/", recursive: bool = True) -> list[dict]:
         vol = self._get(volume_id)

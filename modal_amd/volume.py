@@ -160,11 +160,16 @@ class _Volume(_Object, type_kind="volume"):
     async def _put_data(self, data: bytes, remote_path: str) -> None:
         """Block-wise upload through the CAS: hash each 8 MiB block (HIP
         kernel above the crossover), store, then commit the manifest."""
+        import asyncio
+
         store = self._client.blob_store
         blocks = [
             bytes(data[off : off + BLOCK_SIZE]) for off in range(0, max(len(data), 1), BLOCK_SIZE)
         ]
-        digests = store.put_many(blocks)  # one batched GPU hash dispatch
+        # hash (one batched GPU dispatch) + CAS writes off the event loop
+        digests = await asyncio.get_running_loop().run_in_executor(
+            None, store.put_many, blocks
+        )
         resp = await self._client.svc.volume_put_file_blocks(
             volume_id=self.object_id,
             rel_path=remote_path,
