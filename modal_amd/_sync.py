@@ -295,8 +295,13 @@ def unwrap(obj: Any) -> Any:
     return obj
 
 
+_WRAP_PASSTHROUGH = frozenset({int, float, str, bytes, bool, type(None)})
+
+
 def wrap(obj: Any) -> Any:
     """Translate impl objects to their public wrappers (shallow containers)."""
+    if type(obj) in _WRAP_PASSTHROUGH:  # the common case on hot paths
+        return obj
     wrapper_cls = _WRAPPER_BY_IMPL.get(type(obj))
     if wrapper_cls is not None:
         return wrapper_cls._from_impl(obj)

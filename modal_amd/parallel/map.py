@@ -78,6 +78,8 @@ async def map_invocation(
     return_exceptions: bool,
     wrap_returned_exceptions: bool,
 ) -> AsyncGenerator[Any, None]:
+    if not fn._is_hydrated:
+        await fn.hydrate()
     client = fn._client
     svc = client.svc
     resp = await svc.function_map(function_id=fn.object_id, kind="map")
@@ -132,18 +134,43 @@ async def map_invocation(
             )
 
         approx_bytes = 0
+
+        def _add(item: tuple) -> None:
+            nonlocal approx_bytes, total_inputs
+            args, extra_kwargs = item
+            kw = {**kwargs_common, **extra_kwargs} if extra_kwargs else kwargs_common
+            chunk_buf.append((args, kw))
+            for a in args:
+                if type(a) in (bytes, bytearray, str):
+                    approx_bytes += len(a)
+            total_inputs += 1
+
         try:
-            async for args, extra_kwargs in _iterate_maybe_async(input_iter):
-                kw = {**kwargs_common, **extra_kwargs} if extra_kwargs else kwargs_common
-                chunk_buf.append((args, kw))
-                for a in args:
-                    if type(a) in (bytes, bytearray, str):
-                        approx_bytes += len(a)
-                total_inputs += 1
-                if len(chunk_buf) >= CHUNK_ITEMS or approx_bytes > 4 * 1024 * 1024:
-                    approx_bytes = 0
-                    await sem.acquire(len(chunk_buf))
-                    await flush_chunk()
+            if hasattr(input_iter, "__aiter__"):
+                async for item in input_iter:
+                    _add(item)
+                    if len(chunk_buf) >= CHUNK_ITEMS or approx_bytes > 4 * 1024 * 1024:
+                        approx_bytes = 0
+                        await sem.acquire(len(chunk_buf))
+                        await flush_chunk()
+            else:
+                # sync iterators (the common case): pull whole chunks via
+                # islice — no per-item async generator frames
+                from itertools import islice
+
+                iterator = iter(input_iter)
+                while True:
+                    batch = list(islice(iterator, CHUNK_ITEMS - len(chunk_buf) or CHUNK_ITEMS))
+                    if not batch and not chunk_buf:
+                        break
+                    for item in batch:
+                        _add(item)
+                    if len(chunk_buf) >= CHUNK_ITEMS or approx_bytes > 4 * 1024 * 1024:
+                        approx_bytes = 0
+                        await sem.acquire(len(chunk_buf))
+                        await flush_chunk()
+                    elif not batch:
+                        break
             if chunk_buf:
                 await sem.acquire(len(chunk_buf))
             await flush_chunk()
