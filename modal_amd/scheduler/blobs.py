@@ -44,8 +44,7 @@ class BlobStore:
     COMPRESS_MIN = 1024 * 1024  # GPU (de)compression pays above ~1 MiB
 
     def _store(self, digest: str, data: Union[bytes, memoryview]) -> None:
-        path = self._path(digest)
-        if os.path.exists(path):
+        if os.path.exists(self._path(digest)):
             return
         payload = bytes(data)
         if len(payload) >= self.COMPRESS_MIN:
@@ -58,18 +57,7 @@ class BlobStore:
                     payload = compressed
             except Exception:
                 pass
-        os.makedirs(os.path.dirname(path), exist_ok=True)
-        fd, tmp = tempfile.mkstemp(dir=os.path.dirname(path))
-        try:
-            with os.fdopen(fd, "wb") as f:
-                f.write(payload)
-            os.replace(tmp, path)  # atomic publish; concurrent writers converge
-        except BaseException:
-            try:
-                os.unlink(tmp)
-            except OSError:
-                pass
-            raise
+        self._store_prepared(digest, payload)
 
     def put(self, data: Union[bytes, memoryview]) -> str:
         digest = _hash_bytes(data)
@@ -77,14 +65,49 @@ class BlobStore:
         return digest
 
     def put_many(self, buffers: list) -> list[str]:
-        """Batched put: one GPU hash dispatch over all buffers' leaves
-        (the volume-upload hot path; see ops/hashing.content_digests_batch)."""
+        """Batched put: one GPU hash dispatch over all buffers' leaves AND
+        one batched compression pass over the new large blocks (the
+        volume-upload hot path; ops/hashing.content_digests_batch +
+        ops/compress.compress_buffers)."""
         from ..ops.hashing import content_digests_batch
 
         digests = content_digests_batch(buffers)
-        for digest, data in zip(digests, buffers):
-            self._store(digest, data)
+        # compress only blocks that are big enough and not already stored
+        todo = [
+            i for i, (digest, data) in enumerate(zip(digests, buffers))
+            if len(data) >= self.COMPRESS_MIN and not os.path.exists(self._path(digest))
+        ]
+        compressed: dict[int, bytes] = {}
+        if todo:
+            try:
+                from ..ops.compress import compress_buffers
+
+                for i, payload in zip(todo, compress_buffers([buffers[i] for i in todo])):
+                    if payload is not None:
+                        compressed[i] = payload
+            except Exception:
+                pass  # store raw on any kernel/driver hiccup
+        for i, (digest, data) in enumerate(zip(digests, buffers)):
+            self._store_prepared(digest, compressed.get(i, bytes(data)))
         return digests
+
+    def _store_prepared(self, digest: str, payload: bytes) -> None:
+        """Store an already-encoded payload (raw or MALZ41) under digest."""
+        path = self._path(digest)
+        if os.path.exists(path):
+            return
+        os.makedirs(os.path.dirname(path), exist_ok=True)
+        fd, tmp = tempfile.mkstemp(dir=os.path.dirname(path))
+        try:
+            with os.fdopen(fd, "wb") as f:
+                f.write(payload)
+            os.replace(tmp, path)
+        except BaseException:
+            try:
+                os.unlink(tmp)
+            except OSError:
+                pass
+            raise
 
     def put_file(self, src_path: str) -> str:
         h = hashlib.sha256()
