@@ -1042,8 +1042,12 @@ class Scheduler:
     async def volume_get_or_create(self, name=None, environment="main", create_if_missing=False, ephemeral=False) -> dict:
         return await self.volume_service.get_or_create(name, environment, create_if_missing, ephemeral)
 
-    async def volume_put_file_blocks(self, volume_id, rel_path, block_digests, size, mode=0o644) -> dict:
-        return await self.volume_service.put_file_blocks(volume_id, rel_path, block_digests, size, mode)
+    async def volume_put_file_blocks(
+        self, volume_id, rel_path, block_digests, size, mode=0o644, content_tmp=None
+    ) -> dict:
+        return await self.volume_service.put_file_blocks(
+            volume_id, rel_path, block_digests, size, mode, content_tmp=content_tmp
+        )
 
     async def volume_get_file(self, volume_id, rel_path, offset=0, n_bytes=-1) -> bytes:
         return await self.volume_service.get_file(volume_id, rel_path, offset, n_bytes)

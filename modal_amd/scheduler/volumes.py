@@ -91,18 +91,41 @@ class VolumeService:
 
     # -- writes ----------------------------------------------------------
     async def put_file_blocks(
-        self, volume_id: str, rel_path: str, block_digests: list[str], size: int, mode: int = 0o644
+        self,
+        volume_id: str,
+        rel_path: str,
+        block_digests: list[str],
+        size: int,
+        mode: int = 0o644,
+        content_tmp: Optional[str] = None,
     ) -> dict:
         """Phase-2 commit of a file whose blocks are already in the CAS
         (parity: VolumePutFiles2 missing-block protocol, volume.py:1401-1445).
-        Returns any blocks NOT yet in the store (client must upload & retry)."""
+        Returns any blocks NOT yet in the store (client must upload & retry).
+        ``content_tmp``: same-node fast path — a staged raw copy the client
+        already wrote under the run dir; renamed into place instead of
+        re-assembling from CAS blocks."""
         vol = self._get(volume_id)
         missing = [d for d in block_digests if not self.blob_store.has(d)]
         if missing:
+            if content_tmp:
+                try:
+                    os.unlink(content_tmp)
+                except OSError:
+                    pass
             return {"missing_blocks": missing}
         dest = self._safe_path(vol, rel_path)
         os.makedirs(os.path.dirname(dest), exist_ok=True)
-        if len(block_digests) == 1:
+        staged = False
+        if content_tmp and os.path.isfile(content_tmp) and os.path.getsize(content_tmp) == size:
+            try:
+                os.replace(content_tmp, dest)
+                staged = True
+            except OSError:
+                pass  # cross-device etc.: fall through to block assembly
+        if staged:
+            pass
+        elif len(block_digests) == 1:
             # single-block files come straight out of the CAS (hard link when
             # stored raw, decompressed write when LZ4-compressed)
             self.blob_store.materialize(block_digests[0], dest)

@@ -166,15 +166,33 @@ class _Volume(_Object, type_kind="volume"):
         blocks = [
             bytes(data[off : off + BLOCK_SIZE]) for off in range(0, max(len(data), 1), BLOCK_SIZE)
         ]
+        loop = asyncio.get_running_loop()
         # hash (one batched GPU dispatch) + CAS writes off the event loop
-        digests = await asyncio.get_running_loop().run_in_executor(
-            None, store.put_many, blocks
+        digests = await loop.run_in_executor(None, store.put_many, blocks)
+        # same-node fast path: stage the raw file next to the volume tree so
+        # the service renames it into place instead of re-reading every CAS
+        # block (and GPU-decompressing the compressed ones)
+        content_tmp = None
+        run_dir = getattr(self._client, "run_dir", None) or getattr(
+            self._client.svc, "run_dir", None
         )
+        if run_dir and os.path.isdir(run_dir):
+
+            def _stage() -> str:
+                import tempfile as _tf
+
+                fd, tmp = _tf.mkstemp(dir=run_dir, prefix=".volstage-")
+                with os.fdopen(fd, "wb") as f:
+                    f.write(data)
+                return tmp
+
+            content_tmp = await loop.run_in_executor(None, _stage)
         resp = await self._client.svc.volume_put_file_blocks(
             volume_id=self.object_id,
             rel_path=remote_path,
             block_digests=digests,
             size=len(data),
+            content_tmp=content_tmp,
         )
         if resp.get("missing_blocks"):
             raise InvalidError(f"blocks missing after upload: {resp['missing_blocks']}")
