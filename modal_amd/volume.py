@@ -167,12 +167,11 @@ class _Volume(_Object, type_kind="volume"):
             bytes(data[off : off + BLOCK_SIZE]) for off in range(0, max(len(data), 1), BLOCK_SIZE)
         ]
         loop = asyncio.get_running_loop()
-        # hash (one batched GPU dispatch) + CAS writes off the event loop
-        digests = await loop.run_in_executor(None, store.put_many, blocks)
         # same-node fast path: stage the raw file next to the volume tree so
         # the service renames it into place instead of re-reading every CAS
-        # block (and GPU-decompressing the compressed ones)
-        content_tmp = None
+        # block (and GPU-decompressing the compressed ones); the disk write
+        # overlaps the GPU hash+compress pass below
+        stage_task = None
         run_dir = getattr(self._client, "run_dir", None) or getattr(
             self._client.svc, "run_dir", None
         )
@@ -186,7 +185,10 @@ class _Volume(_Object, type_kind="volume"):
                     f.write(data)
                 return tmp
 
-            content_tmp = await loop.run_in_executor(None, _stage)
+            stage_task = loop.run_in_executor(None, _stage)
+        # hash (one batched GPU dispatch) + batched compression + CAS writes
+        digests = await loop.run_in_executor(None, store.put_many, blocks)
+        content_tmp = await stage_task if stage_task is not None else None
         resp = await self._client.svc.volume_put_file_blocks(
             volume_id=self.object_id,
             rel_path=remote_path,
