@@ -52,7 +52,7 @@ def method(_warn_parentheses_missing: Any = None, *, is_generator: Optional[bool
 
 
 def _lifecycle(kind: str) -> Callable:
-    def deco(_warn_parentheses_missing: Any = None) -> Callable:
+    def deco(_warn_parentheses_missing: Any = None, *, snap: bool = False) -> Callable:
         if _warn_parentheses_missing is not None and callable(_warn_parentheses_missing):
             # bare usage @modal.enter without parens
             f = _warn_parentheses_missing
@@ -60,7 +60,10 @@ def _lifecycle(kind: str) -> Callable:
             return f
 
         def wrapper(f: Callable) -> Callable:
-            f._modal_amd_lifecycle = kind
+            # snap=True: runs BEFORE the memory snapshot is taken (parity:
+            # reference _partial_function.py:589 enter(snap=...)); plain
+            # enter hooks run after restore
+            f._modal_amd_lifecycle = f"{kind}_snap" if snap and kind == "enter" else kind
             return f
 
         return wrapper

@@ -124,6 +124,27 @@ class WorkerRPCTarget:
         await self._runtime._app_teardown(app_id)
         return True
 
+    async def gpu_snapshot(self) -> dict:
+        """Page this worker's GPU tensors to host memory (the scaledown
+        alternative for enable_memory_snapshot functions: HBM is freed,
+        the warm process and its loaded user code survive)."""
+        rt = self._runtime
+        if rt.mem_snapshot is not None:
+            return {"state": rt.mem_snapshot.state.value}
+        from .gpu_snapshot import GPUMemorySnapshot
+
+        snap = GPUMemorySnapshot()
+        await asyncio.get_running_loop().run_in_executor(None, snap.checkpoint)
+        rt.mem_snapshot = snap
+        return {"state": snap.state.value}
+
+    async def gpu_restore(self) -> dict:
+        rt = self._runtime
+        snap, rt.mem_snapshot = rt.mem_snapshot, None
+        if snap is not None:
+            await asyncio.get_running_loop().run_in_executor(None, snap.restore)
+        return {"state": "running"}
+
 
 class FunctionRuntime:
     """Worker-side state for one registered function."""
@@ -221,6 +242,11 @@ class _ClsService:
                 self.instance = inst
         if not self._entered:
             self._entered = True
+            # snap=True hooks run first — their state is what a memory
+            # snapshot captures; plain enter hooks model post-restore work
+            # (parity: enter pre/post-snapshot, _partial_function.py:589)
+            for hook_name in _lifecycle_hooks(self.cls, "enter_snap"):
+                getattr(self.instance, hook_name)()
             for hook_name in _lifecycle_hooks(self.cls, "enter"):
                 getattr(self.instance, hook_name)()
         return self.instance
@@ -287,6 +313,7 @@ class WorkerRuntime:
         self.conn: Optional[Connection] = None
         self.task_id: str = ""
         self.functions: dict[str, FunctionRuntime] = {}
+        self.mem_snapshot: Any = None  # GPUMemorySnapshot while paged out
         self.executor = ThreadPoolExecutor(
             max_workers=int(os.environ.get("MODAL_AMD_WORKER_THREADS", "16")),
             thread_name_prefix="modal-amd-input",

@@ -359,7 +359,14 @@ class Scheduler:
             max_containers=int(options.get("max_containers", 0)),
             buffer_containers=int(options.get("buffer_containers", 0)),
             scaledown_window=float(options.get("scaledown_window", 60.0)),
-            metadata=options.get("metadata") or {},
+            metadata={
+                **(options.get("metadata") or {}),
+                **(
+                    {"enable_memory_snapshot": True}
+                    if options.get("enable_memory_snapshot")
+                    else {}
+                ),
+            },
             web_config=options.get("web_config"),
             secret_ids=list(options.get("secret_ids") or []),
             volume_mounts=dict(options.get("volume_mounts") or {}),

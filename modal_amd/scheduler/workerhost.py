@@ -57,6 +57,7 @@ class WorkerHandle:
         self.last_heartbeat = time.time()
         self.last_active = time.time()  # last input assignment (scaledown clock)
         self.gpu_stats: Optional[dict] = None  # HBM gauges from heartbeats
+        self.paged = False  # GPU memory snapshotted to host (scaledown)
         # tokens of inputs currently assigned here, mapped to their records
         self.inflight: dict[str, InputRecord] = {}
         # per-function outstanding count (for credit computation)
@@ -133,9 +134,38 @@ class WorkerPool:
         alive = sum(1 for w in self.workers.values() if w.alive)
         excess = alive - max(floor, 1)
         for w in idle[: max(excess, 0)]:
+            if not w.paged and self._hosts_snapshot_fn(w):
+                # enable_memory_snapshot: page GPU memory to host instead of
+                # reaping — HBM frees, the warm import/enter state survives
+                # (parity: the reference's memory-snapshot cold-start
+                # elimination, gpu_memory_snapshot.py:230-300)
+                try:
+                    await w.conn.call("gpu_snapshot", timeout=120)
+                    w.paged = True
+                except Exception:
+                    pass
+                continue
+            if w.paged:
+                continue  # already costs ~no HBM; keep it warm
             w.draining = True
             try:
                 await w.conn.send({"t": "shutdown"})
+            except Exception:
+                pass
+
+    def _hosts_snapshot_fn(self, w: WorkerHandle) -> bool:
+        for fid in w.functions_loaded:
+            fdef = self.scheduler.functions.get(fid)
+            if fdef is not None and fdef.metadata.get("enable_memory_snapshot"):
+                return True
+        return False
+
+    async def _ensure_unpaged(self, w: WorkerHandle) -> None:
+        """Restore a paged-out worker's GPU state before giving it work."""
+        if w.paged:
+            w.paged = False
+            try:
+                await w.conn.call("gpu_restore", timeout=300)
             except Exception:
                 pass
 
@@ -642,6 +672,7 @@ class WorkerPool:
 
     async def _send_batch(self, w: WorkerHandle, fdef: FunctionDef, batch: list[InputRecord]) -> None:
         try:
+            await self._ensure_unpaged(w)
             if (
                 fdef.function_id not in w.functions_loaded
                 or w.defs_version.get(fdef.function_id, 0) != fdef.definition_version
@@ -694,6 +725,7 @@ class WorkerPool:
     async def _send_group(self, w: WorkerHandle, record: Any, group: Any, fdef: FunctionDef) -> None:
         """One frame carries a whole chunk (range protocol)."""
         try:
+            await self._ensure_unpaged(w)
             if (
                 fdef.function_id not in w.functions_loaded
                 or w.defs_version.get(fdef.function_id, 0) != fdef.definition_version
