@@ -131,14 +131,14 @@ class WorkerPool:
             and not w.external  # torchrun-owned workers are not ours to reap
             and now - w.last_active > window
         ]
-        alive = sum(1 for w in self.workers.values() if w.alive)
-        excess = alive - max(floor, 1)
-        for w in idle[: max(excess, 0)]:
+        reapable = []
+        for w in idle:
             if not w.paged and self._hosts_snapshot_fn(w):
                 # enable_memory_snapshot: page GPU memory to host instead of
                 # reaping — HBM frees, the warm import/enter state survives
                 # (parity: the reference's memory-snapshot cold-start
-                # elimination, gpu_memory_snapshot.py:230-300)
+                # elimination, gpu_memory_snapshot.py:230-300). Reversible,
+                # so not subject to the warm floor.
                 try:
                     await w.conn.call("gpu_snapshot", timeout=120)
                     w.paged = True
@@ -147,6 +147,10 @@ class WorkerPool:
                 continue
             if w.paged:
                 continue  # already costs ~no HBM; keep it warm
+            reapable.append(w)
+        alive = sum(1 for w in self.workers.values() if w.alive)
+        excess = alive - max(floor, 1)
+        for w in reapable[: max(excess, 0)]:
             w.draining = True
             try:
                 await w.conn.send({"t": "shutdown"})
