@@ -804,7 +804,18 @@ class Scheduler:
         if rec is None or rec.final or rec.retry_count != retry_count:
             return  # stale attempt (dedup parity: parallel_map.py:1416-1431)
         fdef = self.functions.get(record.function_id)
-        if status == GENERIC_STATUS_FAILURE and fdef is not None:
+        if rec.cluster is not None and status in (
+            GENERIC_STATUS_FAILURE,
+            GENERIC_STATUS_INTERNAL_FAILURE,
+        ):
+            # gang fate-sharing: a solo retry of one rank can never
+            # rendezvous with peers that already ran, so a failed gang
+            # member fails its whole gang (parity: clustered containers
+            # fate-share in the reference's scheduler)
+            asyncio.get_running_loop().create_task(
+                self._fail_gang_siblings(record, rec, exc_repr)
+            )
+        elif status == GENERIC_STATUS_FAILURE and fdef is not None:
             policy = fdef.retry_policy
             if rec.retry_count < policy.max_retries and not rec.cancelled:
                 rec.retry_count += 1
@@ -823,6 +834,36 @@ class Scheduler:
             rec, status, output, output_format, exc_repr, retry_count, output_blob,
             out_chunk, out_ci,
         )
+
+    async def _fail_gang_siblings(self, record: CallRecord, failed: InputRecord, exc_repr: Optional[str]) -> None:
+        """Terminate the other ranks of a failed gang member's cluster:
+        cancel their in-flight executions and finalize them so the caller
+        unblocks (stale ranks would otherwise hang in collectives)."""
+        cluster_id = failed.cluster["cluster_id"]
+        siblings = [
+            r
+            for r in record.inputs.values()
+            if r.cluster is not None
+            and r.cluster["cluster_id"] == cluster_id
+            and r.idx != failed.idx
+            and not r.final
+        ]
+        tokens = []
+        for r in siblings:
+            r.cancelled = True
+            tokens.append(r.token)
+        if tokens:
+            await self.pool.cancel_inputs(tokens, terminate=True)
+        for r in siblings:
+            if not r.final:
+                self.finalize_input(
+                    r,
+                    GENERIC_STATUS_TERMINATED,
+                    None,
+                    0,
+                    f"gang member rank {failed.cluster['rank']} failed: {exc_repr}",
+                    r.retry_count,
+                )
 
     def finalize_input(
         self,

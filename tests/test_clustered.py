@@ -62,3 +62,32 @@ def test_clustered_rccl_allreduce(client):
     allsum = _make_allreduce_fn(app, "nccl")
     with app.run(client=client):
         assert allsum.remote(3) == 7.0
+
+
+def test_gang_member_failure_fails_gang(client):
+    """A failed gang member terminates its whole gang instead of retrying
+    solo (a lone rank can never rendezvous; gang fate-sharing)."""
+    import time
+
+    app = modal.App("gang-fail-app")
+
+    @app.function()
+    @clustered(size=2)
+    def failer():
+        import time as _t
+
+        from modal_amd.experimental import get_cluster_info
+
+        if get_cluster_info().rank == 1:
+            raise RuntimeError("boom-rank1")
+        _t.sleep(30)  # rank 0 parked, as if waiting in a collective
+        return "rank0-done"
+
+    with app.run(client=client):
+        t0 = time.time()
+        with pytest.raises(Exception) as excinfo:
+            failer.remote()
+        elapsed = time.time() - t0
+        assert elapsed < 20, "gang should be torn down, not awaited to completion"
+        msg = str(excinfo.value)
+        assert "boom-rank1" in msg or "gang member" in msg or "cancel" in msg.lower()

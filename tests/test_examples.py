@@ -38,3 +38,20 @@ def test_example_web_endpoint(runner, client):
     result = runner.invoke(entrypoint_cli, ["run", "examples/web_endpoint.py::app.main"])
     assert result.exit_code == 0, result.output
     assert "'hello': 'MI355X'" in result.output or '"hello":"MI355X"' in result.output
+
+
+@pytest.mark.timeout(300)
+def test_example_train_ddp(runner, client):
+    """DDP training example: 2-rank gang on CPU (gloo), params stay in
+    sync through bucketed gradient all-reduce."""
+    import os
+
+    os.environ["MODAL_AMD_FORCE_CPU"] = "1"
+    try:
+        result = runner.invoke(
+            entrypoint_cli, ["run", "examples/train_ddp.py::app.main", "--steps=4"]
+        )
+    finally:
+        os.environ.pop("MODAL_AMD_FORCE_CPU", None)
+    assert result.exit_code == 0, result.output
+    assert "ranks in sync: True" in result.output
