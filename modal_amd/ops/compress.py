@@ -104,7 +104,9 @@ def compress_buffer_gpu(data: bytes) -> Optional[bytes]:
     n_seg = (n + SEG_SIZE - 1) // SEG_SIZE
     if n_seg == 0:
         return None
-    src = torch.frombuffer(bytearray(data), dtype=torch.uint8).cuda()
+    from .staging import stage_to_gpu
+
+    src = stage_to_gpu(data)
     stride_buf = torch.empty(n_seg * OUT_STRIDE, dtype=torch.uint8, device="cuda")
     comp_lens_d = torch.zeros(n_seg, dtype=torch.int32, device="cuda")
     rc = lib.ma_lz4_compress(
@@ -136,7 +138,9 @@ def compress_buffer_gpu(data: bytes) -> Optional[bytes]:
     offsets = torch.arange(n_seg, dtype=torch.int64) * OUT_STRIDE
     packed, _ = pack_gpu(stride_buf, offsets, torch.tensor(eff_lens, dtype=torch.int64))
     torch.cuda.synchronize()
-    payload = packed.cpu().numpy().tobytes()
+    from .staging import fetch_from_gpu
+
+    payload = fetch_from_gpu(packed)
     return _header(n, comp_lens) + payload
 
 
@@ -166,7 +170,9 @@ def compress_buffers_gpu(buffers: list) -> list:
     big = bytearray(total_padded)
     for (src_off, n, _ns, _sb), data in zip(metas, buffers):
         big[src_off : src_off + n] = data
-    src = torch.frombuffer(big, dtype=torch.uint8).cuda()  # one H2D
+    from .staging import stage_to_gpu
+
+    src = stage_to_gpu(big)  # one pinned H2D on the side stream
     stride_buf = torch.empty(total_seg * OUT_STRIDE, dtype=torch.uint8, device="cuda")
     comp_lens_d = torch.zeros(total_seg, dtype=torch.int32, device="cuda")
     stream = torch.cuda.current_stream().cuda_stream
@@ -210,7 +216,9 @@ def compress_buffers_gpu(buffers: list) -> list:
         offsets = (torch.arange(n_seg, dtype=torch.int64) + sb) * OUT_STRIDE
         packed, _ = pack_gpu(stride_buf, offsets, torch.tensor(eff_lens, dtype=torch.int64))
         torch.cuda.synchronize()
-        out.append(_header(n, comp_lens) + packed.cpu().numpy().tobytes())
+        from .staging import fetch_from_gpu
+
+        out.append(_header(n, comp_lens) + fetch_from_gpu(packed))
     return out
 
 
@@ -228,7 +236,9 @@ def decompress_buffer_gpu(blob: bytes) -> bytes:
 
     raw_len, comp_lens, off = parse_header(blob)
     n_seg = len(comp_lens)
-    payload = torch.frombuffer(bytearray(blob[off:]), dtype=torch.uint8).cuda()
+    from .staging import stage_to_gpu
+
+    payload = stage_to_gpu(blob[off:])
     offs = []
     pos = 0
     for i, clen in enumerate(comp_lens):
@@ -249,7 +259,9 @@ def decompress_buffer_gpu(blob: bytes) -> bytes:
     bad = int(status.item())
     if bad:
         raise ValueError(f"LZ4 container corrupt at segment {bad - 1}")
-    return dst[:raw_len].cpu().numpy().tobytes()
+    from .staging import fetch_from_gpu
+
+    return fetch_from_gpu(dst[:raw_len])
 
 
 # ---------------------------------------------------------------------------

@@ -55,7 +55,9 @@ def _tree_sha256_gpu(data: Buffer) -> Optional[bytes]:
 
     n = len(data)
     n_leaves = (n + LEAF_SIZE - 1) // LEAF_SIZE
-    src = torch.frombuffer(bytearray(data), dtype=torch.uint8).cuda()
+    from .staging import stage_to_gpu
+
+    src = stage_to_gpu(data)  # pinned arena + side-stream hipMemcpyAsync
     offsets = torch.arange(0, n_leaves, dtype=torch.int64) * LEAF_SIZE
     lengths = torch.full((n_leaves,), LEAF_SIZE, dtype=torch.int64)
     if n % LEAF_SIZE:
@@ -144,7 +146,9 @@ def content_digests_batch(buffers: list, gpu_threshold: int = GPU_MIN_BYTES) -> 
             lengths.append(min(LEAF_SIZE, len(data) - leaf * LEAF_SIZE))
     if not offsets:
         return [small_idx[i] for i in range(len(buffers))]
-    src = torch.frombuffer(blob, dtype=torch.uint8).cuda()
+    from .staging import stage_to_gpu
+
+    src = stage_to_gpu(blob)
     out = torch.empty((len(offsets), 32), dtype=torch.uint8, device="cuda")
     off_d = torch.tensor(offsets, dtype=torch.int64).cuda()
     len_d = torch.tensor(lengths, dtype=torch.int64).cuda()
