@@ -45,6 +45,7 @@ from ..scheduler.calls import (
     GENERIC_STATUS_TERMINATED,
     GENERIC_STATUS_TIMEOUT,
 )
+from ..exception import InputCancellation
 from ..scheduler.transport import Connection
 from .execution_context import _reset_current_context, _set_current_context
 
@@ -314,6 +315,9 @@ class WorkerRuntime:
         self.task_id: str = ""
         self.functions: dict[str, FunctionRuntime] = {}
         self.mem_snapshot: Any = None  # GPUMemorySnapshot while paged out
+        # token -> executor-thread idents running that input's sync code
+        self._sync_threads: dict[str, set] = {}
+        self._sync_threads_lock = _threading.Lock()
         self.executor = ThreadPoolExecutor(
             max_workers=int(os.environ.get("MODAL_AMD_WORKER_THREADS", "16")),
             thread_name_prefix="modal-amd-input",
@@ -507,6 +511,10 @@ class WorkerRuntime:
                     task.cancel()
                 else:
                     self._abandoned.add(token)
+                # sync user code on executor threads: deliver the
+                # cancellation as an async exception at the next bytecode
+                # (SURVEY hard part 4: the SIGUSR1 analog for thread pools)
+                self._inject_cancel(token)
             if msg.get("terminate"):
                 self._shutdown.set()
         elif kind == "mesh_init":
@@ -657,7 +665,7 @@ class WorkerRuntime:
                 if is_gen:
                     await self._run_generator(frt, fn, token, args, kwargs)
                     return
-                result = await self._execute(frt, fn, args, kwargs)
+                result = await self._execute(frt, fn, args, kwargs, token)
                 data = serialize(result)
                 self.post_output(token, frt.function_id, GENERIC_STATUS_SUCCESS, data, DataFormat.PICKLE)
             except asyncio.TimeoutError:
@@ -670,7 +678,7 @@ class WorkerRuntime:
                     f"Function exceeded timeout of {frt.timeout}s "
                     f"(ran {time.monotonic() - started:.1f}s)",
                 )
-            except asyncio.CancelledError:
+            except (asyncio.CancelledError, InputCancellation):
                 self.post_output(
                     token, frt.function_id, GENERIC_STATUS_TERMINATED, None, 0, "input cancelled"
                 )
@@ -757,6 +765,7 @@ class WorkerRuntime:
                 return
             item = {"chunk": msg["chunk_id"], "ci": 0}
             tok_c = _current_function_call_id.set(call_id)
+            self._register_sync_thread(msg["token"])
             try:
                 for ci in range(start, end):
                     item["ci"] = ci
@@ -764,6 +773,8 @@ class WorkerRuntime:
                     try:
                         args, kwargs = self._resolve_item_args(item)
                         values[ci] = _invoke_on_slot_stream(fn, args, kwargs)
+                    except InputCancellation:
+                        raise  # injected cancel: abort the whole range
                     except BaseException as exc:
                         errors[ci] = (
                             self._serialize_exception(exc),
@@ -772,6 +783,7 @@ class WorkerRuntime:
                     finally:
                         _current_input_id.reset(tok_i)
             finally:
+                self._unregister_sync_thread(msg["token"])
                 _current_function_call_id.reset(tok_c)
 
         from .execution_context import _current_function_call_id, _current_input_id
@@ -957,7 +969,39 @@ class WorkerRuntime:
         if not self._outbox_flush_scheduled:
             await self._flush_outbox()
 
-    async def _execute(self, frt: FunctionRuntime, fn: Any, args: tuple, kwargs: dict) -> Any:
+    def _register_sync_thread(self, token: str) -> None:
+        if not token:
+            return
+        with self._sync_threads_lock:
+            self._sync_threads.setdefault(token, set()).add(_threading.get_ident())
+
+    def _unregister_sync_thread(self, token: str) -> None:
+        if not token:
+            return
+        with self._sync_threads_lock:
+            idents = self._sync_threads.get(token)
+            if idents is not None:
+                idents.discard(_threading.get_ident())
+                if not idents:
+                    self._sync_threads.pop(token, None)
+
+    def _inject_cancel(self, token: str) -> None:
+        """Raise InputCancellation inside executor threads running this
+        input's sync user code (delivered at the next Python bytecode;
+        blocking C calls finish first). The thread-pool analog of the
+        reference's SIGUSR1 cancellation (container_io_manager.py:890)."""
+        import ctypes
+
+        with self._sync_threads_lock:
+            idents = list(self._sync_threads.get(token, ()))
+        for ident in idents:
+            ctypes.pythonapi.PyThreadState_SetAsyncExc(
+                ctypes.c_ulong(ident), ctypes.py_object(InputCancellation)
+            )
+
+    async def _execute(
+        self, frt: FunctionRuntime, fn: Any, args: tuple, kwargs: dict, token: str = ""
+    ) -> Any:
         from ..utils.tracing import enabled as trace_enabled, trace_range
 
         if inspect.iscoroutinefunction(fn):
@@ -968,10 +1012,14 @@ class WorkerRuntime:
         ctx = contextvars.copy_context()
 
         def invoke() -> Any:
-            if trace_enabled():
-                with trace_range(f"modal_amd::{frt.name}"):
-                    return ctx.run(_invoke_on_slot_stream, fn, args, kwargs)
-            return ctx.run(_invoke_on_slot_stream, fn, args, kwargs)
+            self._register_sync_thread(token)
+            try:
+                if trace_enabled():
+                    with trace_range(f"modal_amd::{frt.name}"):
+                        return ctx.run(_invoke_on_slot_stream, fn, args, kwargs)
+                return ctx.run(_invoke_on_slot_stream, fn, args, kwargs)
+            finally:
+                self._unregister_sync_thread(token)
 
         fut = asyncio.get_running_loop().run_in_executor(self.executor, invoke)
         if frt.timeout:

@@ -258,3 +258,38 @@ def test_chunked_map_worker_death_requeues_chunk(client, run_dir):
         out = sorted(kill_once.map(range(150), kwargs={"path": marker}, order_outputs=False))
         assert out == list(range(150))
         assert os.path.exists(marker)
+
+
+def test_cancel_interrupts_sync_user_code(client):
+    """Cancelling a running input stops pure-Python sync user code via
+    async-exception injection WITHOUT killing the worker (SURVEY hard
+    part 4: the thread-pool analog of SIGUSR1 cancellation)."""
+    import time
+
+    app = modal.App("cancel-sync-app")
+
+    @app.function()
+    def spinner():
+        deadline = time.time() + 60
+        x = 0
+        while time.time() < deadline:  # pure-Python loop: injectable
+            x += 1
+        return x
+
+    @app.function()
+    def quick(v):
+        return v * 2
+
+    with app.run(client=client):
+        fc = spinner.spawn()
+        time.sleep(1.0)  # let it start spinning
+        workers_before = {w.task_id for w in client.svc.pool.workers.values()}
+        t0 = time.time()
+        fc.cancel()
+        with pytest.raises(Exception):
+            fc.get(timeout=20)
+        assert time.time() - t0 < 15, "cancel should interrupt the loop promptly"
+        # the worker survived (no termination, no respawn needed)
+        assert quick.remote(21) == 42
+        workers_after = {w.task_id for w in client.svc.pool.workers.values()}
+        assert workers_before & workers_after, "cancel should not kill workers"
