@@ -191,11 +191,15 @@ def compress_buffers_gpu(buffers: list) -> list:
     comp_lens_all = comp_lens_d.cpu().tolist()  # one D2H
 
     from .packing import pack_gpu
+    from .staging import fetch_from_gpu
 
-    out: list = []
-    for src_off, n, n_seg, sb in metas:
+    # decide per buffer, then compact EVERY kept buffer's segments with ONE
+    # pack launch and ONE pinned D2H (a per-buffer pack+sync+readback loop
+    # was 6x the kernel time in the config-4 trace)
+    kept: list = []  # (meta, comp_lens, eff_lens, total)
+    out: list = [None] * len(buffers)
+    for bi, (src_off, n, n_seg, sb) in enumerate(metas):
         if n_seg == 0:
-            out.append(None)
             continue
         comp_lens = comp_lens_all[sb : sb + n_seg]
         eff_lens, total = [], 0
@@ -205,7 +209,6 @@ def compress_buffers_gpu(buffers: list) -> list:
             eff_lens.append(eff)
             total += eff
         if total >= n * MIN_GAIN:
-            out.append(None)
             continue
         for i, clen in enumerate(comp_lens):
             if clen == 0:
@@ -213,12 +216,24 @@ def compress_buffers_gpu(buffers: list) -> list:
                 dst0 = (sb + i) * OUT_STRIDE
                 src0 = src_off + i * SEG_SIZE
                 stride_buf[dst0 : dst0 + seg_raw] = src[src0 : src0 + seg_raw]
-        offsets = (torch.arange(n_seg, dtype=torch.int64) + sb) * OUT_STRIDE
-        packed, _ = pack_gpu(stride_buf, offsets, torch.tensor(eff_lens, dtype=torch.int64))
-        torch.cuda.synchronize()
-        from .staging import fetch_from_gpu
-
-        out.append(_header(n, comp_lens) + fetch_from_gpu(packed))
+        kept.append((bi, n, sb, n_seg, comp_lens, eff_lens, total))
+    if not kept:
+        return out
+    all_offsets = torch.cat(
+        [(torch.arange(n_seg, dtype=torch.int64) + sb) * OUT_STRIDE
+         for _bi, _n, sb, n_seg, _cl, _el, _t in kept]
+    )
+    all_eff = torch.tensor(
+        [e for _bi, _n, _sb, _ns, _cl, eff_lens, _t in kept for e in eff_lens],
+        dtype=torch.int64,
+    )
+    packed, _ = pack_gpu(stride_buf, all_offsets, all_eff)
+    torch.cuda.synchronize()
+    payload_all = fetch_from_gpu(packed)
+    pos = 0
+    for bi, n, _sb, _ns, comp_lens, _eff_lens, total in kept:
+        out[bi] = _header(n, comp_lens) + payload_all[pos : pos + total]
+        pos += total
     return out
 
 
