@@ -66,10 +66,12 @@ def _invoke_on_slot_stream(fn: Any, args: tuple, kwargs: dict) -> Any:
     Concurrency slots each get a side stream, so one item's device sync
     (``.item()``, ``.cpu()``) waits only its own work instead of convoying
     every slot's kernels on the default stream. OFF by default: measured
-    2x SLOWER on the map bench (21.5k vs 44.3k items/s, same box) — per-item
-    ``stream.synchronize()`` busy-spins cost more than the convoy for short
-    per-item ops. Opt in with MODAL_AMD_SLOT_STREAMS=1 for workloads with
-    long per-item kernels.
+    2x SLOWER on the map bench in BOTH variants on MI355X — with a host
+    ``stream.synchronize()`` per item (21.5k vs 44.3k items/s) AND with
+    device-side event ordering only (20.8k/17.6k vs 41.7k/34.0k, same
+    box) — many active HSA queues cost more than the default-stream
+    convoy for microsecond ops. Opt in with MODAL_AMD_SLOT_STREAMS=1 for
+    workloads with long per-item kernels.
     """
     torch = sys.modules.get("torch")
     # sys.modules can expose a *partially initialized* torch while another
@@ -90,7 +92,12 @@ def _invoke_on_slot_stream(fn: Any, args: tuple, kwargs: dict) -> Any:
         _slot_tls.stream = stream
     with torch.cuda.stream(stream):
         result = fn(*args, **kwargs)
-    stream.synchronize()  # results must be safe to read/serialize anywhere
+    # order the default stream after this slot's work WITHOUT a host sync
+    # (the first attempt host-blocked in stream.synchronize() per item and
+    # measured 2x slower; an event wait is device-side only)
+    ev = torch.cuda.Event()
+    ev.record(stream)
+    torch.cuda.default_stream().wait_event(ev)
     return result
 
 #: the live worker runtime of this process (None in client processes);
