@@ -383,6 +383,26 @@ def metrics() -> None:
     click.echo(synchronizer.run(client.svc.node_metrics()), nl=False)
 
 
+@container_cli.command(name="exec", context_settings={"ignore_unknown_options": True})
+@click.argument("task_id")
+@click.argument("cmd", nargs=-1, required=True)
+def container_exec(task_id: str, cmd: tuple) -> None:
+    """Run a command in a worker's context (parity: modal container exec)."""
+    client = _get_client()
+    resp = synchronizer.run(client.svc.container_exec(task_id=task_id, cmd=list(cmd)))
+    click.echo(resp["output"], nl=False)
+    sys.exit(resp["returncode"] or 0)
+
+
+@container_cli.command(name="stop")
+@click.argument("task_id")
+def container_stop(task_id: str) -> None:
+    """Stop one worker (parity: modal container stop)."""
+    client = _get_client()
+    synchronizer.run(client.svc.container_stop(task_id=task_id))
+    click.echo(f"Stopped {task_id}")
+
+
 @container_cli.command(name="list")
 def container_list() -> None:
     client = _get_client()

@@ -146,6 +146,23 @@ class WorkerRPCTarget:
         rt.mem_snapshot = snap
         return {"state": snap.state.value}
 
+    async def exec_command(self, cmd: list, timeout: float = 60.0) -> dict:
+        """Run a command in this worker's context (parity: `modal
+        container exec`, reference cli/container.py:297 — the local
+        analog executes in the worker process's env/cwd)."""
+        proc = await asyncio.create_subprocess_exec(
+            *cmd,
+            stdout=asyncio.subprocess.PIPE,
+            stderr=asyncio.subprocess.STDOUT,
+            env=dict(os.environ),
+        )
+        try:
+            out, _ = await asyncio.wait_for(proc.communicate(), timeout)
+        except asyncio.TimeoutError:
+            proc.kill()
+            return {"returncode": -1, "output": "(timed out)"}
+        return {"returncode": proc.returncode, "output": out[-65536:].decode(errors="replace")}
+
     async def gpu_restore(self) -> dict:
         rt = self._runtime
         snap, rt.mem_snapshot = rt.mem_snapshot, None

@@ -73,6 +73,7 @@ class RPCAdapter:
         "function_finish_inputs",
         "function_get_outputs", "function_call_cancel", "function_call_info",
         "function_get_current_stats", "generator_poll", "node_stats", "node_metrics",
+        "container_exec", "container_stop",
         "app_lookup", "app_get_layout", "cluster_hello",
         "volume_get_or_create", "volume_put_file_blocks", "volume_get_file", "volume_list_files",
         "volume_remove_file", "volume_copy_files", "volume_commit", "volume_reload",
@@ -424,6 +425,24 @@ class Scheduler:
                 1 for c in self.calls.values() if not c.done_event.is_set()
             ),
         }
+
+    def _worker_by_task(self, task_id: str) -> Any:
+        for w in self.pool.workers.values():
+            if w.task_id == task_id:
+                return w
+        raise NotFoundError(f"Container {task_id} not found")
+
+    async def container_exec(self, task_id: str, cmd: list, timeout: float = 60.0) -> dict:
+        """Run a command inside a worker's context (parity: modal
+        container exec, reference cli/container.py:297)."""
+        w = self._worker_by_task(task_id)
+        return await w.conn.call("exec_command", {"cmd": cmd, "timeout": timeout}, timeout=timeout + 10)
+
+    async def container_stop(self, task_id: str) -> None:
+        """Stop one worker (parity: modal container stop, cli/container.py:318)."""
+        w = self._worker_by_task(task_id)
+        w.draining = True
+        await w.conn.send({"t": "shutdown"})
 
     async def node_metrics(self) -> str:
         """Prometheus-exposition snapshot of the gauges/counters SURVEY §5.5

@@ -116,3 +116,29 @@ def test_cli_shell_piped(runner, client):
     )
     assert result.exit_code == 0, result.output
     assert "piped-5" in result.output
+
+
+def test_cli_container_exec_and_stop(runner, client):
+    """`container exec` runs in a worker's context; `container stop`
+    drains it (parity: reference cli/container.py:297,318)."""
+    import modal_amd as modal
+
+    app = modal.App("cexec-app")
+
+    @app.function()
+    def noop():
+        return 1
+
+    with app.run(client=client):
+        assert noop.remote() == 1
+        workers = list(client.svc.pool.workers.values())
+        assert workers
+        task_id = workers[0].task_id
+        result = runner.invoke(
+            entrypoint_cli, ["container", "exec", task_id, "python3", "-c", "print(6*7)"]
+        )
+        assert result.exit_code == 0, result.output
+        assert "42" in result.output
+        result = runner.invoke(entrypoint_cli, ["container", "stop", task_id])
+        assert result.exit_code == 0, result.output
+        assert "Stopped" in result.output
