@@ -375,6 +375,22 @@ def container_stats() -> None:
         )
 
 
+@entrypoint_cli.group()
+def endpoint() -> None:
+    """Manage web endpoints (parity: modal endpoint)."""
+
+
+@endpoint.command(name="list")
+def endpoint_list() -> None:
+    client = _get_client()
+    rows = synchronizer.run(client.svc.endpoint_list())
+    if not rows:
+        click.echo("No web endpoints registered.")
+        return
+    for row in rows:
+        click.echo(f"{row['label']}  {row['function_id']}  {row['url'] or '(gateway not started)'}")
+
+
 @entrypoint_cli.command()
 def metrics() -> None:
     """Node metrics in Prometheus exposition format (also served at the

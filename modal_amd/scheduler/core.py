@@ -73,7 +73,7 @@ class RPCAdapter:
         "function_finish_inputs",
         "function_get_outputs", "function_call_cancel", "function_call_info",
         "function_get_current_stats", "generator_poll", "node_stats", "node_metrics",
-        "container_exec", "container_stop",
+        "container_exec", "container_stop", "endpoint_list",
         "app_lookup", "app_get_layout", "cluster_hello",
         "volume_get_or_create", "volume_put_file_blocks", "volume_get_file", "volume_list_files",
         "volume_remove_file", "volume_copy_files", "volume_commit", "volume_reload",
@@ -443,6 +443,16 @@ class Scheduler:
         w = self._worker_by_task(task_id)
         w.draining = True
         await w.conn.send({"t": "shutdown"})
+
+    async def endpoint_list(self) -> list:
+        """Registered web endpoints (parity: modal endpoint list,
+        reference cli/endpoint.py)."""
+        gw = self.web_gateway
+        return [
+            {"label": label, "function_id": fid,
+             "url": gw.url_for(label) if gw.port else None}
+            for label, fid in gw.routes.items()
+        ]
 
     async def node_metrics(self) -> str:
         """Prometheus-exposition snapshot of the gauges/counters SURVEY §5.5

@@ -142,3 +142,19 @@ def test_cli_container_exec_and_stop(runner, client):
         result = runner.invoke(entrypoint_cli, ["container", "stop", task_id])
         assert result.exit_code == 0, result.output
         assert "Stopped" in result.output
+
+
+def test_cli_endpoint_list(runner, client):
+    import modal_amd as modal
+
+    app = modal.App("ep-app")
+
+    @app.function()
+    @modal.fastapi_endpoint()
+    def hello_ep():
+        return {"ok": True}
+
+    with app.run(client=client):
+        result = runner.invoke(entrypoint_cli, ["endpoint", "list"])
+        assert result.exit_code == 0, result.output
+        assert "hello-ep" in result.output or "hello_ep" in result.output
