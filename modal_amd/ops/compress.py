@@ -154,25 +154,20 @@ def compress_buffers_gpu(buffers: list) -> list:
     lib = load_lib(required=True)
     import torch
 
+    from .staging import stage_many_to_gpu
+
+    # one pinned H2D for all buffers, 4 KiB-aligned (no concat bytearray)
+    src, src_offsets = stage_many_to_gpu(buffers, align=SEG_SIZE)
     metas = []  # (src_off, n, n_seg, seg_base)
-    off = 0
     seg_base = 0
-    for data in buffers:
+    for data, src_off in zip(buffers, src_offsets):
         n = len(data)
         n_seg = (n + SEG_SIZE - 1) // SEG_SIZE
-        metas.append((off, n, n_seg, seg_base))
-        off += (n + SEG_SIZE - 1) // SEG_SIZE * SEG_SIZE  # keep 4 KiB alignment
+        metas.append((src_off, n, n_seg, seg_base))
         seg_base += n_seg
-    total_padded, total_seg = off, seg_base
+    total_seg = seg_base
     if total_seg == 0:
         return [None] * len(buffers)
-
-    big = bytearray(total_padded)
-    for (src_off, n, _ns, _sb), data in zip(metas, buffers):
-        big[src_off : src_off + n] = data
-    from .staging import stage_to_gpu
-
-    src = stage_to_gpu(big)  # one pinned H2D on the side stream
     stride_buf = torch.empty(total_seg * OUT_STRIDE, dtype=torch.uint8, device="cuda")
     comp_lens_d = torch.zeros(total_seg, dtype=torch.int32, device="cuda")
     stream = torch.cuda.current_stream().cuda_stream

@@ -123,11 +123,12 @@ def content_digests_batch(buffers: list, gpu_threshold: int = GPU_MIN_BYTES) -> 
     lib = load_lib(required=True)
     import torch
 
-    blob = bytearray()
     spans: list[tuple[int, int]] = []  # (first_leaf_index, n_leaves) per buffer
     offsets: list[int] = []
     lengths: list[int] = []
     small_idx: dict[int, str] = {}
+    big_bufs: list = []
+    big_meta: list[int] = []  # index into spans to fix up with real offsets
     for i, data in enumerate(buffers):
         if len(data) < gpu_threshold:
             # digest form must depend only on the buffer's own size so every
@@ -137,18 +138,23 @@ def content_digests_batch(buffers: list, gpu_threshold: int = GPU_MIN_BYTES) -> 
             small_idx[i] = _hl.sha256(bytes(data)).hexdigest()
             spans.append((len(offsets), 0))
             continue
-        base = len(blob)
-        blob += bytes(data)
         n_leaves = (len(data) + LEAF_SIZE - 1) // LEAF_SIZE
         spans.append((len(offsets), n_leaves))
+        big_meta.append((len(big_bufs), i, len(offsets), n_leaves))
+        big_bufs.append(data)
+        offsets.extend([0] * n_leaves)  # placeholders until staged
         for leaf in range(n_leaves):
-            offsets.append(base + leaf * LEAF_SIZE)
             lengths.append(min(LEAF_SIZE, len(data) - leaf * LEAF_SIZE))
-    if not offsets:
+    if not lengths:
         return [small_idx[i] for i in range(len(buffers))]
-    from .staging import stage_to_gpu
+    from .staging import stage_many_to_gpu
 
-    src = stage_to_gpu(blob)
+    # one pinned H2D for all large buffers (no concat bytearray pass)
+    src, base_offsets = stage_many_to_gpu(big_bufs, align=LEAF_SIZE)
+    for bk, _i, first, n_leaves in big_meta:
+        base = base_offsets[bk]
+        for leaf in range(n_leaves):
+            offsets[first + leaf] = base + leaf * LEAF_SIZE
     out = torch.empty((len(offsets), 32), dtype=torch.uint8, device="cuda")
     off_d = torch.tensor(offsets, dtype=torch.int64).cuda()
     len_d = torch.tensor(lengths, dtype=torch.int64).cuda()

@@ -60,6 +60,39 @@ def stage_to_gpu(data: Any) -> Any:
     return dst
 
 
+def stage_many_to_gpu(buffers: list, align: int = 1) -> tuple:
+    """Stage several host buffers into ONE device tensor with a single
+    pinned H2D, each placed at an ``align``-aligned offset (skips the
+    intermediate concat bytearray a caller would otherwise build — a
+    full extra pass over the payload). Returns (device_tensor, offsets)."""
+    import numpy as np
+    import torch
+
+    offsets = []
+    off = 0
+    for data in buffers:
+        offsets.append(off)
+        off += -(-len(data) // align) * align
+    total = off
+    if total == 0:
+        return torch.empty(0, dtype=torch.uint8, device="cuda"), offsets
+    if total > (1 << 30):
+        big = bytearray(total)
+        for data, o in zip(buffers, offsets):
+            big[o : o + len(data)] = data
+        return torch.frombuffer(big, dtype=torch.uint8).cuda(), offsets
+    with _lock:
+        _ensure(total)
+        for data, o in zip(buffers, offsets):
+            _pin_np[o : o + len(data)] = np.frombuffer(data, dtype=np.uint8)
+        dst = torch.empty(total, dtype=torch.uint8, device="cuda")
+        with torch.cuda.stream(_h2d_stream):
+            dst.copy_(_pin_buf[:total], non_blocking=True)
+        _h2d_stream.synchronize()
+    torch.cuda.current_stream().wait_stream(_h2d_stream)
+    return dst, offsets
+
+
 def fetch_from_gpu(tensor: Any) -> bytes:
     """Device uint8 tensor -> bytes via the pinned arena + side stream."""
     import torch
