@@ -144,7 +144,7 @@ def compress_buffer_gpu(data: bytes) -> Optional[bytes]:
     return _header(n, comp_lens) + payload
 
 
-def compress_buffers_gpu(buffers: list) -> list:
+def compress_buffers_gpu(buffers: list, staged: "object" = None, only: "object" = None) -> list:
     """Batched compression: ONE host->device transfer and ONE sync for a
     whole upload's blocks (volume v2 8 MiB blocks are exact multiples of
     SEG_SIZE, so per-buffer kernel launches share the staged source at
@@ -154,15 +154,19 @@ def compress_buffers_gpu(buffers: list) -> list:
     lib = load_lib(required=True)
     import torch
 
-    from .staging import stage_many_to_gpu
+    pick = set(range(len(buffers))) if only is None else set(only)
+    if staged is not None:
+        src, src_offsets = staged  # shared with the digest pass
+    else:
+        from .staging import stage_many_to_gpu
 
-    # one pinned H2D for all buffers, 4 KiB-aligned (no concat bytearray)
-    src, src_offsets = stage_many_to_gpu(buffers, align=SEG_SIZE)
-    metas = []  # (src_off, n, n_seg, seg_base)
+        # one pinned H2D for all buffers, 4 KiB-aligned (no concat bytearray)
+        src, src_offsets = stage_many_to_gpu(buffers, align=SEG_SIZE)
+    metas = []  # (src_off, n, n_seg, seg_base); n_seg=0 for skipped buffers
     seg_base = 0
-    for data, src_off in zip(buffers, src_offsets):
+    for bi, (data, src_off) in enumerate(zip(buffers, src_offsets)):
         n = len(data)
-        n_seg = (n + SEG_SIZE - 1) // SEG_SIZE
+        n_seg = (n + SEG_SIZE - 1) // SEG_SIZE if bi in pick else 0
         metas.append((src_off, n, n_seg, seg_base))
         seg_base += n_seg
     total_seg = seg_base
@@ -232,11 +236,11 @@ def compress_buffers_gpu(buffers: list) -> list:
     return out
 
 
-def compress_buffers(buffers: list) -> list:
+def compress_buffers(buffers: list, staged: "object" = None, only: "object" = None) -> list:
     """GPU-batched when present; None entries otherwise (CPU compression
     is not worth its cost on the storage path)."""
     if gpu_available():
-        return compress_buffers_gpu(buffers)
+        return compress_buffers_gpu(buffers, staged=staged, only=only)
     return [None] * len(buffers)
 
 

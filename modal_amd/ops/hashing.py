@@ -110,7 +110,9 @@ def tree_sha256(data: Buffer) -> bytes:
     return tree_sha256_cpu(data)
 
 
-def content_digests_batch(buffers: list, gpu_threshold: int = GPU_MIN_BYTES) -> list[str]:
+def content_digests_batch(
+    buffers: list, gpu_threshold: int = GPU_MIN_BYTES, staged: "object" = None
+) -> list[str]:
     """CAS keys for many buffers with ONE kernel launch over all leaves.
 
     This is the volume-upload hot path: a multi-GiB file's 8 MiB blocks hash
@@ -147,12 +149,17 @@ def content_digests_batch(buffers: list, gpu_threshold: int = GPU_MIN_BYTES) -> 
             lengths.append(min(LEAF_SIZE, len(data) - leaf * LEAF_SIZE))
     if not lengths:
         return [small_idx[i] for i in range(len(buffers))]
-    from .staging import stage_many_to_gpu
+    if staged is not None:
+        # caller pre-staged ALL buffers (one pinned H2D shared with the
+        # compression pass, blobs.put_many); map big-buffer offsets from it
+        src, all_offsets = staged
+        base_offsets = [all_offsets[i] for _bk, i, _f, _n in big_meta]
+    else:
+        from .staging import stage_many_to_gpu
 
-    # one pinned H2D for all large buffers (no concat bytearray pass)
-    src, base_offsets = stage_many_to_gpu(big_bufs, align=LEAF_SIZE)
-    for bk, _i, first, n_leaves in big_meta:
-        base = base_offsets[bk]
+        # one pinned H2D for all large buffers (no concat bytearray pass)
+        src, base_offsets = stage_many_to_gpu(big_bufs, align=LEAF_SIZE)
+    for (bk, _i, first, n_leaves), base in zip(big_meta, base_offsets):
         for leaf in range(n_leaves):
             offsets[first + leaf] = base + leaf * LEAF_SIZE
     out = torch.empty((len(offsets), 32), dtype=torch.uint8, device="cuda")

@@ -69,9 +69,22 @@ class BlobStore:
         one batched compression pass over the new large blocks (the
         volume-upload hot path; ops/hashing.content_digests_batch +
         ops/compress.compress_buffers)."""
-        from ..ops.hashing import content_digests_batch
+        from ..ops import gpu_available
+        from ..ops.hashing import GPU_MIN_BYTES, content_digests_batch
 
-        digests = content_digests_batch(buffers)
+        # stage once: the digest and compression passes share one pinned
+        # H2D of all buffers (LEAF_SIZE alignment satisfies both kernels)
+        staged = None
+        total = sum(len(b) for b in buffers)
+        if total >= GPU_MIN_BYTES and gpu_available():
+            try:
+                from ..ops.hashing import LEAF_SIZE
+                from ..ops.staging import stage_many_to_gpu
+
+                staged = stage_many_to_gpu(buffers, align=LEAF_SIZE)
+            except Exception:
+                staged = None
+        digests = content_digests_batch(buffers, staged=staged)
         # compress only blocks that are big enough and not already stored
         todo = [
             i for i, (digest, data) in enumerate(zip(digests, buffers))
@@ -82,9 +95,15 @@ class BlobStore:
             try:
                 from ..ops.compress import compress_buffers
 
-                for i, payload in zip(todo, compress_buffers([buffers[i] for i in todo])):
-                    if payload is not None:
-                        compressed[i] = payload
+                if staged is not None:
+                    payloads = compress_buffers(buffers, staged=staged, only=todo)
+                else:
+                    payloads = [None] * len(buffers)
+                    for i, payload in zip(todo, compress_buffers([buffers[i] for i in todo])):
+                        payloads[i] = payload
+                for i in todo:
+                    if payloads[i] is not None:
+                        compressed[i] = payloads[i]
             except Exception:
                 pass  # store raw on any kernel/driver hiccup
         for i, (digest, data) in enumerate(zip(digests, buffers)):
