@@ -105,14 +105,17 @@ class WebGateway:
             if done_flag:
                 return True
             value = deserialize(data)
-            if stream is None:
-                headers = {k: v for k, v in value.get("headers", [])}
-                headers.pop("Content-Length", None)
-                headers.pop("content-length", None)
-                stream = web.StreamResponse(status=value.get("status", 200), headers=headers)
-                await stream.prepare(request)
-            else:
-                await stream.write(value)
+            try:
+                if stream is None:
+                    headers = {k: v for k, v in value.get("headers", [])}
+                    headers.pop("Content-Length", None)
+                    headers.pop("content-length", None)
+                    stream = web.StreamResponse(status=value.get("status", 200), headers=headers)
+                    await stream.prepare(request)
+                else:
+                    await stream.write(value)
+            except (ConnectionResetError, OSError):
+                return True  # client went away: stop forwarding
             return False
 
         try:
@@ -140,5 +143,8 @@ class WebGateway:
                 output_task.cancel()
         if stream is None:
             return web.Response(status=500, text="web function produced no response")
-        await stream.write_eof()
+        try:
+            await stream.write_eof()
+        except ConnectionResetError:
+            pass  # client went away mid-stream
         return stream
