@@ -116,7 +116,9 @@ async def process_output_item(item: dict, client: Any) -> Any:
             except Exception:
                 raise RemoteError(item.get("exc") or "remote exception (undeserializable)") from None
             if isinstance(exc, BaseException):
-                raise exc
+                from .utils.tb import attach_remote_frames
+
+                raise attach_remote_frames(exc)
         raise RemoteError(item.get("exc") or "remote exception")
     raise ExecutionError(f"Unknown output status {status}")
 

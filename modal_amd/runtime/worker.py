@@ -1162,9 +1162,15 @@ class WorkerRuntime:
 
     # ---- outputs ---------------------------------------------------------
     def _serialize_exception(self, exc: BaseException) -> bytes:
-        from ..utils.tb import clean_traceback
+        from ..utils.tb import clean_traceback, extract_frames
 
         clean_traceback(exc)
+        try:
+            # pickle drops __traceback__: carry the user frames explicitly
+            # so the client re-synthesizes them (utils/tb.py forge)
+            exc.__modal_amd_tb__ = extract_frames(exc)
+        except Exception:
+            pass
         try:
             return serialize(exc)
         except BaseException:

@@ -19,6 +19,71 @@ def _is_internal(frame_file: str) -> bool:
     return frame_file.startswith(_PKG_ROOT) or "asyncio" in frame_file.split(os.sep)
 
 
+def extract_frames(exc: BaseException) -> list:
+    """Summarize user frames of exc.__traceback__ for the wire (pickle
+    drops tracebacks; the reference vendors tblib for this —
+    _traceback.py:27 extract_traceback)."""
+    out = []
+    tb = exc.__traceback__
+    while tb is not None:
+        code = tb.tb_frame.f_code
+        out.append((code.co_filename, tb.tb_lineno, code.co_name))
+        tb = tb.tb_next
+    return out
+
+
+_frame_cache: dict = {}
+
+
+def _forge_frame(filename: str, name: str) -> Optional[object]:
+    """Manufacture a frame whose code claims (filename, name) — the tblib
+    technique: raise inside exec'd code compiled with that filename."""
+    key = (filename, name)
+    if key in _frame_cache:
+        return _frame_cache[key]
+    try:
+        code = compile("raise ValueError()", filename, "exec")
+        if name.isidentifier():
+            code = code.replace(co_name=name)
+        frame = None
+        try:
+            exec(code, {})
+        except ValueError as e:
+            frame = e.__traceback__.tb_next.tb_frame
+        _frame_cache[key] = frame
+        return frame
+    except Exception:
+        _frame_cache[key] = None
+        return None
+
+
+def forge_traceback(frames: list) -> Optional[types.TracebackType]:
+    """Rebuild a TracebackType chain from extract_frames output. On a
+    single node the filenames resolve locally, so linecache shows real
+    source lines in the re-synthesized stack."""
+    tb: Optional[types.TracebackType] = None
+    for filename, lineno, name in reversed(frames):
+        frame = _forge_frame(filename, name)
+        if frame is None:
+            continue
+        try:
+            tb = types.TracebackType(tb, frame, -1, lineno)
+        except Exception:
+            continue
+    return tb
+
+
+def attach_remote_frames(exc: BaseException) -> BaseException:
+    """Client side: splice the worker-recorded frames back under exc so
+    `raise` shows local frames then the remote user stack."""
+    frames = getattr(exc, "__modal_amd_tb__", None)
+    if frames:
+        tb = forge_traceback(frames)
+        if tb is not None:
+            return exc.with_traceback(tb)
+    return exc
+
+
 def clean_traceback(exc: BaseException) -> None:
     """Drop framework/asyncio frames from exc.__traceback__ in place."""
     tb = exc.__traceback__

@@ -138,3 +138,49 @@ def test_output_steps_and_map_progress(client, capsys):
     p.update(3, 10)
     assert (p.completed, p.submitted) == (3, 10)
     p.close()
+
+
+def test_remote_traceback_shows_user_frames_only(client):
+    """Remote exceptions re-raise locally with the USER's frames and
+    without framework/asyncio noise (parity: reference _traceback.py
+    re-synthesis + frame suppression)."""
+    import traceback
+
+    import modal_amd as modal
+
+    app = modal.App("tb-app")
+
+    @app.function()
+    def outer_fails():
+        def deep_inner():
+            raise ValueError("tb-sentinel")
+
+        deep_inner()
+
+    with app.run(client=client):
+        try:
+            outer_fails.remote()
+            raise AssertionError("should have raised")
+        except ValueError as exc:
+            tb_text = "".join(traceback.format_exception(type(exc), exc, exc.__traceback__))
+    assert "tb-sentinel" in tb_text
+    assert "deep_inner" in tb_text            # the remote user frame survives
+    assert "outer_fails" in tb_text
+    assert "/modal_amd/runtime/worker.py" not in tb_text  # framework frames dropped
+
+
+def test_retry_policy_clamp_and_backoff():
+    """Retries delay math: 1 s floor, 24 h ceiling, exponential factor
+    (parity: reference retries.py:8-9 clamp)."""
+    import pytest
+
+    import modal_amd as modal
+    from modal_amd.exception import InvalidError
+
+    r = modal.Retries(max_retries=5, initial_delay=2.0, backoff_coefficient=2.0, max_delay=10.0)
+    policy = r._to_policy_dict()
+    assert policy["max_retries"] == 5
+    with pytest.raises(InvalidError):
+        modal.Retries(max_retries=5, initial_delay=0.1)  # below the 1 s floor
+    with pytest.raises(InvalidError):
+        modal.Retries(max_retries=5, initial_delay=2.0, max_delay=100_000.0)  # > 24 h
