@@ -27,8 +27,10 @@ async def usage_summary(client: Any = None) -> list[dict]:
         row = per_function.setdefault(
             name, {"function": name, "inputs": 0, "runtime_seconds": 0.0, "gpu": bool(fdef and fdef.needs_gpu)}
         )
+        # range-protocol successes never materialize per-item records, so
+        # the call-level completion counter is the authoritative count
+        row["inputs"] += max(record.completed, len(record.inputs))
         for rec in record.inputs.values():
-            row["inputs"] += 1
             if rec.finished_at and rec.started_at:
                 row["runtime_seconds"] += rec.finished_at - rec.started_at
     return sorted(per_function.values(), key=lambda r: -r["runtime_seconds"])
