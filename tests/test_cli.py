@@ -158,3 +158,24 @@ def test_cli_endpoint_list(runner, client):
         result = runner.invoke(entrypoint_cli, ["endpoint", "list"])
         assert result.exit_code == 0, result.output
         assert "hello-ep" in result.output or "hello_ep" in result.output
+
+
+def test_cli_app_logs(runner, client):
+    """`app logs` replays worker stdout captured by the log plane."""
+    import modal_amd as modal
+
+    app = modal.App("logs-app")
+
+    @app.function()
+    def chatty():
+        print("log-line-sentinel")
+        return 1
+
+    with app.run(client=client):
+        assert chatty.remote() == 1
+        import time
+
+        time.sleep(0.3)  # log forwarding is async
+        result = runner.invoke(entrypoint_cli, ["app", "logs", app.app_id])
+    assert result.exit_code == 0, result.output
+    assert "log-line-sentinel" in result.output
