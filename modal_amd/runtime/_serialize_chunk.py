@@ -1,15 +1,61 @@
 """Output-chunk serialization for the worker fast path.
 
 One C-pickler pass over a whole frame's result values (the mirror of input
-chunking). Values containing tensors or live handles need the hook-aware
-pickler per item, so those frames fall back to per-item bytes.
-"""
+chunking). Map items that return small CUDA tensors get a **batched
+readback**: one device gather + one pinned D2H for the whole chunk instead
+of a ~24 us `.cpu()`/`.item()` sync per item (the
+batched tensor-pack path; measured sync latency dominates per-item GPU
+ops). Values with nested/oversized device tensors fall back to the
+hook-aware per-item pickler (mesh export / host staging)."""
 
 from __future__ import annotations
 
 import pickle
 import sys
 from typing import Optional
+
+# above this the per-item blob/mesh export path is the right tool
+BATCH_READBACK_MAX_BYTES = 64 * 1024 * 1024
+
+
+def _batch_cuda_to_host(values: list, torch: "object") -> None:
+    """Replace top-level small CUDA tensors with host copies using one
+    device gather + one pinned D2H (in place; no-op when none qualify)."""
+    idxs = [
+        i for i, v in enumerate(values)
+        if type(v) is torch.Tensor and v.is_cuda
+    ]
+    if not idxs:
+        return
+    tensors = [values[i].detach() for i in idxs]
+    total = sum(t.numel() * t.element_size() for t in tensors)
+    if total == 0 or total > BATCH_READBACK_MAX_BYTES:
+        return
+    try:
+        flat = torch.cat([t.contiguous().reshape(-1).view(torch.uint8) for t in tensors])
+        from ..ops.staging import fetch_from_gpu
+
+        buf = fetch_from_gpu(flat)  # ONE sync for the whole chunk
+        off = 0
+        for i, t in zip(idxs, tensors):
+            nbytes = t.numel() * t.element_size()
+            host = torch.frombuffer(
+                bytearray(buf[off : off + nbytes]), dtype=t.dtype
+            ).reshape(t.shape)
+            values[i] = host
+            off += nbytes
+    except Exception:
+        return  # keep device tensors; the per-item path handles them
+
+
+def _has_cuda(obj: "object", tensor_cls: type) -> bool:
+    if isinstance(obj, tensor_cls):
+        return bool(obj.is_cuda)
+    if type(obj) in (list, tuple):
+        return any(_has_cuda(v, tensor_cls) for v in obj)
+    if type(obj) is dict:
+        return any(_has_cuda(v, tensor_cls) for v in obj.values())
+    return False
 
 
 def serialize_value_chunk(values: list) -> tuple[Optional[bytes], Optional[list]]:
@@ -19,9 +65,13 @@ def serialize_value_chunk(values: list) -> tuple[Optional[bytes], Optional[list]
         from .._serialization import _walk_for_tensors
 
         if _walk_for_tensors(values, torch.Tensor):
-            from .._serialization import serialize
+            _batch_cuda_to_host(values, torch)
+            if _has_cuda(values, torch.Tensor):
+                # nested or oversized device tensors: hook-aware per item
+                from .._serialization import serialize
 
-            return None, [serialize(v) for v in values]
+                return None, [serialize(v) for v in values]
+            # host tensors pickle natively; fall through to one C pass
     try:
         data = pickle.dumps(values, 4)
     except Exception:

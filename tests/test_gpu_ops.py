@@ -202,3 +202,29 @@ def test_volume_gpu_hash_upload(client):
             batch.put_file(io.BytesIO(payload), "model.bin")
         data = b"".join(chunk for chunk in vol.read_file("model.bin"))
         assert data == payload
+
+
+@pytest.mark.gpu
+def test_map_returning_gpu_tensors_batched_readback(client):
+    """Map items returning CUDA tensors arrive as host tensors via ONE
+    batched D2H per chunk (not a per-item sync); values are exact."""
+    torch = pytest.importorskip("torch")
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    import modal_amd as modal
+
+    app = modal.App("tensor-map-app")
+
+    @app.function(gpu=1)
+    def make(i):
+        import torch
+
+        return torch.full((8,), float(i), device="cuda", dtype=torch.bfloat16)
+
+    with app.run(client=client):
+        outs = list(make.map(range(200), order_outputs=True))
+    assert len(outs) == 200
+    for i, t in enumerate(outs):
+        assert not t.is_cuda          # host tensor on arrival
+        assert t.dtype == torch.bfloat16 and t.shape == (8,)
+        assert float(t[0]) == float(i)
