@@ -159,3 +159,29 @@ def test_aio_variants_for_resources(client):
         assert entries == []
 
     asyncio.run(main())
+
+
+def test_update_autoscaler_and_stats(client):
+    """update_autoscaler / keep_warm adjust the live FunctionDef;
+    get_current_stats reports backlog+runners (parity: reference
+    _functions.py:1195-1292, :2021)."""
+    import modal_amd as modal
+
+    app = modal.App("scaler-app")
+
+    @app.function()
+    def f(x):
+        return x
+
+    with app.run(client=client):
+        assert f.remote(1) == 1
+        f.update_autoscaler(min_containers=2, max_containers=5, buffer_containers=1,
+                            scaledown_window=120)
+        fdef = client.svc.functions[f.object_id]
+        assert (fdef.min_containers, fdef.max_containers) == (2, 5)
+        assert fdef.buffer_containers == 1 and fdef.scaledown_window == 120
+        f.keep_warm(3)
+        assert client.svc.functions[f.object_id].min_containers == 3
+        stats = f.get_current_stats()
+        assert stats["backlog"] == 0
+        assert "num_total_tasks" in stats
