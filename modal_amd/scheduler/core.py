@@ -27,7 +27,6 @@ from .calls import (
     GENERIC_STATUS_INTERNAL_FAILURE,
     GENERIC_STATUS_SUCCESS,
     GENERIC_STATUS_TERMINATED,
-    GENERIC_STATUS_TIMEOUT,
     MAX_INPUTS_OUTSTANDING_DEFAULT,
     MAX_INTERNAL_FAILURE_COUNT,
     CallRecord,
