@@ -13,6 +13,7 @@ reference documents it.
 from __future__ import annotations
 
 import asyncio
+import logging
 import os
 import tempfile
 import time
