@@ -97,3 +97,26 @@ def test_ephemeral_state_not_persisted(run_dir):
     finally:
         synchronizer.run(c2.close())
         _Client._singleton = None
+
+
+def test_corrupt_state_file_degrades_gracefully(run_dir):
+    """A torn/corrupt state.pkl must not prevent scheduler startup."""
+    import os
+
+    with open(os.path.join(run_dir, "state.pkl"), "wb") as f:
+        f.write(b"\x80\x05garbage-not-a-pickle")
+    c = _start_client(run_dir)
+    try:
+        import modal_amd as modal
+
+        app = modal.App("after-corrupt")
+
+        @app.function()
+        def ok():
+            return "fine"
+
+        with app.run(client=c):
+            assert ok.remote() == "fine"
+    finally:
+        synchronizer.run(c.close())
+        _Client._singleton = None
