@@ -220,3 +220,29 @@ def test_map_large_outputs(client):
         out = list(inflate.map(range(3)))
         for i, blob in enumerate(out):
             assert blob == bytes([i % 256]) * (3 * 1024 * 1024)
+
+
+def test_nested_invocations_inside_worker(client):
+    """Functions calling other functions (unary, map, remote_gen) from
+    inside a worker — the thread-local blocking path end-to-end."""
+    app = modal.App("nested-app")
+
+    @app.function()
+    def leaf(x):
+        return x + 1
+
+    @app.function()
+    def streamer(n):
+        for i in range(n):
+            yield i * 10
+
+    @app.function()
+    def orchestrator(n):
+        unary = leaf.remote(n)                      # nested unary
+        mapped = sorted(leaf.map(range(3)))         # nested map
+        streamed = list(streamer.remote_gen(3))     # nested generator
+        return {"unary": unary, "mapped": mapped, "streamed": streamed}
+
+    with app.run(client=client):
+        out = orchestrator.remote(41)
+    assert out == {"unary": 42, "mapped": [1, 2, 3], "streamed": [0, 10, 20]}
