@@ -246,3 +246,28 @@ def test_nested_invocations_inside_worker(client):
     with app.run(client=client):
         out = orchestrator.remote(41)
     assert out == {"unary": 42, "mapped": [1, 2, 3], "streamed": [0, 10, 20]}
+
+
+def test_async_function_uses_handles_aio(client):
+    """Async user functions use .aio handle APIs on the worker's own
+    loop (per-loop connection path)."""
+    app = modal.App("async-handles-app")
+
+    @app.function()
+    def helper(x):
+        return x * 3
+
+    @app.function()
+    async def async_orchestrator(n):
+        d = modal.Dict.from_name("async-d", create_if_missing=True)
+        await d.put.aio("k", n)
+        v = await d.get.aio("k")
+        r = await helper.remote.aio(v)
+        total = 0
+        async for out in helper.map.aio(range(3)):
+            total += out
+        return {"dict": v, "unary": r, "map_sum": total}
+
+    with app.run(client=client):
+        out = async_orchestrator.remote(7)
+    assert out == {"dict": 7, "unary": 21, "map_sum": 0 + 3 + 6}
