@@ -206,6 +206,7 @@ class CallRecord:
         self.completed: int = 0
         self.cancelled = False
         self.created_at = time.time()
+        self.finished_at: Optional[float] = None
         # completion-order queue of idx for streaming GetOutputs
         self.output_ready: asyncio.Queue[int] = asyncio.Queue()
         # per-input completion events for unary waits
@@ -292,6 +293,8 @@ class CallRecord:
 
     def _check_done(self) -> None:
         if self.num_inputs_final is not None and self.completed >= self.num_inputs_final:
+            if not self.done_event.is_set():
+                self.finished_at = time.time()  # call-record GC clock
             self.done_event.set()
 
     # -- completion ------------------------------------------------------

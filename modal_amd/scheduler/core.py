@@ -159,6 +159,7 @@ class Scheduler:
             self.schedule_runner = ScheduleRunner(self)
             self.schedule_runner.start()
             self._persist_task = asyncio.get_running_loop().create_task(self._persist_loop())
+            self._call_gc_task = asyncio.get_running_loop().create_task(self._call_gc_loop())
             self._started = True
 
     async def stop(self) -> None:
@@ -169,6 +170,9 @@ class Scheduler:
         if self._persist_task is not None:
             self._persist_task.cancel()
             self._persist_task = None
+        if getattr(self, "_call_gc_task", None) is not None:
+            self._call_gc_task.cancel()
+            self._call_gc_task = None
         from . import persist
 
         try:  # final snapshot so a clean stop never loses deployments
@@ -179,6 +183,33 @@ class Scheduler:
         await self.web_gateway.stop()
         await self.pool.stop()
         self._started = False
+
+    #: completed call records are kept this long for late .get()/gather
+    #: (parity: the reference expires spawn results server-side too)
+    CALL_RETENTION_SECONDS = float(os.environ.get("MODAL_AMD_CALL_RETENTION", "3600"))
+
+    def _gc_calls_once(self) -> int:
+        now = time.time()
+        dropped = 0
+        for call_id, rec in list(self.calls.items()):
+            if (
+                rec.done_event.is_set()
+                and rec.finished_at is not None
+                and now - rec.finished_at > self.CALL_RETENTION_SECONDS
+            ):
+                self.calls.pop(call_id, None)
+                dropped += 1
+        return dropped
+
+    async def _call_gc_loop(self) -> None:
+        """Drop long-completed call records so a long-lived daemon does
+        not accumulate them unboundedly."""
+        while True:
+            await asyncio.sleep(60.0)
+            try:
+                self._gc_calls_once()
+            except Exception:
+                pass
 
     async def _persist_loop(self) -> None:
         """Snapshot named/deployed state every 2 s when it changed

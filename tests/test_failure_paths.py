@@ -293,3 +293,33 @@ def test_cancel_interrupts_sync_user_code(client):
         assert quick.remote(21) == 42
         workers_after = {w.task_id for w in client.svc.pool.workers.values()}
         assert workers_before & workers_after, "cancel should not kill workers"
+
+
+def test_completed_call_records_are_garbage_collected(client):
+    """Long-completed CallRecords drop after the retention window so a
+    daemon does not grow unboundedly; recent calls survive for late
+    .get()/gather."""
+    import time
+
+    app = modal.App("gc-app")
+
+    @app.function()
+    def f(x):
+        return x
+
+    with app.run(client=client):
+        svc = client.svc
+        assert f.remote(1) == 1
+        assert sorted(f.map(range(5))) == list(range(5))
+        n_before = len(svc.calls)
+        assert n_before >= 2
+        # age the finished records past retention, then run one GC pass
+        for rec in svc.calls.values():
+            if rec.finished_at is not None:
+                rec.finished_at = time.time() - svc.CALL_RETENTION_SECONDS - 1
+        dropped = svc._gc_calls_once()
+        assert dropped >= 2
+        # fresh calls still work and their records are retained
+        fc = f.spawn(9)
+        assert fc.get(timeout=30) == 9
+        assert svc._gc_calls_once() == 0  # recent: kept
