@@ -199,6 +199,24 @@ class Scheduler:
             ):
                 self.calls.pop(call_id, None)
                 dropped += 1
+        # stopped ephemeral apps: logs deques (10k entries each) are the cost
+        for app_id, app in list(self.apps.items()):
+            if (
+                app.state == "stopped"
+                and now - app.last_heartbeat > self.CALL_RETENTION_SECONDS
+            ):
+                self.apps.pop(app_id, None)
+                dropped += 1
+        # finished sandboxes keep stdio buffers: drop them after retention
+        svc = self.sandbox_service
+        for sb_id, sb in list(svc.sandboxes.items()):
+            if (
+                sb.main.returncode is not None
+                and all(e.returncode is not None for e in sb.execs.values())
+                and now - sb.created_at > self.CALL_RETENTION_SECONDS
+            ):
+                svc.sandboxes.pop(sb_id, None)
+                dropped += 1
         return dropped
 
     async def _call_gc_loop(self) -> None:
