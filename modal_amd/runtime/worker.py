@@ -773,6 +773,9 @@ class WorkerRuntime:
         call_id = msg["call_id"]
         self._chunk_cache[msg["chunk_id"]] = {"raw": msg.get("payload"), "decoded": None}
         self._chunk_order.append(msg["chunk_id"])
+        while len(self._chunk_order) > 512:  # bound the payload cache
+            old = self._chunk_order.pop(0)
+            self._chunk_cache.pop(old, None)
         loop = asyncio.get_running_loop()
 
         values: dict[int, Any] = {}
