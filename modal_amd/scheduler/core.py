@@ -207,6 +207,14 @@ class Scheduler:
             ):
                 self.apps.pop(app_id, None)
                 dropped += 1
+        # function rows of dropped apps hold cloudpickled definitions
+        for fid, fdef in list(self.functions.items()):
+            if fdef.app_id and fdef.app_id not in self.apps:
+                self.functions.pop(fid, None)
+                dropped += 1
+        self.function_names = {
+            k: v for k, v in self.function_names.items() if v in self.functions
+        }
         # finished sandboxes keep stdio buffers: drop them after retention
         svc = self.sandbox_service
         for sb_id, sb in list(svc.sandboxes.items()):
