@@ -108,6 +108,11 @@ class UserCodeProxy:
     async def _ensure(self) -> Connection:
         loop = asyncio.get_running_loop()
         fut = self._conns.get(loop)
+        if fut is not None and fut.done() and not fut.cancelled():
+            conn = fut.result() if fut.exception() is None else None
+            if conn is None or conn.closed:
+                self._conns.pop(loop, None)  # reconnect after scheduler restart
+                fut = None
         if fut is None:
             fut = loop.create_future()
             self._conns[loop] = fut

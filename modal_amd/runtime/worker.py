@@ -137,14 +137,16 @@ class WorkerRPCTarget:
         alternative for enable_memory_snapshot functions: HBM is freed,
         the warm process and its loaded user code survive)."""
         rt = self._runtime
-        if rt.mem_snapshot is not None:
-            return {"state": rt.mem_snapshot.state.value}
-        from .gpu_snapshot import GPUMemorySnapshot
+        async with rt.snapshot_lock:  # RPCs serve concurrently: a restore
+            # racing this page-out must strictly order after it
+            if rt.mem_snapshot is not None:
+                return {"state": rt.mem_snapshot.state.value}
+            from .gpu_snapshot import GPUMemorySnapshot
 
-        snap = GPUMemorySnapshot()
-        await asyncio.get_running_loop().run_in_executor(None, snap.checkpoint)
-        rt.mem_snapshot = snap
-        return {"state": snap.state.value}
+            snap = GPUMemorySnapshot()
+            await asyncio.get_running_loop().run_in_executor(None, snap.checkpoint)
+            rt.mem_snapshot = snap
+            return {"state": snap.state.value}
 
     async def exec_command(self, cmd: list, timeout: float = 60.0) -> dict:
         """Run a command in this worker's context (parity: `modal
@@ -165,10 +167,11 @@ class WorkerRPCTarget:
 
     async def gpu_restore(self) -> dict:
         rt = self._runtime
-        snap, rt.mem_snapshot = rt.mem_snapshot, None
-        if snap is not None:
-            await asyncio.get_running_loop().run_in_executor(None, snap.restore)
-        return {"state": "running"}
+        async with rt.snapshot_lock:
+            snap, rt.mem_snapshot = rt.mem_snapshot, None
+            if snap is not None:
+                await asyncio.get_running_loop().run_in_executor(None, snap.restore)
+            return {"state": "running"}
 
 
 class FunctionRuntime:
@@ -339,6 +342,7 @@ class WorkerRuntime:
         self.task_id: str = ""
         self.functions: dict[str, FunctionRuntime] = {}
         self.mem_snapshot: Any = None  # GPUMemorySnapshot while paged out
+        self.snapshot_lock = asyncio.Lock()  # orders page-out vs restore RPCs
         # token -> executor-thread idents running that input's sync code
         self._sync_threads: dict[str, set] = {}
         self._sync_threads_lock = _threading.Lock()

@@ -139,11 +139,13 @@ class WorkerPool:
                 # (parity: the reference's memory-snapshot cold-start
                 # elimination, gpu_memory_snapshot.py:230-300). Reversible,
                 # so not subject to the warm floor.
+                w.paged = True  # BEFORE the await: a concurrent dispatch
+                # must see it and queue a gpu_restore, which the worker's
+                # snapshot lock orders strictly after this page-out
                 try:
                     await w.conn.call("gpu_snapshot", timeout=120)
-                    w.paged = True
                 except Exception:
-                    pass
+                    w.paged = False
                 continue
             if w.paged:
                 continue  # already costs ~no HBM; keep it warm
