@@ -925,6 +925,16 @@ class Scheduler:
         """Terminate the other ranks of a failed gang member's cluster:
         cancel their in-flight executions and finalize them so the caller
         unblocks (stale ranks would otherwise hang in collectives)."""
+        try:
+            await self._fail_gang_siblings_inner(record, failed, exc_repr)
+        except Exception:
+            logging.getLogger("modal_amd.scheduler").warning(
+                "gang teardown failed", exc_info=True
+            )
+
+    async def _fail_gang_siblings_inner(
+        self, record: CallRecord, failed: InputRecord, exc_repr: Optional[str]
+    ) -> None:
         cluster_id = failed.cluster["cluster_id"]
         siblings = [
             r
