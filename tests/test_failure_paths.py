@@ -323,3 +323,28 @@ def test_completed_call_records_are_garbage_collected(client):
         fc = f.spawn(9)
         assert fc.get(timeout=30) == 9
         assert svc._gc_calls_once() == 0  # recent: kept
+
+
+def test_map_spreads_across_workers(client):
+    """Chunk dispatch load-balances across the pool (the 8-GPU scaling
+    premise: no worker starves while others queue)."""
+    from modal_amd._sync import synchronizer
+
+    async def spawn_more():
+        for _ in range(3):
+            await client.svc.pool.spawn_worker()
+
+    synchronizer.run(spawn_more())
+    app = modal.App("spread-app")
+
+    @app.function()
+    def work(x):
+        import os
+        import time
+
+        time.sleep(0.001)  # long enough that one worker cannot take all
+        return os.getpid()
+
+    with app.run(client=client):
+        pids = set(work.map(range(2000)))
+        assert len(pids) >= 3, f"work concentrated on {len(pids)} worker(s)"
