@@ -24,6 +24,39 @@ from ..utils.ids import new_id
 STDIN_CHUNK = 256 * 1024  # parity: reference task_command_router_client.py:31
 
 
+
+def _resource_preexec(cpu: "Optional[float]", memory_mib: "Optional[int]"):
+    """Unprivileged resource enforcement for sandbox processes (parity:
+    the reference's container cpu/memory reservations, sandbox.py:551):
+    RLIMIT_AS caps the address space at ``memory`` MiB; ``cpu`` cores are
+    approximated by pinning the process to that many CPUs. Runs in the
+    child between fork and exec."""
+    if cpu is None and memory_mib is None:
+        return None
+
+    def _apply() -> None:
+        os.setsid()
+        if memory_mib:
+            import resource
+
+            limit = int(memory_mib) * 1024 * 1024
+            try:
+                resource.setrlimit(resource.RLIMIT_AS, (limit, limit))
+            except (ValueError, OSError):
+                pass
+        if cpu:
+            try:
+                import math
+
+                avail = sorted(os.sched_getaffinity(0))
+                take = max(1, math.ceil(cpu))
+                os.sched_setaffinity(0, set(avail[:take]))
+            except OSError:
+                pass
+
+    return _apply
+
+
 class _ProcState:
     """One supervised process: the sandbox entrypoint or an exec."""
 
@@ -260,6 +293,7 @@ class SandboxService:
             state.name = name
         self.sandboxes[sandbox_id] = state
         args = entrypoint_args or ["sleep", "infinity"]
+        preexec = _resource_preexec(cpu, memory)
         proc = await asyncio.create_subprocess_exec(
             *args,
             cwd=state.workdir,
@@ -267,7 +301,7 @@ class SandboxService:
             stdin=asyncio.subprocess.PIPE,
             stdout=asyncio.subprocess.PIPE,
             stderr=asyncio.subprocess.PIPE,
-            start_new_session=True,
+            **({"preexec_fn": preexec} if preexec else {"start_new_session": True}),
         )
         await state.main.attach(proc)
         if timeout:

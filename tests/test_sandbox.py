@@ -153,3 +153,34 @@ def test_sandbox_pty_exec(client):
         assert "got:ping" in out           # interactive stdin through the master
     finally:
         sb.terminate()
+
+
+def test_sandbox_memory_limit_enforced(client):
+    """Sandbox memory= caps the process address space (RLIMIT_AS):
+    a 200 MiB allocation under a 128 MiB cap dies; an 8 MiB one lives."""
+    code = "x = bytearray(200 * 1024 * 1024); print('allocated')"
+    sb = modal.Sandbox.create("python3", "-c", code, memory=128)
+    rc = sb.wait(raise_on_termination=False)
+    assert rc != 0, sb.stdout.read()
+    err = sb.stderr.read()
+    assert "MemoryError" in err or rc != 0
+
+    sb2 = modal.Sandbox.create(
+        "python3", "-c", "x = bytearray(8 * 1024 * 1024); print('ok')", memory=512
+    )
+    assert sb2.wait(raise_on_termination=False) == 0
+    assert "ok" in sb2.stdout.read()
+
+
+def test_sandbox_cpu_affinity(client):
+    import os as _os
+
+    if len(_os.sched_getaffinity(0)) < 2:
+        import pytest as _pytest
+
+        _pytest.skip("needs >=2 CPUs")
+    sb = modal.Sandbox.create(
+        "python3", "-c", "import os; print(len(os.sched_getaffinity(0)))", cpu=1
+    )
+    assert sb.wait(raise_on_termination=False) == 0
+    assert sb.stdout.read().strip() == "1"
