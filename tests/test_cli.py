@@ -179,3 +179,12 @@ def test_cli_app_logs(runner, client):
         result = runner.invoke(entrypoint_cli, ["app", "logs", app.app_id])
     assert result.exit_code == 0, result.output
     assert "log-line-sentinel" in result.output
+
+
+def test_cli_environment_and_profile(runner, client):
+    result = runner.invoke(entrypoint_cli, ["environment", "list"])
+    assert result.exit_code == 0, result.output
+    result = runner.invoke(entrypoint_cli, ["profile", "current"])
+    assert result.exit_code == 0, result.output
+    result = runner.invoke(entrypoint_cli, ["config", "show"])
+    assert result.exit_code == 0, result.output
