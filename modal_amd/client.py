@@ -52,9 +52,11 @@ def map_remote_error(exc: RemoteRPCError) -> Exception:
 class SchedulerProxy:
     """Socket-backed scheduler: async methods forwarded as RPC frames."""
 
-    #: class attribute (normal lookup beats __getattr__): lets callers
-    #: distinguish the proxy from the in-process Scheduler
+    #: class attributes (normal lookup beats __getattr__): distinguish the
+    #: proxy from the in-process Scheduler, and keep attribute probes like
+    #: client.blob_store from resolving to RPC stubs
     is_proxy = True
+    blob_store = None
 
     def __init__(self, conn: Connection):
         self._conn = conn
@@ -86,6 +88,7 @@ class UserCodeProxy:
     """
 
     is_proxy = True
+    blob_store = None  # resolved via run_dir (shared filesystem), not RPC
 
     def __init__(self, socket_path: str):
         self._socket_path = socket_path
@@ -161,7 +164,7 @@ class _Client:
         """The shared content-addressed store (same filesystem on all sides)."""
         if getattr(self, "_blob_store", None) is None:
             svc = self.svc
-            if hasattr(svc, "blob_store"):
+            if getattr(svc, "blob_store", None) is not None:
                 self._blob_store = svc.blob_store
             else:
                 import os as _os

@@ -410,7 +410,10 @@ class WorkerRuntime:
         _synchronizer.thread_local_blocking = os.environ.get(
             "MODAL_AMD_TL_BLOCKING", "1"
         ) not in ("0", "false")
-        proxy_client = _Client(client_mod.UserCodeProxy(self.socket_path), "container")
+        proxy_client = _Client(
+            client_mod.UserCodeProxy(self.socket_path), "container",
+            run_dir=os.path.dirname(self.socket_path),  # shared CAS for big args
+        )
         _Client.set_default(proxy_client)
         set_client_context(proxy_client)
 

@@ -271,3 +271,31 @@ def test_async_function_uses_handles_aio(client):
     with app.run(client=client):
         out = async_orchestrator.remote(7)
     assert out == {"dict": 7, "unary": 21, "map_sum": 0 + 3 + 6}
+
+
+def test_nested_big_args_spill_to_cas(client):
+    """A worker passing a >2 MiB argument to another function spills it
+    through the shared CAS instead of crashing on the proxy client
+    (client.blob_store must be a real store inside workers)."""
+    import hashlib
+
+    app = modal.App("bigarg-app")
+
+    @app.function()
+    def digest(blob: bytes) -> str:
+        import hashlib as h
+
+        return h.sha256(blob).hexdigest()
+
+    @app.function()
+    def orchestrate(n: int) -> str:
+        import os as _os
+
+        blob = _os.urandom(n)
+        import hashlib as h
+
+        assert digest.remote(blob) == h.sha256(blob).hexdigest()
+        return "ok"
+
+    with app.run(client=client):
+        assert orchestrate.remote(3 * 1024 * 1024) == "ok"
