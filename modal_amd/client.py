@@ -57,6 +57,7 @@ class SchedulerProxy:
     #: client.blob_store from resolving to RPC stubs
     is_proxy = True
     blob_store = None
+    run_dir = None
 
     def __init__(self, conn: Connection):
         self._conn = conn
@@ -89,6 +90,7 @@ class UserCodeProxy:
 
     is_proxy = True
     blob_store = None  # resolved via run_dir (shared filesystem), not RPC
+    run_dir = None
 
     def __init__(self, socket_path: str):
         self._socket_path = socket_path
