@@ -188,3 +188,21 @@ def test_cli_environment_and_profile(runner, client):
     assert result.exit_code == 0, result.output
     result = runner.invoke(entrypoint_cli, ["config", "show"])
     assert result.exit_code == 0, result.output
+
+
+def test_cli_curl_and_launch(runner, client):
+    import modal_amd as modal
+
+    app = modal.App("curl-app")
+
+    @app.function()
+    @modal.fastapi_endpoint()
+    def hello_curl():
+        return {"via": "curl"}
+
+    with app.run(client=client):
+        result = runner.invoke(entrypoint_cli, ["curl", hello_curl.web_url])
+        assert result.exit_code == 0, result.output
+        assert "curl" in result.output
+    result = runner.invoke(entrypoint_cli, ["launch", "--help"])
+    assert result.exit_code == 0
