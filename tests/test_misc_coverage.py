@@ -227,3 +227,31 @@ def test_tunnel_forward_and_billing(client, capsys):
     rows = usage_summary(client)
     mine = [r for r in rows if r["function"] == "unit"]
     assert mine and mine[0]["inputs"] >= 20
+
+
+def test_iteration_apis(client):
+    """Dict keys/values/items, Queue.iterate, Volume.iterdir/listdir."""
+    import io
+
+    import modal_amd as modal
+
+    d = modal.Dict.from_name("iter-d", create_if_missing=True)
+    d.put("a", 1)
+    d.put("b", 2)
+    assert sorted(d.keys()) == ["a", "b"]
+    assert sorted(d.values()) == [1, 2]
+    assert sorted(d.items()) == [("a", 1), ("b", 2)]
+    assert len(d) == 2 and "a" in d
+
+    with modal.Queue.ephemeral() as q:
+        q.put_many([10, 20, 30])
+        assert list(q.iterate(item_poll_timeout=0.2)) == [10, 20, 30]
+
+    vol = modal.Volume.from_name("iter-v", create_if_missing=True)
+    with vol.batch_upload(force=True) as b:
+        b.put_file(io.BytesIO(b"x"), "/d1/f1.txt")
+        b.put_file(io.BytesIO(b"y"), "/d1/f2.txt")
+    names = [e.path for e in vol.listdir("/d1")]
+    assert sorted(names) == ["d1/f1.txt", "d1/f2.txt"]
+    all_entries = [e.path for e in vol.iterdir("/", recursive=True)]
+    assert "d1/f1.txt" in all_entries
