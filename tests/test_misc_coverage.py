@@ -255,3 +255,26 @@ def test_iteration_apis(client):
     assert sorted(names) == ["d1/f1.txt", "d1/f2.txt"]
     all_entries = [e.path for e in vol.iterdir("/", recursive=True)]
     assert "d1/f1.txt" in all_entries
+
+
+def test_execution_context_helpers(client):
+    """modal.is_local / current_input_id / current_function_call_id."""
+    import modal_amd as modal
+
+    assert modal.is_local() is True
+
+    app = modal.App("ctx-app")
+
+    @app.function()
+    def introspect():
+        return {
+            "local": modal.is_local(),
+            "input_id": modal.current_input_id(),
+            "call_id": modal.current_function_call_id(),
+        }
+
+    with app.run(client=client):
+        out = introspect.remote()
+    assert out["local"] is False
+    assert out["input_id"] and out["input_id"].startswith("in-")
+    assert out["call_id"] and out["call_id"].startswith("fc-")
