@@ -174,6 +174,11 @@ class _Image(_Object, type_kind="image"):
     def cmd(self, cmd: list[str]) -> "_Image":
         return self._extend({"kind": "cmd", "args": list(cmd)})
 
+    def shell(self, shell_commands: list[str]) -> "_Image":
+        """Overwrite the default shell used by later run_commands
+        (parity: reference _image.py:1990)."""
+        return self._extend({"kind": "shell", "args": list(shell_commands)})
+
     def add_local_file(self, local_path: Union[str, os.PathLike], remote_path: str, *, copy: bool = False) -> "_Image":
         import hashlib
 

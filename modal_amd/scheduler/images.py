@@ -95,15 +95,23 @@ class ImageService:
         log: list[str] = []
         site = os.path.join(state.root, "site-packages")
 
+        shell_argv: list = []  # Image.shell() override (SHELL json-array form)
+
         def run_shell(cmd: str, env: Optional[dict] = None) -> None:
             full_env = dict(os.environ)
             full_env.update(state.env)
             if env:
                 full_env.update(env)
-            proc = subprocess.run(
-                cmd, shell=True, cwd=state.root, env=full_env,
-                capture_output=True, text=True, timeout=600,
-            )
+            if shell_argv:
+                proc = subprocess.run(
+                    [*shell_argv, cmd], cwd=state.root, env=full_env,
+                    capture_output=True, text=True, timeout=600,
+                )
+            else:
+                proc = subprocess.run(
+                    cmd, shell=True, cwd=state.root, env=full_env,
+                    capture_output=True, text=True, timeout=600,
+                )
             log.append(f"$ {cmd}\n{proc.stdout}{proc.stderr}")
             if proc.returncode != 0:
                 raise ExecutionError(f"Image build step failed ({cmd!r}):\n{proc.stderr[-2000:]}")
@@ -121,6 +129,8 @@ class ImageService:
                 state.entrypoint = list(layer.get("args", []))
             elif kind == "cmd":
                 state.cmd = list(layer.get("args", []))
+            elif kind == "shell":
+                shell_argv[:] = layer.get("args", [])
             elif kind == "run_commands":
                 for cmd in layer.get("commands", []):
                     await loop.run_in_executor(None, run_shell, cmd)

@@ -174,3 +174,16 @@ def test_exit_hook_writes_land_before_volume_commit(client):
         assert Svc().bump.remote() == 1
     v2 = modal.Volume.from_name("exit-vol")
     assert b"".join(v2.read_file("final.txt")) == b"count=1"
+
+
+def test_image_shell_override(client, tmp_path):
+    """Image.shell() sets the SHELL for later run_commands (parity:
+    reference _image.py:1990)."""
+    marker = tmp_path / "shellname.txt"
+    img = (
+        modal.Image.debian_slim()
+        .shell(["/bin/bash", "-c"])
+        .run_commands(f"echo $0 > {marker}")
+    )
+    img.hydrate()
+    assert "bash" in marker.read_text()
