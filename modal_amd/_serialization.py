@@ -103,7 +103,9 @@ class Pickler(cloudpickle.CloudPickler):
                 )
             return ("modal-amd-object", obj.object_id, obj._get_metadata())
         tensor_cls = _torch_tensor_cls()
-        if tensor_cls is not None and type(obj) is tensor_cls:
+        # isinstance, not exact type: nn.Parameter and other tensor
+        # subclasses must ride the same device-export hooks
+        if tensor_cls is not None and isinstance(obj, tensor_cls):
             import os as _os
             import sys as _sys
 
@@ -190,15 +192,30 @@ def serialize(obj: Any) -> bytes:
     return buf.getvalue()
 
 
+_FAST_UNSAFE_MARKERS = (b"torch", b"modal_amd")
+
+
 def serialize_fast(obj: Any) -> bytes:
     """Hot-path serialize: try the C pickler first (no hooks), fall back to
-    the full cloudpickle path. Safe because live handles and closures make
-    plain pickle raise (their graphs contain the scheduler/locks), which
-    routes them to the hook-aware pickler."""
+    the full cloudpickle path when the payload needs the hooks.
+
+    Fallback triggers two ways: (a) plain pickle raises (closures, live
+    handles whose graphs contain the scheduler/locks), or (b) the pickle
+    stream references a ``torch.*`` / ``modal_amd.*`` global — any tensor
+    (including subclasses like nn.Parameter, and tensors nested inside user
+    objects) that plain pickle manages to serialize necessarily emits such a
+    global to rebuild itself, so a byte scan is an exact detector at C-scan
+    speed. Those payloads are re-serialized with the hook-aware pickler so
+    device tensors export through the mesh instead of baking in the
+    producer's device index."""
     try:
-        return pickle.dumps(obj, PICKLE_PROTOCOL)
+        data = pickle.dumps(obj, PICKLE_PROTOCOL)
     except Exception:
         return serialize(obj)
+    for marker in _FAST_UNSAFE_MARKERS:
+        if marker in data:
+            return serialize(obj)
+    return data
 
 
 def deserialize(data: bytes) -> Any:

@@ -129,7 +129,11 @@ class UserCodeProxy:
 
                 conn = Connection(reader, writer, handler)
                 conn.start()
-                await conn.send({"t": "hello", "role": "client"})
+                from .scheduler.core import read_auth_token
+
+                await conn.send(
+                    {"t": "hello", "role": "client", "auth": read_auth_token(self._socket_path)}
+                )
                 fut.set_result(conn)
             except BaseException as exc:
                 self._conns.pop(loop, None)
@@ -219,7 +223,11 @@ class _Client:
 
         conn = Connection(reader, writer, handler)
         conn.start()
-        await conn.send({"t": "hello", "role": "client"})
+        from .scheduler.core import read_auth_token
+
+        await conn.send(
+            {"t": "hello", "role": "client", "auth": read_auth_token(socket_path)}
+        )
         return cls(SchedulerProxy(conn), client_type, run_dir=os.path.dirname(socket_path))
 
     @classmethod

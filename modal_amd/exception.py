@@ -116,9 +116,18 @@ class ModuleNotMountable(Error):
     """A Python module could not be packaged for a worker."""
 
 
-class QueueEmptyError(Error):
-    """Non-blocking Queue.get on an empty queue."""
+import queue as _stdlib_queue
 
 
-class QueueFullError(Error):
-    """Queue partition is at capacity (5,000 items per partition)."""
+class QueueEmptyError(Error, _stdlib_queue.Empty):
+    """Non-blocking Queue.get on an empty queue.
+
+    Subclasses stdlib ``queue.Empty`` for drop-in parity: the reference
+    raises ``queue.Empty`` (reference queue.py), so user code ported from it
+    catching ``queue.Empty`` keeps working."""
+
+
+class QueueFullError(Error, _stdlib_queue.Full):
+    """Queue partition is at capacity (5,000 items per partition).
+
+    Subclasses stdlib ``queue.Full`` for drop-in parity with the reference."""

@@ -51,21 +51,13 @@ class FunctionCallCancelledError(Error):
 
 
 def make_payload_item(client: Any, args: tuple, kwargs: dict) -> dict:
-    """Serialize one input; offload to the CAS above the inline limit."""
-    import sys as _sys
+    """Serialize one input; offload to the CAS above the inline limit.
 
+    serialize_fast routes any payload touching torch/modal_amd globals (incl.
+    tensors nested in user objects) through the hook-aware pickler itself, so
+    no pre-scan is needed here."""
     from ._serialization import serialize_fast
 
-    if "torch" in _sys.modules:
-        # tensors must go through the hook-aware pickler (mesh export /
-        # host staging) — plain pickle would bake in the client's device index
-        from ._serialization import contains_tensors
-
-        if contains_tensors(args, kwargs):
-            payload = serialize(("P", (args, kwargs)))
-            if len(payload) > INLINE_LIMIT and client.blob_store is not None:
-                return {"payload": b"", "payload_blob": client.blob_store.put(payload)}
-            return {"payload": payload}
     payload = serialize_fast(("P", (args, kwargs)))
     if len(payload) > INLINE_LIMIT:
         store = client.blob_store
@@ -162,7 +154,7 @@ async def await_output_item(
             function_call_id=call_id, max_values=16, timeout=remaining
         )
         for out in outs:
-            if out["idx"] == idx:
+            if out.get("idx") == idx:
                 return out
 
 

@@ -98,16 +98,9 @@ async def map_invocation(
         chunk_buf: list = []
         chunk_seq = 0
 
-        import sys as _sys
-
         def _chunk_serialize(buf: list) -> bytes:
-            # tensors need the hook-aware pickler (device staging / mesh export)
-            if "torch" in _sys.modules:
-                from .._serialization import _walk_for_tensors, serialize
-
-                tensor_cls = _sys.modules["torch"].Tensor
-                if _walk_for_tensors(buf, tensor_cls):
-                    return serialize(("C", buf))
+            # serialize_fast detects tensors/handles in the stream itself and
+            # reroutes to the hook-aware pickler (device staging / mesh export)
             return serialize_fast(("C", buf))
 
         async def flush_chunk() -> None:
@@ -239,11 +232,15 @@ async def map_invocation(
                     values = _pickle.loads(out["chunk_data"])
                     cis = out["cis"]
                     base = out["idx_base"]
+                    # val_off: the position in `values` of cis[0] — nonzero
+                    # when the scheduler split a group to honor max_values
+                    voff = out.get("val_off", 0)
                     pairs = (
-                        enumerate(values) if cis is None else zip(cis, values)
+                        enumerate(values) if cis is None else zip(cis, values[voff:])
                     )
-                    received += len(values)
-                    sem.release(len(values))
+                    n_here = len(values) if cis is None else len(cis)
+                    received += n_here
+                    sem.release(n_here)
                     if order_outputs:
                         for ci, value in pairs:
                             ordering_buffer[base + ci] = value

@@ -387,6 +387,8 @@ class WorkerRuntime:
                 await asyncio.sleep(0.05)
         self.conn = Connection(reader, writer, self._handle, rpc_target=WorkerRPCTarget(self))
         self.conn.start()
+        from ..scheduler.core import read_auth_token
+
         await self.conn.send(
             {
                 "t": "hello",
@@ -395,6 +397,7 @@ class WorkerRuntime:
                 "gpu_index": self.gpu_index,
                 "external": self.external,
                 "pid": os.getpid(),
+                "auth": read_auth_token(self.socket_path),
             }
         )
         # install the process-default client so user code's handles bind here

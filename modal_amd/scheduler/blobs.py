@@ -41,23 +41,31 @@ class BlobStore:
     def _path(self, digest: str) -> str:
         return os.path.join(self.root, digest[:2], digest)
 
+    def _zpath(self, digest: str) -> str:
+        # the compressed/raw bit lives in the FILENAME (".z" suffix), never in
+        # the payload: raw content that happens to begin with the MALZ41 magic
+        # round-trips untouched (advisor finding, round 1)
+        return os.path.join(self.root, digest[:2], digest + ".z")
+
     COMPRESS_MIN = 1024 * 1024  # GPU (de)compression pays above ~1 MiB
 
     def _store(self, digest: str, data: Union[bytes, memoryview]) -> None:
-        if os.path.exists(self._path(digest)):
+        if self.has(digest):
             return
         payload = bytes(data)
+        compressed = False
         if len(payload) >= self.COMPRESS_MIN:
             # HIP LZ4 kernels; digest stays the digest of the RAW content
             try:
                 from ..ops.compress import compress_buffer
 
-                compressed = compress_buffer(payload)
-                if compressed is not None:
-                    payload = compressed
+                z = compress_buffer(payload)
+                if z is not None:
+                    payload = z
+                    compressed = True
             except Exception:
                 pass
-        self._store_prepared(digest, payload)
+        self._store_prepared(digest, payload, compressed)
 
     def put(self, data: Union[bytes, memoryview]) -> str:
         digest = _hash_bytes(data)
@@ -88,7 +96,7 @@ class BlobStore:
         # compress only blocks that are big enough and not already stored
         todo = [
             i for i, (digest, data) in enumerate(zip(digests, buffers))
-            if len(data) >= self.COMPRESS_MIN and not os.path.exists(self._path(digest))
+            if len(data) >= self.COMPRESS_MIN and not self.has(digest)
         ]
         compressed: dict[int, bytes] = {}
         if todo:
@@ -107,12 +115,17 @@ class BlobStore:
             except Exception:
                 pass  # store raw on any kernel/driver hiccup
         for i, (digest, data) in enumerate(zip(digests, buffers)):
-            self._store_prepared(digest, compressed.get(i, bytes(data)))
+            if i in compressed:
+                self._store_prepared(digest, compressed[i], True)
+            else:
+                self._store_prepared(digest, bytes(data), False)
         return digests
 
-    def _store_prepared(self, digest: str, payload: bytes) -> None:
-        """Store an already-encoded payload (raw or MALZ41) under digest."""
-        path = self._path(digest)
+    def _store_prepared(self, digest: str, payload: bytes, compressed: bool = False) -> None:
+        """Store an already-encoded payload under digest; ``compressed`` says
+        whether it is an LZ4 container (recorded as a ``.z`` filename suffix,
+        out-of-band of the bytes)."""
+        path = self._zpath(digest) if compressed else self._path(digest)
         if os.path.exists(path):
             return
         os.makedirs(os.path.dirname(path), exist_ok=True)
@@ -129,14 +142,27 @@ class BlobStore:
             raise
 
     def put_file(self, src_path: str) -> str:
-        h = hashlib.sha256()
-        with open(src_path, "rb") as f:
-            while True:
-                chunk = f.read(1 << 20)
-                if not chunk:
-                    break
-                h.update(chunk)
-        digest = h.hexdigest()
+        # Same CAS key as put(): plain SHA-256 below the GPU threshold, the
+        # domain-separated tree digest above (advisor finding: the two paths
+        # previously diverged, breaking dedup for identical content).
+        from ..ops.hashing import GPU_MIN_BYTES, content_digest
+
+        size = os.path.getsize(src_path)
+        if size < GPU_MIN_BYTES:
+            h = hashlib.sha256()
+            with open(src_path, "rb") as f:
+                while True:
+                    chunk = f.read(1 << 20)
+                    if not chunk:
+                        break
+                    h.update(chunk)
+            digest = h.hexdigest()
+        else:
+            import mmap
+
+            with open(src_path, "rb") as f:
+                with mmap.mmap(f.fileno(), 0, access=mmap.ACCESS_READ) as mm:
+                    digest = content_digest(memoryview(mm))
         path = self._path(digest)
         if not os.path.exists(path):
             os.makedirs(os.path.dirname(path), exist_ok=True)
@@ -149,28 +175,39 @@ class BlobStore:
         return digest
 
     def get(self, digest: str) -> bytes:
-        with open(self._path(digest), "rb") as f:
+        path = self._path(digest)
+        if os.path.exists(path):
+            with open(path, "rb") as f:
+                return f.read()
+        zpath = self._zpath(digest)
+        with open(zpath, "rb") as f:
             blob = f.read()
-        if blob.startswith(b"MALZ41"):
-            from ..ops.compress import decompress_buffer
+        from ..ops.compress import decompress_buffer
 
-            return decompress_buffer(blob)
-        return blob
+        return decompress_buffer(blob)
 
     def open_path(self, digest: str) -> str:
+        """Path to the RAW content on disk (consumers sendfile/mmap it).
+
+        A compressed-only blob is decompressed once into the raw path and
+        cached there; subsequent calls hit the raw file directly."""
         path = self._path(digest)
-        if not os.path.exists(path):
+        if os.path.exists(path):
+            return path
+        zpath = self._zpath(digest)
+        if not os.path.exists(zpath):
             raise FileNotFoundError(f"blob {digest} not in store")
+        self._store_prepared(digest, self.get(digest), False)
         return path
 
     def materialize(self, digest: str, dest: str) -> None:
         """Produce the RAW content at dest: hard-link when the stored form is
-        raw, write a decompressed copy when it is a MALZ41 container."""
-        src = self.open_path(digest)
-        with open(src, "rb") as f:
-            head = f.read(6)
+        raw, write a decompressed copy when only the ``.z`` form exists."""
         os.makedirs(os.path.dirname(dest), exist_ok=True)
-        if head == b"MALZ41":
+        src = self._path(digest)
+        if not os.path.exists(src):
+            if not os.path.exists(self._zpath(digest)):
+                raise FileNotFoundError(f"blob {digest} not in store")
             data = self.get(digest)
             with open(dest, "wb") as f:
                 f.write(data)
@@ -185,7 +222,11 @@ class BlobStore:
             shutil.copyfile(src, dest)
 
     def has(self, digest: str) -> bool:
-        return os.path.exists(self._path(digest))
+        return os.path.exists(self._path(digest)) or os.path.exists(self._zpath(digest))
 
     def size(self, digest: str) -> int:
-        return os.stat(self._path(digest)).st_size
+        """On-disk (stored) size — compressed size for ``.z`` blobs."""
+        path = self._path(digest)
+        if os.path.exists(path):
+            return os.stat(path).st_size
+        return os.stat(self._zpath(digest)).st_size

@@ -13,6 +13,7 @@ from __future__ import annotations
 
 import asyncio
 import time
+from collections import deque
 from dataclasses import dataclass, field
 from typing import Any, Optional
 
@@ -209,6 +210,9 @@ class CallRecord:
         self.finished_at: Optional[float] = None
         # completion-order queue of idx for streaming GetOutputs
         self.output_ready: asyncio.Queue[int] = asyncio.Queue()
+        # entries split/deferred by a bounded GetOutputs response; consumed
+        # before the queue so completion order is preserved
+        self.output_pushback: deque = deque()
         # per-input completion events for unary waits
         self._waiters: dict[int, asyncio.Future] = {}
         # generator data-out channel, per input idx

@@ -41,6 +41,37 @@ from .workerhost import WorkerPool
 OUTPUT_POLL_TIMEOUT = 55.0  # parity: function_utils.py:498 (55 s long-poll)
 
 
+def _load_or_create_auth_token(run_dir: str) -> str:
+    """Socket-handshake secret, stored 0600 inside the (0700) run_dir.
+
+    Possessing the token == being able to read the run_dir, so the Unix
+    socket no longer grants the RPC surface to arbitrary local processes."""
+    path = os.path.join(run_dir, "auth.token")
+    try:
+        with open(path) as f:
+            token = f.read().strip()
+        if token:
+            return token
+    except OSError:
+        pass
+    import secrets
+
+    token = secrets.token_hex(16)
+    fd = os.open(path, os.O_WRONLY | os.O_CREAT | os.O_TRUNC, 0o600)
+    with os.fdopen(fd, "w") as f:
+        f.write(token)
+    return token
+
+
+def read_auth_token(socket_path: str) -> str:
+    """Client-side: the token lives next to the socket."""
+    try:
+        with open(os.path.join(os.path.dirname(socket_path), "auth.token")) as f:
+            return f.read().strip()
+    except OSError:
+        return ""
+
+
 class AppState:
     def __init__(self, app_id: str, description: str, ephemeral: bool, environment: str):
         self.app_id = app_id
@@ -100,6 +131,14 @@ class Scheduler:
 
         self.run_dir = run_dir or config.get("run_dir") or tempfile.mkdtemp(prefix="modal-amd-")
         os.makedirs(self.run_dir, exist_ok=True)
+        try:
+            os.chmod(self.run_dir, 0o700)  # the socket grants full control
+        except OSError:
+            pass
+        # connection handshake token: any process that can READ the run_dir
+        # may use the control socket; anything else is rejected at hello
+        # (round-1 review: the socket was an unauthenticated control plane)
+        self.auth_token = _load_or_create_auth_token(self.run_dir)
         self.apps: dict[str, AppState] = {}
         self.app_names: dict[tuple[str, str], str] = {}  # (env, name) -> app_id
         self.functions: dict[str, FunctionDef] = {}
@@ -737,39 +776,82 @@ class Scheduler:
         (parity: FunctionGetOutputs long-poll, reference _functions.py:224-263)."""
         record = self._call(function_call_id)
         out: list[dict] = []
+        n_vals = 0  # TRUE number of outputs in this response (groups expanded)
 
         def expand(entry: Any) -> None:
+            # honest max_values: a chunk-group entry counts as its item count;
+            # a group that does not fit is SPLIT — the head ships now, the
+            # tail goes to output_pushback for the next poll
+            nonlocal n_vals
             if type(entry) is tuple and entry and entry[0] == "g":
-                _tag, base, count, out_chunk, cis = entry
+                _tag, base, count, out_chunk, cis = entry[:5]
+                val_off = entry[5] if len(entry) > 5 else 0
+                n = count if cis is None else len(cis)
+                budget = max_values - n_vals
+                if n > budget:
+                    cis_full = list(range(count)) if cis is None else cis
+                    head, tail = cis_full[:budget], cis_full[budget:]
+                    if head:
+                        out.append(
+                            {
+                                "group": True,
+                                "idx_base": base,
+                                "count": count,
+                                "cis": head,
+                                "val_off": val_off,
+                                "out_chunk": out_chunk,
+                                "status": 1,
+                            }
+                        )
+                        n_vals += len(head)
+                    record.output_pushback.append(
+                        ("g", base, count, out_chunk, tail, val_off + len(head))
+                    )
+                    return
                 out.append(
                     {
                         "group": True,
                         "idx_base": base,
                         "count": count,
                         "cis": cis,
+                        "val_off": val_off,
                         "out_chunk": out_chunk,
                         "status": 1,
                     }
                 )
+                n_vals += n
             elif isinstance(entry, list):
-                for idx in entry:
+                for pos, idx in enumerate(entry):
+                    if n_vals >= max_values:
+                        record.output_pushback.append(entry[pos:])
+                        return
                     out.append(self._output_item(record, idx))
+                    n_vals += 1
             else:
                 out.append(self._output_item(record, entry))
+                n_vals += 1
+
+        def take_nowait() -> Any:
+            if record.output_pushback:
+                return record.output_pushback.popleft()
+            return record.output_ready.get_nowait()
 
         deadline = time.time() + timeout
         while not out:
-            try:
-                remaining = deadline - time.time()
-                if remaining <= 0:
-                    break
-                entry = await asyncio.wait_for(record.output_ready.get(), remaining)
-            except asyncio.TimeoutError:
-                break
-            expand(entry)
-            while len(out) < max_values:
+            if record.output_pushback:
+                entry = record.output_pushback.popleft()
+            else:
                 try:
-                    entry = record.output_ready.get_nowait()
+                    remaining = deadline - time.time()
+                    if remaining <= 0:
+                        break
+                    entry = await asyncio.wait_for(record.output_ready.get(), remaining)
+                except asyncio.TimeoutError:
+                    break
+            expand(entry)
+            while n_vals < max_values:
+                try:
+                    entry = take_nowait()
                 except asyncio.QueueEmpty:
                     break
                 expand(entry)

@@ -38,14 +38,24 @@ def _match_field(expr: str, value: int, base: int = 0) -> bool:
 
 
 def cron_matches(cron_string: str, dt: datetime) -> bool:
-    """Standard 5-field cron: minute hour day-of-month month day-of-week."""
+    """Standard 5-field cron: minute hour day-of-month month day-of-week.
+
+    POSIX day semantics: when BOTH day-of-month and day-of-week are
+    restricted (neither is ``*``), the date matches if EITHER matches —
+    ``0 0 1,15 * 1`` fires on the 1st, the 15th, and every Monday."""
     minute, hour, dom, month, dow = cron_string.split()
+    dow_value = dt.weekday() + 1 if dt.weekday() < 6 else 0  # 0=Sunday
+    dom_star = dom.strip() == "*"
+    dow_star = dow.strip() == "*"
+    if dom_star or dow_star:
+        day_ok = _match_field(dom, dt.day, base=1) and _match_field(dow, dow_value)
+    else:
+        day_ok = _match_field(dom, dt.day, base=1) or _match_field(dow, dow_value)
     return (
         _match_field(minute, dt.minute)
         and _match_field(hour, dt.hour)
-        and _match_field(dom, dt.day, base=1)
         and _match_field(month, dt.month, base=1)
-        and _match_field(dow, dt.weekday() + 1 if dt.weekday() < 6 else 0)  # 0=Sunday
+        and day_ok
     )
 
 

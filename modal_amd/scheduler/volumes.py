@@ -138,12 +138,10 @@ class VolumeService:
             def _concat() -> None:
                 with open(dest, "wb") as f:
                     for digest in block_digests:
+                        # open_path always yields the RAW form (decompressing
+                        # a .z-stored blob once, cached); sendfile from there
                         src = self.blob_store.open_path(digest)
                         with open(src, "rb") as bf:
-                            if bf.read(6) == b"MALZ41":
-                                f.write(self.blob_store.get(digest))  # GPU decompress
-                                continue
-                            bf.seek(0)
                             n = os.fstat(bf.fileno()).st_size
                             off = 0
                             while off < n:

@@ -106,6 +106,10 @@ class WorkerPool:
     async def start(self) -> None:
         os.makedirs(self.run_dir, exist_ok=True)
         self._server = await asyncio.start_unix_server(self._on_connect, path=self.socket_path)
+        try:
+            os.chmod(self.socket_path, 0o600)
+        except OSError:
+            pass
         self._dispatch_task = asyncio.get_running_loop().create_task(self._dispatch_loop())
         self._retry_task = asyncio.get_running_loop().create_task(self._retry_loop())
         self._health_task = asyncio.get_running_loop().create_task(self._health_loop())
@@ -325,8 +329,19 @@ class WorkerPool:
 
         handle_holder: dict = {}
 
+        authed = False
+
         async def handler(msg: dict) -> None:
+            nonlocal authed
             kind = msg.get("t")
+            if not authed:
+                # handshake gate: nothing (RPCs included) is served until a
+                # hello carrying the run_dir token arrives
+                if kind != "hello" or msg.get("auth") != self.scheduler.auth_token:
+                    await conn.close()
+                    return
+                authed = True
+                conn.rpc_target = self.scheduler.rpc_adapter
             if kind == "hello":
                 role = msg.get("role", "worker")
                 if role == "worker":
@@ -384,7 +399,7 @@ class WorkerPool:
             elif kind == "log":
                 self.scheduler.on_worker_log(handle, msg)
 
-        conn = Connection(reader, writer, handler, rpc_target=self.scheduler.rpc_adapter)
+        conn = Connection(reader, writer, handler, rpc_target=None)  # set after auth
         conn.start()
 
     async def _watch_worker(self, handle: WorkerHandle) -> None:

@@ -51,7 +51,7 @@ def test_cas_transparent_compression(tmp_path, monkeypatch):
     monkeypatch.setattr("modal_amd.ops.compress.compress_buffer", C.compress_buffer_cpu)
     data = b"A repetitive volume block. " * 60_000  # ~1.6 MiB, compressible
     digest = store.put(data)
-    stored = open(store.open_path(digest), "rb").read()
+    stored = open(store._zpath(digest), "rb").read()  # stored under the .z name
     assert stored.startswith(b"MALZ41")
     assert len(stored) < len(data) // 3
     assert store.get(digest) == data
@@ -59,6 +59,39 @@ def test_cas_transparent_compression(tmp_path, monkeypatch):
     dest = str(tmp_path / "materialized")
     store.materialize(digest, dest)
     assert open(dest, "rb").read() == data
+    # open_path always yields the RAW form
+    assert open(store.open_path(digest), "rb").read() == data
+
+
+def test_raw_payload_with_magic_prefix_roundtrips(tmp_path):
+    """A raw-stored payload that itself begins with the MALZ41 magic must not
+    be mis-decompressed: the compressed bit is out-of-band (.z filename), not
+    sniffed from content (advisor finding, round 1)."""
+    from modal_amd.scheduler.blobs import BlobStore
+
+    store = BlobStore(str(tmp_path))
+    for data in (b"MALZ41", b"MALZ41" + os.urandom(4096), b"MALZ41\x00\x00\x00\x00garbage"):
+        digest = store.put(data)
+        assert store.get(digest) == data
+        assert open(store.open_path(digest), "rb").read() == data
+        dest = str(tmp_path / "m")
+        store.materialize(digest, dest)
+        assert open(dest, "rb").read() == data
+
+
+def test_put_file_matches_put_digest(tmp_path):
+    """put_file and put() compute the same CAS key for identical content on
+    both sides of the tree-digest threshold (advisor finding, round 1)."""
+    from modal_amd.ops.hashing import GPU_MIN_BYTES
+    from modal_amd.scheduler.blobs import BlobStore
+
+    store = BlobStore(str(tmp_path / "cas"))
+    for size in (4096, GPU_MIN_BYTES + 12345):
+        data = (b"dedup-check-%d " % size) * (size // 16 + 1)
+        data = data[:size]
+        p = tmp_path / f"f{size}"
+        p.write_bytes(data)
+        assert store.put_file(str(p)) == store.put(data)
 
 
 @pytest.mark.gpu

@@ -1,0 +1,131 @@
+"""RPC-contract honesty: GetOutputs max_values bound + socket handshake.
+
+Round-1 review items: (a) chunk-group expansion could overshoot max_values
+(scheduler/core.py function_get_outputs); (b) the scheduler socket was an
+unauthenticated full-control plane.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import os
+
+import modal_amd as modal
+from modal_amd._sync import synchronizer
+
+
+def test_get_outputs_honors_max_values(client):
+    """Every GetOutputs response carries at most max_values TRUE outputs,
+    even when completions arrive as ~64-item chunk groups; union of split
+    responses still yields every output exactly once."""
+    app = modal.App("test-maxvals")
+
+    @app.function()
+    def ident(x):
+        return x
+
+    n = 200
+    with app.run(client=client):
+        svc = client.svc
+
+        async def drain():
+            # drive the chunk (range) protocol directly so completions arrive
+            # as ~64-item group entries — the overshoot case
+            import pickle as _pickle
+
+            resp = await svc.function_map(function_id=ident.object_id, kind="map")
+            call_id = resp["function_call_id"]
+            for seq, base in enumerate(range(0, n, 64)):
+                buf = [((i,), {}) for i in range(base, min(base + 64, n))]
+                await svc.function_put_chunk(
+                    function_call_id=call_id,
+                    chunk_id=f"{call_id}.c{seq}",
+                    payload=_pickle.dumps(("C", buf)),
+                    count=len(buf),
+                    method="",
+                )
+            await svc.function_finish_inputs(function_call_id=call_id)
+            rec = svc.calls[call_id]
+            await asyncio.wait_for(rec.done_event.wait(), 60)
+            seen: dict[int, int] = {}
+            while True:
+                outs = await svc.function_get_outputs(
+                    function_call_id=call_id, max_values=10, timeout=0.2
+                )
+                if not outs:
+                    break
+                n_vals = 0
+                for item in outs:
+                    if item.get("group"):
+                        import pickle
+
+                        values = pickle.loads(item["chunk_data"])
+                        cis = item["cis"]
+                        voff = item.get("val_off", 0)
+                        pairs = (
+                            enumerate(values) if cis is None else zip(cis, values[voff:])
+                        )
+                        for ci, value in pairs:
+                            idx = item["idx_base"] + ci
+                            seen[idx] = seen.get(idx, 0) + 1
+                            assert value == idx  # input order == idx here
+                            n_vals += 1
+                    else:
+                        seen[item["idx"]] = seen.get(item["idx"], 0) + 1
+                        n_vals += 1
+                assert n_vals <= 10, f"response carried {n_vals} values (> max_values)"
+            return seen
+
+        seen = synchronizer.run(drain())
+        assert sorted(seen) == list(range(n))
+        assert all(v == 1 for v in seen.values()), "an output was delivered twice"
+
+
+def test_socket_rejects_unauthenticated_peer(client, run_dir):
+    """A local process that connects without the run_dir token gets nothing:
+    no RPC responses, connection closed."""
+    sock = os.path.join(run_dir, "scheduler.sock")
+    assert os.path.exists(sock)
+
+    async def probe() -> str:
+        from modal_amd.scheduler.transport import Connection
+
+        reader, writer = await asyncio.open_unix_connection(sock)
+        conn = Connection(reader, writer, lambda msg: asyncio.sleep(0))
+        conn.start()
+        # no hello / wrong token: RPC must never answer
+        await conn.send({"t": "hello", "role": "client", "auth": "wrong-token"})
+        try:
+            await conn.call("node_stats", {}, timeout=1.0)
+            return "answered"
+        except (ConnectionError, asyncio.TimeoutError):
+            return "rejected"
+        finally:
+            await conn.close()
+
+    assert synchronizer.run(probe()) == "rejected"
+
+
+def test_socket_accepts_token_from_run_dir(client, run_dir):
+    """The documented handshake (token file next to the socket) works."""
+    sock = os.path.join(run_dir, "scheduler.sock")
+
+    async def probe():
+        from modal_amd.scheduler.core import read_auth_token
+        from modal_amd.scheduler.transport import Connection
+
+        reader, writer = await asyncio.open_unix_connection(sock)
+        conn = Connection(reader, writer, lambda msg: asyncio.sleep(0))
+        conn.start()
+        await conn.send({"t": "hello", "role": "client", "auth": read_auth_token(sock)})
+        try:
+            return await conn.call("node_stats", {}, timeout=5.0)
+        finally:
+            await conn.close()
+
+    stats = synchronizer.run(probe())
+    assert isinstance(stats, dict)
+
+    # permissions: run_dir 0700, token 0600
+    assert os.stat(run_dir).st_mode & 0o777 == 0o700
+    assert os.stat(os.path.join(run_dir, "auth.token")).st_mode & 0o777 == 0o600

@@ -80,6 +80,18 @@ class TestCron:
         assert not cron_matches("0 0 * * 1-5", datetime(2026, 9, 13, 0, 0))
         assert cron_matches("0 0 * * 1,3,5", datetime(2026, 9, 14, 0, 0))  # Monday
 
+    def test_posix_dom_dow_or_semantics(self):
+        # both dom and dow restricted -> OR (POSIX cron): fires on the 1st,
+        # the 15th, AND every Monday
+        expr = "0 0 1,15 * 1"
+        assert cron_matches(expr, datetime(2026, 9, 1, 0, 0))  # 1st (a Tuesday)
+        assert cron_matches(expr, datetime(2026, 9, 15, 0, 0))  # 15th
+        assert cron_matches(expr, datetime(2026, 9, 14, 0, 0))  # a Monday, not 1st/15th
+        assert not cron_matches(expr, datetime(2026, 9, 13, 0, 0))  # Sunday the 13th
+        # one field unrestricted -> AND as before
+        assert cron_matches("0 0 13 * *", datetime(2026, 9, 13, 0, 0))
+        assert not cron_matches("0 0 * * 1", datetime(2026, 9, 13, 0, 0))
+
 
 class TestFilePatterns:
     def test_basic_globs(self):

@@ -31,6 +31,20 @@ def test_queue_partitions(client):
         assert q.get(partition="other") == 2
 
 
+def test_queue_exceptions_are_stdlib_subclasses(client):
+    """Parity: the reference raises stdlib queue.Empty/queue.Full; ported user
+    code catching those must keep working (advisor finding, round 1)."""
+    import queue as stdlib_queue
+
+    with modal.Queue.ephemeral() as q:
+        with pytest.raises(stdlib_queue.Empty):
+            q.get(timeout=0.05)
+        for i in range(5000):
+            q.put(i)
+        with pytest.raises(stdlib_queue.Full):
+            q.put(5001, timeout=0.05)
+
+
 def test_queue_blocking_timeout(client):
     with modal.Queue.ephemeral() as q:
         t0 = time.time()

@@ -100,3 +100,49 @@ def test_unhydrated_handle_rejected(client):
     q = modal.Queue.from_name("never-hydrated")
     with pytest.raises(SerializationError):
         serialize(q)
+
+
+def test_serialize_fast_detects_nested_framework_payloads(monkeypatch):
+    """serialize_fast must route payloads whose pickle stream touches
+    torch/modal_amd globals through the hook-aware pickler — including
+    tensors nested inside user objects and tensor subclasses
+    (advisor finding, round 1)."""
+    import modal_amd._serialization as S
+
+    calls = []
+    real = S.serialize
+    monkeypatch.setattr(S, "serialize", lambda obj: (calls.append(1), real(obj))[1])
+
+    # pure-primitive payloads stay on the raw C pickler
+    out = S.serialize_fast(("P", ((1, b"x", "y"), {})))
+    assert calls == []
+    import pickle
+
+    assert pickle.loads(out) == ("P", ((1, b"x", "y"), {}))
+
+    torch = pytest.importorskip("torch")
+
+    class Holder:
+        def __init__(self, t):
+            self.t = t
+
+    # tensor nested in a user object: plain pickle CAN serialize it, but the
+    # stream references torch.* -> must fall back to the hooked pickler
+    h = Holder(torch.ones(3))
+    S.serialize_fast(("P", ((h,), {})))
+    assert calls, "nested tensor payload did not take the hook-aware path"
+
+    calls.clear()
+    p = torch.nn.Parameter(torch.ones(2))
+    S.serialize_fast(("P", ((p,), {})))
+    assert calls, "tensor subclass payload did not take the hook-aware path"
+
+
+def test_persistent_id_handles_tensor_subclasses():
+    """nn.Parameter goes through the same persistent-id branch as Tensor."""
+    torch = pytest.importorskip("torch")
+    from modal_amd._serialization import deserialize, serialize
+
+    p = torch.nn.Parameter(torch.arange(4.0), requires_grad=False)
+    out = deserialize(serialize({"p": p}))["p"]
+    assert torch.equal(out.detach(), torch.arange(4.0))
