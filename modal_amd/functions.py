@@ -374,6 +374,24 @@ class _Function(_Object, type_kind="function"):
             self, input_iter, kwargs, order_outputs, return_exceptions, wrap_returned_exceptions
         )
 
+    def _map_inner_batches(
+        self,
+        input_iter: Any,
+        kwargs: dict,
+        order_outputs: bool,
+        return_exceptions: bool,
+        wrap_returned_exceptions: bool,
+        fast_zip_iters: Any = None,
+    ) -> AsyncGenerator[list, None]:
+        """Batch view (lists of values) — the engine's native granularity;
+        outer API layers flatten once, so per-item cost stays a plain loop."""
+        from .parallel.map import map_invocation_batches
+
+        return map_invocation_batches(
+            self, input_iter, kwargs, order_outputs, return_exceptions,
+            wrap_returned_exceptions, fast_zip_iters=fast_zip_iters,
+        )
+
     def map_async(
         self,
         *input_iterators: Any,
@@ -408,10 +426,53 @@ class _Function(_Object, type_kind="function"):
             gen_args(), kwargs or {}, order_outputs, return_exceptions, wrap_returned_exceptions
         )
 
+    def map_async_batches(
+        self,
+        *input_iterators: Any,
+        kwargs: Optional[dict] = None,
+        order_outputs: bool = True,
+        return_exceptions: bool = False,
+        wrap_returned_exceptions: bool = True,
+    ) -> AsyncGenerator[list, None]:
+        if all(not hasattr(it, "__aiter__") for it in input_iterators):
+            # sync iterables (the overwhelmingly common case): the pump
+            # builds whole chunks via C-level zip+islice — zero per-item
+            # Python frames on the input side
+            return self._map_inner_batches(
+                None, kwargs or {}, order_outputs, return_exceptions,
+                wrap_returned_exceptions, fast_zip_iters=input_iterators,
+            )
+
+        def gen_args() -> Any:
+            for combo in zip(*input_iterators):
+                yield (combo, {})
+
+        return self._map_inner_batches(
+            gen_args(), kwargs or {}, order_outputs, return_exceptions, wrap_returned_exceptions
+        )
+
+    def starmap_async_batches(
+        self,
+        input_iterator: Any,
+        *,
+        kwargs: Optional[dict] = None,
+        order_outputs: bool = True,
+        return_exceptions: bool = False,
+        wrap_returned_exceptions: bool = True,
+    ) -> AsyncGenerator[list, None]:
+        def gen_args() -> Any:
+            for item in input_iterator:
+                args = tuple(item) if isinstance(item, (list, tuple)) else (item,)
+                yield (args, {})
+
+        return self._map_inner_batches(
+            gen_args(), kwargs or {}, order_outputs, return_exceptions, wrap_returned_exceptions
+        )
+
     async def for_each_async(
         self, *input_iterators: Any, kwargs: Optional[dict] = None, ignore_exceptions: bool = False
     ) -> None:
-        async for _ in self.map_async(
+        async for _ in self.map_async_batches(
             *input_iterators,
             kwargs=kwargs,
             order_outputs=False,
@@ -552,7 +613,7 @@ def _install_sync_map_methods() -> None:
     """
     from ._sync import synchronizer, unwrap, wrap
 
-    def make(name: str, async_name: str) -> Any:
+    def make(name: str, batches_name: str, item_name: str) -> Any:
         class _MapDescriptor:
             def __get__(self, obj: Any, objtype: Any = None) -> Any:
                 if obj is None:
@@ -560,28 +621,33 @@ def _install_sync_map_methods() -> None:
                 impl = obj._impl
 
                 def blocking(*args: Any, **kwargs: Any) -> Any:
-                    agen = getattr(impl, async_name)(*unwrap(args), **unwrap(kwargs))
-                    for item in synchronizer.run_generator_sync(agen):
-                        yield wrap(item)
+                    # batches cross the sync bridge, so one thread handoff
+                    # serves ~64 items instead of one
+                    agen = getattr(impl, batches_name)(*unwrap(args), **unwrap(kwargs))
+                    for batch in synchronizer.run_generator_sync(agen):
+                        for item in batch:
+                            yield wrap(item)
 
                 async def aio(*args: Any, **kwargs: Any) -> Any:
-                    agen = getattr(impl, async_name)(*unwrap(args), **unwrap(kwargs))
+                    agen = getattr(impl, batches_name)(*unwrap(args), **unwrap(kwargs))
                     if synchronizer.in_loop_thread():
                         # already on the framework loop: no bridge frame
-                        async for item in agen:
-                            yield wrap(item)
+                        async for batch in agen:
+                            for item in batch:
+                                yield wrap(item)
                         return
-                    async for item in synchronizer.run_generator_async(agen):
-                        yield wrap(item)
+                    async for batch in synchronizer.run_generator_async(agen):
+                        for item in batch:
+                            yield wrap(item)
 
                 from ._sync import _AioCallable
 
-                return _AioCallable(blocking, aio, getattr(impl, async_name))
+                return _AioCallable(blocking, aio, getattr(impl, item_name))
 
         return _MapDescriptor()
 
-    setattr(Function, "map", make("map", "map_async"))
-    setattr(Function, "starmap", make("starmap", "starmap_async"))
+    setattr(Function, "map", make("map", "map_async_batches", "map_async"))
+    setattr(Function, "starmap", make("starmap", "starmap_async_batches", "starmap_async"))
 
     class _ForEachDescriptor:
         def __get__(self, obj: Any, objtype: Any = None) -> Any:

@@ -79,6 +79,31 @@ async def map_invocation(
     return_exceptions: bool,
     wrap_returned_exceptions: bool,
 ) -> AsyncGenerator[Any, None]:
+    """Per-item view over map_invocation_batches (one flat loop, no extra
+    async-generator layer per item)."""
+    async for batch in map_invocation_batches(
+        fn, input_iter, kwargs_common, order_outputs, return_exceptions,
+        wrap_returned_exceptions,
+    ):
+        for value in batch:
+            yield value
+
+
+async def map_invocation_batches(
+    fn: Any,
+    input_iter: Any,
+    kwargs_common: dict,
+    order_outputs: bool,
+    return_exceptions: bool,
+    wrap_returned_exceptions: bool,
+    fast_zip_iters: Any = None,
+) -> AsyncGenerator[list, None]:
+    """The map engine proper. Yields LISTS of decoded output values — the
+    natural unit is the ~64-item range-protocol group, so per-item work in
+    the engine is a plain list append/extend, not an async-generator frame.
+    Callers that need per-item semantics flatten at their own (single)
+    layer; the sync bridge crosses threads once per batch instead of once
+    per item."""
     if not fn._is_hydrated:
         await fn.hydrate()
     client = fn._client
@@ -139,7 +164,52 @@ async def map_invocation(
                     approx_bytes += len(a)
             total_inputs += 1
 
+        async def flush_args_chunk(argsbatch: list) -> None:
+            # "C2" wire form: common kwargs factored out, args list built by
+            # C-level zip+islice — no per-item Python in the pump at all
+            nonlocal chunk_seq, total_inputs
+            total_inputs += len(argsbatch)
+            await sem.acquire(len(argsbatch))
+            payload = serialize_fast(("C2", kwargs_common, argsbatch))
+            chunk_id = f"{call_id}.c{chunk_seq}"
+            chunk_seq += 1
+            if len(payload) > 2 * 1024 * 1024:
+                store = client.blob_store
+                if store is not None:
+                    payload = {"blob": store.put(payload)}
+            await svc.function_put_chunk(
+                function_call_id=call_id,
+                chunk_id=chunk_id,
+                payload=payload,
+                count=len(argsbatch),
+                method=fn._method_name or "",
+            )
+
         try:
+            if fast_zip_iters is not None:
+                from itertools import islice
+
+                iterator = zip(*fast_zip_iters)
+                n_flushed = 0
+                while True:
+                    argsbatch = list(islice(iterator, CHUNK_ITEMS))
+                    if not argsbatch:
+                        break
+                    # byte-size guard sampled from the first item: split big-
+                    # payload chunks so one frame stays well under the CAS
+                    # spill threshold
+                    est = sum(
+                        len(a) for a in argsbatch[0] if type(a) in (bytes, bytearray, str)
+                    )
+                    if est * len(argsbatch) > 4 * 1024 * 1024 and len(argsbatch) > 1:
+                        step = max(1, (4 * 1024 * 1024) // max(est, 1))
+                        for s in range(0, len(argsbatch), step):
+                            await flush_args_chunk(argsbatch[s : s + step])
+                    else:
+                        await flush_args_chunk(argsbatch)
+                    n_flushed += 1
+                await svc.function_finish_inputs(function_call_id=call_id)
+                return
             if hasattr(input_iter, "__aiter__"):
                 async for item in input_iter:
                     _add(item)
@@ -224,6 +294,7 @@ async def map_invocation(
                 raise pump_error[0]
             if progress is not None:
                 progress.update(received, total_inputs, pump_done.is_set())
+            batch_out: list = []
             for out in outs:
                 if out.get("group"):
                     # range-protocol group: one pickled value list for ~64 idxs
@@ -235,21 +306,23 @@ async def map_invocation(
                     # val_off: the position in `values` of cis[0] — nonzero
                     # when the scheduler split a group to honor max_values
                     voff = out.get("val_off", 0)
-                    pairs = (
-                        enumerate(values) if cis is None else zip(cis, values[voff:])
-                    )
                     n_here = len(values) if cis is None else len(cis)
                     received += n_here
                     sem.release(n_here)
                     if order_outputs:
-                        for ci, value in pairs:
-                            ordering_buffer[base + ci] = value
+                        if cis is None:
+                            for ci, value in enumerate(values):
+                                ordering_buffer[base + ci] = value
+                        else:
+                            for ci, value in zip(cis, values[voff:]):
+                                ordering_buffer[base + ci] = value
                         while next_output_idx in ordering_buffer:
-                            yield ordering_buffer.pop(next_output_idx)
+                            batch_out.append(ordering_buffer.pop(next_output_idx))
                             next_output_idx += 1
+                    elif cis is None:
+                        batch_out.extend(values)
                     else:
-                        for _ci, value in pairs:
-                            yield value
+                        batch_out.extend(values[voff : voff + len(cis)])
                     continue
                 received += 1
                 sem.release()
@@ -257,10 +330,12 @@ async def map_invocation(
                 if order_outputs:
                     ordering_buffer[out["idx"]] = value
                     while next_output_idx in ordering_buffer:
-                        yield ordering_buffer.pop(next_output_idx)
+                        batch_out.append(ordering_buffer.pop(next_output_idx))
                         next_output_idx += 1
                 else:
-                    yield value
+                    batch_out.append(value)
+            if batch_out:
+                yield batch_out
     finally:
         if progress is not None:
             progress.close()

@@ -621,12 +621,22 @@ class WorkerRuntime:
                         raw = entry["raw"]
                         if isinstance(raw, dict):  # CAS-spilled chunk (>2 MiB)
                             raw = self.blob_store.get(raw["blob"])
-                        kind, decoded = deserialize(raw)
-                        assert kind == "C"
+                        obj = deserialize(raw)
+                        if obj[0] == "C":
+                            decoded = (obj[1], None)  # list of (args, kwargs)
+                        elif obj[0] == "C2":
+                            # common kwargs factored out client-side: the
+                            # chunk body is a bare args list
+                            decoded = (obj[2], obj[1])
+                        else:
+                            raise RuntimeError(f"unknown chunk form {obj[0]!r}")
                         entry["decoded"] = decoded
                         entry["raw"] = None
-            args, kwargs = decoded[item.get("ci", 0)]
-            return args, kwargs
+            pairs, kw_common = decoded
+            if kw_common is None:
+                args, kwargs = pairs[item.get("ci", 0)]
+                return args, kwargs
+            return pairs[item.get("ci", 0)], kw_common
         return self._decode_args(item)
 
     def _decode_args(self, item: dict) -> tuple[tuple, dict]:
