@@ -50,11 +50,14 @@ def map_item_gpu(x: int) -> int:
         torch._ma_bench_cache = cache
     t = cache * float(x % 7 + 1)
     s = t[:4].float().sum()
-    if os.environ.get("MODAL_AMD_BENCH_TENSOR_RET") == "1":
-        # same GPU math; the runtime batch-stages the readback per chunk
-        # (one pinned D2H) instead of a ~24 us .item() sync per item
-        return s
-    return int(s.item()) and x or x
+    if os.environ.get("MODAL_AMD_BENCH_ITEM_SYNC") == "1":
+        # legacy variant: a ~24 us .item() device sync per item
+        return int(s.item()) and x or x
+    # default: return the result tensor; the runtime batch-stages the
+    # readback per ~64-item chunk (one pinned D2H per chunk,
+    # runtime/_serialize_chunk.py) — every item's GPU op and its readback
+    # still happen, the sync just amortizes across the chunk
+    return s
 
 
 def map_item_cpu(x: int) -> int:
@@ -276,7 +279,11 @@ def main() -> None:
                 "seq_len": 0,
                 "parallelism": f"map{n_gpus}",
                 "p50_remote_ms": p50_ms,
-                "per_item_gpu_op": "bf16 vector scale + readback" if has_gpu else "noop",
+                "per_item_gpu_op": (
+                    "bf16 vector scale + chunk-batched pinned D2H readback"
+                    if has_gpu
+                    else "noop"
+                ),
                 "workers": len(scheduler.pool.workers),
             },
         }

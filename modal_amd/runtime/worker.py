@@ -60,6 +60,20 @@ import threading as _threading
 _slot_tls = _threading.local()
 
 
+def _slot_streams_enabled() -> bool:
+    """One-time-per-range guard mirroring _invoke_on_slot_stream's checks."""
+    torch = sys.modules.get("torch")
+    cuda = getattr(torch, "cuda", None) if torch is not None else None
+    try:
+        return (
+            cuda is not None
+            and os.environ.get("MODAL_AMD_SLOT_STREAMS", "0") not in ("0", "false")
+            and cuda.is_initialized()
+        )
+    except AttributeError:
+        return False
+
+
 def _invoke_on_slot_stream(fn: Any, args: tuple, kwargs: dict) -> Any:
     """Run a user callable on this executor thread's own HIP stream.
 
@@ -608,36 +622,42 @@ class WorkerRuntime:
         serve every item of the chunk."""
         cid = item.get("chunk")
         if cid is not None:
-            entry = self._chunk_cache.get(cid)
-            if entry is None:
-                raise RuntimeError(f"chunk {cid} not delivered to this worker")
-            decoded = entry["decoded"]
-            if decoded is None:
-                # one decode serves every range-thread of the chunk
-                lock = entry.setdefault("lock", __import__("threading").Lock())
-                with lock:
-                    decoded = entry["decoded"]
-                    if decoded is None:
-                        raw = entry["raw"]
-                        if isinstance(raw, dict):  # CAS-spilled chunk (>2 MiB)
-                            raw = self.blob_store.get(raw["blob"])
-                        obj = deserialize(raw)
-                        if obj[0] == "C":
-                            decoded = (obj[1], None)  # list of (args, kwargs)
-                        elif obj[0] == "C2":
-                            # common kwargs factored out client-side: the
-                            # chunk body is a bare args list
-                            decoded = (obj[2], obj[1])
-                        else:
-                            raise RuntimeError(f"unknown chunk form {obj[0]!r}")
-                        entry["decoded"] = decoded
-                        entry["raw"] = None
-            pairs, kw_common = decoded
+            pairs, kw_common = self._chunk_pairs(cid)
             if kw_common is None:
                 args, kwargs = pairs[item.get("ci", 0)]
                 return args, kwargs
             return pairs[item.get("ci", 0)], kw_common
         return self._decode_args(item)
+
+    def _chunk_pairs(self, cid: str) -> tuple[list, Any]:
+        """Decoded chunk body: (items, kw_common). kw_common None means each
+        item is an (args, kwargs) pair (C form); otherwise items are bare
+        args tuples sharing kw_common (C2 form)."""
+        entry = self._chunk_cache.get(cid)
+        if entry is None:
+            raise RuntimeError(f"chunk {cid} not delivered to this worker")
+        decoded = entry["decoded"]
+        if decoded is None:
+            # one decode serves every range-thread of the chunk
+            lock = entry.setdefault("lock", __import__("threading").Lock())
+            with lock:
+                decoded = entry["decoded"]
+                if decoded is None:
+                    raw = entry["raw"]
+                    if isinstance(raw, dict):  # CAS-spilled chunk (>2 MiB)
+                        raw = self.blob_store.get(raw["blob"])
+                    obj = deserialize(raw)
+                    if obj[0] == "C":
+                        decoded = (obj[1], None)  # list of (args, kwargs)
+                    elif obj[0] == "C2":
+                        # common kwargs factored out client-side: the
+                        # chunk body is a bare args list
+                        decoded = (obj[2], obj[1])
+                    else:
+                        raise RuntimeError(f"unknown chunk form {obj[0]!r}")
+                    entry["decoded"] = decoded
+                    entry["raw"] = None
+        return decoded
 
     def _decode_args(self, item: dict) -> tuple[tuple, dict]:
         if item.get("payload_blob"):
@@ -804,22 +824,31 @@ class WorkerRuntime:
         def run_range(start: int, end: int) -> None:
             try:
                 fn = frt.get_callable(msg.get("method", ""))
+                pairs, kw_common = self._chunk_pairs(msg["chunk_id"])
             except BaseException as exc:
                 data = self._serialize_exception(exc)
                 err = "".join(traceback.format_exception_only(type(exc), exc)).strip()
                 for ci in range(start, end):
                     errors[ci] = (data, err)
                 return
-            item = {"chunk": msg["chunk_id"], "ci": 0}
+            # tight loop: chunk decode + slot-stream decision hoisted out;
+            # per item only the contextvar, the call, and the result store
+            prefix = f"in-{call_id[3:]}-"
+            plain_call = not _slot_streams_enabled()
             tok_c = _current_function_call_id.set(call_id)
             self._register_sync_thread(msg["token"])
             try:
                 for ci in range(start, end):
-                    item["ci"] = ci
-                    tok_i = _current_input_id.set(f"in-{call_id[3:]}-{ci}")
+                    tok_i = _current_input_id.set(prefix + str(ci))
                     try:
-                        args, kwargs = self._resolve_item_args(item)
-                        values[ci] = _invoke_on_slot_stream(fn, args, kwargs)
+                        if kw_common is None:
+                            args, kwargs = pairs[ci]
+                        else:
+                            args, kwargs = pairs[ci], kw_common
+                        if plain_call:
+                            values[ci] = fn(*args, **kwargs)
+                        else:
+                            values[ci] = _invoke_on_slot_stream(fn, args, kwargs)
                     except InputCancellation:
                         raise  # injected cancel: abort the whole range
                     except BaseException as exc:
