@@ -18,9 +18,36 @@ from typing import Optional
 BATCH_READBACK_MAX_BYTES = 64 * 1024 * 1024
 
 
+def _rebuild_shared(buf: bytes, off: int, nbytes: int, dtype_str: str, shape: tuple):
+    """Unpickle hook: rebuild a host tensor from a chunk-shared byte buffer.
+
+    All tensors of one output chunk reference the SAME bytes object, which
+    pickle memoizes — the buffer serializes once per chunk and each tensor
+    costs only a small tuple on the wire."""
+    import torch
+
+    dtype = getattr(torch, dtype_str)
+    t = torch.frombuffer(bytearray(buf[off : off + nbytes]), dtype=dtype)
+    return t.reshape(shape) if shape else t.reshape(())
+
+
+class _SharedBufTensor:
+    """Stand-in whose pickled form is (`_rebuild_shared`, args): ~0.3 us to
+    pickle vs ~15 us for torch's native tensor reduce on small tensors."""
+
+    __slots__ = ("_args",)
+
+    def __init__(self, buf: bytes, off: int, nbytes: int, dtype_str: str, shape: tuple):
+        self._args = (buf, off, nbytes, dtype_str, shape)
+
+    def __reduce__(self):
+        return (_rebuild_shared, self._args)
+
+
 def _batch_cuda_to_host(values: list, torch: "object") -> None:
-    """Replace top-level small CUDA tensors with host copies using one
-    device gather + one pinned D2H (in place; no-op when none qualify)."""
+    """Replace top-level small CUDA tensors with host-rebuildable stand-ins
+    using one device gather + one pinned D2H (in place; no-op when none
+    qualify). The stand-ins unpickle as ordinary host torch.Tensors."""
     idxs = [
         i for i, v in enumerate(values)
         if type(v) is torch.Tensor and v.is_cuda
@@ -39,10 +66,9 @@ def _batch_cuda_to_host(values: list, torch: "object") -> None:
         off = 0
         for i, t in zip(idxs, tensors):
             nbytes = t.numel() * t.element_size()
-            host = torch.frombuffer(
-                bytearray(buf[off : off + nbytes]), dtype=t.dtype
-            ).reshape(t.shape)
-            values[i] = host
+            values[i] = _SharedBufTensor(
+                buf, off, nbytes, str(t.dtype).removeprefix("torch."), tuple(t.shape)
+            )
             off += nbytes
     except Exception:
         return  # keep device tensors; the per-item path handles them

@@ -1,53 +1,97 @@
-"""Isolate per-item worker-side costs for the map hot path (GPU box)."""
+"""Decompose the per-item worker cost of the map bench GPU op.
+
+Measures, per 64-item chunk on one MI355X:
+  A. the bare GPU op (3 launches/item), no readback
+  B. op + per-item .item() sync             (bench ITEM_SYNC variant)
+  C. op + batched cat+pinned D2H readback   (runtime _serialize_chunk path)
+  D. pickle cost of 64 small HOST tensors vs 64 ints vs _FastTensor wrapper
+Run on a GPU box: python scripts/worker_microbench.py
+"""
+
+from __future__ import annotations
+
 import pickle
-import sys
 import time
 
-sys.path.insert(0, "/root/repo")
 import torch
 
-assert torch.cuda.is_available()
-N = 20_000
-cache = torch.ones(4096, device="cuda", dtype=torch.bfloat16)
 
-def op(x: int) -> int:
-    t = cache * float(x % 7 + 1)
-    return int(t[:4].float().sum().item()) and x or x
+def timeit(label: str, fn, iters: int = 50) -> float:
+    fn()  # warm
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / iters
+    per_item = dt / 64 * 1e6
+    print(f"{label:44s} {dt * 1e3:8.3f} ms/chunk  {per_item:7.2f} us/item")
+    return dt
 
-# warm
-for i in range(200): op(i)
-torch.cuda.synchronize()
 
-t0 = time.perf_counter()
-for i in range(N): op(i)
-el = time.perf_counter() - t0
-print(f"gpu op alone: {el/N*1e6:.1f} us/item")
+def main() -> None:
+    cache = torch.ones(4096, device="cuda", dtype=torch.bfloat16)
 
-# kernel-only (no readback)
-t0 = time.perf_counter()
-for i in range(N):
-    t = cache * float(i % 7 + 1)
-torch.cuda.synchronize()
-el = time.perf_counter() - t0
-print(f"gpu op no readback: {el/N*1e6:.1f} us/item")
+    def op(x: int):
+        t = cache * float(x % 7 + 1)
+        return t[:4].float().sum()
 
-# readback alone
-t = cache * 2.0
-t0 = time.perf_counter()
-for i in range(N):
-    t[:4].float().sum().item()
-el = time.perf_counter() - t0
-print(f"readback alone: {el/N*1e6:.1f} us/item")
+    def chunk_op_only():
+        vals = [op(x) for x in range(64)]
+        torch.cuda.synchronize()
+        return vals
 
-# worker-pipeline mimicry without GPU: chunk decode + dispatch + result chunk
-chunk = pickle.dumps(("C", [((i,), {}) for i in range(64)]))
-t0 = time.perf_counter()
-reps = N // 64
-for r in range(reps):
-    _tag, items = pickle.loads(chunk)
-    values = [None] * 64
-    for ci, (args, kwargs) in enumerate(items):
-        values[ci] = args[0]
-    out = pickle.dumps(values)
-el = time.perf_counter() - t0
-print(f"chunk decode+loop+encode (no op): {el/(reps*64)*1e6:.2f} us/item")
+    def chunk_item_sync():
+        return [int(op(x).item()) for x in range(64)]
+
+    from modal_amd.runtime._serialize_chunk import _batch_cuda_to_host
+
+    def chunk_batched_readback():
+        vals = [op(x) for x in range(64)]
+        _batch_cuda_to_host(vals, torch)
+        return vals
+
+    def chunk_batched_plus_pickle():
+        vals = [op(x) for x in range(64)]
+        _batch_cuda_to_host(vals, torch)
+        return pickle.dumps(vals)
+
+    timeit("A op only (3 launches/item) + sync", chunk_op_only)
+    timeit("B op + per-item .item()", chunk_item_sync)
+    timeit("C op + batched readback", chunk_batched_readback)
+    timeit("C' op + batched readback + pickle", chunk_batched_plus_pickle)
+
+    # host-side pickle costs in isolation
+    host_tensors = chunk_batched_readback()
+    ints = list(range(64))
+    t0 = time.perf_counter()
+    for _ in range(200):
+        pickle.dumps(host_tensors)
+    print(f"pickle 64 host tensors: {(time.perf_counter() - t0) / 200 * 1e6:.1f} us/chunk")
+    t0 = time.perf_counter()
+    for _ in range(200):
+        pickle.dumps(ints)
+    print(f"pickle 64 ints:         {(time.perf_counter() - t0) / 200 * 1e6:.1f} us/chunk")
+
+    # one-kernel-per-item op variant (single fused launch count)
+    def op1(x: int):
+        return (cache * float(x % 7 + 1)).sum()
+
+    def chunk_op1():
+        vals = [op1(x) for x in range(64)]
+        torch.cuda.synchronize()
+        return vals
+
+    timeit("E 1-launch... (2 kernels: mul+sum)", chunk_op1)
+
+    # launch overhead floor: 1 trivial kernel per item
+    def chunk_single_launch():
+        vals = [cache.sum() for _ in range(64)]
+        torch.cuda.synchronize()
+        return vals
+
+    timeit("F floor: one sum launch per item", chunk_single_launch)
+
+
+if __name__ == "__main__":
+    main()

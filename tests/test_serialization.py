@@ -146,3 +146,31 @@ def test_persistent_id_handles_tensor_subclasses():
     p = torch.nn.Parameter(torch.arange(4.0), requires_grad=False)
     out = deserialize(serialize({"p": p}))["p"]
     assert torch.equal(out.detach(), torch.arange(4.0))
+
+
+def test_shared_buf_tensor_roundtrip():
+    """Output-chunk tensor stand-ins unpickle as host tensors; the shared
+    buffer pickles once per chunk (memoized), so per-tensor wire cost is a
+    small tuple."""
+    torch = pytest.importorskip("torch")
+    import pickle
+
+    from modal_amd.runtime._serialize_chunk import _SharedBufTensor
+
+    a = torch.arange(6, dtype=torch.float32)
+    b = torch.tensor(3.5, dtype=torch.float32)
+    buf = a.numpy().tobytes() + b.reshape(1).numpy().tobytes()
+    values = [
+        _SharedBufTensor(buf, 0, 24, "float32", (2, 3)),
+        _SharedBufTensor(buf, 24, 4, "float32", ()),
+        "plain",
+    ]
+    data = pickle.dumps(values, 4)
+    out = pickle.loads(data)
+    assert torch.equal(out[0], a.reshape(2, 3))
+    assert out[1].item() == 3.5 and out[1].shape == ()
+    assert out[2] == "plain"
+    # shared buffer memoized: doubling the tensor count adds ~tuple bytes,
+    # not another copy of the buffer
+    values2 = values[:2] * 8
+    assert len(pickle.dumps(values2, 4)) < len(buf) + 16 * 64
