@@ -293,6 +293,13 @@ def serialize_payload(args: tuple, kwargs: dict) -> bytes:
 
 
 def deserialize_payload(data: bytes, tensors: Optional[list] = None) -> tuple[tuple, dict]:
-    kind, payload = deserialize(data)
-    args, kwargs = payload
-    return args, kwargs
+    obj = deserialize(data)
+    first = obj[0]
+    if isinstance(first, str):  # native envelope: ("P", (args, kwargs))
+        args, kwargs = obj[1]
+        return args, kwargs
+    # reference wire form: a bare pickled (args, kwargs) 2-tuple
+    # (reference _serialization.py serialize((args, kwargs)) — the gRPC
+    # bridge passes such payloads through untranslated)
+    args, kwargs = obj
+    return tuple(args), dict(kwargs)

@@ -277,6 +277,11 @@ def daemon(run_dir: Optional[str]) -> None:
         scheduler = Scheduler(run_dir=run_dir)
         await scheduler.start()
         click.echo(f"modal-amd scheduler on {scheduler.pool.socket_path}")
+        try:
+            grpc_sock = await scheduler.start_grpc_bridge()
+            click.echo(f"modal-amd api.proto gRPC plane on {grpc_sock}")
+        except Exception as exc:  # grpc missing: msgpack plane still serves
+            click.echo(f"gRPC plane unavailable: {exc}")
         while True:
             await asyncio.sleep(3600)
 

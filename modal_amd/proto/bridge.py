@@ -1,0 +1,389 @@
+"""api.proto gRPC plane over the in-process scheduler.
+
+A ``grpc.aio`` server on ``<run_dir>/grpc.sock`` speaking the ModalClient
+service (modal_proto/api.proto:4680) for the App*/Function*/Queue*/Dict*/
+Secret*/Blob* groups, adapting protos to the scheduler's native surface.
+The msgpack socket remains the fast path; this plane exists so
+reference-style clients, mock-servicer tests, and wire-format checks work
+against the real scheduler (round-1 review, Missing #1).
+
+Payload translation: reference FunctionInput.args is a pickled
+``(args, kwargs)`` 2-tuple (reference _serialization.py); the native worker
+accepts that form directly (runtime/worker.py deserialize_payload handles
+both it and the native ``("P", ...)`` envelope).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import os
+import pickle
+from typing import Any, Optional
+
+from .compiler import load
+
+
+class GrpcBridge:
+    def __init__(self, scheduler: Any):
+        self.scheduler = scheduler
+        self.api, self.router_pb = load()
+        self._server: Any = None
+        self.socket_path = os.path.join(scheduler.run_dir, "grpc.sock")
+        self._blob_dir = os.path.join(scheduler.run_dir, "grpcblobs")
+        self._out_chunk_cache: dict[str, list] = {}
+
+    # -- lifecycle ---------------------------------------------------------
+
+    async def start(self) -> str:
+        import grpc
+
+        os.makedirs(self._blob_dir, exist_ok=True)
+        server = grpc.aio.server()
+        api = self.api
+        pool = api.DESCRIPTOR.pool
+        from google.protobuf import message_factory
+
+        handlers = {}
+        svc = api.SERVICES["ModalClient"]
+        for method in svc.method:
+            impl = getattr(self, method.name, None)
+            if impl is None:
+                continue
+            req_cls = message_factory.GetMessageClass(
+                pool.FindMessageTypeByName(method.input_type.lstrip("."))
+            )
+            if method.server_streaming:
+                handlers[method.name] = grpc.unary_stream_rpc_method_handler(
+                    impl,
+                    request_deserializer=req_cls.FromString,
+                    response_serializer=lambda msg: msg.SerializeToString(),
+                )
+            else:
+                handlers[method.name] = grpc.unary_unary_rpc_method_handler(
+                    impl,
+                    request_deserializer=req_cls.FromString,
+                    response_serializer=lambda msg: msg.SerializeToString(),
+                )
+        server.add_generic_rpc_handlers(
+            (grpc.method_handlers_generic_handler("modal.client.ModalClient", handlers),)
+        )
+        server.add_insecure_port(f"unix:{self.socket_path}")
+        await server.start()
+        self._server = server
+        return self.socket_path
+
+    async def stop(self) -> None:
+        if self._server is not None:
+            await self._server.stop(0.5)
+            self._server = None
+
+    # -- App ---------------------------------------------------------------
+
+    async def AppCreate(self, request: Any, context: Any) -> Any:
+        resp = await self.scheduler.app_create(
+            description=request.description,
+            ephemeral=True,
+            environment=request.environment_name,
+        )
+        return self.api.AppCreateResponse(app_id=resp["app_id"])
+
+    async def AppClientDisconnect(self, request: Any, context: Any) -> Any:
+        from google.protobuf import empty_pb2
+
+        await self.scheduler.app_client_disconnect(request.app_id)
+        return empty_pb2.Empty()
+
+    async def AppStop(self, request: Any, context: Any) -> Any:
+        from google.protobuf import empty_pb2
+
+        await self.scheduler.app_stop(request.app_id)
+        return empty_pb2.Empty()
+
+    # -- Function ------------------------------------------------------------
+
+    async def FunctionCreate(self, request: Any, context: Any) -> Any:
+        fn = request.function
+        is_gen = fn.function_type == self.api.Function.FunctionType.FUNCTION_TYPE_GENERATOR
+        options = {
+            "definition_kind": "serialized",
+            "is_generator": is_gen,
+            "timeout": fn.timeout_secs or None,
+            "max_concurrent_inputs": max(
+                getattr(fn, "max_concurrent_inputs", 0) or 0, 1
+            ),
+        }
+        if fn.resources.gpu_config.count:
+            options["needs_gpu"] = True
+            options["gpu_count"] = fn.resources.gpu_config.count
+        if fn.retry_policy.retries:
+            options["retries"] = {
+                "max_retries": fn.retry_policy.retries,
+                "initial_delay": fn.retry_policy.initial_delay_ms / 1000.0,
+                "backoff_coefficient": fn.retry_policy.backoff_coefficient or 1.0,
+            }
+        resp = await self.scheduler.function_create(
+            app_id=request.app_id,
+            name=fn.function_name or fn.module_name or "grpc-function",
+            definition=bytes(fn.function_serialized),
+            options=options,
+        )
+        out = self.api.FunctionCreateResponse(function_id=resp["function_id"])
+        out.handle_metadata.function_name = fn.function_name
+        return out
+
+    def _put_item_to_native(self, item: Any) -> dict:
+        native: dict = {}
+        which = item.input.WhichOneof("args_oneof")
+        if which == "args_blob_id":
+            native["payload_blob"] = item.input.args_blob_id
+            native["payload"] = b""
+        else:
+            native["payload"] = bytes(item.input.args)
+        if item.input.HasField("method_name"):
+            native["method"] = item.input.method_name
+        return native
+
+    async def FunctionMap(self, request: Any, context: Any) -> Any:
+        kind_map = {
+            self.api.FUNCTION_CALL_TYPE_UNARY: "unary",
+            self.api.FUNCTION_CALL_TYPE_MAP: "map",
+        }
+        kind = kind_map.get(request.function_call_type, "map")
+        if request.from_spawn_map:
+            kind = "spawn_map"
+        items = [self._put_item_to_native(i) for i in request.pipelined_inputs]
+        try:
+            resp = await self.scheduler.function_map(
+                function_id=request.function_id,
+                kind=kind,
+                pipelined_inputs=items or None,
+                return_exceptions=request.return_exceptions,
+            )
+        except Exception as exc:
+            await self._abort_not_found(context, exc)
+            raise
+        out = self.api.FunctionMapResponse(
+            function_call_id=resp["function_call_id"],
+            sync_client_retries_enabled=bool(resp.get("sync_client_retries_enabled")),
+            max_inputs_outstanding=resp.get("max_inputs_outstanding") or 1000,
+            function_call_jwt=resp["function_call_id"],  # local: the id is the token
+        )
+        retries = resp.get("retry_policy") or {}
+        out.retry_policy.retries = int(retries.get("max_retries", 0))
+        out.retry_policy.initial_delay_ms = int(
+            float(retries.get("initial_delay", 1.0)) * 1000
+        )
+        out.retry_policy.backoff_coefficient = float(
+            retries.get("backoff_coefficient", 1.0)
+        )
+        for idx, item in enumerate(items):
+            pi = out.pipelined_inputs.add()
+            pi.idx = idx
+            pi.input_id = f"in-{resp['function_call_id'][3:]}-{idx}"
+            pi.input_jwt = pi.input_id
+        return out
+
+    async def FunctionPutInputs(self, request: Any, context: Any) -> Any:
+        items = [self._put_item_to_native(i) for i in request.inputs]
+        try:
+            placed = await self.scheduler.function_put_inputs(
+                function_call_id=request.function_call_id, items=items
+            )
+        except Exception as exc:
+            await self._abort_not_found(context, exc)
+            raise
+        out = self.api.FunctionPutInputsResponse()
+        for rec, req_item in zip(placed, request.inputs):
+            ri = out.inputs.add()
+            ri.idx = req_item.idx
+            ri.input_id = rec["input_id"]
+            ri.input_jwt = rec["input_id"]
+        # the reference marks the end of inputs with final_input on the last
+        # item (function_utils.py should_upload / FunctionInput.final_input)
+        if any(i.input.final_input for i in request.inputs):
+            await self.scheduler.function_finish_inputs(
+                function_call_id=request.function_call_id
+            )
+        return out
+
+    async def FunctionGetOutputs(self, request: Any, context: Any) -> Any:
+        native = await self.scheduler.function_get_outputs(
+            function_call_id=request.function_call_id,
+            max_values=request.max_values or 256,
+            timeout=min(request.timeout or 55.0, 55.0),
+            clear_on_success=request.clear_on_success,
+        )
+        out = self.api.FunctionGetOutputsResponse()
+        record = self.scheduler.calls.get(request.function_call_id)
+        call_suffix = request.function_call_id[3:]
+        for item in native:
+            if item.get("group"):
+                values = pickle.loads(item["chunk_data"])
+                cis = item["cis"]
+                voff = item.get("val_off", 0)
+                base = item["idx_base"]
+                pairs = (
+                    enumerate(values) if cis is None else zip(cis, values[voff:])
+                )
+                for ci, value in pairs:
+                    o = out.outputs.add()
+                    o.idx = base + ci
+                    o.input_id = f"in-{call_suffix}-{base + ci}"
+                    o.result.status = (
+                        self.api.GenericResult.GenericStatus.GENERIC_STATUS_SUCCESS
+                    )
+                    o.result.data = pickle.dumps(value, 4)
+                    o.data_format = self.api.DATA_FORMAT_PICKLE
+                    out.idxs.append(o.idx)
+                continue
+            o = out.outputs.add()
+            o.idx = item.get("idx", 0)
+            o.input_id = f"in-{call_suffix}-{o.idx}"
+            o.retry_count = item.get("retry_count", 0)
+            o.result.status = item.get("status", 0)
+            data = item.get("data")
+            cid = item.get("out_chunk")
+            if cid:
+                # shared worker output chunk: bytes attach once per chunk per
+                # response; cache the decoded list for that chunk's siblings
+                values = self._out_chunk_cache.get(cid)
+                if values is None and item.get("chunk_data") is not None:
+                    values = pickle.loads(item["chunk_data"])
+                    self._out_chunk_cache[cid] = values
+                    while len(self._out_chunk_cache) > 64:
+                        self._out_chunk_cache.pop(next(iter(self._out_chunk_cache)))
+                if values is not None:
+                    data = pickle.dumps(values[item.get("out_ci", 0)], 4)
+            if data is not None:
+                o.result.data = data
+            if item.get("data_blob"):
+                o.result.data_blob_id = item["data_blob"]
+            if item.get("exc"):
+                o.result.exception = item["exc"]
+            o.data_format = self.api.DATA_FORMAT_PICKLE
+            out.idxs.append(o.idx)
+        if record is not None and record.num_inputs_final is not None:
+            out.num_unfinished_inputs = max(
+                0, record.num_inputs_final - record.completed
+            )
+        return out
+
+    async def FunctionCallCancel(self, request: Any, context: Any) -> Any:
+        from google.protobuf import empty_pb2
+
+        await self.scheduler.function_call_cancel(
+            function_call_id=request.function_call_id,
+            terminate_containers=request.terminate_containers,
+        )
+        return empty_pb2.Empty()
+
+    # -- Queue ---------------------------------------------------------------
+
+    async def QueueGetOrCreate(self, request: Any, context: Any) -> Any:
+        qid = await self.scheduler.queue_get_or_create(
+            name=request.deployment_name or None,
+            environment=request.environment_name or "main",
+            create_if_missing=True,
+            ephemeral=not request.deployment_name,
+        )
+        return self.api.QueueGetOrCreateResponse(queue_id=qid)
+
+    async def QueuePut(self, request: Any, context: Any) -> Any:
+        from google.protobuf import empty_pb2
+
+        await self.scheduler.queue_put(
+            queue_id=request.queue_id,
+            values=[bytes(v) for v in request.values],
+            partition=bytes(request.partition_key) or None,
+            block=False,
+            deadline=None,
+        )
+        return empty_pb2.Empty()
+
+    async def QueueGet(self, request: Any, context: Any) -> Any:
+        import time as _time
+
+        deadline = _time.time() + request.timeout if request.timeout else None
+        values = await self.scheduler.queue_get(
+            queue_id=request.queue_id,
+            partition=bytes(request.partition_key) or None,
+            n_values=request.n_values or 1,
+            block=bool(request.timeout),
+            deadline=deadline,
+        )
+        return self.api.QueueGetResponse(values=values)
+
+    async def QueueLen(self, request: Any, context: Any) -> Any:
+        n = self.scheduler.services.queue_len(
+            request.queue_id, bytes(request.partition_key) or None, request.total
+        )
+        return self.api.QueueLenResponse(len=n)
+
+    # -- Dict ----------------------------------------------------------------
+
+    async def DictGetOrCreate(self, request: Any, context: Any) -> Any:
+        did = await self.scheduler.dict_get_or_create(
+            name=request.deployment_name or None,
+            environment=request.environment_name or "main",
+            create_if_missing=True,
+            ephemeral=not request.deployment_name,
+        )
+        return self.api.DictGetOrCreateResponse(dict_id=did)
+
+    async def DictUpdate(self, request: Any, context: Any) -> Any:
+        updates = {bytes(e.key): bytes(e.value) for e in request.updates}
+        created = await self.scheduler.dict_update(
+            dict_id=request.dict_id,
+            updates=updates,
+            if_not_exists=request.if_not_exists,
+        )
+        return self.api.DictUpdateResponse(created=bool(created))
+
+    async def DictGet(self, request: Any, context: Any) -> Any:
+        value = await self.scheduler.dict_get(request.dict_id, bytes(request.key))
+        if value is None:
+            return self.api.DictGetResponse(found=False)
+        return self.api.DictGetResponse(found=True, value=value)
+
+    # -- Secret --------------------------------------------------------------
+
+    async def SecretGetOrCreate(self, request: Any, context: Any) -> Any:
+        sid = await self.scheduler.secret_get_or_create(
+            name=request.deployment_name or None,
+            env=dict(request.env_dict),
+            environment=request.environment_name or "main",
+        )
+        return self.api.SecretGetOrCreateResponse(secret_id=sid)
+
+    # -- Blob ----------------------------------------------------------------
+
+    async def BlobCreate(self, request: Any, context: Any) -> Any:
+        from ..utils.ids import new_id
+
+        blob_id = new_id("blob")
+        path = os.path.join(self._blob_dir, blob_id)
+        return self.api.BlobCreateResponse(
+            blob_id=blob_id, upload_url=f"file://{path}"
+        )
+
+    async def BlobGet(self, request: Any, context: Any) -> Any:
+        path = os.path.join(self._blob_dir, request.blob_id)
+        if not os.path.exists(path):
+            # fall through to the CAS (native blob ids are digests)
+            try:
+                path = await self.scheduler.blob_path(request.blob_id)
+            except Exception:
+                import grpc
+
+                await context.abort(grpc.StatusCode.NOT_FOUND, "blob not found")
+        return self.api.BlobGetResponse(download_url=f"file://{path}")
+
+    # -- helpers -------------------------------------------------------------
+
+    async def _abort_not_found(self, context: Any, exc: Exception) -> None:
+        import grpc
+
+        from ..exception import NotFoundError
+
+        if isinstance(exc, NotFoundError):
+            await context.abort(grpc.StatusCode.NOT_FOUND, str(exc))

@@ -199,11 +199,31 @@ class Scheduler:
             self.schedule_runner.start()
             self._persist_task = asyncio.get_running_loop().create_task(self._persist_loop())
             self._call_gc_task = asyncio.get_running_loop().create_task(self._call_gc_loop())
+            if os.environ.get("MODAL_AMD_GRPC") == "1":
+                await self.start_grpc_bridge()
             self._started = True
+
+    async def start_grpc_bridge(self) -> str:
+        """api.proto gRPC plane on <run_dir>/grpc.sock (proto/bridge.py);
+        lazy because grpc import costs ~0.5 s the fast path never pays."""
+        bridge = self._extra.get("grpc_bridge")
+        if bridge is None:
+            from ..proto.bridge import GrpcBridge
+
+            bridge = GrpcBridge(self)
+            await bridge.start()
+            self._extra["grpc_bridge"] = bridge
+        return bridge.socket_path
 
     async def stop(self) -> None:
         if not self._started:
             return
+        bridge = self._extra.pop("grpc_bridge", None)
+        if bridge is not None:
+            try:
+                await bridge.stop()
+            except Exception:
+                pass
         if getattr(self, "schedule_runner", None) is not None:
             self.schedule_runner.stop()
         if self._persist_task is not None:
