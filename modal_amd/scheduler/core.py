@@ -1330,12 +1330,19 @@ class Scheduler:
         }
         restore_image = kwargs.pop("restore_image_id", None)
         restore_blob = None
+        image_fsroot = None
         if restore_image:
             # image ids minted by sandbox_snapshot_fs carry their tar blob
             restore_blob = self._extra.get("snapshot_blobs", {}).get(restore_image)
+            # built images contribute their filesystem layer as an extra
+            # overlay lower for the sandbox root (isolation.py)
+            img = self.image_service.by_id.get(restore_image)
+            if img is not None and os.path.isdir(img.fsroot) and os.listdir(img.fsroot):
+                image_fsroot = img.fsroot
         return await self.sandbox_service.create(
             volume_paths=volume_paths,
             restore_blob=restore_blob,
+            image_fsroot=image_fsroot,
             blob_store=self.blob_store,
             **kwargs,
         )

@@ -176,9 +176,12 @@ def test_exit_hook_writes_land_before_volume_commit(client):
     assert b"".join(v2.read_file("final.txt")) == b"count=1"
 
 
-def test_image_shell_override(client, tmp_path):
+def test_image_shell_override(client, tmp_path, run_dir):
     """Image.shell() sets the SHELL for later run_commands (parity:
-    reference _image.py:1990)."""
+    reference _image.py:1990). With isolation on, absolute-path writes
+    land in the image's fs layer, so look there first."""
+    import os
+
     marker = tmp_path / "shellname.txt"
     img = (
         modal.Image.debian_slim()
@@ -186,4 +189,13 @@ def test_image_shell_override(client, tmp_path):
         .run_commands(f"echo $0 > {marker}")
     )
     img.hydrate()
-    assert "bash" in marker.read_text()
+    candidates = [str(marker)]
+    images_root = os.path.join(run_dir, "images")
+    if os.path.isdir(images_root):
+        candidates += [
+            os.path.join(images_root, d, "fsdiff", str(marker).lstrip("/"))
+            for d in os.listdir(images_root)
+        ]
+    hits = [p for p in candidates if os.path.exists(p)]
+    assert hits, "run_commands produced no marker file"
+    assert "bash" in open(hits[0]).read()
