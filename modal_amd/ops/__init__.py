@@ -58,6 +58,9 @@ def load_lib(required: bool = False) -> Optional[ctypes.CDLL]:
                 ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                 ctypes.c_long, ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
             ]
+            from .rawmem import _bind as _bind_rawmem
+
+            _bind_rawmem(lib)
             _lib = lib
             return lib
         except OSError as exc:

@@ -12,7 +12,7 @@ import sys
 
 _THIS_DIR = os.path.dirname(os.path.abspath(__file__))
 _CSRC = os.path.join(_THIS_DIR, "csrc")
-SOURCES = ["sha256.hip", "pack.hip", "lz4.hip"]
+SOURCES = ["sha256.hip", "pack.hip", "lz4.hip", "rawmem.hip"]
 
 
 def _hipcc() -> str:

@@ -187,6 +187,14 @@ class WorkerRPCTarget:
                 await asyncio.get_running_loop().run_in_executor(None, snap.restore)
             return {"state": "running"}
 
+    async def snapshot_state(self) -> bytes:
+        """Cross-process snapshot payload: registered tensors, tracked raw
+        hipMalloc allocations, RNG state (gpu_snapshot.capture_payload).
+        The scheduler stores it and can restore it into a FRESH worker."""
+        from .gpu_snapshot import capture_payload
+
+        return await asyncio.get_running_loop().run_in_executor(None, capture_payload)
+
 
 class FunctionRuntime:
     """Worker-side state for one registered function."""
@@ -388,6 +396,15 @@ class WorkerRuntime:
         global RUNTIME
         RUNTIME = self
         self.loop = asyncio.get_running_loop()
+        if os.environ.get("MODAL_AMD_RESTORE_STATE_PATH"):
+            # restored worker: busy-wait for restore-state.json, rehydrate
+            # tracked GPU state, exit 222 on failure (parity: reference
+            # task_lifecycle_manager.py:146-193 + sentinel contract)
+            from .gpu_snapshot import wait_and_restore_from_state_file
+
+            await asyncio.get_running_loop().run_in_executor(
+                None, wait_and_restore_from_state_file
+            )
         deadline = time.time() + (60.0 if self.external else 0.0)
         while True:
             try:

@@ -114,6 +114,7 @@ class RPCAdapter:
         "sandbox_from_name", "sandbox_snapshot_fs", "sandbox_fs_op", "sandbox_resize",
         "image_get_or_create", "image_info", "mount_get_or_create",
         "device_transfer", "tensor_pull_relay",
+        "worker_snapshot", "worker_restore", "start_grpc_bridge",
     }
 
     def __init__(self, scheduler: "Scheduler"):
@@ -1264,6 +1265,67 @@ class Scheduler:
             "world_size": world_size,
             "addrs": [state["members"].get(r, "") for r in range(world_size)],
         }
+
+    # -- worker GPU snapshots ---------------------------------------------
+    async def worker_snapshot(self, worker_id: int) -> dict:
+        """Capture a worker's snapshot-visible GPU state (registered
+        tensors + tracked raw hipMalloc allocations + RNG) into the CAS.
+        Parity: the snapshot half of ContainerCheckpoint
+        (reference task_lifecycle_manager.py:195)."""
+        w = self.pool.workers.get(int(worker_id))
+        if w is None:
+            raise NotFoundError(f"worker {worker_id} not connected")
+        data = await w.conn.call("snapshot_state", timeout=300)
+        blob_id = self.blob_store.put(data)
+        return {"snapshot_id": blob_id}
+
+    async def worker_restore(
+        self, snapshot_id: str, gpu_index: Optional[int] = None
+    ) -> dict:
+        """Spawn a FRESH worker that rehydrates a snapshot via the
+        restore-state.json contract; on sentinel exit 222 the spawn is
+        retried WITHOUT the snapshot (degraded, parity: reference
+        gpu_memory_snapshot.py:20-23)."""
+        import json as _json
+
+        state_path = os.path.join(self.run_dir, f"restore-{new_id('task')[3:]}.json")
+        state = {
+            "task_id": new_id("task"),
+            "snapshot_id": snapshot_id,
+            "snapshot_path": self.blob_store.open_path(snapshot_id),
+            "env": {},
+            "snapshot_debug": False,
+        }
+        with open(state_path, "w") as f:
+            _json.dump(state, f)
+        before = set(self.pool.workers)
+        proc = await self.pool.spawn_worker(
+            gpu_index=gpu_index,
+            extra_env={"MODAL_AMD_RESTORE_STATE_PATH": state_path},
+        )
+        deadline = time.time() + 120
+        while time.time() < deadline:
+            fresh = set(self.pool.workers) - before
+            if fresh:
+                return {"worker_id": fresh.pop(), "degraded": False}
+            rc = proc.poll() if proc is not None else None
+            if rc is not None:
+                # exit-222 contract: retry without the snapshot
+                self.pool._pending_spawns = max(0, self.pool._pending_spawns - 1)
+                before2 = set(self.pool.workers)
+                await self.pool.spawn_worker(gpu_index=gpu_index)
+                while time.time() < deadline:
+                    fresh = set(self.pool.workers) - before2
+                    if fresh:
+                        return {
+                            "worker_id": fresh.pop(),
+                            "degraded": True,
+                            "exit_code": rc,
+                        }
+                    await asyncio.sleep(0.1)
+                break
+            await asyncio.sleep(0.1)
+        raise InvalidError("restored worker never connected")
 
     # -- blobs -------------------------------------------------------------
     async def blob_put(self, data: bytes) -> dict:

@@ -261,7 +261,7 @@ class WorkerPool:
             pass
         return 0
 
-    async def spawn_worker(self, gpu_index: Optional[int] = None, extra_env: Optional[dict] = None) -> None:
+    async def spawn_worker(self, gpu_index: Optional[int] = None, extra_env: Optional[dict] = None) -> Any:
         """Spawn one worker process; it will connect back to our socket."""
         worker_id = self._next_worker_id
         self._next_worker_id += 1
@@ -283,6 +283,7 @@ class WorkerPool:
             start_new_session=True,
         )
         self._procs.append(proc)
+        return proc
 
     async def ensure_workers(self, needs_gpu: bool) -> None:
         """Lazily bring up the pool sized to the hardware (288 GB/GPU MI355X:

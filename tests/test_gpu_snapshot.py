@@ -1,0 +1,243 @@
+"""Process-grade GPU snapshot/restore (round-1 review Missing #4).
+
+CPU tests cover the payload format, the restore-state.json + exit-222
+contract, and the cross-worker flow with CPU tensors; the @gpu test proves
+the full bar: snapshot, kill the worker, restore into a fresh process —
+tensor contents AND a raw (non-torch) hipMalloc buffer both survive.
+"""
+
+from __future__ import annotations
+
+import os
+import time
+
+import pytest
+
+import modal_amd as modal
+from modal_amd._sync import synchronizer
+
+
+def test_capture_restore_payload_cpu():
+    import torch
+
+    from modal_amd.runtime import gpu_snapshot as gs
+
+    gs._registered_tensors.clear()
+    gs._restored_tensors.clear()
+    t = torch.arange(16, dtype=torch.float32).reshape(4, 4)
+    gs.register_tensor("weights", t)
+    import random
+
+    random.seed(1234)
+    payload = gs.capture_payload()
+
+    # simulate a fresh process: clear registries, perturb RNG
+    gs._registered_tensors.clear()
+    gs._restored_tensors.clear()
+    random.seed(999)
+    gs.restore_payload(payload)
+    out = gs.restored_tensor("weights")
+    assert torch.equal(out, t)
+    assert random.random() == (random.seed(1234) or random.random())
+
+
+def test_restore_state_file_contract(tmp_path, monkeypatch):
+    """Busy-wait for restore-state.json, apply env overrides, load the
+    snapshot payload (parity: task_lifecycle_manager.py:146-193)."""
+    import json
+    import threading
+
+    import torch
+
+    from modal_amd.runtime import gpu_snapshot as gs
+
+    gs._registered_tensors.clear()
+    gs._restored_tensors.clear()
+    gs.register_tensor("w", torch.ones(3))
+    snap_path = tmp_path / "snap.bin"
+    snap_path.write_bytes(gs.capture_payload())
+    gs._registered_tensors.clear()
+    gs._restored_tensors.clear()
+
+    state_path = tmp_path / "restore-state.json"
+    monkeypatch.setenv("MODAL_AMD_RESTORE_STATE_PATH", str(state_path))
+
+    def write_late():
+        time.sleep(0.3)
+        state_path.write_text(
+            json.dumps(
+                {
+                    "task_id": "ta-restored",
+                    "snapshot_path": str(snap_path),
+                    "env": {"RESTORED_MARKER": "yes"},
+                }
+            )
+        )
+
+    threading.Thread(target=write_late, daemon=True).start()
+    state = gs.wait_and_restore_from_state_file(timeout=10)
+    assert state["task_id"] == "ta-restored"
+    assert os.environ.get("RESTORED_MARKER") == "yes"
+    assert torch.equal(gs.restored_tensor("w"), torch.ones(3))
+
+
+def test_restore_failure_exits_with_sentinel(tmp_path, monkeypatch):
+    import json
+
+    from modal_amd.runtime import gpu_snapshot as gs
+
+    state_path = tmp_path / "restore-state.json"
+    state_path.write_text(json.dumps({"snapshot_path": str(tmp_path / "missing.bin")}))
+    monkeypatch.setenv("MODAL_AMD_RESTORE_STATE_PATH", str(state_path))
+    with pytest.raises(SystemExit) as err:
+        gs.wait_and_restore_from_state_file(timeout=5)
+    assert err.value.code == gs.CUDA_CHECKPOINT_SENTINEL_EXIT
+
+
+def test_worker_snapshot_restore_cross_process_cpu(client):
+    """snapshot -> kill worker -> restore into a FRESH worker process:
+    registered tensor state survives (CPU half of the contract; the raw
+    hipMalloc half is the @gpu test)."""
+    app = modal.App("snap-app")
+
+    @app.function()
+    def seed_state():
+        import os as _os
+
+        import torch
+
+        from modal_amd.runtime import gpu_snapshot as gs
+
+        gs.register_tensor("model-state", torch.arange(8, dtype=torch.float32) * 2)
+        return _os.getpid()
+
+    @app.function()
+    def read_state():
+        import os as _os
+
+        from modal_amd.runtime import gpu_snapshot as gs
+
+        t = gs.restored_tensor("model-state")
+        return (_os.getpid(), None if t is None else t.tolist())
+
+    with app.run(client=client):
+        victim_pid = seed_state.remote()
+        svc = client.svc
+        # find the worker that owns the state
+        victim = None
+        for wid, w in svc.pool.workers.items():
+            if getattr(w, "pid", None) == victim_pid or True:
+                pass
+        # snapshot every worker; the one with the tensor carries it
+        snaps = {}
+        for wid in list(svc.pool.workers):
+            resp = synchronizer.run(svc.worker_snapshot(wid))
+            snaps[wid] = resp["snapshot_id"]
+        # kill all current workers (worker death)
+        import signal
+
+        for proc in list(svc.pool._procs):
+            try:
+                os.kill(proc.pid, signal.SIGKILL)
+            except OSError:
+                pass
+        # restore each snapshot into fresh workers
+        restored_ids = []
+        for snap_id in snaps.values():
+            resp = synchronizer.run(svc.worker_restore(snap_id))
+            assert resp["degraded"] is False
+            restored_ids.append(resp["worker_id"])
+        # some restored worker must yield the tensor, from a NEW pid
+        deadline = time.time() + 30
+        seen = []
+        while time.time() < deadline:
+            pid, state = read_state.remote()
+            seen.append((pid, state))
+            if state is not None:
+                assert state == [0.0, 2.0, 4.0, 6.0, 8.0, 10.0, 12.0, 14.0]
+                assert pid != victim_pid
+                return
+            time.sleep(0.2)
+        raise AssertionError(f"restored state never observed; saw {seen[-5:]}")
+
+
+@pytest.mark.gpu
+def test_gpu_snapshot_kill_restore_end_to_end(client):
+    """The round-1 review's 'Done' bar: snapshot, kill worker, restore into
+    a fresh process; tensor contents AND a non-torch hipMalloc buffer both
+    survive."""
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    app = modal.App("snap-gpu")
+
+    @app.function(gpu=1)
+    def seed_gpu_state():
+        import os as _os
+
+        import torch
+
+        from modal_amd.ops.rawmem import RawDeviceBuffer
+        from modal_amd.runtime import gpu_snapshot as gs
+
+        t = torch.arange(1024, dtype=torch.bfloat16, device="cuda") * 3
+        gs.register_tensor("gpu-weights", t)
+        buf = RawDeviceBuffer("raw-scratch", 4096)
+        buf.write(bytes(range(256)) * 16)
+        gs._keepalive_raw = buf  # hold the allocation
+        return _os.getpid()
+
+    @app.function(gpu=1)
+    def read_gpu_state():
+        import os as _os
+
+        import torch
+
+        from modal_amd.ops import rawmem
+        from modal_amd.runtime import gpu_snapshot as gs
+
+        t = gs.restored_tensor("gpu-weights")
+        raw = rawmem.lookup("raw-scratch")
+        raw_ok = None
+        if raw is not None:
+            lib_data = rawmem.snapshot_all().get("raw-scratch")
+            raw_ok = lib_data == bytes(range(256)) * 16
+        return (
+            _os.getpid(),
+            None if t is None else (t.is_cuda, t.float().sum().item()),
+            raw_ok,
+        )
+
+    with app.run(client=client):
+        victim_pid = seed_gpu_state.remote()
+        svc = client.svc
+        snaps = [
+            synchronizer.run(svc.worker_snapshot(wid))["snapshot_id"]
+            for wid in list(svc.pool.workers)
+        ]
+        import signal
+
+        for proc in list(svc.pool._procs):
+            try:
+                os.kill(proc.pid, signal.SIGKILL)
+            except OSError:
+                pass
+        for snap_id in snaps:
+            resp = synchronizer.run(svc.worker_restore(snap_id, gpu_index=0))
+            assert resp["degraded"] is False
+        deadline = time.time() + 60
+        while time.time() < deadline:
+            pid, tensor_state, raw_ok = read_gpu_state.remote()
+            if tensor_state is not None:
+                is_cuda, total = tensor_state
+                assert is_cuda
+                expected = float(
+                    (torch.arange(1024, dtype=torch.bfloat16) * 3).float().sum()
+                )
+                assert abs(total - expected) < 1e-3
+                assert raw_ok is True
+                assert pid != victim_pid
+                return
+            time.sleep(0.2)
+        raise AssertionError("restored GPU state never observed")
