@@ -235,19 +235,19 @@ def capture_payload() -> bytes:
     from ..ops import rawmem
 
     tensors: dict[str, tuple] = {}
-    try:
+    if _registered_tensors:
         import torch
 
         for key, t in _registered_tensors.items():
-            host = t.detach().cpu()
+            host = t.detach().cpu().contiguous()
+            # byte-level extraction: numpy() would reject bf16/fp8 dtypes
+            raw = host.reshape(-1).view(torch.uint8).numpy().tobytes() if host.numel() else b""
             tensors[key] = (
-                host.numpy().tobytes(),
+                raw,
                 str(t.dtype).removeprefix("torch."),
                 tuple(t.shape),
                 t.is_cuda,
             )
-    except Exception:
-        pass
     payload = {
         "version": 1,
         "tensors": tensors,
@@ -270,8 +270,12 @@ def restore_payload(data: bytes) -> dict:
         import torch
 
         for key, (raw, dtype_s, shape, was_cuda) in payload.get("tensors", {}).items():
-            t = torch.frombuffer(bytearray(raw), dtype=getattr(torch, dtype_s))
-            t = t.reshape(shape) if shape else t.reshape(())
+            dtype = getattr(torch, dtype_s)
+            if raw:
+                t = torch.frombuffer(bytearray(raw), dtype=torch.uint8).view(dtype)
+                t = t.reshape(shape) if shape else t.reshape(())
+            else:
+                t = torch.empty(shape or (0,), dtype=dtype)
             if was_cuda and torch.cuda.is_available():
                 t = t.cuda()
             _restored_tensors[key] = t

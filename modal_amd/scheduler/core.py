@@ -1298,30 +1298,27 @@ class Scheduler:
         }
         with open(state_path, "w") as f:
             _json.dump(state, f)
-        before = set(self.pool.workers)
+        # spawn_worker stamps MODAL_AMD_WORKER_ID from _next_worker_id, so
+        # the restored worker's id is known a priori (autoscaled respawns
+        # racing this cannot be confused with it)
+        wid = self.pool._next_worker_id
         proc = await self.pool.spawn_worker(
             gpu_index=gpu_index,
             extra_env={"MODAL_AMD_RESTORE_STATE_PATH": state_path},
         )
         deadline = time.time() + 120
         while time.time() < deadline:
-            fresh = set(self.pool.workers) - before
-            if fresh:
-                return {"worker_id": fresh.pop(), "degraded": False}
+            if wid in self.pool.workers:
+                return {"worker_id": wid, "degraded": False}
             rc = proc.poll() if proc is not None else None
             if rc is not None:
                 # exit-222 contract: retry without the snapshot
                 self.pool._pending_spawns = max(0, self.pool._pending_spawns - 1)
-                before2 = set(self.pool.workers)
+                wid2 = self.pool._next_worker_id
                 await self.pool.spawn_worker(gpu_index=gpu_index)
                 while time.time() < deadline:
-                    fresh = set(self.pool.workers) - before2
-                    if fresh:
-                        return {
-                            "worker_id": fresh.pop(),
-                            "degraded": True,
-                            "exit_code": rc,
-                        }
+                    if wid2 in self.pool.workers:
+                        return {"worker_id": wid2, "degraded": True, "exit_code": rc}
                     await asyncio.sleep(0.1)
                 break
             await asyncio.sleep(0.1)
