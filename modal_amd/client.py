@@ -166,6 +166,20 @@ class _Client:
         return getattr(sched, "run_dir", None)
 
     @property
+    def xfer_dir(self) -> Optional[str]:
+        """One-shot payload handoff directory (same filesystem on all
+        sides): spilled map chunks are TRANSPORT, not content-addressed
+        storage — a direct file handoff skips hashing and compression
+        entirely (round-1 review Weak #3: the per-chunk CAS spill did
+        per-chunk GPU round trips at 0.54 GiB/s)."""
+        run_dir = self.run_dir
+        if not run_dir:
+            return None
+        path = os.path.join(run_dir, "xfer")
+        os.makedirs(path, exist_ok=True)
+        return path
+
+    @property
     def blob_store(self) -> Any:
         """The shared content-addressed store (same filesystem on all sides)."""
         if getattr(self, "_blob_store", None) is None:

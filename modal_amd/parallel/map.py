@@ -138,12 +138,12 @@ async def map_invocation_batches(
             count = len(chunk_buf)
             chunk_buf = []
             if len(payload) > 2 * 1024 * 1024:
-                # big-item chunks spill to the CAS (parity: the 2 MiB inline
-                # payload limit, blob_utils.py:36) — workers read from the
-                # shared store instead of the wire
-                store = client.blob_store
-                if store is not None:
-                    payload = {"blob": store.put(payload)}
+                # big-item chunks spill off the wire (parity: the 2 MiB
+                # inline payload limit, blob_utils.py:36) — workers read
+                # from the shared filesystem instead of the socket
+                payload = await asyncio.get_running_loop().run_in_executor(
+                    None, _spill, payload, chunk_id
+                )
             await svc.function_put_chunk(
                 function_call_id=call_id,
                 chunk_id=chunk_id,
@@ -164,6 +164,22 @@ async def map_invocation_batches(
                     approx_bytes += len(a)
             total_inputs += 1
 
+        def _spill(payload: bytes, chunk_id: str) -> Any:
+            """Oversized chunk payloads leave the control socket. Transport
+            spill = one-shot file handoff (no hash, no compress — the
+            scheduler unlinks it when the chunk completes); CAS fallback
+            when no shared run_dir exists."""
+            xfer = client.xfer_dir
+            if xfer is not None:
+                path = os.path.join(xfer, chunk_id)
+                with open(path, "wb") as f:
+                    f.write(payload)
+                return {"xfer": path}
+            store = client.blob_store
+            if store is not None:
+                return {"blob": store.put(payload)}
+            return payload
+
         async def flush_args_chunk(argsbatch: list) -> None:
             # "C2" wire form: common kwargs factored out, args list built by
             # C-level zip+islice — no per-item Python in the pump at all
@@ -174,9 +190,9 @@ async def map_invocation_batches(
             chunk_id = f"{call_id}.c{chunk_seq}"
             chunk_seq += 1
             if len(payload) > 2 * 1024 * 1024:
-                store = client.blob_store
-                if store is not None:
-                    payload = {"blob": store.put(payload)}
+                payload = await asyncio.get_running_loop().run_in_executor(
+                    None, _spill, payload, chunk_id
+                )
             await svc.function_put_chunk(
                 function_call_id=call_id,
                 chunk_id=chunk_id,

@@ -661,8 +661,14 @@ class WorkerRuntime:
                 decoded = entry["decoded"]
                 if decoded is None:
                     raw = entry["raw"]
-                    if isinstance(raw, dict):  # CAS-spilled chunk (>2 MiB)
-                        raw = self.blob_store.get(raw["blob"])
+                    if isinstance(raw, dict):  # spilled chunk (>2 MiB)
+                        if raw.get("xfer"):
+                            # one-shot file handoff (scheduler unlinks it
+                            # when the chunk completes)
+                            with open(raw["xfer"], "rb") as f:
+                                raw = f.read()
+                        else:
+                            raw = self.blob_store.get(raw["blob"])
                     obj = deserialize(raw)
                     if obj[0] == "C":
                         decoded = (obj[1], None)  # list of (args, kwargs)

@@ -72,13 +72,23 @@ class BlobStore:
         self._store(digest, data)
         return digest
 
+    PIPELINE_MIN_BYTES = 96 * 1024 * 1024  # pipelined windows pay above this
+
     def put_many(self, buffers: list) -> list[str]:
         """Batched put: one GPU hash dispatch over all buffers' leaves AND
         one batched compression pass over the new large blocks (the
-        volume-upload hot path; ops/hashing.content_digests_batch +
-        ops/compress.compress_buffers)."""
+        volume-upload hot path). Large uploads take the PIPELINED path
+        (ops/pipeline.py): H2D / digest+compress kernels / D2H / container
+        assembly + CAS writes overlapped across 64 MiB windows."""
         from ..ops import gpu_available
         from ..ops.hashing import GPU_MIN_BYTES, content_digests_batch
+
+        total_bytes = sum(len(b) for b in buffers)
+        if total_bytes >= self.PIPELINE_MIN_BYTES and gpu_available():
+            try:
+                return self._put_many_pipelined(buffers)
+            except Exception:
+                pass  # serial fallback below
 
         # stage once: the digest and compression passes share one pinned
         # H2D of all buffers (LEAF_SIZE alignment satisfies both kernels)
@@ -119,6 +129,21 @@ class BlobStore:
                 self._store_prepared(digest, compressed[i], True)
             else:
                 self._store_prepared(digest, bytes(data), False)
+        return digests
+
+    def _put_many_pipelined(self, buffers: list) -> list[str]:
+        from ..ops.pipeline import hash_compress_blocks
+
+        def store(i: int, digest: str, payload: bytes, compressed: bool) -> None:
+            if self.has(digest):
+                return
+            if len(buffers[i]) < self.COMPRESS_MIN and compressed:
+                # parity with the serial path: tiny blocks store raw
+                self._store_prepared(digest, buffers[i], False)
+                return
+            self._store_prepared(digest, payload, compressed)
+
+        digests, _containers = hash_compress_blocks(buffers, compress=True, store=store)
         return digests
 
     def _store_prepared(self, digest: str, payload: bytes, compressed: bool = False) -> None:

@@ -219,6 +219,22 @@ class CallRecord:
         self.gen_queues: dict[int, asyncio.Queue] = {}
         self.done_event = asyncio.Event()
 
+
+    def _drop_chunk(self, chunk_id: str) -> None:
+        """Release a fully-consumed input chunk; one-shot xfer spill files
+        (parallel/map.py _spill) are unlinked here."""
+        chunk = self.chunks.pop(chunk_id, None)
+        if chunk is None:
+            return
+        data = chunk.get("data")
+        if isinstance(data, dict) and data.get("xfer"):
+            import os
+
+            try:
+                os.unlink(data["xfer"])
+            except OSError:
+                pass
+
     # -- input intake ----------------------------------------------------
     def add_input(
         self,
@@ -280,7 +296,7 @@ class CallRecord:
         if chunk is not None:
             chunk["refs"] -= n
             if chunk["refs"] <= 0:
-                del self.chunks[group.chunk_id]
+                self._drop_chunk(group.chunk_id)
         self.completed += n
         if cis is None:
             group.state = "done"
@@ -336,7 +352,7 @@ class CallRecord:
             if chunk is not None:
                 chunk["refs"] -= 1
                 if chunk["refs"] <= 0:
-                    del self.chunks[rec.chunk_id]
+                    self._drop_chunk(rec.chunk_id)
         self.completed += 1
         self.output_ready.put_nowait(idx)
         waiter = self._waiters.pop(idx, None)
@@ -365,7 +381,7 @@ class CallRecord:
                 if chunk is not None:
                     chunk["refs"] -= 1
                     if chunk["refs"] <= 0:
-                        del self.chunks[rec.chunk_id]
+                        self._drop_chunk(rec.chunk_id)
             ready.append(idx)
             waiter = self._waiters.pop(idx, None)
             if waiter is not None and not waiter.done():

@@ -163,8 +163,12 @@ class _Volume(_Object, type_kind="volume"):
         import asyncio
 
         store = self._client.blob_store
+        # memoryview slices: no second copy of the whole payload (every
+        # consumer — np.frombuffer staging, hashlib, file writes — takes
+        # buffer objects)
+        mv = memoryview(data)
         blocks = [
-            bytes(data[off : off + BLOCK_SIZE]) for off in range(0, max(len(data), 1), BLOCK_SIZE)
+            mv[off : off + BLOCK_SIZE] for off in range(0, max(len(data), 1), BLOCK_SIZE)
         ]
         loop = asyncio.get_running_loop()
         # same-node fast path: stage the raw file next to the volume tree so

@@ -28,3 +28,15 @@ t5=time.perf_counter(); print(f"full upload: {t5-t4:.3f}s = {256/1024/(t5-t4):.3
 with vol.batch_upload(force=True) as b:
     b.put_file(io.BytesIO(blob), "/p2.bin")
 t6=time.perf_counter(); print(f"re-upload same blocks (CAS hit): {t6-t5:.3f}s")
+
+# warm path: fresh content, pipeline slots already allocated
+blob2 = (os.urandom(1024) + b"\x01"*3072) * (256*256)
+t7=time.perf_counter()
+with vol.batch_upload(force=True) as b:
+    b.put_file(io.BytesIO(blob2), "/p3.bin")
+t8=time.perf_counter(); print(f"warm upload 256 MiB: {t8-t7:.3f}s = {256/1024/(t8-t7):.3f} GiB/s")
+blob3 = (os.urandom(1024) + b"\x02"*3072) * (4*256*256)  # 1 GiB
+t9=time.perf_counter()
+with vol.batch_upload(force=True) as b:
+    b.put_file(io.BytesIO(blob3), "/p4.bin")
+t10=time.perf_counter(); print(f"warm upload 1 GiB: {t10-t9:.3f}s = {1024/1024/(t10-t9):.3f} GiB/s")

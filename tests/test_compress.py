@@ -159,3 +159,37 @@ def test_put_many_batched_compression_roundtrip():
         assert store.get(digest) == data
     # compressible blocks got stored as MALZ41 (smaller on disk)
     assert store.size(digests[0]) < len(blocks[0]) // 2
+
+
+@pytest.mark.gpu
+def test_pipelined_put_many_bit_identical():
+    """The pipelined window path (ops/pipeline.py) produces the same CAS
+    keys and round-trips the same bytes as the serial path."""
+    import tempfile
+
+    torch = pytest.importorskip("torch")
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from modal_amd.scheduler.blobs import BlobStore
+
+    BLOCK = 8 * 1024 * 1024
+    blocks = []
+    for i in range(20):  # 160 MiB > PIPELINE_MIN_BYTES
+        if i % 3 == 0:
+            blocks.append(os.urandom(BLOCK))  # incompressible
+        else:
+            blocks.append((b"compressible %d " % i) * (BLOCK // 16))
+    blocks.append(b"tail-block" * 1000)  # small odd-size tail
+
+    store_a = BlobStore(tempfile.mkdtemp())
+    # call the pipelined path DIRECTLY so a failure can't silently fall
+    # back to the serial path
+    digests_pipelined = store_a._put_many_pipelined(blocks)
+
+    store_b = BlobStore(tempfile.mkdtemp())
+    store_b.PIPELINE_MIN_BYTES = 1 << 60  # force the serial path
+    digests_serial = store_b.put_many(blocks)
+
+    assert digests_pipelined == digests_serial
+    for digest, block in zip(digests_pipelined, blocks):
+        assert store_a.get(digest) == block

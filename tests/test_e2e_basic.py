@@ -299,3 +299,27 @@ def test_nested_big_args_spill_to_cas(client):
 
     with app.run(client=client):
         assert orchestrate.remote(3 * 1024 * 1024) == "ok"
+
+
+def test_big_args_xfer_path(client, run_dir):
+    """>2 MiB chunk payloads ride the one-shot file handoff (no hashing,
+    no compression) and are unlinked when the chunk completes."""
+    import hashlib
+    import os
+
+    app = modal.App("bigargs")
+
+    @app.function()
+    def digest(blob):
+        import hashlib as h
+
+        return h.sha256(blob).hexdigest()
+
+    payloads = [os.urandom(1024 * 1024) for _ in range(12)]
+    expected = sorted(hashlib.sha256(p).hexdigest() for p in payloads)
+    with app.run(client=client):
+        outs = sorted(digest.map(payloads, order_outputs=False))
+    assert outs == expected
+    xfer_dir = os.path.join(run_dir, "xfer")
+    leftovers = os.listdir(xfer_dir) if os.path.isdir(xfer_dir) else []
+    assert leftovers == [], f"xfer files leaked: {leftovers}"
