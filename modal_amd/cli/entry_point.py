@@ -314,7 +314,20 @@ def app_stop(app_id: str) -> None:
 
 @app_cli.command(name="logs")
 @click.argument("app_id")
-def app_logs(app_id: str) -> None:
+@click.option("-f", "--follow", is_flag=True, help="Tail live (offset-resumable long-poll).")
+@click.option("--timeout", type=float, default=None, help="Stop following after N seconds.")
+def app_logs(app_id: str, follow: bool, timeout: float) -> None:
+    """Print an app's logs; with -f, tail them live — including from a
+    daemon-hosted app over the attach socket (parity: `modal app logs`)."""
+    if follow:
+        from ..logs_manager import tail_app_logs
+
+        try:
+            for entry in tail_app_logs(app_id, timeout=timeout):
+                click.echo(entry.get("data", ""), nl=False)
+        except KeyboardInterrupt:
+            pass
+        return
     from ..logs_manager import fetch_app_logs
 
     for entry in fetch_app_logs(app_id):

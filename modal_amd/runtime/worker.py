@@ -859,6 +859,9 @@ class WorkerRuntime:
             prefix = f"in-{call_id[3:]}-"
             plain_call = not _slot_streams_enabled()
             tok_c = _current_function_call_id.set(call_id)
+            # executor threads have their OWN context: the app id must be
+            # set here for user prints to route to the right app's logs
+            tok_a = _app_id_var.set(frt.app_id)
             self._register_sync_thread(msg["token"])
             try:
                 for ci in range(start, end):
@@ -884,6 +887,7 @@ class WorkerRuntime:
             finally:
                 self._unregister_sync_thread(msg["token"])
                 _current_function_call_id.reset(tok_c)
+                _app_id_var.reset(tok_a)
 
         from .execution_context import _current_function_call_id, _current_input_id
 
@@ -1313,7 +1317,9 @@ class WorkerRuntime:
     def post_log(self, fd: int, data: str) -> None:
         if self.loop is None or self.conn is None or self.conn.closed:
             return
-        self._log_buffer.append((fd, data))
+        # capture the app id HERE (the writer's context — executor threads
+        # carry it); the flush task's context would read an empty default
+        self._log_buffer.append((fd, data, _app_id_var.get()))
         if not self._log_flush_scheduled:
             self._log_flush_scheduled = True
             try:
@@ -1329,11 +1335,10 @@ class WorkerRuntime:
         buf, self._log_buffer = self._log_buffer, []
         if not buf:
             return
-        app_id = _app_id_var.get()
-        by_fd: dict[int, list[str]] = {}
-        for fd, data in buf:
-            by_fd.setdefault(fd, []).append(data)
-        for fd, chunks in by_fd.items():
+        by_key: dict[tuple[int, str], list[str]] = {}
+        for fd, data, app_id in buf:
+            by_key.setdefault((fd, app_id), []).append(data)
+        for (fd, app_id), chunks in by_key.items():
             asyncio.get_running_loop().create_task(
                 self._send_log(fd, "".join(chunks), app_id)
             )

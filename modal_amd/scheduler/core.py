@@ -85,6 +85,8 @@ class AppState:
         self.deployment_name: Optional[str] = None
         self.logs: deque = deque(maxlen=10_000)
         self.log_subscribers: list[asyncio.Queue] = []
+        self.log_seq = 0  # total entries EVER appended (offset-resume base)
+        self.log_event = asyncio.Event()  # pulsed on each append (long-poll)
 
 
 class RPCAdapter:
@@ -114,7 +116,7 @@ class RPCAdapter:
         "sandbox_from_name", "sandbox_snapshot_fs", "sandbox_fs_op", "sandbox_resize",
         "image_get_or_create", "image_info", "mount_get_or_create",
         "device_transfer", "tensor_pull_relay",
-        "worker_snapshot", "worker_restore", "start_grpc_bridge",
+        "worker_snapshot", "worker_restore", "start_grpc_bridge", "app_get_logs",
     }
 
     def __init__(self, scheduler: "Scheduler"):
@@ -1126,6 +1128,8 @@ class Scheduler:
         app = self.apps.get(app_id)
         if app is not None:
             app.logs.append(entry)
+            app.log_seq += 1
+            app.log_event.set()
             for sub in app.log_subscribers:
                 sub.put_nowait(entry)
         else:
@@ -1264,6 +1268,43 @@ class Scheduler:
             "rank": rank,
             "world_size": world_size,
             "addrs": [state["members"].get(r, "") for r in range(world_size)],
+        }
+
+    async def app_get_logs(
+        self,
+        app_id: str,
+        offset: int = 0,
+        timeout: float = 55.0,
+        max_entries: int = 1000,
+    ) -> dict:
+        """Offset-resumable log long-poll (parity: the reference's
+        reconnecting deadline'd AppGetLogs streams, _logs_manager.py:24-26).
+
+        ``offset`` is an absolute sequence number; entries older than the
+        ring keeps are reported as ``dropped``. Works over the proxy
+        transport, so a daemon-attached CLI can tail live."""
+        app = self._app(app_id)
+        deadline = time.time() + min(timeout, OUTPUT_POLL_TIMEOUT)
+        while app.log_seq <= offset:
+            remaining = deadline - time.time()
+            if remaining <= 0 or app.state == "stopped":
+                return {"entries": [], "next_offset": offset, "dropped": 0,
+                        "app_state": app.state}
+            app.log_event.clear()
+            try:
+                await asyncio.wait_for(app.log_event.wait(), remaining)
+            except asyncio.TimeoutError:
+                pass
+        oldest = app.log_seq - len(app.logs)
+        start = max(offset, oldest)
+        dropped = start - offset
+        skip = start - oldest
+        entries = list(app.logs)[skip : skip + max_entries]
+        return {
+            "entries": entries,
+            "next_offset": start + len(entries),
+            "dropped": dropped,
+            "app_state": app.state,
         }
 
     # -- worker GPU snapshots ---------------------------------------------
