@@ -39,8 +39,7 @@ def test_queue_exceptions_are_stdlib_subclasses(client):
     with modal.Queue.ephemeral() as q:
         with pytest.raises(stdlib_queue.Empty):
             q.get(timeout=0.05)
-        for i in range(5000):
-            q.put(i)
+        q.put_many(list(range(5000)))
         with pytest.raises(stdlib_queue.Full):
             q.put(5001, timeout=0.05)
 
