@@ -126,7 +126,31 @@ class Pickler(cloudpickle.CloudPickler):
         return None
 
 
+#: module-path redirects for payloads produced by OTHER serializers —
+#: reference clients pickle functions with globals rooted at their vendored
+#: cloudpickle (modal._vendor.cloudpickle); map those onto the installed
+#: cloudpickle, whose _make_function/_make_skeleton_class protocol matches
+_MODULE_COMPAT = {
+    "modal._vendor.cloudpickle": "cloudpickle.cloudpickle",
+    "modal._vendor.cloudpickle_ext": "cloudpickle.cloudpickle",
+}
+
+
 class Unpickler(pickle.Unpickler):
+    def find_class(self, module: str, name: str) -> Any:
+        mapped = _MODULE_COMPAT.get(module)
+        if mapped is not None:
+            import importlib
+
+            for candidate in (mapped, "cloudpickle"):
+                try:
+                    mod = importlib.import_module(candidate)
+                    if hasattr(mod, name):
+                        return getattr(mod, name)
+                except ImportError:
+                    continue
+        return super().find_class(module, name)
+
     def persistent_load(self, pid: Any) -> Any:
         tag = pid[0]
         if tag == "modal-amd-object":

@@ -109,3 +109,152 @@ class FlashAutoscaler:
             self._pending_desired = None
             return applied
         return current_replicas
+
+
+def scrape_prometheus_metric(url: str, metric_name: str, timeout: float = 2.0) -> Optional[float]:
+    """Sum of a metric's samples from one Prometheus text endpoint
+    (parity: the reference autoscaler's per-container metrics scrape,
+    experimental/flash.py:522)."""
+    import urllib.request
+
+    try:
+        with urllib.request.urlopen(url, timeout=timeout) as resp:
+            text = resp.read().decode()
+    except Exception:
+        return None
+    total = None
+    for line in text.splitlines():
+        if line.startswith("#") or not line.startswith(metric_name):
+            continue
+        # "<name>{labels} <value>" or "<name> <value>"
+        head = line.split("{", 1)[0].split(None, 1)[0]
+        if head != metric_name:
+            continue
+        try:
+            value = float(line.rsplit(None, 1)[1])
+        except (ValueError, IndexError):
+            continue
+        total = value if total is None else total + value
+    return total
+
+
+class FlashAutoscalerLoop:
+    """The production wiring the round-1 review flagged as missing: a loop
+    that feeds FlashAutoscaler.decide() and APPLIES its decisions on the
+    worker pool (parity: reference experimental/flash.py:281-641 —
+    scrape -> decide -> set replica count).
+
+    Metric sources, in precedence order:
+      * ``metric_urls`` + ``metric_name``: mean of a Prometheus metric
+        scraped across replica endpoints
+      * ``get_metric`` callable
+      * default: the function's backlog per live worker (the scheduler's
+        own signal — no HTTP needed)
+    """
+
+    def __init__(
+        self,
+        scheduler: "object",
+        function_id: str,
+        target_value: float,
+        *,
+        metric_urls: Optional[list] = None,
+        metric_name: Optional[str] = None,
+        get_metric: Optional[Callable[[], float]] = None,
+        interval: float = 2.0,
+        min_replicas: int = 1,
+        max_replicas: int = 8,
+        scale_up_stabilization: float = 0.0,
+        scale_down_stabilization: float = 60.0,
+        tolerance: float = 0.1,
+    ):
+        self.scheduler = scheduler
+        self.function_id = function_id
+        self.interval = interval
+        self._task = None
+
+        def _metric() -> float:
+            if metric_urls and metric_name:
+                values = [
+                    v
+                    for v in (
+                        scrape_prometheus_metric(u, metric_name) for u in metric_urls
+                    )
+                    if v is not None
+                ]
+                return sum(values) / len(values) if values else 0.0
+            if get_metric is not None:
+                return get_metric()
+            pool = scheduler.pool
+            backlog = len(pool.pending.get(function_id, ()))
+            inflight = sum(
+                w.outstanding.get(function_id, 0) for w in pool.workers.values()
+            )
+            replicas = max(1, self.current_replicas())
+            return (backlog + inflight) / replicas
+
+        self.autoscaler = FlashAutoscaler(
+            _metric,
+            target_value,
+            min_replicas=min_replicas,
+            max_replicas=max_replicas,
+            scale_up_stabilization=scale_up_stabilization,
+            scale_down_stabilization=scale_down_stabilization,
+            tolerance=tolerance,
+        )
+
+    def current_replicas(self) -> int:
+        fdef = self.scheduler.functions.get(self.function_id)
+        pool = self.scheduler.pool
+        return sum(
+            1
+            for w in pool.workers.values()
+            if w.alive and not w.draining and (fdef is None or w.has_gpu or not fdef.needs_gpu)
+        )
+
+    async def tick(self, now: Optional[float] = None) -> int:
+        """One scrape->decide->apply round. Returns the applied count."""
+        current = self.current_replicas()
+        desired = self.autoscaler.decide(current, now)
+        if desired > current:
+            fdef = self.scheduler.functions.get(self.function_id)
+            gpu = 0 if (fdef is not None and fdef.needs_gpu) else None
+            for _ in range(desired - current):
+                await self.scheduler.pool.spawn_worker(gpu_index=gpu)
+        elif desired < current:
+            pool = self.scheduler.pool
+            victims = [
+                w
+                for w in pool.workers.values()
+                if w.alive and not w.draining and not w.inflight
+            ][: current - desired]
+            for w in victims:
+                w.draining = True
+                try:
+                    await w.conn.send({"t": "shutdown"})
+                except Exception:
+                    pass
+        return desired
+
+    async def _loop(self) -> None:
+        import asyncio
+
+        while True:
+            try:
+                await self.tick()
+            except asyncio.CancelledError:
+                raise
+            except Exception:
+                pass
+            await asyncio.sleep(self.interval)
+
+    def start(self) -> None:
+        import asyncio
+
+        if self._task is None:
+            self._task = asyncio.get_running_loop().create_task(self._loop())
+
+    def stop(self) -> None:
+        if self._task is not None:
+            self._task.cancel()
+            self._task = None

@@ -174,3 +174,49 @@ def test_shared_buf_tensor_roundtrip():
     # not another copy of the buffer
     values2 = values[:2] * 8
     assert len(pickle.dumps(values2, 4)) < len(buf) + 16 * 64
+
+
+def test_reference_cloudpickle_fixtures():
+    """Payloads produced by the REFERENCE's vendored cloudpickle (fixture
+    dir generated by make_ref_pickle_fixtures.py) deserialize correctly
+    through modal_amd's deserializer (round-1 review Missing #7 /
+    SURVEY hard part 6: wire formats are language/serializer-neutral)."""
+    import os
+
+    fixtures = os.path.join(os.path.dirname(__file__), "fixtures", "ref_pickles")
+    if not os.path.isdir(fixtures):
+        pytest.skip("fixture dir not generated")
+
+    def load(name):
+        with open(os.path.join(fixtures, name), "rb") as f:
+            return deserialize(f.read())
+
+    assert load("closure_fn.pkl")(6) == 43          # 6*7+1: closure survived
+    assert load("lambda.pkl")(5) == 15
+    exc = load("exception.pkl")
+    assert isinstance(exc, ValueError) and "outer-message" in str(exc)
+    assert isinstance(exc.__cause__, KeyError)
+    args, kwargs = load("args_kwargs.pkl")
+    assert args == (1, "two", b"three") and kwargs == {"k": [4, 5]}
+    rec = load("recursive.pkl")
+    assert rec[0] is rec[1] and rec[2] is rec       # shared refs + cycle kept
+    model = load("class_instance.pkl")
+    assert model.predict(4) == 12
+    np_payload = load("numpy.pkl")
+    assert np_payload["arr"].tolist() == [[0, 1, 2], [3, 4, 5]]
+    assert load("plain_protocol4.pkl") == {"items_total": 5}
+
+
+def test_reference_args_form_runs_through_worker(client):
+    """A reference-serialized (args, kwargs) payload executes end-to-end
+    (deserialize_payload accepts the bare 2-tuple wire form)."""
+    from modal_amd._serialization import deserialize_payload
+
+    import os
+
+    fixtures = os.path.join(os.path.dirname(__file__), "fixtures", "ref_pickles")
+    if not os.path.isdir(fixtures):
+        pytest.skip("fixture dir not generated")
+    with open(os.path.join(fixtures, "args_kwargs.pkl"), "rb") as f:
+        args, kwargs = deserialize_payload(f.read())
+    assert args == (1, "two", b"three") and kwargs == {"k": [4, 5]}
