@@ -305,9 +305,27 @@ class App:
 
         return decorator
 
-    def server(self, **kwargs: Any) -> Callable:
-        """Register a server class (reference app.py:1280). Maps to cls()."""
-        return self.cls(**kwargs)
+    def server(
+        self,
+        _warn_parentheses_missing: Any = None,
+        *,
+        port: int = 8000,
+        startup_timeout: float = 30.0,
+        **function_kwargs: Any,
+    ) -> Callable:
+        """Register a server class (reference app.py:1280 + _server.py):
+        the class's @modal.enter starts an HTTP server on ``port``; the
+        returned handle's start() waits for readiness (``startup_timeout``)
+        inside the worker, then ``url`` serves on 127.0.0.1."""
+        if _warn_parentheses_missing is not None:
+            raise InvalidError("Use @app.server() with parentheses")
+
+        def decorator(user_cls: type) -> Any:
+            from .server import make_server
+
+            return make_server(self, user_cls, port, startup_timeout, function_kwargs)
+
+        return decorator
 
     def include(self, other: "App") -> "App":
         """Merge another app's registrations (reference app.py:1480)."""

@@ -168,3 +168,59 @@ def test_batched_actually_groups(client):
         out = list(sizes.map(range(64), order_outputs=False))
         assert len(out) == 64
         assert max(out) > 1, "no batching happened"
+
+
+def test_app_server_lifecycle(client):
+    """@app.server: @enter starts an HTTP process; start() probes the port
+    inside the worker; the URL serves; @exit tears the process down
+    (round-1 review: Server was a bare Cls alias)."""
+    import urllib.request
+
+    import modal_amd as modal
+
+    app = modal.App("server-app")
+
+    @app.server(port=18431, startup_timeout=20)
+    class Httpd:
+        @modal.enter()
+        def boot(self):
+            import subprocess
+            import sys
+
+            self.proc = subprocess.Popen(
+                [sys.executable, "-c",
+                 "import http.server;"
+                 "h=type('H',(http.server.BaseHTTPRequestHandler,),"
+                 "{'do_GET':lambda s:(s.send_response(200),s.end_headers(),"
+                 "s.wfile.write(b'served-by-modal-amd')),"
+                 "'log_message':lambda s,*a:None});"
+                 "http.server.HTTPServer(('127.0.0.1',18431),h).serve_forever()"],
+            )
+
+        @modal.exit()
+        def shutdown(self):
+            self.proc.terminate()
+
+    with app.run(client=client):
+        server = Httpd.start()
+        assert server.url == "http://127.0.0.1:18431"
+        with urllib.request.urlopen(server.url, timeout=10) as resp:
+            assert resp.read() == b"served-by-modal-amd"
+        server.stop()
+
+
+def test_server_config_validation():
+    import modal_amd as modal
+    from modal_amd.exception import InvalidError
+
+    app = modal.App("server-val")
+    import pytest as _pytest
+
+    with _pytest.raises(InvalidError):
+        @app.server(port=0)
+        class Bad1:
+            pass
+    with _pytest.raises(InvalidError):
+        @app.server(port=80, startup_timeout=0)
+        class Bad2:
+            pass

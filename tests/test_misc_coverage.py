@@ -294,3 +294,46 @@ def test_function_call_from_id_and_num_inputs(client):
         fc = add.spawn(20, 22)
         fc2 = FunctionCall.from_id(fc.object_id)
         assert fc2.get(timeout=30) == 42
+
+
+def test_tunnel_is_a_real_relay():
+    """The tunnel listens on a DISTINCT port and proxies bytes; closing it
+    kills the listener while the service stays reachable (round-1 review:
+    the identity mapping had no forwarding logic)."""
+    import socket
+    import socketserver
+    import threading
+
+    import modal_amd as modal
+
+    class Echo(socketserver.BaseRequestHandler):
+        def handle(self):
+            data = self.request.recv(1024)
+            self.request.sendall(b"echo:" + data)
+
+    with socketserver.TCPServer(("127.0.0.1", 0), Echo) as srv:
+        port = srv.server_address[1]
+        threading.Thread(target=srv.serve_forever, daemon=True).start()
+        with modal.forward(port, unencrypted=True) as tunnel:
+            host, tport = tunnel.tcp_socket
+            assert tport != port, "tunnel must not be an identity mapping"
+            with socket.create_connection((host, tport), timeout=5) as s:
+                s.sendall(b"ping")
+                assert s.recv(1024) == b"echo:ping"
+            assert tunnel._relay.connections_served == 1
+        # tunnel closed: no relay serves its port anymore (a loopback
+        # connect may still "succeed" via the kernel's ephemeral-port
+        # self-connect artifact — that is not our listener)
+        try:
+            s = socket.create_connection((host, tport), timeout=1)
+        except OSError:
+            pass  # refused: the common case
+        else:
+            with s:
+                assert s.getsockname() == s.getpeername(), (
+                    "something still proxies the closed tunnel port"
+                )
+        with socket.create_connection(("127.0.0.1", port), timeout=5) as s:
+            s.sendall(b"direct")
+            assert s.recv(1024) == b"echo:direct"
+        srv.shutdown()
