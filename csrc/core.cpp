@@ -18,7 +18,10 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include <algorithm>
 #include <atomic>
+#include <functional>
+#include <thread>
 #include <cerrno>
 #include <cstdint>
 #include <cstring>
@@ -219,6 +222,236 @@ py::list unpack_payloads(py::buffer buf) {
 
 }  // namespace
 
+
+// ---------------------------------------------------------------------------
+// LZ4 block codec + MALZ41 container (CPU side of ops/csrc/lz4.hip).
+//
+// The CPU fallback previously ran through a pure-Python codec
+// (utils/lz4ref.py, ~tens of MB/s); GPU-less consumers decompressing .z CAS
+// entries and image/mount materialization now run at native speed. The
+// container format is identical to the HIP kernels': 4 KiB segments, each a
+// standard LZ4 block (or stored raw, comp_len 0). Segments compress and
+// decompress in parallel across a small thread pool; the GIL is released
+// for the whole pass.
+
+static constexpr int kSegSize = 4096;
+static constexpr int kMinMatch = 4;
+static constexpr int kMFLimit = 12;
+static constexpr int kLastLiterals = 5;
+
+static inline uint32_t lz4_hash(uint32_t v) { return (v * 2654435761u) >> (32 - 12); }
+
+// compress one block; returns compressed size or 0 when dst_cap exceeded
+static size_t lz4_compress_one(const uint8_t* src, size_t n, uint8_t* dst, size_t dst_cap) {
+  uint16_t table[1 << 12];
+  std::memset(table, 0, sizeof(table));
+  size_t op = 0, ip = 0, anchor = 0;
+  auto emit = [&](size_t lit_start, size_t lit_len, int match_len) -> bool {
+    size_t need = 1 + lit_len + (lit_len >= 15 ? 1 + (lit_len - 15) / 255 : 0) +
+                  (match_len >= 0 ? 2 + 1 + 16 : 0);
+    if (op + need + 8 > dst_cap) return false;
+    int ml = match_len >= 0 ? match_len - kMinMatch : 0;
+    uint8_t token = (uint8_t)((lit_len < 15 ? lit_len : 15) << 4);
+    if (match_len >= 0) token |= (uint8_t)(ml < 15 ? ml : 15);
+    dst[op++] = token;
+    if (lit_len >= 15) {
+      size_t rest = lit_len - 15;
+      while (rest >= 255) { dst[op++] = 255; rest -= 255; }
+      dst[op++] = (uint8_t)rest;
+    }
+    std::memcpy(dst + op, src + lit_start, lit_len);
+    op += lit_len;
+    return true;
+  };
+  if (n >= (size_t)(kMinMatch + kLastLiterals)) {
+    size_t mflimit = n >= kMFLimit ? n - kMFLimit : 0;
+    while (ip < mflimit) {
+      uint32_t seq;
+      std::memcpy(&seq, src + ip, 4);
+      uint32_t h = lz4_hash(seq);
+      size_t cand = table[h];
+      table[h] = (uint16_t)ip;  // n <= 4096 so 16 bits suffice
+      uint32_t cseq;
+      bool ok = cand < ip && ip - cand <= 0xFFFF;
+      if (ok) { std::memcpy(&cseq, src + cand, 4); ok = cseq == seq; }
+      if (ok) {
+        size_t mlen = kMinMatch;
+        size_t maxm = n - kLastLiterals - ip;
+        while (mlen < maxm && src[cand + mlen] == src[ip + mlen]) mlen++;
+        if (!emit(anchor, ip - anchor, (int)mlen)) return 0;
+        size_t off = ip - cand;
+        dst[op++] = (uint8_t)(off & 0xFF);
+        dst[op++] = (uint8_t)(off >> 8);
+        int ml = (int)mlen - kMinMatch;
+        if (ml >= 15) {
+          int rest = ml - 15;
+          while (rest >= 255) { dst[op++] = 255; rest -= 255; }
+          dst[op++] = (uint8_t)rest;
+        }
+        ip += mlen;
+        anchor = ip;
+      } else {
+        ip++;
+      }
+    }
+  }
+  if (!emit(anchor, n - anchor, -1)) return 0;
+  return op;
+}
+
+static bool lz4_decompress_one(const uint8_t* src, size_t comp_len,
+                               uint8_t* dst, size_t raw_len) {
+  size_t ip = 0, op = 0;
+  while (ip < comp_len) {
+    uint8_t token = src[ip++];
+    size_t lit = token >> 4;
+    if (lit == 15) {
+      uint8_t b;
+      do { if (ip >= comp_len) return false; b = src[ip++]; lit += b; } while (b == 255);
+    }
+    if (ip + lit > comp_len || op + lit > raw_len) return false;
+    std::memcpy(dst + op, src + ip, lit);
+    ip += lit; op += lit;
+    if (ip >= comp_len) break;  // last literals
+    if (ip + 2 > comp_len) return false;
+    size_t off = src[ip] | ((size_t)src[ip + 1] << 8);
+    ip += 2;
+    if (off == 0 || off > op) return false;
+    size_t mlen = (token & 0xF) + kMinMatch;
+    if ((token & 0xF) == 15) {
+      uint8_t b;
+      do { if (ip >= comp_len) return false; b = src[ip++]; mlen += b; } while (b == 255);
+    }
+    if (op + mlen > raw_len) return false;
+    // overlapping copy: byte-by-byte when ranges overlap
+    if (off >= mlen) {
+      std::memcpy(dst + op, dst + op - off, mlen);
+    } else {
+      for (size_t k = 0; k < mlen; k++) dst[op + k] = dst[op - off + k];
+    }
+    op += mlen;
+  }
+  return op == raw_len;
+}
+
+static void parallel_for(size_t n, const std::function<void(size_t, size_t)>& body) {
+  unsigned hw = std::thread::hardware_concurrency();
+  size_t n_threads = std::min<size_t>(std::max(1u, hw / 2), 8);
+  if (n < 8 || n_threads <= 1) { body(0, n); return; }
+  std::vector<std::thread> threads;
+  size_t chunk = (n + n_threads - 1) / n_threads;
+  for (size_t t = 0; t < n_threads; t++) {
+    size_t lo = t * chunk, hi = std::min(n, lo + chunk);
+    if (lo >= hi) break;
+    threads.emplace_back([&body, lo, hi] { body(lo, hi); });
+  }
+  for (auto& th : threads) th.join();
+}
+
+// MALZ41 container compress; returns None when incompressible (>= min_gain)
+py::object malz_compress(py::bytes data, double min_gain) {
+  char* buf; ssize_t n_s;
+  if (PyBytes_AsStringAndSize(data.ptr(), &buf, &n_s) != 0) throw py::error_already_set();
+  size_t n = (size_t)n_s;
+  size_t n_seg = (n + kSegSize - 1) / kSegSize;
+  if (n_seg == 0) return py::none();
+  std::vector<uint32_t> comp_lens(n_seg, 0);
+  std::vector<std::vector<uint8_t>> outs(n_seg);
+  const uint8_t* src = (const uint8_t*)buf;
+  {
+    py::gil_scoped_release release;
+    parallel_for(n_seg, [&](size_t lo, size_t hi) {
+      for (size_t i = lo; i < hi; i++) {
+        size_t seg_off = i * kSegSize;
+        size_t seg_len = std::min((size_t)kSegSize, n - seg_off);
+        outs[i].resize(kSegSize);  // only keep if it SAVES space
+        size_t c = lz4_compress_one(src + seg_off, seg_len, outs[i].data(), seg_len > 1 ? seg_len - 1 : 0);
+        if (c > 0 && c < seg_len) {
+          comp_lens[i] = (uint32_t)c;
+          outs[i].resize(c);
+        } else {
+          comp_lens[i] = 0;
+          outs[i].clear();
+        }
+      }
+    });
+  }
+  size_t total = 0;
+  for (size_t i = 0; i < n_seg; i++) {
+    size_t seg_len = std::min((size_t)kSegSize, n - i * kSegSize);
+    total += comp_lens[i] ? comp_lens[i] : seg_len;
+  }
+  if ((double)total >= (double)n * min_gain) return py::none();
+  size_t header = 6 + 12 + 4 * n_seg;
+  py::bytes out_obj(nullptr, (ssize_t)(header + total));
+  uint8_t* out = (uint8_t*)PyBytes_AsString(out_obj.ptr());
+  std::memcpy(out, "MALZ41", 6);
+  uint64_t n64 = n;
+  uint32_t ns32 = (uint32_t)n_seg;
+  std::memcpy(out + 6, &n64, 8);
+  std::memcpy(out + 14, &ns32, 4);
+  std::memcpy(out + 18, comp_lens.data(), 4 * n_seg);
+  size_t op = header;
+  for (size_t i = 0; i < n_seg; i++) {
+    size_t seg_off = i * kSegSize;
+    size_t seg_len = std::min((size_t)kSegSize, n - seg_off);
+    if (comp_lens[i]) {
+      std::memcpy(out + op, outs[i].data(), comp_lens[i]);
+      op += comp_lens[i];
+    } else {
+      std::memcpy(out + op, src + seg_off, seg_len);
+      op += seg_len;
+    }
+  }
+  return out_obj;
+}
+
+py::bytes malz_decompress(py::bytes blob) {
+  char* buf; ssize_t bn;
+  if (PyBytes_AsStringAndSize(blob.ptr(), &buf, &bn) != 0) throw py::error_already_set();
+  if (bn < 18 || std::memcmp(buf, "MALZ41", 6) != 0)
+    throw std::runtime_error("not a MALZ41 container");
+  uint64_t raw_len;
+  uint32_t n_seg;
+  std::memcpy(&raw_len, buf + 6, 8);
+  std::memcpy(&n_seg, buf + 14, 4);
+  size_t header = 6 + 12 + 4 * (size_t)n_seg;
+  if ((size_t)bn < header) throw std::runtime_error("truncated container header");
+  std::vector<uint32_t> comp_lens(n_seg);
+  std::memcpy(comp_lens.data(), buf + 18, 4 * (size_t)n_seg);
+  // per-segment payload offsets
+  std::vector<size_t> seg_payload_off(n_seg);
+  size_t pos = header;
+  for (size_t i = 0; i < n_seg; i++) {
+    seg_payload_off[i] = pos;
+    size_t seg_len = std::min((size_t)kSegSize, (size_t)raw_len - i * kSegSize);
+    pos += comp_lens[i] ? comp_lens[i] : seg_len;
+  }
+  if (pos > (size_t)bn) throw std::runtime_error("truncated container payload");
+  py::bytes out_obj(nullptr, (ssize_t)raw_len);
+  uint8_t* out = (uint8_t*)PyBytes_AsString(out_obj.ptr());
+  const uint8_t* src = (const uint8_t*)buf;
+  bool ok = true;
+  {
+    py::gil_scoped_release release;
+    parallel_for(n_seg, [&](size_t lo, size_t hi) {
+      for (size_t i = lo; i < hi; i++) {
+        size_t seg_off = i * kSegSize;
+        size_t seg_len = std::min((size_t)kSegSize, (size_t)raw_len - seg_off);
+        if (comp_lens[i] == 0) {
+          std::memcpy(out + seg_off, src + seg_payload_off[i], seg_len);
+        } else if (!lz4_decompress_one(src + seg_payload_off[i], comp_lens[i],
+                                       out + seg_off, seg_len)) {
+          ok = false;
+        }
+      }
+    });
+  }
+  if (!ok) throw std::runtime_error("corrupt LZ4 segment");
+  return out_obj;
+}
+
+
 PYBIND11_MODULE(_core, m) {
   m.doc() = "modal_amd native core: shm ring transport + batch framing";
   py::class_<ShmRing>(m, "ShmRing")
@@ -229,5 +462,7 @@ PYBIND11_MODULE(_core, m) {
       .def("pending_bytes", &ShmRing::pending_bytes)
       .def("capacity", &ShmRing::capacity);
   m.def("pack_payloads", &pack_payloads);
+  m.def("malz_compress", &malz_compress, py::arg("data"), py::arg("min_gain") = 0.95);
+  m.def("malz_decompress", &malz_decompress);
   m.def("unpack_payloads", &unpack_payloads);
 }

@@ -193,3 +193,60 @@ def test_pipelined_put_many_bit_identical():
     assert digests_pipelined == digests_serial
     for digest, block in zip(digests_pipelined, blocks):
         assert store_a.get(digest) == block
+
+
+def test_native_codec_cross_compatible_with_reference_codec():
+    """csrc/core.cpp's segment-parallel LZ4 codec and the pure-Python
+    reference (utils/lz4ref.py) decode each other's containers and agree on
+    compress/raw decisions."""
+    from modal_amd.ops.compress import _native_core
+    from modal_amd.utils import lz4ref
+
+    core = _native_core()
+    if core is None:
+        pytest.skip("native core not built")
+
+    def py_container(data):
+        # the pre-native pure-python container builder
+        n = len(data)
+        comp_lens, payload = [], bytearray()
+        for start in range(0, n, C.SEG_SIZE):
+            seg = data[start : start + C.SEG_SIZE]
+            comp = lz4ref.compress_block(seg)
+            if len(comp) < len(seg):
+                comp_lens.append(len(comp))
+                payload += comp
+            else:
+                comp_lens.append(0)
+                payload += seg
+        if not comp_lens or len(payload) >= n * C.MIN_GAIN:
+            return None
+        return C._header(n, comp_lens) + bytes(payload)
+
+    cases = [
+        b"",
+        b"abcd" * 3000,
+        os.urandom(60_000),
+        (b"mixed \x00\x01 payload " * 10_000) + os.urandom(9000),
+        bytes(range(256)) * 40,
+    ]
+    for data in cases:
+        native = core.malz_compress(data, C.MIN_GAIN)
+        ref = py_container(data)
+        assert (native is None) == (ref is None)
+        if native is not None:
+            assert core.malz_decompress(bytes(ref)) == data   # C++ reads py
+        if ref is not None:
+            # py reference decoder reads the C++ container
+            raw_len, comp_lens, off = C.parse_header(native)
+            out = bytearray()
+            pos = off
+            for i, clen in enumerate(comp_lens):
+                seg_raw = min(C.SEG_SIZE, raw_len - i * C.SEG_SIZE)
+                if clen == 0:
+                    out += native[pos : pos + seg_raw]
+                    pos += seg_raw
+                else:
+                    out += lz4ref.decompress_block(native[pos : pos + clen], seg_raw)
+                    pos += clen
+            assert bytes(out) == data
