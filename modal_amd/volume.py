@@ -92,9 +92,32 @@ class _Volume(_Object, type_kind="volume"):
     # -- reads -----------------------------------------------------------
     async def read_file(self, path: str) -> AsyncGenerator[bytes, None]:
         """Stream a file's contents in 8 MiB blocks (parity: reference
-        read_file block streaming, volume.py:837-963)."""
+        read_file block streaming, volume.py:837-963).
+
+        Same-node fast path: the volume tree lives on this filesystem, so
+        read straight from it (file-cache speed) instead of shuttling every
+        block through the control socket; the RPC loop remains the
+        fallback for any client without filesystem access."""
         if not self._is_hydrated:
             await self.hydrate()
+        try:
+            vol_dir = await self._client.svc.volume_dir(volume_id=self.object_id)
+        except Exception:
+            vol_dir = None
+        if vol_dir and os.path.isdir(vol_dir):
+            full = os.path.normpath(os.path.join(vol_dir, path.lstrip("/")))
+            if full.startswith(os.path.abspath(vol_dir)) and os.path.isfile(full):
+                import asyncio as _asyncio
+
+                loop = _asyncio.get_running_loop()
+                with open(full, "rb") as f:
+                    while True:
+                        chunk = await loop.run_in_executor(None, f.read, BLOCK_SIZE)
+                        if not chunk:
+                            return
+                        yield chunk
+                        if len(chunk) < BLOCK_SIZE:
+                            return
         offset = 0
         while True:
             chunk = await self._client.svc.volume_get_file(

@@ -22,7 +22,7 @@ async def boot():
     await s.start()
     c = _Client(s, "client")
     _Client.set_default(c)
-    for _ in range(2):
+    for _ in range(int(os.environ.get("SOAK_WORKERS", "4"))):
         await s.pool.spawn_worker(gpu_index=0)
     return s, c
 
