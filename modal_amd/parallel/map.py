@@ -180,15 +180,46 @@ async def map_invocation_batches(
                 return {"blob": store.put(payload)}
             return payload
 
+        # big-payload chunks overlap: serialize+spill on executor threads in
+        # a bounded task window, so the pump assembles chunk N+1 while N's
+        # 4 MiB pickle+file-write is still in flight
+        spill_tasks: list = []
+
+        async def _spill_and_put(argsbatch: list, chunk_id: str) -> None:
+            loop = asyncio.get_running_loop()
+            payload = await loop.run_in_executor(
+                None, serialize_fast, ("C2", kwargs_common, argsbatch)
+            )
+            if len(payload) > 2 * 1024 * 1024:
+                payload = await loop.run_in_executor(None, _spill, payload, chunk_id)
+            await svc.function_put_chunk(
+                function_call_id=call_id,
+                chunk_id=chunk_id,
+                payload=payload,
+                count=len(argsbatch),
+                method=fn._method_name or "",
+            )
+
         async def flush_args_chunk(argsbatch: list) -> None:
             # "C2" wire form: common kwargs factored out, args list built by
             # C-level zip+islice — no per-item Python in the pump at all
             nonlocal chunk_seq, total_inputs
             total_inputs += len(argsbatch)
             await sem.acquire(len(argsbatch))
-            payload = serialize_fast(("C2", kwargs_common, argsbatch))
             chunk_id = f"{call_id}.c{chunk_seq}"
             chunk_seq += 1
+            approx = sum(
+                len(a) for a in argsbatch[0] if type(a) in (bytes, bytearray, str)
+            ) * len(argsbatch)
+            if approx > 1024 * 1024:
+                task = asyncio.get_running_loop().create_task(
+                    _spill_and_put(argsbatch, chunk_id)
+                )
+                spill_tasks.append(task)
+                while len(spill_tasks) > 4:
+                    await spill_tasks.pop(0)
+                return
+            payload = serialize_fast(("C2", kwargs_common, argsbatch))
             if len(payload) > 2 * 1024 * 1024:
                 payload = await asyncio.get_running_loop().run_in_executor(
                     None, _spill, payload, chunk_id
@@ -224,6 +255,8 @@ async def map_invocation_batches(
                     else:
                         await flush_args_chunk(argsbatch)
                     n_flushed += 1
+                if spill_tasks:
+                    await asyncio.gather(*spill_tasks)
                 await svc.function_finish_inputs(function_call_id=call_id)
                 return
             if hasattr(input_iter, "__aiter__"):
