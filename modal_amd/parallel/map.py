@@ -180,25 +180,32 @@ async def map_invocation_batches(
                 return {"blob": store.put(payload)}
             return payload
 
-        # big-payload chunks overlap: serialize+spill on executor threads in
-        # a bounded task window, so the pump assembles chunk N+1 while N's
-        # 4 MiB pickle+file-write is still in flight
-        spill_tasks: list = []
+        # Big-payload chunks overlap: serialize+spill run on executor
+        # threads in a bounded window, but the put_chunk RPCs DRAIN IN
+        # ORDER (chunk index bases are assigned at put time — out-of-order
+        # puts would scramble map output ordering).
+        prepare_fifo: list = []  # (prepare_task, chunk_id, count)
 
-        async def _spill_and_put(argsbatch: list, chunk_id: str) -> None:
-            loop = asyncio.get_running_loop()
-            payload = await loop.run_in_executor(
-                None, serialize_fast, ("C2", kwargs_common, argsbatch)
-            )
+        def _prepare(argsbatch: list, chunk_id: str) -> Any:
+            payload = serialize_fast(("C2", kwargs_common, argsbatch))
             if len(payload) > 2 * 1024 * 1024:
-                payload = await loop.run_in_executor(None, _spill, payload, chunk_id)
-            await svc.function_put_chunk(
-                function_call_id=call_id,
-                chunk_id=chunk_id,
-                payload=payload,
-                count=len(argsbatch),
-                method=fn._method_name or "",
-            )
+                payload = _spill(payload, chunk_id)
+            return payload
+
+        async def _drain_prepared(block: bool) -> None:
+            while prepare_fifo:
+                task, chunk_id, count = prepare_fifo[0]
+                if not block and not task.done():
+                    return
+                payload = await task
+                prepare_fifo.pop(0)
+                await svc.function_put_chunk(
+                    function_call_id=call_id,
+                    chunk_id=chunk_id,
+                    payload=payload,
+                    count=count,
+                    method=fn._method_name or "",
+                )
 
         async def flush_args_chunk(argsbatch: list) -> None:
             # "C2" wire form: common kwargs factored out, args list built by
@@ -211,19 +218,16 @@ async def map_invocation_batches(
             approx = sum(
                 len(a) for a in argsbatch[0] if type(a) in (bytes, bytearray, str)
             ) * len(argsbatch)
+            loop = asyncio.get_running_loop()
             if approx > 1024 * 1024:
-                task = asyncio.get_running_loop().create_task(
-                    _spill_and_put(argsbatch, chunk_id)
-                )
-                spill_tasks.append(task)
-                while len(spill_tasks) > 4:
-                    await spill_tasks.pop(0)
+                fut = loop.run_in_executor(None, _prepare, argsbatch, chunk_id)
+                prepare_fifo.append((asyncio.ensure_future(fut), chunk_id, len(argsbatch)))
+                await _drain_prepared(block=len(prepare_fifo) > 4)
                 return
+            await _drain_prepared(block=True)  # keep put order across sizes
             payload = serialize_fast(("C2", kwargs_common, argsbatch))
             if len(payload) > 2 * 1024 * 1024:
-                payload = await asyncio.get_running_loop().run_in_executor(
-                    None, _spill, payload, chunk_id
-                )
+                payload = await loop.run_in_executor(None, _spill, payload, chunk_id)
             await svc.function_put_chunk(
                 function_call_id=call_id,
                 chunk_id=chunk_id,
@@ -255,8 +259,7 @@ async def map_invocation_batches(
                     else:
                         await flush_args_chunk(argsbatch)
                     n_flushed += 1
-                if spill_tasks:
-                    await asyncio.gather(*spill_tasks)
+                await _drain_prepared(block=True)
                 await svc.function_finish_inputs(function_call_id=call_id)
                 return
             if hasattr(input_iter, "__aiter__"):
