@@ -252,7 +252,7 @@ async def map_invocation_batches(
                     est = sum(
                         len(a) for a in argsbatch[0] if type(a) in (bytes, bytearray, str)
                     )
-                    if est * len(argsbatch) > 4 * 1024 * 1024 and len(argsbatch) > 1:
+                    if est * len(argsbatch) > 16 * 1024 * 1024 and len(argsbatch) > 1:
                         step = max(1, (4 * 1024 * 1024) // max(est, 1))
                         for s in range(0, len(argsbatch), step):
                             await flush_args_chunk(argsbatch[s : s + step])
