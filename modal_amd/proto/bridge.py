@@ -378,6 +378,132 @@ class GrpcBridge:
                 await context.abort(grpc.StatusCode.NOT_FOUND, "blob not found")
         return self.api.BlobGetResponse(download_url=f"file://{path}")
 
+    # -- Sandbox -------------------------------------------------------------
+
+    async def SandboxCreate(self, request: Any, context: Any) -> Any:
+        d = request.definition
+        res = d.resources
+        resp = await self.scheduler.sandbox_create(
+            entrypoint_args=list(d.entrypoint_args),
+            workdir=d.workdir if d.HasField("workdir") else None,
+            timeout=d.timeout_secs or None,
+            app_id=request.app_id,
+            cpu=(res.milli_cpu / 1000.0) if res.milli_cpu else None,
+            memory=res.memory_mb or None,
+            gpu=0 if res.gpu_config.count else None,
+        )
+        out = self.api.SandboxCreateResponse(sandbox_id=resp["sandbox_id"])
+        return out
+
+    async def SandboxWait(self, request: Any, context: Any) -> Any:
+        native = await self.scheduler.sandbox_wait(
+            request.sandbox_id,
+            timeout=request.timeout or None,
+            raise_on_timeout=False,
+        )
+        out = self.api.SandboxWaitResponse()
+        rc = native.get("returncode")
+        if rc is not None:
+            out.result.status = (
+                self.api.GenericResult.GenericStatus.GENERIC_STATUS_SUCCESS
+                if rc == 0
+                else self.api.GenericResult.GenericStatus.GENERIC_STATUS_FAILURE
+            )
+            out.result.exitcode = rc
+        return out
+
+    async def SandboxTerminate(self, request: Any, context: Any) -> Any:
+        await self.scheduler.sandbox_terminate(request.sandbox_id)
+        return self.api.SandboxTerminateResponse()
+
+    async def SandboxGetTaskId(self, request: Any, context: Any) -> Any:
+        sb = self.scheduler.sandbox_service.sandboxes.get(request.sandbox_id)
+        task_id = sb.task_id if sb is not None else ""
+        return self.api.SandboxGetTaskIdResponse(task_id=task_id)
+
+    async def SandboxList(self, request: Any, context: Any) -> Any:
+        rows = await self.scheduler.sandbox_list(app_id=request.app_id or None)
+        out = self.api.SandboxListResponse()
+        for row in rows:
+            info = out.sandboxes.add()
+            info.id = row.get("sandbox_id", "")
+            info.created_at = row.get("created_at", 0.0)
+        return out
+
+    # -- Volume --------------------------------------------------------------
+
+    async def VolumeGetOrCreate(self, request: Any, context: Any) -> Any:
+        resp = await self.scheduler.volume_get_or_create(
+            name=request.deployment_name or None,
+            environment=request.environment_name or "main",
+            create_if_missing=True,
+            ephemeral=not request.deployment_name,
+        )
+        return self.api.VolumeGetOrCreateResponse(volume_id=resp["volume_id"])
+
+    async def VolumeListFiles(self, request: Any, context: Any) -> Any:
+        entries = await self.scheduler.volume_list_files(
+            request.volume_id, request.path or "/", recursive=request.recursive
+        )
+        resp = self.api.VolumeListFilesResponse()
+        type_map = {"file": self.api.FileEntry.FileType.FILE,
+                    "dir": self.api.FileEntry.FileType.DIRECTORY}
+        for e in entries:
+            fe = resp.entries.add()
+            fe.path = e["path"]
+            fe.type = type_map.get(e.get("type", "file"), self.api.FileEntry.FileType.FILE)
+            fe.size = int(e.get("size", 0))
+            fe.mtime = int(e.get("mtime", 0))
+        yield resp  # server-streaming: one batch
+
+    async def VolumeCommit(self, request: Any, context: Any) -> Any:
+        await self.scheduler.volume_commit(request.volume_id)
+        return self.api.VolumeCommitResponse()
+
+    async def VolumeReload(self, request: Any, context: Any) -> Any:
+        from google.protobuf import empty_pb2
+
+        await self.scheduler.volume_reload(request.volume_id)
+        return empty_pb2.Empty()
+
+    # -- Image ---------------------------------------------------------------
+
+    async def ImageGetOrCreate(self, request: Any, context: Any) -> Any:
+        # translate the reference's dockerfile-command image definition into
+        # the local layer recipe (scheduler/images.py)
+        recipe: list = [{"kind": "base", "name": "local"}]
+        for cmd in request.image.dockerfile_commands:
+            line = cmd.strip()
+            upper = line.upper()
+            if upper.startswith("RUN "):
+                recipe.append({"kind": "run_commands", "commands": [line[4:]]})
+            elif upper.startswith("ENV "):
+                body = line[4:]
+                if "=" in body:
+                    key, _, value = body.partition("=")
+                    recipe.append({"kind": "env", "vars": {key.strip(): value.strip()}})
+            elif upper.startswith("WORKDIR "):
+                recipe.append({"kind": "workdir", "path": line[8:].strip()})
+            else:
+                recipe.append({"kind": "dockerfile_commands", "commands": [line]})
+        resp = await self.scheduler.image_get_or_create(recipe)
+        out = self.api.ImageGetOrCreateResponse(image_id=resp["image_id"])
+        out.metadata.image_builder_version = "local"
+        return out
+
+    async def ImageJoinStreaming(self, request: Any, context: Any) -> Any:
+        # local builds are synchronous: report the (finished) result once
+        info = await self.scheduler.image_info(request.image_id)
+        resp = self.api.ImageJoinStreamingResponse()
+        resp.result.status = (
+            self.api.GenericResult.GenericStatus.GENERIC_STATUS_SUCCESS
+            if info.get("built")
+            else self.api.GenericResult.GenericStatus.GENERIC_STATUS_FAILURE
+        )
+        entry = resp.task_logs.add()
+        entry.data = info.get("build_log", "")
+        yield resp
+
     # -- helpers -------------------------------------------------------------
 
     async def _abort_not_found(self, context: Any, exc: Exception) -> None:

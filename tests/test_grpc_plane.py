@@ -243,3 +243,113 @@ def test_not_found_maps_to_grpc_status(grpc_plane):
             api.FunctionMapResponse,
         )
     assert err.value.code() == grpc.StatusCode.NOT_FOUND
+
+
+@pytest.fixture()
+def grpc_stream(client):
+    from modal_amd.proto.compiler import load
+
+    api, _router = load()
+    path = synchronizer.run(client.svc.start_grpc_bridge())
+    channel = grpc.insecure_channel(f"unix:{path}")
+
+    def invoke(method, request, response_cls):
+        rpc = channel.unary_unary(
+            f"/modal.client.ModalClient/{method}",
+            request_serializer=lambda m: m.SerializeToString(),
+            response_deserializer=response_cls.FromString,
+        )
+        return rpc(request, timeout=60)
+
+    def stream(method, request, response_cls):
+        rpc = channel.unary_stream(
+            f"/modal.client.ModalClient/{method}",
+            request_serializer=lambda m: m.SerializeToString(),
+            response_deserializer=response_cls.FromString,
+        )
+        return list(rpc(request, timeout=60))
+
+    yield api, invoke, stream
+    channel.close()
+
+
+def test_sandbox_via_protos(grpc_stream):
+    api, invoke, _stream = grpc_stream
+    req = api.SandboxCreateRequest()
+    req.definition.entrypoint_args.extend(["sh", "-c", "echo sandboxed; exit 3"])
+    req.definition.timeout_secs = 30
+    resp = invoke("SandboxCreate", req, api.SandboxCreateResponse)
+    assert resp.sandbox_id.startswith("sb-")
+    wait_resp = invoke(
+        "SandboxWait",
+        api.SandboxWaitRequest(sandbox_id=resp.sandbox_id, timeout=30),
+        api.SandboxWaitResponse,
+    )
+    assert wait_resp.result.exitcode == 3
+    assert (
+        wait_resp.result.status
+        == api.GenericResult.GenericStatus.GENERIC_STATUS_FAILURE
+    )
+    tid = invoke(
+        "SandboxGetTaskId",
+        api.SandboxGetTaskIdRequest(sandbox_id=resp.sandbox_id),
+        api.SandboxGetTaskIdResponse,
+    )
+    assert tid.task_id.startswith("ta-")
+    listed = invoke("SandboxList", api.SandboxListRequest(), api.SandboxListResponse)
+    assert any(s.id == resp.sandbox_id for s in listed.sandboxes)
+    invoke(
+        "SandboxTerminate",
+        api.SandboxTerminateRequest(sandbox_id=resp.sandbox_id),
+        api.SandboxTerminateResponse,
+    )
+
+
+def test_volume_and_image_via_protos(grpc_stream, run_dir):
+    api, invoke, stream = grpc_stream
+    vol = invoke(
+        "VolumeGetOrCreate",
+        api.VolumeGetOrCreateRequest(deployment_name="proto-vol"),
+        api.VolumeGetOrCreateResponse,
+    )
+    assert vol.volume_id.startswith("vo-")
+    # write a file via the native side, then list through the proto stream
+    import os
+
+    vol_dir = synchronizer.run(client_svc_volume_dir(run_dir, vol.volume_id))
+    with open(os.path.join(vol_dir, "hello.txt"), "w") as f:
+        f.write("proto")
+    batches = stream(
+        "VolumeListFiles",
+        api.VolumeListFilesRequest(volume_id=vol.volume_id, path="/", recursive=True),
+        api.VolumeListFilesResponse,
+    )
+    entries = [e for b in batches for e in b.entries]
+    assert any(e.path == "hello.txt" and e.size == 5 for e in entries)
+    invoke(
+        "VolumeCommit",
+        api.VolumeCommitRequest(volume_id=vol.volume_id),
+        api.VolumeCommitResponse,
+    )
+
+    img_req = api.ImageGetOrCreateRequest()
+    img_req.image.dockerfile_commands.extend(
+        ["ENV PROTO_MARK=1", "RUN true"]
+    )
+    img = invoke("ImageGetOrCreate", img_req, api.ImageGetOrCreateResponse)
+    assert img.image_id.startswith("im-")
+    joined = stream(
+        "ImageJoinStreaming",
+        api.ImageJoinStreamingRequest(image_id=img.image_id),
+        api.ImageJoinStreamingResponse,
+    )
+    assert joined and joined[0].result.status == (
+        api.GenericResult.GenericStatus.GENERIC_STATUS_SUCCESS
+    )
+
+
+async def client_svc_volume_dir(run_dir, volume_id):
+    from modal_amd.client import _Client
+
+    client = _Client._singleton
+    return await client.svc.volume_dir(volume_id=volume_id)
