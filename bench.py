@@ -32,8 +32,9 @@ if _WORLD_SIZE > 1 and "HIP_VISIBLE_DEVICES" not in os.environ:
 
 ITEMS_PER_GPU = 12_500  # x8 GPUs = the 100k-input config of BASELINE.json
 # 288 GB HBM3E per MI355X easily hosts several payload processes; swept
-# per-box (profiles/README.md); override with MODAL_AMD_BENCH_WPG
-WORKERS_PER_GPU = int(os.environ.get("MODAL_AMD_BENCH_WPG", "3"))
+# per-box (profiles/README.md, r2: 4 workers x 128-item chunks ~1.8x the
+# round-1 3x64 on the same box); override with MODAL_AMD_BENCH_WPG
+WORKERS_PER_GPU = int(os.environ.get("MODAL_AMD_BENCH_WPG", "4"))
 
 
 def _bench_run_dir() -> str:

@@ -30,7 +30,7 @@ OUTPUT_FETCH_MAX = 1024
 # items per shared pickle chunk: one C-pickler pass serves ~64 inputs on the
 # client AND one unpickle serves them on the worker (SURVEY §2 row 6's
 # "tensor-aware fast path" generalized to all small map payloads)
-CHUNK_ITEMS = int(os.environ.get("MODAL_AMD_CHUNK_ITEMS", "64"))
+CHUNK_ITEMS = int(os.environ.get("MODAL_AMD_CHUNK_ITEMS", "128"))
 
 
 class BulkSemaphore:
