@@ -523,6 +523,29 @@ def volume_rm(name: str, remote_path: str, recursive: bool) -> None:
     modal.Volume.from_name(name).remove_file(remote_path, recursive=recursive)
 
 
+@volume_cli.command(name="cp")
+@click.argument("name")
+@click.argument("src_path")
+@click.argument("dst_path")
+def volume_cp(name: str, src_path: str, dst_path: str) -> None:
+    """Copy a file within a volume (parity: `modal volume cp`)."""
+    import modal_amd as modal
+
+    modal.Volume.from_name(name).copy_files([src_path], dst_path)
+    click.echo(f"Copied {src_path} -> {dst_path}")
+
+
+@volume_cli.command(name="rename")
+@click.argument("old_name")
+@click.argument("new_name")
+def volume_rename(old_name: str, new_name: str) -> None:
+    """Rename a volume (parity: `modal volume rename`)."""
+    import modal_amd as modal
+
+    modal.Volume.rename(old_name, new_name)
+    click.echo(f"Renamed volume '{old_name}' -> '{new_name}'")
+
+
 @volume_cli.command(name="delete")
 @click.argument("name")
 @click.option("--yes", "-y", is_flag=True)

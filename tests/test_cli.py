@@ -206,3 +206,24 @@ def test_cli_curl_and_launch(runner, client):
         assert "curl" in result.output
     result = runner.invoke(entrypoint_cli, ["launch", "--help"])
     assert result.exit_code == 0
+
+
+def test_cli_volume_cp_and_rename(client, tmp_path):
+    import modal_amd as modal
+    from modal_amd.cli.entry_point import entrypoint_cli
+
+    from click.testing import CliRunner
+
+    vol = modal.Volume.from_name("cpvol", create_if_missing=True)
+    src = tmp_path / "a.txt"
+    src.write_text("copy-me")
+    with vol.batch_upload() as b:
+        b.put_file(str(src), "/a.txt")
+    runner = CliRunner()
+    result = runner.invoke(entrypoint_cli, ["volume", "cp", "cpvol", "/a.txt", "/b.txt"])
+    assert result.exit_code == 0, result.output
+    assert b"copy-me" == b"".join(vol.read_file("b.txt"))
+    result = runner.invoke(entrypoint_cli, ["volume", "rename", "cpvol", "cpvol2"])
+    assert result.exit_code == 0, result.output
+    vol2 = modal.Volume.from_name("cpvol2")
+    assert b"copy-me" == b"".join(vol2.read_file("a.txt"))
