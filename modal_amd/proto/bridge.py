@@ -43,30 +43,38 @@ class GrpcBridge:
         pool = api.DESCRIPTOR.pool
         from google.protobuf import message_factory
 
-        handlers = {}
-        svc = api.SERVICES["ModalClient"]
-        for method in svc.method:
-            impl = getattr(self, method.name, None)
-            if impl is None:
-                continue
-            req_cls = message_factory.GetMessageClass(
-                pool.FindMessageTypeByName(method.input_type.lstrip("."))
-            )
-            if method.server_streaming:
-                handlers[method.name] = grpc.unary_stream_rpc_method_handler(
-                    impl,
-                    request_deserializer=req_cls.FromString,
-                    response_serializer=lambda msg: msg.SerializeToString(),
+        def make_handlers(svc: Any) -> dict:
+            handlers = {}
+            for method in svc.method:
+                impl = getattr(self, method.name, None)
+                if impl is None:
+                    continue
+                req_cls = message_factory.GetMessageClass(
+                    pool.FindMessageTypeByName(method.input_type.lstrip("."))
                 )
-            else:
-                handlers[method.name] = grpc.unary_unary_rpc_method_handler(
-                    impl,
-                    request_deserializer=req_cls.FromString,
-                    response_serializer=lambda msg: msg.SerializeToString(),
-                )
-        server.add_generic_rpc_handlers(
-            (grpc.method_handlers_generic_handler("modal.client.ModalClient", handlers),)
-        )
+                if method.server_streaming:
+                    handlers[method.name] = grpc.unary_stream_rpc_method_handler(
+                        impl,
+                        request_deserializer=req_cls.FromString,
+                        response_serializer=lambda msg: msg.SerializeToString(),
+                    )
+                else:
+                    handlers[method.name] = grpc.unary_unary_rpc_method_handler(
+                        impl,
+                        request_deserializer=req_cls.FromString,
+                        response_serializer=lambda msg: msg.SerializeToString(),
+                    )
+            return handlers
+
+        server.add_generic_rpc_handlers((
+            grpc.method_handlers_generic_handler(
+                "modal.client.ModalClient", make_handlers(api.SERVICES["ModalClient"])
+            ),
+            grpc.method_handlers_generic_handler(
+                "modal.task_command_router.TaskCommandRouter",
+                make_handlers(self.router_pb.SERVICES["TaskCommandRouter"]),
+            ),
+        ))
         server.add_insecure_port(f"unix:{self.socket_path}")
         await server.start()
         self._server = server
@@ -503,6 +511,82 @@ class GrpcBridge:
         entry = resp.task_logs.add()
         entry.data = info.get("build_log", "")
         yield resp
+
+    # -- TaskCommandRouter (the second gRPC plane) -----------------------------
+    # The reference client dials the WORKER directly for exec/stdio
+    # (task_command_router.proto:373, task_command_router_client.py:211).
+    # Locally the same service answers on the scheduler's grpc socket; the
+    # offset-resume semantics ride the native sandbox stdio buffers.
+
+    def _sandbox_by_task(self, task_id: str) -> Any:
+        for sb in self.scheduler.sandbox_service.sandboxes.values():
+            if sb.task_id == task_id:
+                return sb
+        return None
+
+    async def TaskExecStart(self, request: Any, context: Any) -> Any:
+        sb = self._sandbox_by_task(request.task_id)
+        if sb is None:
+            import grpc
+
+            await context.abort(grpc.StatusCode.NOT_FOUND, "task not found")
+        await self.scheduler.sandbox_exec(
+            sb.sandbox_id,
+            list(request.command_args),
+            env=dict(request.env) or None,
+            workdir=request.workdir if request.HasField("workdir") else None,
+            timeout=request.timeout_secs if request.HasField("timeout_secs") else None,
+            exec_id=request.exec_id,
+        )
+        return self.router_pb.TaskExecStartResponse()
+
+    async def TaskExecStdioRead(self, request: Any, context: Any) -> Any:
+        fd = 2 if request.file_descriptor == 1 else 1  # enum: 0=stdout, 1=stderr
+        offset = request.offset
+        while True:
+            resp = await self.scheduler.sandbox_stdio_read(
+                target_id=request.exec_id, fd=fd, offset=offset, timeout=5.0
+            )
+            if resp["data"]:
+                out = self.router_pb.TaskExecStdioReadResponse()
+                out.data = resp["data"]
+                offset = resp["next_offset"]
+                yield out
+            if resp["eof"]:
+                return
+
+    async def TaskExecStdinWrite(self, request: Any, context: Any) -> Any:
+        await self.scheduler.sandbox_stdin_write(
+            target_id=request.exec_id,
+            offset=request.offset,
+            data=bytes(request.data),
+            eof=request.eof,
+        )
+        return self.router_pb.TaskExecStdinWriteResponse()
+
+    async def TaskExecWait(self, request: Any, context: Any) -> Any:
+        native = await self.scheduler.sandbox_wait(
+            request.exec_id, timeout=None, raise_on_timeout=False
+        )
+        out = self.router_pb.TaskExecWaitResponse()
+        rc = native.get("returncode")
+        if rc is not None:
+            if rc < 0:
+                out.signal = -rc
+            else:
+                out.code = rc
+        return out
+
+    async def TaskExecPoll(self, request: Any, context: Any) -> Any:
+        native = await self.scheduler.sandbox_poll(request.exec_id)
+        out = self.router_pb.TaskExecPollResponse()
+        rc = native.get("returncode")
+        if rc is not None:
+            if rc < 0:
+                out.signal = -rc
+            else:
+                out.code = rc
+        return out
 
     # -- helpers -------------------------------------------------------------
 
