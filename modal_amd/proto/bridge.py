@@ -276,6 +276,47 @@ class GrpcBridge:
             )
         return out
 
+    async def FunctionCallGetDataOut(self, request: Any, context: Any) -> Any:
+        """Generator data stream (parity: FunctionCallGetDataOut,
+        reference function_utils.py:437-495): DataChunks in index order,
+        ending after the GENERATOR_DONE sentinel."""
+        call_id = request.function_call_id or request.attempt_token
+        next_index = int(request.last_index)
+        buffered: dict[int, tuple] = {}
+        done_index: Any = None
+        while True:
+            entries = await self.scheduler.generator_poll(
+                function_call_id=call_id, idx=0, timeout=5.0
+            )
+            if not entries:
+                info = await self.scheduler.function_call_info(function_call_id=call_id)
+                if info.get("completed", 0) >= info.get("total", 0) and info.get("final"):
+                    return  # call finished without (more) generator data
+                continue
+            for index, data, fmt, is_done in entries:
+                buffered[index] = (data, fmt, is_done)
+            while next_index in buffered:
+                data, fmt, is_done = buffered.pop(next_index)
+                chunk = self.api.DataChunk(index=next_index)
+                if is_done and data is None:
+                    # terminal sentinel (parity: GeneratorDone chunk)
+                    from .._serialization import (
+                        DataFormat, GeneratorDone, serialize_data_format,
+                    )
+
+                    chunk.data_format = self.api.DATA_FORMAT_GENERATOR_DONE
+                    chunk.data = serialize_data_format(
+                        GeneratorDone(items_total=next_index), DataFormat.GENERATOR_DONE
+                    )
+                else:
+                    chunk.data_format = fmt or self.api.DATA_FORMAT_PICKLE
+                    if data is not None:
+                        chunk.data = data
+                yield chunk
+                next_index += 1
+                if is_done:
+                    return
+
     async def FunctionCallCancel(self, request: Any, context: Any) -> Any:
         from google.protobuf import empty_pb2
 

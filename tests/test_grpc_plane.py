@@ -440,3 +440,46 @@ def test_task_command_router_exec_via_protos(grpc_stream, client):
         api.SandboxTerminateResponse,
     )
     channel.close()
+
+
+def test_generator_stream_via_protos(grpc_stream):
+    """FunctionCallGetDataOut streams generator items as DataChunks in
+    index order, terminated by a GENERATOR_DONE chunk."""
+    api, invoke, stream = grpc_stream
+
+    def gen(n):
+        for i in range(n):
+            yield i * 10
+
+    _app_id, function_id = _create_function(api, invoke, gen)
+    # mark as generator via FunctionCreate (function_type)
+    req = api.FunctionCreateRequest(app_id=_app_id)
+    import cloudpickle
+
+    req.function.function_name = "gen_fn"
+    req.function.function_serialized = cloudpickle.dumps(gen)
+    req.function.function_type = api.Function.FunctionType.FUNCTION_TYPE_GENERATOR
+    function_id = invoke("FunctionCreate", req, api.FunctionCreateResponse).function_id
+
+    map_req = api.FunctionMapRequest(
+        function_id=function_id, function_call_type=api.FUNCTION_CALL_TYPE_UNARY
+    )
+    item = map_req.pipelined_inputs.add()
+    item.input.args = pickle.dumps(((4,), {}))
+    item.input.final_input = True
+    map_resp = invoke("FunctionMap", map_req, api.FunctionMapResponse)
+
+    chunks = stream(
+        "FunctionCallGetDataOut",
+        api.FunctionCallGetDataRequest(function_call_id=map_resp.function_call_id),
+        api.DataChunk,
+    )
+    values = []
+    saw_done = False
+    for chunk in chunks:
+        if chunk.data_format == api.DATA_FORMAT_GENERATOR_DONE:
+            saw_done = True
+            break
+        values.append(pickle.loads(chunk.data))
+    assert values == [0, 10, 20, 30]
+    assert saw_done
