@@ -326,6 +326,66 @@ class GrpcBridge:
         )
         return empty_pb2.Empty()
 
+    # -- input plane (Attempt trio) -------------------------------------------
+    # parity: _InputPlaneInvocation (reference _functions.py:396-549) —
+    # locally the "input plane" IS the scheduler, and the attempt token is
+    # the call id (SURVEY §2 row 11's stated plan).
+
+    async def AttemptStart(self, request: Any, context: Any) -> Any:
+        item = self._put_item_to_native(request.input)
+        try:
+            resp = await self.scheduler.function_map(
+                function_id=request.function_id,
+                kind="unary",
+                pipelined_inputs=[item],
+            )
+        except Exception as exc:
+            await self._abort_not_found(context, exc)
+            raise
+        await self.scheduler.function_finish_inputs(
+            function_call_id=resp["function_call_id"]
+        )
+        out = self.api.AttemptStartResponse(attempt_token=resp["function_call_id"])
+        retries = resp.get("retry_policy") or {}
+        out.retry_policy.retries = int(retries.get("max_retries", 0))
+        return out
+
+    async def AttemptAwait(self, request: Any, context: Any) -> Any:
+        native = await self.scheduler.function_get_outputs(
+            function_call_id=request.attempt_token,
+            max_values=1,
+            timeout=min(request.timeout_secs or 55.0, 55.0),
+            clear_on_success=False,
+        )
+        out = self.api.AttemptAwaitResponse()
+        if native:
+            item = native[0]
+            o = out.output
+            o.idx = item.get("idx", 0)
+            o.result.status = item.get("status", 0)
+            data = item.get("data")
+            cid = item.get("out_chunk")
+            if cid and item.get("chunk_data") is not None:
+                data = pickle.dumps(
+                    pickle.loads(item["chunk_data"])[item.get("out_ci", 0)], 4
+                )
+            if data is not None:
+                o.result.data = data
+            if item.get("exc"):
+                o.result.exception = item["exc"]
+            o.data_format = self.api.DATA_FORMAT_PICKLE
+        return out
+
+    async def AttemptRetry(self, request: Any, context: Any) -> Any:
+        # a fresh attempt for the same logical input (new local call)
+        resp = await self.AttemptStart(
+            self.api.AttemptStartRequest(
+                function_id=request.function_id, input=request.input
+            ),
+            context,
+        )
+        return self.api.AttemptRetryResponse(attempt_token=resp.attempt_token)
+
     # -- Queue ---------------------------------------------------------------
 
     async def QueueGetOrCreate(self, request: Any, context: Any) -> Any:

@@ -483,3 +483,47 @@ def test_generator_stream_via_protos(grpc_stream):
         values.append(pickle.loads(chunk.data))
     assert values == [0, 10, 20, 30]
     assert saw_done
+
+
+def test_attempt_trio_via_protos(grpc_plane):
+    """Input-plane variant: AttemptStart/AttemptAwait/AttemptRetry
+    (parity: _InputPlaneInvocation, reference _functions.py:396-549)."""
+    api, invoke = grpc_plane
+    _app_id, function_id = _create_function(api, invoke, lambda x: x * 11)
+
+    start_req = api.AttemptStartRequest(function_id=function_id)
+    start_req.input.input.args = pickle.dumps(((4,), {}))
+    start_req.input.input.final_input = True
+    start = invoke("AttemptStart", start_req, api.AttemptStartResponse)
+    assert start.attempt_token.startswith("fc-")
+
+    deadline = time.time() + 30
+    value = None
+    while value is None and time.time() < deadline:
+        out = invoke(
+            "AttemptAwait",
+            api.AttemptAwaitRequest(attempt_token=start.attempt_token, timeout_secs=10),
+            api.AttemptAwaitResponse,
+        )
+        if out.HasField("output"):
+            value = pickle.loads(out.output.result.data)
+    assert value == 44
+
+    retry_req = api.AttemptRetryRequest(
+        function_id=function_id, attempt_token=start.attempt_token
+    )
+    retry_req.input.input.args = pickle.dumps(((5,), {}))
+    retry_req.input.input.final_input = True
+    retry = invoke("AttemptRetry", retry_req, api.AttemptRetryResponse)
+    assert retry.attempt_token != start.attempt_token
+    deadline = time.time() + 30
+    value = None
+    while value is None and time.time() < deadline:
+        out = invoke(
+            "AttemptAwait",
+            api.AttemptAwaitRequest(attempt_token=retry.attempt_token, timeout_secs=10),
+            api.AttemptAwaitResponse,
+        )
+        if out.HasField("output"):
+            value = pickle.loads(out.output.result.data)
+    assert value == 55
