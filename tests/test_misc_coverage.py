@@ -337,3 +337,25 @@ def test_tunnel_is_a_real_relay():
             s.sendall(b"direct")
             assert s.recv(1024) == b"echo:direct"
         srv.shutdown()
+
+
+def test_type_stub_generation(tmp_path):
+    """scripts/gen_stubs.py emits a parseable .pyi covering __all__
+    (synchronicity-aware stub generation)."""
+    import ast
+    import subprocess
+    import sys as _sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [_sys.executable, os.path.join(repo, "scripts", "gen_stubs.py")],
+        capture_output=True, text=True, timeout=120,
+    )
+    assert out.returncode == 0, out.stderr
+    stub_path = os.path.join(repo, "modal_amd", "__init__.pyi")
+    src = open(stub_path).read()
+    ast.parse(src)
+    import modal_amd
+
+    for name in modal_amd.__all__:
+        assert name in src, f"{name} missing from stubs"
