@@ -348,3 +348,37 @@ def test_map_spreads_across_workers(client):
     with app.run(client=client):
         pids = set(work.map(range(2000)))
         assert len(pids) >= 3, f"work concentrated on {len(pids)} worker(s)"
+
+
+def test_bigarg_xfer_survives_worker_death(client, run_dir):
+    """A worker dying mid-chunk on the big-arg (xfer-spill) path: the chunk
+    redelivers from the SAME spill file (it is unlinked only on chunk
+    completion), payloads stay bit-exact, and no xfer files leak."""
+    import hashlib
+
+    app = modal.App("xfer-death")
+    marker = os.path.join(run_dir, "xfer-killed")
+
+    @app.function()
+    def digest_or_die(blob, path):
+        import hashlib as h
+        import os as _os
+
+        d = h.sha256(blob).hexdigest()
+        # the odd-length payload kills its first worker mid-chunk
+        if len(blob) % 2 == 1 and not _os.path.exists(path):
+            open(path, "w").write("x")
+            _os._exit(1)
+        return d
+
+    payloads = [os.urandom(1 << 20) for _ in range(24)]  # 24 x 1 MiB: spills
+    payloads[10] = os.urandom((1 << 20) + 1)  # the deterministic killer
+    expected = sorted(hashlib.sha256(p).hexdigest() for p in payloads)
+    with app.run(client=client):
+        out = sorted(
+            digest_or_die.map(payloads, kwargs={"path": marker}, order_outputs=False)
+        )
+    assert out == expected
+    assert os.path.exists(marker), "kill never triggered (fixture too lucky)"
+    leftovers = os.listdir(os.path.join(run_dir, "xfer"))
+    assert leftovers == [], f"xfer files leaked after worker death: {leftovers}"
