@@ -1173,7 +1173,11 @@ class WorkerRuntime:
         )
 
     async def _send_gen_item(self, token: str, index: int, value: Any) -> None:
-        data = serialize(value)
+        from .._serialization import serialize_fast
+
+        # generator/web items are usually primitives; serialize_fast falls
+        # back to the hook-aware pickler for tensors/handles itself
+        data = serialize_fast(value)
         await self.conn.send(
             {"t": "gen_data", "token": token, "index": index, "data": data, "format": int(DataFormat.PICKLE)}
         )

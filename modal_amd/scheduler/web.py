@@ -14,7 +14,7 @@ from __future__ import annotations
 import asyncio
 from typing import TYPE_CHECKING, Any, Optional
 
-from .._serialization import serialize
+from .._serialization import serialize_fast
 from .calls import GENERIC_STATUS_SUCCESS
 
 if TYPE_CHECKING:
@@ -80,7 +80,7 @@ class WebGateway:
             "headers": [[k, v] for k, v in request.headers.items()],
             "body": body,
         }
-        payload = serialize(("P", ((req,), {})))
+        payload = serialize_fast(("P", ((req,), {})))  # primitives only: C pickler
         resp = await self.scheduler.function_map(
             function_id=function_id, kind="unary", pipelined_inputs=[{"payload": payload, "method": "__web__"}]
         )
