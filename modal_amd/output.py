@@ -57,8 +57,44 @@ class OutputManager:
 
             ts = datetime.datetime.fromtimestamp(entry.get("ts", 0)).strftime("%H:%M:%S")
             data = "".join(f"{ts} {line}\n" for line in data.splitlines())
+        if self._console is not None and entry.get("task_id"):
+            # rich TTY: dim per-task prefix, stderr tinted (parity: the
+            # reference's colored container-log rendering, _output/rich.py)
+            tag = entry["task_id"][-6:]
+            style = "red" if entry.get("fd") == 2 else ""
+            for line in data.splitlines():
+                self._console.print(
+                    f"[dim]\\[{tag}][/dim] {line}",
+                    style=style,
+                    highlight=False,
+                    markup=True,
+                )
+            return
         stream.write(data)
         stream.flush()
+
+    @contextlib.contextmanager
+    def status(self, message: str) -> Any:
+        """Spinner for a long-running step (image build, object creation);
+        plain begin/end lines off-TTY (parity: reference status.py)."""
+        cm = None
+        if self._console is not None:
+            try:
+                cm = self._console.status(message)
+                cm.__enter__()
+            except Exception:
+                cm = None
+        if cm is None:
+            sys.stdout.write(f"... {message}\n")
+            sys.stdout.flush()
+        try:
+            yield self
+        finally:
+            if cm is not None:
+                try:
+                    cm.__exit__(None, None, None)
+                except Exception:
+                    pass
 
 
 class MapProgress:

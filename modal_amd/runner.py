@@ -147,10 +147,14 @@ class AppRunContext:
         )
         app._app_id = resp["app_id"]
         app._running_client = client
-        await _register_functions(app, client, app._app_id)
         from .output import get_output_manager
 
         manager = get_output_manager()
+        if manager is not None:
+            with manager.status("Creating objects..."):
+                await _register_functions(app, client, app._app_id)
+        else:
+            await _register_functions(app, client, app._app_id)
         if manager is not None:  # parity: OutputManager step lines
             manager.print_step(f"Initialized app {app.description or app._app_id}.")
             names = sorted(app._functions.keys()) if getattr(app, "_functions", None) else []
@@ -182,12 +186,32 @@ class AppRunContext:
 
     def _start_log_stream(self) -> None:
         svc = self.client.svc
-        if getattr(svc, "is_proxy", False):
-            return  # remote attach: log streaming handled by logs manager
         from .output import get_output_manager
 
         manager = get_output_manager()
         if manager is None:
+            return
+        if getattr(svc, "is_proxy", False):
+            # daemon-attached: tail over the offset-resumable long-poll RPC
+            app_id = self.app._app_id
+
+            async def drain_remote() -> None:
+                offset = 0
+                while True:
+                    try:
+                        resp = await svc.app_get_logs(
+                            app_id=app_id, offset=offset, timeout=10.0
+                        )
+                    except asyncio.CancelledError:
+                        raise
+                    except Exception:
+                        await asyncio.sleep(0.5)
+                        continue
+                    for entry in resp["entries"]:
+                        manager.print_log(entry)
+                    offset = resp["next_offset"]
+
+            self._log_task = asyncio.get_running_loop().create_task(drain_remote())
             return
         state = svc.apps.get(self.app._app_id)
         if state is None:

@@ -111,3 +111,55 @@ print("TAIL_OK")
             daemon.wait(timeout=5)
         except subprocess.TimeoutExpired:
             daemon.kill()
+
+
+def test_daemon_attached_app_run_streams_logs(tmp_path):
+    """app.run() with enable_output against a DAEMON scheduler streams the
+    app's worker prints back to the attached terminal (round-1: the proxy
+    path returned nothing)."""
+    run_dir = str(tmp_path / "daemon")
+    daemon = subprocess.Popen(
+        [sys.executable, "-m", "modal_amd.cli.entry_point", "daemon", "--run-dir", run_dir],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL, start_new_session=True,
+    )
+    sock = os.path.join(run_dir, "scheduler.sock")
+    try:
+        deadline = time.time() + 30
+        while not os.path.exists(sock):
+            assert daemon.poll() is None
+            assert time.time() < deadline
+            time.sleep(0.05)
+        env = dict(os.environ)
+        env["MODAL_AMD_ATTACH_SOCKET"] = sock
+        script = r"""
+import time
+import modal_amd as modal
+
+app = modal.App("stream-app")
+
+@app.function()
+def talk(i):
+    print(f"streamed-line-{i}", flush=True)
+    return i
+
+with modal.enable_output():
+    with app.run():
+        for i in range(4):
+            talk.remote(i)
+        time.sleep(2.0)  # let the long-poll round-trip deliver
+print("RUN_DONE")
+"""
+        proc = subprocess.run(
+            [sys.executable, "-c", script], env=env, capture_output=True, text=True,
+            timeout=120,
+        )
+        assert proc.returncode == 0, proc.stdout + proc.stderr
+        assert "RUN_DONE" in proc.stdout
+        missing = [i for i in range(4) if f"streamed-line-{i}" not in proc.stdout]
+        assert not missing, f"attached app.run missed logs {missing}: {proc.stdout!r}"
+    finally:
+        daemon.terminate()
+        try:
+            daemon.wait(timeout=5)
+        except subprocess.TimeoutExpired:
+            daemon.kill()
