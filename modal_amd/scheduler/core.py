@@ -189,6 +189,14 @@ class Scheduler:
             logging.getLogger("modal_amd.scheduler").warning(
                 "state restore failed", exc_info=True
             )
+        try:  # unfinished detached spawn calls re-enter the dispatch queue
+            from . import wal
+
+            wal.replay(self)
+        except Exception:
+            logging.getLogger("modal_amd.scheduler").warning(
+                "WAL replay failed", exc_info=True
+            )
 
     # -- lifecycle -------------------------------------------------------
     async def start(self) -> None:
@@ -664,6 +672,13 @@ class Scheduler:
             raise NotFoundError(f"Function {function_id} not found")
         record = CallRecord(function_id, kind, return_exceptions)
         self.calls[record.call_id] = record
+        from . import wal
+
+        if wal.journaled(self, record):
+            # detached spawn work on a deployed app survives scheduler
+            # restarts (SURVEY hard part 5: WAL in scheduler)
+            record.durable = True
+            wal.journal_created(self, record)
         if pipelined_inputs:
             await self.function_put_inputs(record.call_id, pipelined_inputs)
         return {
@@ -700,6 +715,10 @@ class Scheduler:
                 tensors=item.get("tensors"),
                 payload_blob=item.get("payload_blob"),
             )
+            if getattr(record, "durable", False):
+                from . import wal
+
+                wal.journal_input(self, record, rec)
             chunk_id = item.get("chunk")
             if chunk_id:
                 rec.chunk_id = chunk_id
@@ -780,7 +799,12 @@ class Scheduler:
         return {"idx_base": group.base_idx, "count": count}
 
     async def function_finish_inputs(self, function_call_id: str) -> None:
-        self._call(function_call_id).finish_inputs()
+        record = self._call(function_call_id)
+        record.finish_inputs()
+        if getattr(record, "durable", False):
+            from . import wal
+
+            wal.journal_finish(self, record)
 
     async def function_wait_output(
         self, function_call_id: str, idx: int = 0, timeout: Optional[float] = None
@@ -1090,6 +1114,15 @@ class Scheduler:
             rec.idx, status, output, output_format, exc_repr, retry_count, output_blob,
             out_chunk, out_ci,
         )
+        if getattr(record, "durable", False) and rec.final:
+            from . import wal
+
+            wal.journal_result(self, record, rec)
+            if (
+                record.num_inputs_final is not None
+                and record.completed >= record.num_inputs_final
+            ):
+                wal.drop(self, record.call_id)
 
     def on_generator_data(self, msg: dict) -> None:
         call_id, idx_s, _ = msg["token"].rsplit(":", 2)

@@ -495,6 +495,27 @@ class WorkerPool:
             record = self.scheduler.calls.get(call_id)
             if record is not None:
                 record.post_outputs_bulk(triples, chunk_id)
+                if getattr(record, "durable", False):
+                    # durable spawn calls journal results with the bytes
+                    # EXTRACTED (the shared chunk dies with this process)
+                    import pickle as _pickle
+
+                    from . import wal
+
+                    values = _pickle.loads(msg["data"])
+                    for idx, _retry, ci in triples:
+                        rec = record.inputs.get(idx)
+                        if rec is not None and rec.final:
+                            wal.journal_result(
+                                self.scheduler, record, rec,
+                                output=_pickle.dumps(values[ci], 4),
+                                output_format=1,
+                            )
+                    if (
+                        record.num_inputs_final is not None
+                        and record.completed >= record.num_inputs_final
+                    ):
+                        wal.drop(self.scheduler, record.call_id)
         self._dispatch_wake.set()
 
     # -- dispatch --------------------------------------------------------
