@@ -101,6 +101,63 @@ class GrpcBridge:
         await self.scheduler.app_client_disconnect(request.app_id)
         return empty_pb2.Empty()
 
+    async def AppGetOrCreate(self, request: Any, context: Any) -> Any:
+        try:
+            resp = await self.scheduler.app_lookup(
+                request.app_name, request.environment_name or "main"
+            )
+            return self.api.AppGetOrCreateResponse(app_id=resp["app_id"])
+        except Exception:
+            pass
+        resp = await self.scheduler.app_create(
+            description=request.app_name, ephemeral=False,
+            environment=request.environment_name or "main",
+        )
+        await self.scheduler.app_publish(resp["app_id"], request.app_name)
+        return self.api.AppGetOrCreateResponse(app_id=resp["app_id"])
+
+    async def AppDeploy(self, request: Any, context: Any) -> Any:
+        await self.scheduler.app_publish(request.app_id, request.name)
+        return self.api.AppDeployResponse(url=f"local://{request.name}")
+
+    async def AppLookup(self, request: Any, context: Any) -> Any:
+        try:
+            resp = await self.scheduler.app_lookup(
+                request.app_name, request.environment_name or "main"
+            )
+        except Exception as exc:
+            await self._abort_not_found(context, exc)
+            raise
+        return self.api.AppLookupResponse(app_id=resp["app_id"])
+
+    async def FunctionGet(self, request: Any, context: Any) -> Any:
+        try:
+            resp = await self.scheduler.function_lookup(
+                request.app_name, request.object_tag, request.environment_name or "main"
+            )
+        except Exception as exc:
+            await self._abort_not_found(context, exc)
+            raise
+        out = self.api.FunctionGetResponse(function_id=resp["function_id"])
+        meta = resp.get("metadata") or {}
+        out.handle_metadata.function_name = meta.get("function_name", "")
+        return out
+
+    async def FunctionGetCurrentStats(self, request: Any, context: Any) -> Any:
+        native = await self.scheduler.function_get_current_stats(request.function_id)
+        return self.api.FunctionStats(
+            backlog=native.get("backlog", 0),
+            num_total_tasks=native.get("num_total_tasks", 0),
+        )
+
+    async def DictContents(self, request: Any, context: Any) -> Any:
+        items = await self.scheduler.dict_items(request.dict_id)
+        for key, value in items:
+            entry = self.api.DictEntry()
+            entry.key = bytes(key)
+            entry.value = bytes(value)
+            yield entry
+
     async def AppStop(self, request: Any, context: Any) -> Any:
         from google.protobuf import empty_pb2
 

@@ -527,3 +527,58 @@ def test_attempt_trio_via_protos(grpc_plane):
         if out.HasField("output"):
             value = pickle.loads(out.output.result.data)
     assert value == 55
+
+
+def test_deployment_flow_via_protos(grpc_stream):
+    """AppGetOrCreate/AppDeploy/AppLookup/FunctionGet: a wire client can
+    deploy and later look functions up by (app, name)."""
+    import cloudpickle
+
+    api, invoke, stream = grpc_stream
+    app_id = invoke(
+        "AppGetOrCreate",
+        api.AppGetOrCreateRequest(app_name="proto-deployed"),
+        api.AppGetOrCreateResponse,
+    ).app_id
+    # idempotent
+    assert invoke(
+        "AppGetOrCreate",
+        api.AppGetOrCreateRequest(app_name="proto-deployed"),
+        api.AppGetOrCreateResponse,
+    ).app_id == app_id
+
+    req = api.FunctionCreateRequest(app_id=app_id)
+    req.function.function_name = "quadruple"
+    req.function.function_serialized = cloudpickle.dumps(lambda x: x * 4)
+    fid = invoke("FunctionCreate", req, api.FunctionCreateResponse).function_id
+    invoke("AppDeploy", api.AppDeployRequest(app_id=app_id, name="proto-deployed"),
+           api.AppDeployResponse)
+
+    looked = invoke(
+        "AppLookup", api.AppLookupRequest(app_name="proto-deployed"), api.AppLookupResponse
+    )
+    assert looked.app_id == app_id
+    got = invoke(
+        "FunctionGet",
+        api.FunctionGetRequest(app_name="proto-deployed", object_tag="quadruple"),
+        api.FunctionGetResponse,
+    )
+    assert got.function_id == fid
+    stats = invoke(
+        "FunctionGetCurrentStats",
+        api.FunctionGetCurrentStatsRequest(function_id=fid),
+        api.FunctionStats,
+    )
+    assert stats.backlog == 0
+
+    # dict contents stream
+    d = invoke("DictGetOrCreate", api.DictGetOrCreateRequest(), api.DictGetOrCreateResponse)
+    upd = api.DictUpdateRequest(dict_id=d.dict_id)
+    for k, v in ((b"k1", b"v1"), (b"k2", b"v2")):
+        e = upd.updates.add()
+        e.key, e.value = k, v
+    invoke("DictUpdate", upd, api.DictUpdateResponse)
+    entries = stream(
+        "DictContents", api.DictContentsRequest(dict_id=d.dict_id), api.DictEntry
+    )
+    assert sorted((e.key, e.value) for e in entries) == [(b"k1", b"v1"), (b"k2", b"v2")]
