@@ -511,6 +511,73 @@ class GrpcBridge:
             return self.api.DictGetResponse(found=False)
         return self.api.DictGetResponse(found=True, value=value)
 
+    async def QueueClear(self, request: Any, context: Any) -> Any:
+        from google.protobuf import empty_pb2
+
+        await self.scheduler.queue_clear(
+            request.queue_id, bytes(request.partition_key) or None, request.all_partitions
+        )
+        return empty_pb2.Empty()
+
+    async def QueueDelete(self, request: Any, context: Any) -> Any:
+        from google.protobuf import empty_pb2
+
+        await self.scheduler.queue_delete(request.queue_id)
+        return empty_pb2.Empty()
+
+    async def DictLen(self, request: Any, context: Any) -> Any:
+        n = await self.scheduler.dict_len(request.dict_id)
+        return self.api.DictLenResponse(len=n)
+
+    async def DictClear(self, request: Any, context: Any) -> Any:
+        from google.protobuf import empty_pb2
+
+        await self.scheduler.dict_clear(request.dict_id)
+        return empty_pb2.Empty()
+
+    async def DictDelete(self, request: Any, context: Any) -> Any:
+        from google.protobuf import empty_pb2
+
+        await self.scheduler.dict_delete(request.dict_id)
+        return empty_pb2.Empty()
+
+    async def VolumeRemoveFile(self, request: Any, context: Any) -> Any:
+        from google.protobuf import empty_pb2
+
+        await self.scheduler.volume_remove_file(
+            request.volume_id, request.path, recursive=request.recursive
+        )
+        return empty_pb2.Empty()
+
+    async def VolumeDelete(self, request: Any, context: Any) -> Any:
+        from google.protobuf import empty_pb2
+
+        await self.scheduler.volume_delete(request.volume_id)
+        return empty_pb2.Empty()
+
+    async def SandboxGetLogs(self, request: Any, context: Any) -> Any:
+        """Sandbox stdout/stderr as TaskLogsBatch stream (offset tracked in
+        last_entry_id for reconnect parity)."""
+        fd = 2 if request.file_descriptor == 2 else 1
+        offset = int(request.last_entry_id) if request.last_entry_id else 0
+        while True:
+            resp = await self.scheduler.sandbox_stdio_read(
+                target_id=request.sandbox_id, fd=fd, offset=offset,
+                timeout=min(request.timeout or 5.0, 55.0),
+            )
+            if resp["data"]:
+                batch = self.api.TaskLogsBatch(entry_id=str(resp["next_offset"]))
+                item = batch.items.add()
+                item.data = resp["data"].decode("utf-8", errors="replace")
+                offset = resp["next_offset"]
+                yield batch
+            if resp["eof"]:
+                out = self.api.TaskLogsBatch(eof=True, entry_id=str(offset))
+                yield out
+                return
+            if not resp["data"]:
+                return  # poll window elapsed: client reconnects at entry_id
+
     # -- Secret --------------------------------------------------------------
 
     async def SecretGetOrCreate(self, request: Any, context: Any) -> Any:

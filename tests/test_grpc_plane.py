@@ -582,3 +582,39 @@ def test_deployment_flow_via_protos(grpc_stream):
         "DictContents", api.DictContentsRequest(dict_id=d.dict_id), api.DictEntry
     )
     assert sorted((e.key, e.value) for e in entries) == [(b"k1", b"v1"), (b"k2", b"v2")]
+
+
+def test_resource_housekeeping_via_protos(grpc_stream):
+    """Clear/delete/len RPCs + SandboxGetLogs streaming."""
+    api, invoke, stream = grpc_stream
+    from google.protobuf import empty_pb2
+
+    q = invoke("QueueGetOrCreate", api.QueueGetOrCreateRequest(), api.QueueGetOrCreateResponse)
+    invoke("QueuePut", api.QueuePutRequest(queue_id=q.queue_id, values=[b"a"]), empty_pb2.Empty)
+    invoke("QueueClear", api.QueueClearRequest(queue_id=q.queue_id, all_partitions=True), empty_pb2.Empty)
+    assert invoke("QueueLen", api.QueueLenRequest(queue_id=q.queue_id), api.QueueLenResponse).len == 0
+    invoke("QueueDelete", api.QueueDeleteRequest(queue_id=q.queue_id), empty_pb2.Empty)
+
+    d = invoke("DictGetOrCreate", api.DictGetOrCreateRequest(), api.DictGetOrCreateResponse)
+    upd = api.DictUpdateRequest(dict_id=d.dict_id)
+    e = upd.updates.add(); e.key, e.value = b"k", b"v"
+    invoke("DictUpdate", upd, api.DictUpdateResponse)
+    assert invoke("DictLen", api.DictLenRequest(dict_id=d.dict_id), api.DictLenResponse).len == 1
+    invoke("DictClear", api.DictClearRequest(dict_id=d.dict_id), empty_pb2.Empty)
+    assert invoke("DictLen", api.DictLenRequest(dict_id=d.dict_id), api.DictLenResponse).len == 0
+    invoke("DictDelete", api.DictDeleteRequest(dict_id=d.dict_id), empty_pb2.Empty)
+
+    # sandbox logs stream
+    sb_req = api.SandboxCreateRequest()
+    sb_req.definition.entrypoint_args.extend(["sh", "-c", "echo log-line-one; echo log-line-two"])
+    sb = invoke("SandboxCreate", sb_req, api.SandboxCreateResponse)
+    invoke("SandboxWait", api.SandboxWaitRequest(sandbox_id=sb.sandbox_id, timeout=30),
+           api.SandboxWaitResponse)
+    batches = stream(
+        "SandboxGetLogs",
+        api.SandboxGetLogsRequest(sandbox_id=sb.sandbox_id, timeout=5),
+        api.TaskLogsBatch,
+    )
+    text = "".join(item.data for b in batches for item in b.items)
+    assert "log-line-one" in text and "log-line-two" in text
+    assert batches[-1].eof
