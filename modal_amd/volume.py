@@ -110,13 +110,14 @@ class _Volume(_Object, type_kind="volume"):
                 import asyncio as _asyncio
 
                 loop = _asyncio.get_running_loop()
+                read_size = 4 * BLOCK_SIZE  # local reads: fewer executor hops
                 with open(full, "rb") as f:
                     while True:
-                        chunk = await loop.run_in_executor(None, f.read, BLOCK_SIZE)
+                        chunk = await loop.run_in_executor(None, f.read, read_size)
                         if not chunk:
                             return
                         yield chunk
-                        if len(chunk) < BLOCK_SIZE:
+                        if len(chunk) < read_size:
                             return
         offset = 0
         while True:
