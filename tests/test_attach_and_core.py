@@ -125,3 +125,20 @@ def test_ring_carries_bulk_frames(client):
         svc = client.svc
         sent = sum(w.conn.ring_frames_sent for w in svc.pool.workers.values())
         assert sent >= 1, "bulk frame never used the shm ring"
+
+
+def test_native_core_asan_clean():
+    """ASan build of csrc/core.cpp exercised over the shm ring, codec and
+    payload packing with no reported errors (C++ sanitizer
+    jobs for the native core)."""
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        ["bash", os.path.join(repo, "scripts", "asan_check.sh")],
+        capture_output=True, text=True, timeout=300,
+    )
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
+    assert "ASAN CHECK OK" in out.stdout
