@@ -268,6 +268,10 @@ class Scheduler:
                 and now - rec.finished_at > self.CALL_RETENTION_SECONDS
             ):
                 self.calls.pop(call_id, None)
+                # release any undrained chunk payloads (unlinks one-shot
+                # xfer spill files a cancelled/abandoned map left behind)
+                for chunk_id in list(rec.chunks):
+                    rec._drop_chunk(chunk_id)
                 dropped += 1
         # stopped ephemeral apps: logs deques (10k entries each) are the cost
         for app_id, app in list(self.apps.items()):
