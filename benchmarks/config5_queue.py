@@ -36,10 +36,34 @@ def consume(q: object, n: int) -> int:
     return total
 
 
+def consume_batched(q: object, n: int) -> int:
+    """get_many(64) variant: amortizes the per-get RPC round trip
+    (reference consumers batch exactly this way with get_many)."""
+    import torch
+
+    have_gpu = torch.cuda.is_available()
+    if have_gpu:
+        cache = torch.ones(2048, device="cuda", dtype=torch.bfloat16)
+    total = 0
+    got = 0
+    while got < n:
+        items = q.get_many(min(64, n - got), timeout=60)
+        if not items:
+            continue
+        got += len(items)
+        for item in items:
+            if have_gpu:
+                total += int((cache * float(item % 5 + 1))[:2].float().sum().item()) and item or item
+            else:
+                total += item
+    return total
+
+
 def main() -> None:
     parser = argparse.ArgumentParser()
     parser.add_argument("--items", type=int, default=20_000)
     parser.add_argument("--consumers", type=int, default=8)
+    parser.add_argument("--batched", action="store_true", help="consumers use get_many(64)")
     args = parser.parse_args()
 
     import torch
@@ -48,7 +72,7 @@ def main() -> None:
 
     app = modal.App("bench-queue")
     gpu = 1 if torch.cuda.is_available() else None
-    consumer = app.function(gpu=gpu)(consume)
+    consumer = app.function(gpu=gpu)(consume_batched if args.batched else consume)
 
     per = args.items // args.consumers
     with app.run():
@@ -76,6 +100,7 @@ def main() -> None:
         "queue_items_per_sec": round(n_done / elapsed, 1),
         "items": n_done,
         "consumers": args.consumers,
+        "batched": args.batched,
         "gpu": bool(gpu),
     }))
 
