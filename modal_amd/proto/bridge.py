@@ -383,6 +383,25 @@ class GrpcBridge:
         )
         return empty_pb2.Empty()
 
+    async def ClientHello(self, request: Any, context: Any) -> Any:
+        """The reference client's first RPC on connect (client.py:209)."""
+        return self.api.ClientHelloResponse(image_builder_version="local")
+
+    async def AppList(self, request: Any, context: Any) -> Any:
+        rows = await self.scheduler.app_list(request.environment_name or "main")
+        out = self.api.AppListResponse()
+        state_map = {
+            "running": self.api.APP_STATE_EPHEMERAL,
+            "deployed": self.api.APP_STATE_DEPLOYED,
+            "stopped": self.api.APP_STATE_STOPPED,
+        }
+        for row in rows:
+            item = out.apps.add()
+            item.app_id = row["app_id"]
+            item.description = row.get("description") or ""
+            item.state = state_map.get(row.get("state"), 0)
+        return out
+
     # -- input plane (Attempt trio) -------------------------------------------
     # parity: _InputPlaneInvocation (reference _functions.py:396-549) —
     # locally the "input plane" IS the scheduler, and the attempt token is

@@ -618,3 +618,14 @@ def test_resource_housekeeping_via_protos(grpc_stream):
     text = "".join(item.data for b in batches for item in b.items)
     assert "log-line-one" in text and "log-line-two" in text
     assert batches[-1].eof
+
+
+def test_client_hello_and_app_list(grpc_plane):
+    api, invoke = grpc_plane
+    from google.protobuf import empty_pb2
+
+    hello = invoke("ClientHello", empty_pb2.Empty(), api.ClientHelloResponse)
+    assert hello.image_builder_version == "local"
+    invoke("AppCreate", api.AppCreateRequest(description="listed-app"), api.AppCreateResponse)
+    apps = invoke("AppList", api.AppListRequest(), api.AppListResponse)
+    assert any(a.description == "listed-app" for a in apps.apps)
