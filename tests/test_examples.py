@@ -55,3 +55,12 @@ def test_example_train_ddp(runner, client):
         os.environ.pop("MODAL_AMD_FORCE_CPU", None)
     assert result.exit_code == 0, result.output
     assert "ranks in sync: True" in result.output
+
+
+def test_example_http_server_class(runner, client):
+    result = runner.invoke(
+        entrypoint_cli, ["run", "examples/http_server_class.py::app.main"]
+    )
+    assert result.exit_code == 0, result.output
+    assert "response: served by EchoServer" in result.output
+    assert "via tunnel: served by EchoServer" in result.output
