@@ -93,6 +93,8 @@ def hash_compress_blocks(
 
     lib = _lib()
     n_blocks = len(blocks)
+    if n_blocks == 0:
+        return [], []
     digests: list = [None] * n_blocks
     containers: list = [None] * n_blocks
 
