@@ -688,7 +688,16 @@ class Scheduler:
         return {
             "function_call_id": record.call_id,
             "retry_policy": self.functions[function_id].retry_policy.to_dict(),
-            "max_inputs_outstanding": MAX_INPUTS_OUTSTANDING_DEFAULT,
+            # server-sized (the reference makes this server-overridable,
+            # parallel_map.py:387): the default 1,000 items is ~7 chunks of
+            # 128 — enough for one worker's pipeline but it STARVES a
+            # multi-GPU pool, so scale with the live worker count
+            "max_inputs_outstanding": max(
+                MAX_INPUTS_OUTSTANDING_DEFAULT,
+                256 * max(
+                    sum(1 for w in self.pool.workers.values() if w.alive), 1
+                ),
+            ),
             "sync_client_retries_enabled": True,
         }
 
