@@ -34,7 +34,7 @@ ITEMS_PER_GPU = 12_500  # x8 GPUs = the 100k-input config of BASELINE.json
 # 288 GB HBM3E per MI355X easily hosts several payload processes; swept
 # per-box (profiles/README.md, r2: 4 workers x 128-item chunks ~1.8x the
 # round-1 3x64 on the same box); override with MODAL_AMD_BENCH_WPG
-WORKERS_PER_GPU = int(os.environ.get("MODAL_AMD_BENCH_WPG", "4"))
+WORKERS_PER_GPU = int(os.environ.get("MODAL_AMD_BENCH_WPG", "8"))
 
 
 def _bench_run_dir() -> str:
