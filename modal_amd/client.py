@@ -75,6 +75,14 @@ class SchedulerProxy:
         call.__name__ = name
         return call
 
+    async def function_put_chunk_oneway(self, **params: Any) -> None:
+        """Fire-and-forget chunk intake: ONE socket frame, no round trip —
+        the map pump stays pipelined across a process boundary (the
+        1000s-of-RTTs put loop was what made the daemon split slower than
+        in-proc in round-2 measurements). Loss model: same socket as
+        everything else; if it drops, the connection is dead anyway."""
+        await self._conn.send({"t": "putc", "p": params})
+
 
 class UserCodeProxy:
     """Scheduler proxy for user code inside worker processes.

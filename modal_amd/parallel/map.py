@@ -120,6 +120,15 @@ async def map_invocation_batches(
         nonlocal total_inputs
         from .._serialization import serialize_fast
 
+        # proxied schedulers take chunk puts as ONE-WAY frames (ordered on
+        # the socket; put_chunk's intake prefix is await-free, so index
+        # bases stay sequential) — a per-chunk RTT would serialize the pump
+        if "function_put_chunk_oneway" in type(svc).__dict__:
+            put_chunk = svc.function_put_chunk_oneway  # real method, not an
+            # RPC stub a __getattr__-based proxy would synthesize
+        else:
+            put_chunk = svc.function_put_chunk
+
         chunk_buf: list = []
         chunk_seq = 0
 
@@ -144,7 +153,7 @@ async def map_invocation_batches(
                 payload = await asyncio.get_running_loop().run_in_executor(
                     None, _spill, payload, chunk_id
                 )
-            await svc.function_put_chunk(
+            await put_chunk(
                 function_call_id=call_id,
                 chunk_id=chunk_id,
                 payload=payload,
@@ -199,7 +208,7 @@ async def map_invocation_batches(
                     return
                 payload = await task
                 prepare_fifo.pop(0)
-                await svc.function_put_chunk(
+                await put_chunk(
                     function_call_id=call_id,
                     chunk_id=chunk_id,
                     payload=payload,
@@ -228,7 +237,7 @@ async def map_invocation_batches(
             payload = serialize_fast(("C2", kwargs_common, argsbatch))
             if len(payload) > 2 * 1024 * 1024:
                 payload = await loop.run_in_executor(None, _spill, payload, chunk_id)
-            await svc.function_put_chunk(
+            await put_chunk(
                 function_call_id=call_id,
                 chunk_id=chunk_id,
                 payload=payload,

@@ -380,6 +380,12 @@ class WorkerPool:
                     # tooling/sandbox client connection: RPC only
                     await conn.send({"t": "hello_ack"})
                 return
+            if kind == "putc":
+                # fire-and-forget chunk intake from a proxied client's pump
+                asyncio.get_running_loop().create_task(
+                    self.scheduler.function_put_chunk(**msg["p"])
+                )
+                return
             handle = handle_holder.get("h")
             if handle is None:
                 return
