@@ -8,6 +8,7 @@ unauthenticated full-control plane.
 from __future__ import annotations
 
 import asyncio
+import time
 import os
 
 import modal_amd as modal
@@ -129,3 +130,35 @@ def test_socket_accepts_token_from_run_dir(client, run_dir):
     # permissions: run_dir 0700, token 0600
     assert os.stat(run_dir).st_mode & 0o777 == 0o700
     assert os.stat(os.path.join(run_dir, "auth.token")).st_mode & 0o777 == 0o600
+
+
+def test_outstanding_cap_scales_with_pool(client):
+    """function_map's max_inputs_outstanding is server-sized to the live
+    worker pool (the reference makes it server-overridable,
+    parallel_map.py:387): the fixed 1,000 default is ~7 chunks of 128 and
+    starves a multi-worker pool."""
+    app = modal.App("cap-app")
+
+    @app.function()
+    def f(x):
+        return x
+
+    with app.run(client=client):
+        f.remote(1)  # ensure at least one worker exists
+        svc = client.svc
+
+        async def probe():
+            resp = await svc.function_map(function_id=f.object_id, kind="map")
+            return resp["max_inputs_outstanding"]
+
+        cap = synchronizer.run(probe())
+        alive = sum(1 for w in client.svc.pool.workers.values() if w.alive)
+        assert cap >= max(1000, 256 * alive)
+        # grow the pool: the cap grows with it
+        synchronizer.run(svc.pool.spawn_worker(gpu_index=None))
+        deadline = time.time() + 30
+        while sum(1 for w in svc.pool.workers.values() if w.alive) <= alive:
+            assert time.time() < deadline
+            time.sleep(0.05)
+        cap2 = synchronizer.run(probe())
+        assert cap2 >= cap + 256
