@@ -304,8 +304,17 @@ class Scheduler:
     async def _call_gc_loop(self) -> None:
         """Drop long-completed call records so a long-lived daemon does
         not accumulate them unboundedly."""
+        gc_rounds = 0
         while True:
             await asyncio.sleep(60.0)
+            gc_rounds += 1
+            if gc_rounds % 10 == 0:
+                try:  # leftover per-sandbox cgroup dirs (members long dead)
+                    from .isolation import cleanup_stale_cgroups
+
+                    cleanup_stale_cgroups()
+                except Exception:
+                    pass
             try:
                 self._gc_calls_once()
             except Exception:
