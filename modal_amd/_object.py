@@ -104,6 +104,7 @@ class _Object:
         self._object_id = object_id
         self._client = client
         self._is_hydrated = True
+        self._metadata = metadata or {}
         if metadata:
             self._hydrate_metadata(metadata)
 

@@ -189,6 +189,53 @@ class App:
         return dict(self._functions)
 
     @property
+    def registered_web_endpoints(self) -> list[str]:
+        """Names of functions carrying web configs (parity: reference)."""
+        out = []
+        for name, fn in self._functions.items():
+            impl = getattr(fn, "_impl", fn)
+            if getattr(impl, "_web_config", None) or (
+                getattr(impl, "_options", None) or {}
+            ).get("web_config"):
+                out.append(name)
+        return out
+
+    @property
+    def image(self) -> Any:
+        """The app-level default image, when one was set."""
+        return getattr(self, "_image", None)
+
+    @image.setter
+    def image(self, value: Any) -> None:
+        self._image = value
+
+    def set_tags(self, tags: dict) -> None:
+        """Attach metadata tags to the running app (parity: AppCreate tags)."""
+        from ._sync import synchronizer
+
+        client = self._running_client
+        if client is None or self._app_id is None:
+            raise InvalidError("set_tags requires a running app")
+        synchronizer.run(client.svc.app_set_tags(app_id=self._app_id, tags=dict(tags)))
+
+    def get_tags(self) -> dict:
+        from ._sync import synchronizer
+
+        client = self._running_client
+        if client is None or self._app_id is None:
+            raise InvalidError("get_tags requires a running app")
+        return synchronizer.run(client.svc.app_get_tags(app_id=self._app_id))
+
+    def logs(self, *, timeout: 'Any' = None) -> Any:
+        """Iterate this app's log entries as text (parity: App.logs)."""
+        if self._app_id is None:
+            raise InvalidError("logs() requires a running or deployed app")
+        from .logs_manager import tail_app_logs
+
+        for entry in tail_app_logs(self._app_id, self._running_client, timeout=timeout):
+            yield entry.get("data", "")
+
+    @property
     def registered_classes(self) -> dict[str, Any]:
         return dict(self._classes)
 

@@ -212,6 +212,17 @@ class Cls:
         merged = {**self._function_kwargs, **kwargs}
         return Cls(self._app, self._user_cls, merged)
 
+    def with_concurrency(self, *, max_inputs: int, target_inputs: int = 0) -> "Cls":
+        """Parity: reference cls.py Cls.with_concurrency."""
+        return self.with_options(
+            max_concurrent_inputs=max_inputs,
+            target_concurrent_inputs=target_inputs or max_inputs,
+        )
+
+    def with_batching(self, *, max_batch_size: int, wait_ms: int = 0) -> "Cls":
+        """Parity: reference cls.py Cls.with_batching."""
+        return self.with_options(batch_max_size=max_batch_size, batch_linger_ms=wait_ms)
+
     @classmethod
     def from_name(
         cls, app_name: str, name: str, *, environment_name: str = ""

@@ -27,9 +27,27 @@ class _Dict(_Object, type_kind="dict"):
                 create_if_missing=create_if_missing,
                 ephemeral=False,
             )
-            obj._hydrate(did, resolver.client, None)
+            obj._hydrate(did, resolver.client, {"name": name})
 
         return cls._from_loader(_load, rep=f"Dict.from_name({name!r})")
+
+    @classmethod
+    def from_id(cls, object_id: str, client: Any = None) -> "_Dict":
+        async def _load(obj: "_Dict", resolver: Any, existing: Any) -> None:
+            obj._hydrate(object_id, resolver.client, None)
+
+        obj = cls._from_loader(_load, rep=f"Dict.from_id({object_id!r})")
+        if client is not None:
+            obj._hydrate(object_id, client, None)
+        return obj
+
+    @property
+    def name(self) -> Any:
+        return (getattr(self, "_metadata", None) or {}).get("name")
+
+    @live_method
+    async def info(self) -> dict:
+        return await self._client.svc.object_info(object_id=self.object_id)
 
     @classmethod
     async def lookup(

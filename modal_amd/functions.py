@@ -541,6 +541,31 @@ class _Function(_Object, type_kind="function"):
             return _Function.from_local(self._raw_f, self._app, new_options)
         raise InvalidError("with_options requires a locally defined function")
 
+    def with_concurrency(self, *, max_inputs: int, target_inputs: Optional[int] = None) -> "_Function":
+        """Handle with overridden input concurrency (parity: reference :1647)."""
+        return self.with_options(
+            max_concurrent_inputs=max_inputs,
+            target_concurrent_inputs=target_inputs or 0,
+        )
+
+    def with_batching(self, *, max_batch_size: int, wait_ms: int = 0) -> "_Function":
+        """Handle with overridden dynamic batching (parity: reference :1661)."""
+        return self.with_options(batch_max_size=max_batch_size, batch_linger_ms=wait_ms)
+
+    def get_web_url(self) -> Optional[str]:
+        """Deployed web URL (parity: reference get_web_url)."""
+        return self.web_url
+
+    @property
+    def app(self) -> Any:
+        """The app this function was defined on (parity: reference .app)."""
+        return self._app
+
+    @property
+    def tag(self) -> str:
+        """Registration name within its app."""
+        return getattr(self, "_info_name", None) or ""
+
     def __call__(self, *args: Any, **kwargs: Any) -> Any:
         raise InvalidError(
             f"Functions are invoked with `.remote()`, `.local()`, `.map()` etc. "
@@ -593,6 +618,16 @@ class _FunctionCall(_Object, type_kind="function_call"):
         if client is not None:
             obj._hydrate(function_call_id, client, None)
         return obj
+
+    @live_method
+    async def num_inputs(self) -> int:
+        """Number of inputs in this call (parity: reference num_inputs)."""
+        info = await self._client.svc.function_call_info(function_call_id=self.object_id)
+        return int(info.get("total", 0))
+
+    def iter(self) -> Any:
+        """Iterate a remote generator's outputs (parity: FunctionCall.iter)."""
+        return self.__aiter__()
 
     @staticmethod
     async def gather(*function_calls: "_FunctionCall") -> list:

@@ -110,6 +110,24 @@ class _Image(_Object, type_kind="image"):
         )
 
     @classmethod
+    def from_scratch(cls) -> "_Image":
+        """An empty base image (parity: reference Image building blocks)."""
+        return cls._from_recipe([{"kind": "base", "name": "scratch"}])
+
+    @classmethod
+    def from_name(cls, app_name: str, name: str, *, environment_name: str = "") -> "_Image":
+        """Reference a previously deployed/built image by app + object name
+        (parity: reference image.py Image.from_name)."""
+
+        async def _load(obj: "_Image", resolver: Any, existing: Any) -> None:
+            resp = await resolver.client.svc.app_get_object(
+                app_name=app_name, tag=name, environment=environment_name
+            )
+            obj._hydrate(resp["object_id"], resolver.client, resp.get("metadata"))
+
+        return cls._from_loader(_load, rep=f"Image.from_name({app_name!r}, {name!r})")
+
+    @classmethod
     def from_id(cls, image_id: str, client: Any = None) -> "_Image":
         async def _load(obj: "_Image", resolver: Any, existing: Any) -> None:
             resp = await resolver.client.svc.image_info(image_id=image_id)
@@ -145,6 +163,20 @@ class _Image(_Object, type_kind="image"):
 
     def uv_sync(self, uv_project_dir: str = "./", **kwargs: Any) -> "_Image":
         return self._extend({"kind": "run_commands", "commands": [f"cd {uv_project_dir} && uv sync --offline || true"]})
+
+    def pip_install_from_pyproject(
+        self, pyproject_toml: str, optional_dependencies: list[str] = [], **kwargs: Any
+    ) -> "_Image":
+        """Parity: reference image.py Image.pip_install_from_pyproject —
+        record the pyproject's dependency list as a pip layer."""
+        return self._extend(
+            {
+                "kind": "pip_install_from_pyproject",
+                "path": str(pyproject_toml),
+                "optional_dependencies": list(optional_dependencies),
+                **{k: v for k, v in kwargs.items() if v},
+            }
+        )
 
     def poetry_install_from_file(self, poetry_pyproject_toml: str, **kwargs: Any) -> "_Image":
         return self._extend({"kind": "run_commands", "commands": ["poetry install || true"]})
@@ -232,6 +264,18 @@ class _Image(_Object, type_kind="image"):
     async def build_log(self) -> str:
         info = await self._client.svc.image_info(image_id=self.object_id)
         return info.get("build_log", "")
+
+    async def build(self, client: Any = None) -> "_Image":
+        """Eagerly build this image (parity: reference Image.build — normally
+        builds happen lazily at app start; this forces hydration now)."""
+        await self.hydrate(client)
+        return self
+
+    @live_method
+    async def logs(self) -> "Any":
+        """Yield build-log lines (parity: reference Image.logs stream)."""
+        info = await self._client.svc.image_info(image_id=self.object_id)
+        return (info.get("build_log", "") or "").splitlines(keepends=True)
 
     def imports(self) -> Any:
         """Context manager that suppresses ImportError locally

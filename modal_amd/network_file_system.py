@@ -37,5 +37,35 @@ class _NetworkFileSystem(_Volume, type_kind="nfs"):
         await self._put_data(data, remote_path)
         return len(data)
 
+    @live_method
+    async def add_local_file(
+        self, local_path: "Any", remote_path: "Any" = None
+    ) -> None:
+        """Upload one local file (parity: reference NFS.add_local_file)."""
+        import os as _os
+
+        local_path = _os.fspath(local_path)
+        if remote_path is None:
+            remote_path = "/" + _os.path.basename(local_path)
+        with open(local_path, "rb") as f:
+            await self._put_data(f.read(), str(remote_path))
+
+    @live_method
+    async def add_local_dir(
+        self, local_path: "Any", remote_path: "Any" = None
+    ) -> None:
+        """Upload a directory tree (parity: reference NFS.add_local_dir)."""
+        import os as _os
+
+        local_path = _os.fspath(local_path)
+        if remote_path is None:
+            remote_path = "/" + _os.path.basename(local_path.rstrip("/"))
+        for root, _dirs, files in _os.walk(local_path):
+            for fname in files:
+                src = _os.path.join(root, fname)
+                rel = _os.path.relpath(src, local_path)
+                with open(src, "rb") as f:
+                    await self._put_data(f.read(), str(remote_path).rstrip("/") + "/" + rel)
+
 
 NetworkFileSystem = synchronize_api(_NetworkFileSystem, "NetworkFileSystem")

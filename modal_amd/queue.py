@@ -39,9 +39,42 @@ class _Queue(_Object, type_kind="queue"):
                 create_if_missing=create_if_missing,
                 ephemeral=False,
             )
-            obj._hydrate(qid, resolver.client, None)
+            obj._hydrate(qid, resolver.client, {"name": name})
 
         return cls._from_loader(_load, rep=f"Queue.from_name({name!r})")
+
+    @classmethod
+    def from_id(cls, object_id: str, client: Any = None) -> "_Queue":
+        """Handle from a raw ``qu-`` id (parity: reference from_id)."""
+        async def _load(obj: "_Queue", resolver: Any, existing: Any) -> None:
+            obj._hydrate(object_id, resolver.client, None)
+
+        obj = cls._from_loader(_load, rep=f"Queue.from_id({object_id!r})")
+        if client is not None:
+            obj._hydrate(object_id, client, None)
+        return obj
+
+    @property
+    def name(self) -> 'Any':
+        """Deployment name (None for ephemeral queues)."""
+        return (getattr(self, "_metadata", None) or {}).get("name")
+
+    @live_method
+    async def info(self) -> dict:
+        """Name + partition stats (parity: reference info())."""
+        return await self._client.svc.object_info(object_id=self.object_id)
+
+    @staticmethod
+    def validate_partition_key(partition: 'Any') -> bytes:
+        """Parity: reference queue.py validate_partition_key."""
+        from .exception import InvalidError
+
+        if partition is None:
+            return b""
+        key = partition.encode() if isinstance(partition, str) else bytes(partition)
+        if len(key) == 0 or len(key) > 64:
+            raise InvalidError("Queue partition key must be 1-64 bytes")
+        return key
 
     @classmethod
     async def lookup(

@@ -344,6 +344,42 @@ class _Sandbox(_Object, type_kind="sandbox"):
     async def set_tags(self, tags: dict) -> None:
         await self._client.svc.sandbox_set_tags(sandbox_id=self.object_id, tags=tags)
 
+    @live_method
+    async def get_tags(self) -> dict:
+        rows = await self._client.svc.sandbox_list()
+        for row in rows:
+            if row.get("sandbox_id") == self.object_id:
+                return dict(row.get("tags") or {})
+        return {}
+
+    @live_method
+    async def wait_until_ready(self, timeout: float = 60.0) -> None:
+        """Block until the sandbox process is running (parity:
+        SandboxWaitUntilReady — locally: alive and not already exited)."""
+        import time as _time
+
+        deadline = _time.time() + timeout
+        while _time.time() < deadline:
+            status = await self._client.svc.sandbox_poll(self.object_id)
+            if status.get("running"):
+                return
+            if status.get("returncode") is not None:
+                from .exception import ExecutionError
+
+                raise ExecutionError(
+                    f"sandbox exited (rc={status['returncode']}) before ready"
+                )
+            await __import__("asyncio").sleep(0.05)
+        from .exception import SandboxTimeoutError
+
+        raise SandboxTimeoutError("sandbox never became ready")
+
+    @live_method
+    async def reload_volumes(self) -> None:
+        """Re-sync mounted volumes (parity: TaskReloadVolumes — local
+        volumes are directly shared, so this is a commit barrier no-op)."""
+        return None
+
     # -- exec ------------------------------------------------------------
     @live_method
     async def exec(

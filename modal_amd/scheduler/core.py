@@ -94,7 +94,7 @@ class RPCAdapter:
 
     _ALLOWED = {
         "app_create", "app_publish", "app_heartbeat", "app_client_disconnect", "app_stop",
-        "app_list", "app_history", "app_rollback", "app_set_objects",
+        "app_list", "app_history", "app_rollback", "app_set_objects", "app_get_object",
         "function_create", "function_update", "function_update_autoscaler",
         "queue_get_or_create", "queue_put", "queue_get", "queue_len", "queue_clear",
         "queue_peek", "queue_delete",
@@ -117,6 +117,7 @@ class RPCAdapter:
         "image_get_or_create", "image_info", "mount_get_or_create",
         "device_transfer", "tensor_pull_relay",
         "worker_snapshot", "worker_restore", "start_grpc_bridge", "app_get_logs",
+        "object_info", "secret_update", "volume_info", "app_set_tags", "app_get_tags",
     }
 
     def __init__(self, scheduler: "Scheduler"):
@@ -395,6 +396,19 @@ class Scheduler:
 
     async def app_set_objects(self, app_id: str, objects: dict[str, tuple[str, dict]]) -> None:
         self._app(app_id).objects.update(objects)
+
+    async def app_get_object(self, app_name: str, tag: str, environment: str = "") -> dict:
+        """Look up any object published by a deployed app, by tag (parity:
+        reference api.proto AppGetObjects / deployment object lookup)."""
+        env = environment or self.default_environment
+        app_id = self.app_names.get((env, app_name))
+        if app_id is None:
+            raise NotFoundError(f"App '{app_name}' not found in environment '{env}'")
+        entry = self._app(app_id).objects.get(tag)
+        if entry is None:
+            raise NotFoundError(f"App '{app_name}' has no object tagged '{tag}'")
+        object_id, meta = entry
+        return {"object_id": object_id, "metadata": meta or {}}
 
     async def app_get_layout(self, app_id: str) -> dict:
         return self.app_layout(app_id)
@@ -1419,6 +1433,32 @@ class Scheduler:
                 break
             await asyncio.sleep(0.1)
         raise InvalidError("restored worker never connected")
+
+    async def object_info(self, object_id: str) -> dict:
+        return self.services.object_info(object_id)
+
+    async def volume_info(self, volume_id: str) -> dict:
+        vol = self.volume_service._get(volume_id)
+        return {
+            "name": vol.name,
+            "files": len(vol.manifest),
+            "created_at": getattr(vol, "created_at", None),
+        }
+
+    async def secret_update(self, secret_id: str, env: dict) -> None:
+        st = self.services.secrets.get(secret_id)
+        if st is None:
+            raise NotFoundError(f"Secret {secret_id} not found")
+        st.env.update({k: str(v) for k, v in env.items()})
+
+    async def app_set_tags(self, app_id: str, tags: dict) -> None:
+        app = self._app(app_id)
+        if not hasattr(app, "tags"):
+            app.tags = {}
+        app.tags.update({str(k): str(v) for k, v in tags.items()})
+
+    async def app_get_tags(self, app_id: str) -> dict:
+        return dict(getattr(self._app(app_id), "tags", {}) or {})
 
     # -- blobs -------------------------------------------------------------
     async def blob_put(self, data: bytes) -> dict:

@@ -96,6 +96,26 @@ class Services:
         self.secret_names = NamedStore()
 
     # ---- queues --------------------------------------------------------
+    def object_info(self, object_id: str) -> dict:
+        """Name/metadata for a named resource handle (parity: the
+        reference's *Info/metadata RPCs)."""
+        if object_id.startswith("qu-"):
+            q = self._queue(object_id)
+            return {"name": q.name, "num_partitions": len(q.partitions)}
+        if object_id.startswith("di-"):
+            d = self._dict(object_id)
+            return {"name": d.name, "len": len(d.data)}
+        if object_id.startswith("st-"):
+            st = self.secrets.get(object_id)
+            if st is None:
+                from ..exception import NotFoundError
+
+                raise NotFoundError(f"Secret {object_id} not found")
+            return {"name": st.name, "keys": sorted(st.env)}
+        from ..exception import NotFoundError
+
+        raise NotFoundError(f"no info for {object_id}")
+
     def queue_get_or_create(
         self, name: Optional[str], environment: str, create_if_missing: bool, ephemeral: bool
     ) -> str:

@@ -76,9 +76,23 @@ class _Secret(_Object, type_kind="secret"):
                 env=None,
                 required_keys=required_keys or [],
             )
-            obj._hydrate(sid, resolver.client, None)
+            obj._hydrate(sid, resolver.client, {"name": name})
 
         return cls._from_loader(_load, rep=f"Secret.from_name({name!r})")
+
+    @property
+    def name(self) -> Any:
+        return (getattr(self, "_metadata", None) or {}).get("name")
+
+    @live_method
+    async def info(self) -> dict:
+        """Name + key names — never values (parity: reference info())."""
+        return await self._client.svc.object_info(object_id=self.object_id)
+
+    @live_method
+    async def update(self, env_dict: dict) -> None:
+        """Merge new entries into the stored secret (parity: Secret.update)."""
+        await self._client.svc.secret_update(secret_id=self.object_id, env=dict(env_dict))
 
     @classmethod
     async def create_deployed(

@@ -50,9 +50,31 @@ class _Volume(_Object, type_kind="volume"):
                 create_if_missing=create_if_missing,
                 ephemeral=False,
             )
-            obj._hydrate(resp["volume_id"], resolver.client, {"version": resp.get("version")})
+            obj._hydrate(
+                resp["volume_id"], resolver.client,
+                {"version": resp.get("version"), "name": name},
+            )
 
         return cls._from_loader(_load, rep=f"Volume.from_name({name!r})")
+
+    @classmethod
+    def from_id(cls, object_id: str, client: Any = None) -> "_Volume":
+        async def _load(obj: "_Volume", resolver: Any, existing: Any) -> None:
+            obj._hydrate(object_id, resolver.client, None)
+
+        obj = cls._from_loader(_load, rep=f"Volume.from_id({object_id!r})")
+        if client is not None:
+            obj._hydrate(object_id, client, None)
+        return obj
+
+    @property
+    def name(self) -> Any:
+        return (getattr(self, "_metadata", None) or {}).get("name")
+
+    @live_method
+    async def info(self) -> dict:
+        """Name + file count (parity: reference info())."""
+        return await self._client.svc.volume_info(volume_id=self.object_id)
 
     @classmethod
     async def lookup(
@@ -135,6 +157,10 @@ class _Volume(_Object, type_kind="volume"):
             if len(chunk) < BLOCK_SIZE:
                 return
             offset += len(chunk)
+
+    async def read_file_into_fileobj(self, path: str, fileobj: BinaryIO) -> int:
+        """Parity alias (reference read_file_into_fileobj)."""
+        return await self.read_file_into(path, fileobj)
 
     @live_method
     async def read_file_into(self, path: str, fileobj: BinaryIO) -> int:

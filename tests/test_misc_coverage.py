@@ -359,3 +359,114 @@ def test_type_stub_generation(tmp_path):
 
     for name in modal_amd.__all__:
         assert name in src, f"{name} missing from stubs"
+
+
+def test_api_surface_parity_batch(client, tmp_path):
+    """Round-2 parity batch: members the AST audit flagged missing vs the
+    reference (App tags/logs, Queue/Dict/Secret/Volume from_id+info,
+    Cls.with_concurrency, Image.from_scratch, NFS.add_local_*)."""
+    import io
+
+    # Queue / Dict / Secret / Volume: from_id + info + name
+    q = modal.Queue.from_name("parity-q", create_if_missing=True)
+    q.put(1)
+    q2 = modal.Queue.from_id(q.object_id)
+    assert q2.get() == 1
+    info = q.info()
+    assert info.get("name") == "parity-q"
+    assert q.name == "parity-q"
+    modal.Queue.validate_partition_key(b"x" * 10)
+    with pytest.raises(Exception):
+        modal.Queue.validate_partition_key(b"x" * 5000)
+
+    d = modal.Dict.from_name("parity-d", create_if_missing=True)
+    d["k"] = "v"
+    assert modal.Dict.from_id(d.object_id)["k"] == "v"
+    assert d.info().get("name") == "parity-d"
+
+    modal.Secret.create_deployed("parity-s", {"A": "0"})
+    s = modal.Secret.from_name("parity-s")
+    s.update(env_dict={"A": "1"})
+    assert s.info().get("name") == "parity-s"
+
+    v = modal.Volume.from_name("parity-v", create_if_missing=True)
+    with v.batch_upload() as b:
+        b.put_file(io.BytesIO(b"hello parity"), "/f.bin")
+    buf = io.BytesIO()
+    n = modal.Volume.from_id(v.object_id).read_file_into_fileobj("/f.bin", buf)
+    assert n == 12 and buf.getvalue() == b"hello parity"
+
+    # NFS add_local_file / add_local_dir
+    nfs = modal.NetworkFileSystem.from_name("parity-nfs", create_if_missing=True)
+    local = tmp_path / "one.txt"
+    local.write_text("one")
+    nfs.add_local_file(local)
+    assert b"".join(nfs.read_file("one.txt")) == b"one"
+    tree = tmp_path / "tree"
+    (tree / "sub").mkdir(parents=True)
+    (tree / "sub" / "two.txt").write_text("two")
+    nfs.add_local_dir(tree, "/t")
+    assert b"".join(nfs.read_file("t/sub/two.txt")) == b"two"
+
+    # Image: from_scratch / pip_install_from_pyproject / build / logs / from_name
+    img = modal.Image.from_scratch().pip_install_from_pyproject("pyproject.toml")
+    img.build()
+    assert isinstance(img.logs(), list)
+
+    # App: tags + registered_web_endpoints + image property
+    app = modal.App("parity-app", image=img)
+    assert app.image is img
+
+    @app.function()
+    def plain(x):
+        return x + 1
+
+    @app.function()
+    @modal.fastapi_endpoint(method="GET")
+    def webby():
+        return "hi"
+
+    assert app.registered_web_endpoints == ["webby"]
+    with app.run(client=client):
+        app.set_tags({"team": "amd"})
+        assert app.get_tags() == {"team": "amd"}
+        assert plain.remote(1) == 2
+        # Cls.with_concurrency / with_batching exist and chain
+        from modal_amd.cls import Cls
+        assert hasattr(Cls, "with_concurrency") and hasattr(Cls, "with_batching")
+        # FunctionCall num_inputs + Function accessors
+        fc = plain.spawn(5)
+        assert fc.get() == 6
+        assert fc.num_inputs() >= 1
+        assert plain.app is app or plain.app is not None
+
+
+def test_sandbox_wait_until_ready_and_tags(client):
+    sb = modal.Sandbox.create("sleep", "0.4", client=client)
+    sb.wait_until_ready(timeout=10)
+    sb.set_tags({"k": "v"})
+    assert sb.get_tags() == {"k": "v"}
+    sb.reload_volumes()
+    sb.wait()
+    assert sb.returncode == 0
+
+
+def test_image_from_name_deployed(client):
+    img = modal.Image.from_scratch()
+    app = modal.App("img-holder")
+
+    @app.function(image=img)
+    def f():
+        return 1
+
+    app.deploy(name="img-holder", client=client)
+    from modal_amd._sync import synchronizer
+
+    synchronizer.run(
+        client.svc.app_set_objects(
+            app_id=app.app_id, objects={"base": (img.object_id, {})}
+        )
+    )
+    got = modal.Image.from_name("img-holder", "base")
+    got.build(client=client)
+    assert got.object_id == img.object_id
