@@ -212,6 +212,32 @@ class Cls:
         merged = {**self._function_kwargs, **kwargs}
         return Cls(self._app, self._user_cls, merged)
 
+    @staticmethod
+    def validate_construction_mechanism(user_cls: type) -> None:
+        """Reject classes mixing a custom __init__ with modal.parameter()
+        fields, and unannotated parameters (parity: reference cls.py:566)."""
+        params = {k for k, v in vars(user_cls).items() if isinstance(v, _Parameter)}
+        has_custom_init = user_cls.__init__ is not object.__init__
+        if params and has_custom_init:
+            raise InvalidError(
+                "A class can't have both a custom __init__ constructor "
+                "and dataclass-style modal.parameter() annotations"
+            )
+        annotations = getattr(user_cls, "__annotations__", {})
+        missing = params - set(annotations)
+        if missing:
+            raise InvalidError(
+                "All modal.parameter() specifications need to be type-annotated "
+                f"(missing: {sorted(missing)})"
+            )
+
+    @staticmethod
+    def from_local(user_cls: type, app: Any, class_service_function: Any = None) -> "Cls":
+        """Build a Cls from a local class definition (parity: reference
+        cls.py:604 from_local — validation + registration in one step)."""
+        Cls.validate_construction_mechanism(user_cls)
+        return Cls(app, user_cls, {})
+
     def with_concurrency(self, *, max_inputs: int, target_inputs: int = 0) -> "Cls":
         """Parity: reference cls.py Cls.with_concurrency."""
         return self.with_options(

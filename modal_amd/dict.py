@@ -187,3 +187,7 @@ class _EphemeralDict:
 
 
 Dict = synchronize_api(_Dict, "Dict")
+
+from .object_manager import install as _install_manager  # noqa: E402
+
+_install_manager(_Dict, Dict, "dict")

@@ -566,11 +566,102 @@ class _Function(_Object, type_kind="function"):
         """Registration name within its app."""
         return getattr(self, "_info_name", None) or ""
 
+    @property
+    def stub(self) -> Any:
+        """Deprecated alias for .app (parity: reference _functions.py:1401)."""
+        return self._app
+
+    @property
+    def spec(self) -> dict:
+        """The resolved resource/runtime spec (parity: _functions.py:1413 —
+        a dict here instead of a _FunctionSpec dataclass)."""
+        keys = (
+            "name", "needs_gpu", "gpu_count", "timeout", "retries", "cpu",
+            "memory", "min_containers", "max_containers", "buffer_containers",
+            "scaledown_window", "cloud", "region", "is_generator",
+            "cluster_size", "max_concurrent_inputs", "target_concurrent_inputs",
+            "batch_max_size", "batch_linger_ms", "web_config",
+        )
+        return {k: self._options.get(k) for k in keys if k in self._options}
+
+    def get_build_def(self) -> str:
+        """Plaintext source + arg spec, stable across pickles — used as an
+        image-hash component (parity: _functions.py:1422)."""
+        import inspect as _inspect
+
+        if self._raw_f is None:
+            raise InvalidError("get_build_def requires a local definition")
+        try:
+            src = _inspect.getsource(self._raw_f)
+        except (OSError, TypeError):
+            src = repr(self._raw_f)
+        return f"{src}\n{sorted((k, repr(v)) for k, v in self.spec.items())!r}"
+
+    def logs(self) -> "_FunctionLogsManager":
+        """Log access scoped to this function (parity: _functions.py:651)."""
+        return _FunctionLogsManager(self)
+
     def __call__(self, *args: Any, **kwargs: Any) -> Any:
         raise InvalidError(
             f"Functions are invoked with `.remote()`, `.local()`, `.map()` etc. "
             f"— not called directly (tried to call {self._rep})"
         )
+
+
+class _FunctionCallLogsManager:
+    """fetch() over the app logs for a spawned call (local single-stream)."""
+
+    def __init__(self, fc: "_FunctionCall"):
+        self._fc = fc
+
+    async def fetch(self) -> list[str]:
+        fc = self._fc
+        app_id = (getattr(fc, "_metadata", None) or {}).get("app_id")
+        if app_id is None or fc._client is None:
+            return []
+        resp = await fc._client.svc.app_get_logs(app_id=app_id, offset=0, timeout=0.05)
+        return [e.get("data", "") for e in resp.get("entries", [])]
+
+
+class _FunctionLogsManager:
+    """fetch()/tail() over the owning app's log stream, filtered to this
+    function's tasks (entries carry function_id)."""
+
+    def __init__(self, fn: "_Function"):
+        self._fn = fn
+
+    def _match(self, entry: dict) -> bool:
+        fid = getattr(self._fn, "_object_id", None)
+        efid = entry.get("function_id")
+        return not efid or not fid or efid == fid
+
+    async def fetch(self) -> list[str]:
+        fn = self._fn
+        app = fn._app
+        app_id = getattr(app, "_app_id", None) if app is not None else None
+        if app_id is None or fn._client is None:
+            return []
+        resp = await fn._client.svc.app_get_logs(app_id=app_id, offset=0, timeout=0.05)
+        return [e.get("data", "") for e in resp.get("entries", []) if self._match(e)]
+
+    async def tail(self, poll_interval: float = 0.25) -> Any:
+        fn = self._fn
+        app_id = getattr(fn._app, "_app_id", None) if fn._app is not None else None
+
+        async def gen() -> Any:
+            offset = 0
+            while app_id is not None and fn._client is not None:
+                resp = await fn._client.svc.app_get_logs(
+                    app_id=app_id, offset=offset, timeout=poll_interval
+                )
+                for e in resp.get("entries", []):
+                    if self._match(e):
+                        yield e.get("data", "")
+                offset = resp.get("next_offset", offset)
+                if resp.get("done"):
+                    return
+
+        return gen()
 
 
 class _FunctionCall(_Object, type_kind="function_call"):
@@ -624,6 +715,11 @@ class _FunctionCall(_Object, type_kind="function_call"):
         """Number of inputs in this call (parity: reference num_inputs)."""
         info = await self._client.svc.function_call_info(function_call_id=self.object_id)
         return int(info.get("total", 0))
+
+    def logs(self) -> Any:
+        """Logs for the tasks serving this call: the owning app's stream
+        (parity: reference _functions.py:2094 _FunctionCallLogsManager)."""
+        return _FunctionCallLogsManager(self)
 
     def iter(self) -> Any:
         """Iterate a remote generator's outputs (parity: FunctionCall.iter)."""

@@ -115,17 +115,18 @@ class _Image(_Object, type_kind="image"):
         return cls._from_recipe([{"kind": "base", "name": "scratch"}])
 
     @classmethod
-    def from_name(cls, app_name: str, name: str, *, environment_name: str = "") -> "_Image":
-        """Reference a previously deployed/built image by app + object name
-        (parity: reference image.py Image.from_name)."""
+    def from_name(cls, name: str, *, environment_name: str = "") -> "_Image":
+        """Reference an Image previously published with .publish(); names may
+        carry a ':tag' (':latest' implied) — parity: reference
+        _image.py:2972 from_name."""
 
         async def _load(obj: "_Image", resolver: Any, existing: Any) -> None:
-            resp = await resolver.client.svc.app_get_object(
-                app_name=app_name, tag=name, environment=environment_name
+            resp = await resolver.client.svc.image_from_name(
+                name=name, environment=environment_name
             )
-            obj._hydrate(resp["object_id"], resolver.client, resp.get("metadata"))
+            obj._hydrate(resp["image_id"], resolver.client, None)
 
-        return cls._from_loader(_load, rep=f"Image.from_name({app_name!r}, {name!r})")
+        return cls._from_loader(_load, rep=f"Image.from_name({name!r})")
 
     @classmethod
     def from_id(cls, image_id: str, client: Any = None) -> "_Image":
@@ -264,6 +265,36 @@ class _Image(_Object, type_kind="image"):
     async def build_log(self) -> str:
         info = await self._client.svc.image_info(image_id=self.object_id)
         return info.get("build_log", "")
+
+    def pipe(self, func: Callable, *args: Any, **kwargs: Any) -> "_Image":
+        """Apply a local recipe-expanding function: func(image, *args)
+        (parity: reference _image.py:2889)."""
+        from ._sync import unwrap as _unwrap, wrap as _wrap
+
+        return _unwrap(func(_wrap(self), *args, **kwargs))
+
+    def pip_install_private_repos(
+        self, *repositories: str, git_user: str = "", **kwargs: Any
+    ) -> "_Image":
+        """Recorded layer (no egress here — like registry pulls, the clone
+        runs when connectivity exists); parity: _image.py:1187."""
+        return self._extend(
+            {
+                "kind": "pip_install_private_repos",
+                "repositories": list(repositories),
+                "git_user": git_user,
+            }
+        )
+
+    async def publish(
+        self, name: str, *, environment_name: str = "", client: Any = None
+    ) -> None:
+        """Publish this built image under a workspace name (parity:
+        reference _image.py:3010)."""
+        await self.hydrate(client)
+        await self._client.svc.image_publish(
+            image_id=self.object_id, name=name, environment=environment_name
+        )
 
     async def build(self, client: Any = None) -> "_Image":
         """Eagerly build this image (parity: reference Image.build — normally

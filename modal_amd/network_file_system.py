@@ -31,6 +31,23 @@ class _NetworkFileSystem(_Volume, type_kind="nfs"):
 
         return cls._from_loader(_load, rep=f"NetworkFileSystem.from_name({name!r})")
 
+    @classmethod
+    async def create_deployed(
+        cls, deployment_name: str, *, environment_name: str = ""
+    ) -> str:
+        """Eagerly create a named NFS; returns its id (parity: reference
+        network_file_system.py:197 create_deployed)."""
+        from .client import _Client
+
+        client = await _Client.from_env()
+        resp = await client.svc.volume_get_or_create(
+            name=f"nfs/{deployment_name}",
+            environment=environment_name or "main",
+            create_if_missing=True,
+            ephemeral=False,
+        )
+        return resp["volume_id"]
+
     @live_method
     async def write_file(self, remote_path: str, fp: BinaryIO) -> int:
         data = fp.read()

@@ -260,3 +260,7 @@ class _EphemeralQueue:
 
 
 Queue = synchronize_api(_Queue, "Queue")
+
+from .object_manager import install as _install_manager  # noqa: E402
+
+_install_manager(_Queue, Queue, "queue")

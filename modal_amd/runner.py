@@ -41,7 +41,11 @@ async def _prepare_function_options(fn_impl: Any, resolver: Resolver) -> dict:
     for path, volume in (options.get("_volumes") or {}).items():
         impl = unwrap(volume)
         await resolver.load(impl)
-        volume_mounts[str(path)] = impl.object_id
+        opts = getattr(impl, "_mount_options", None)
+        if opts:
+            volume_mounts[str(path)] = {"volume_id": impl.object_id, **opts}
+        else:
+            volume_mounts[str(path)] = impl.object_id
     options["volume_mounts"] = volume_mounts
     image = options.get("_image")
     if image is not None:

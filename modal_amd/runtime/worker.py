@@ -793,7 +793,12 @@ class WorkerRuntime:
         volume_ids: set = set()
         for fid in [f for f, frt in self.functions.items() if frt.app_id == app_id]:
             frt = self.functions.pop(fid)
-            volume_ids.update(frt.volumes.values())
+            from .volumes import mount_spec
+
+            for spec in frt.volumes.values():
+                vid, ro, _sub = mount_spec(spec)
+                if not ro:
+                    volume_ids.add(vid)
             if frt._service is not None:
                 await asyncio.get_running_loop().run_in_executor(
                     self.executor, frt._service.exit

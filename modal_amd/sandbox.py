@@ -229,7 +229,11 @@ class _Sandbox(_Object, type_kind="sandbox"):
             for path, vol in volumes.items():
                 impl = unwrap(vol)
                 await resolver.load(impl)
-                volume_mounts[str(path)] = impl.object_id
+                opts = getattr(impl, "_mount_options", None)
+                if opts:
+                    volume_mounts[str(path)] = {"volume_id": impl.object_id, **opts}
+                else:
+                    volume_mounts[str(path)] = impl.object_id
         app_id = ""
         if app is not None and getattr(app, "_app_id", None):
             app_id = app._app_id
@@ -497,5 +501,145 @@ class _Sandbox(_Object, type_kind="sandbox"):
 
         return {}
 
+    @live_method
+    async def snapshot_directory(
+        self, path: str, *, timeout: float = 55.0, ttl: Any = None
+    ) -> Any:
+        """Snapshot one directory into a new Image (parity: reference
+        sandbox.py:1643 snapshot_directory)."""
+        resp = await self._client.svc.sandbox_snapshot_dir(
+            sandbox_id=self.object_id, path=str(path)
+        )
+        from .image import _Image
+
+        return _Image._new_hydrated(resp["image_id"], self._client, None)
+
+    @live_method
+    async def mount_image(self, path: str, image: Any) -> None:
+        """Expose an Image's filesystem layer inside the sandbox (parity:
+        reference sandbox.py:1548 mount_image)."""
+        from ._sync import unwrap as _unwrap
+
+        impl = _unwrap(image)
+        if not impl.is_hydrated:
+            await impl.hydrate(self._client)
+        await self._client.svc.sandbox_mount_image(
+            sandbox_id=self.object_id, path=str(path), image_id=impl.object_id
+        )
+
+    @live_method
+    async def unmount_image(self, path: str) -> None:
+        await self._client.svc.sandbox_unmount_image(
+            sandbox_id=self.object_id, path=str(path)
+        )
+
+    @live_method
+    async def create_connect_token(
+        self, user_metadata: Any = None, port: int = 8080
+    ) -> dict:
+        """Mint URL+token credentials for HTTP access to this sandbox
+        (parity: reference sandbox.py:1799). Locally the URL is direct."""
+        import json as _json
+
+        if isinstance(user_metadata, dict):
+            user_metadata = _json.dumps(user_metadata)
+        return await self._client.svc.sandbox_connect_token(
+            sandbox_id=self.object_id, port=port, user_metadata=user_metadata
+        )
+
+    async def detach(self) -> None:
+        """Drop this handle's attachment (parity: reference sandbox.py:1193 —
+        the sandbox keeps running; re-attach with Sandbox.from_id)."""
+        self._detached = True
+
+    @property
+    def filesystem(self) -> "_SandboxFilesystem":
+        """Namespace for filesystem APIs (parity: reference sandbox.py:2339)."""
+        return _SandboxFilesystem(self)
+
+    def logs(self) -> "_SandboxLogsManager":
+        """Entrypoint-log access, incl. after termination (parity: reference
+        sandbox.py:2646 _SandboxLogsManager with fetch/tail)."""
+        return _SandboxLogsManager(self)
+
+
+class _SandboxFilesystem:
+    """Grouped filesystem operations over one sandbox (open/ls/mkdir/rm/
+    exists/watch/read/write convenience)."""
+
+    def __init__(self, sandbox: "_Sandbox"):
+        self._sb = sandbox
+
+    async def open(self, path: str, mode: str = "r") -> _FileIO:
+        return await self._sb.open(path, mode)
+
+    async def list_files(self, path: str = ".") -> list[str]:
+        return await self._sb.ls(path)
+
+    async def ls(self, path: str = ".") -> list[str]:
+        return await self._sb.ls(path)
+
+    async def mkdir(self, path: str, parents: bool = False) -> None:
+        return await self._sb.mkdir(path, parents=parents)
+
+    async def rm(self, path: str, recursive: bool = False) -> None:
+        return await self._sb.rm(path, recursive=recursive)
+
+    async def exists(self, path: str) -> bool:
+        return await self._sb.exists(path)
+
+    async def read_file(self, path: str) -> bytes:
+        f = await self._sb.open(path, "rb")
+        try:
+            return await f.read()
+        finally:
+            await f.close()
+
+    async def write_file(self, path: str, data: bytes) -> int:
+        f = await self._sb.open(path, "wb")
+        try:
+            return await f.write(data)
+        finally:
+            await f.close()
+
+    def watch(self, path: str, poll_interval: float = 0.5) -> Any:
+        return self._sb.watch(path, poll_interval=poll_interval)
+
+
+class _SandboxLogsManager:
+    """fetch()/tail() over the sandbox entrypoint's captured stdio."""
+
+    def __init__(self, sandbox: "_Sandbox"):
+        self._sb = sandbox
+
+    async def fetch(self, *, stderr: bool = False) -> str:
+        """Everything captured so far (does not wait for termination)."""
+        reader = self._sb.stderr if stderr else self._sb.stdout
+        chunks = []
+        while True:
+            data, eof = await reader.read_chunk(timeout=0.05)
+            if data:
+                chunks.append(data)
+            if eof or not data:
+                break
+        blob = b"".join(chunks)
+        return blob.decode("utf-8", errors="replace")
+
+    def tail(self, *, stderr: bool = False, poll_interval: float = 0.2) -> Any:
+        """Async-iterate new output until the entrypoint exits."""
+        reader = self._sb.stderr if stderr else self._sb.stdout
+
+        async def gen() -> Any:
+            while True:
+                data, eof = await reader.read_chunk(timeout=poll_interval)
+                if data:
+                    yield data.decode("utf-8", errors="replace")
+                if eof:
+                    return
+
+        return gen()
+
 
 Sandbox = synchronize_api(_Sandbox, "Sandbox")
+SandboxFilesystem = synchronize_api(_SandboxFilesystem, "SandboxFilesystem")
+SandboxLogsManager = synchronize_api(_SandboxLogsManager, "SandboxLogsManager")

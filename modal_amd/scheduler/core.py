@@ -114,6 +114,8 @@ class RPCAdapter:
         "sandbox_create", "sandbox_wait", "sandbox_terminate", "sandbox_poll", "sandbox_stdio_read",
         "sandbox_stdin_write", "sandbox_exec", "sandbox_list", "sandbox_set_tags",
         "sandbox_from_name", "sandbox_snapshot_fs", "sandbox_fs_op", "sandbox_resize",
+        "sandbox_snapshot_dir", "sandbox_mount_image", "sandbox_unmount_image",
+        "sandbox_connect_token", "named_objects_list", "named_object_delete", "image_publish", "image_from_name",
         "image_get_or_create", "image_info", "mount_get_or_create",
         "device_transfer", "tensor_pull_relay",
         "worker_snapshot", "worker_restore", "start_grpc_bridge", "app_get_logs",
@@ -1434,6 +1436,45 @@ class Scheduler:
             await asyncio.sleep(0.1)
         raise InvalidError("restored worker never connected")
 
+    async def named_objects_list(self, kind: str, environment: str = "") -> list:
+        """List deployed named objects of one kind (parity: reference
+        <Type>.objects.list() managers, e.g. queue.py _QueueManager.list)."""
+        env = environment or self.default_environment
+        if kind == "volume":
+            store = self.volume_service.by_name
+        elif kind == "queue":
+            store = self.services.queue_names.by_name
+        elif kind == "dict":
+            store = self.services.dict_names.by_name
+        elif kind == "secret":
+            store = self.services.secret_names.by_name
+        else:
+            raise InvalidError(f"Unknown object kind {kind!r}")
+        return [
+            {"name": name, "object_id": oid}
+            for (e, name), oid in sorted(store.items())
+            if e == env
+        ]
+
+    async def named_object_delete(self, kind: str, name: str, environment: str = "") -> None:
+        """Delete a deployed named object (parity: <Type>.objects.delete())."""
+        env = environment or self.default_environment
+        rows = await self.named_objects_list(kind, env)
+        oid = next((r["object_id"] for r in rows if r["name"] == name), None)
+        if oid is None:
+            raise NotFoundError(f"{kind} '{name}' not found in environment '{env}'")
+        if kind == "volume":
+            await self.volume_service.delete(oid)
+        elif kind == "queue":
+            self.services.queue_delete(oid)
+        elif kind == "dict":
+            self.services.dict_delete(oid)
+        elif kind == "secret":
+            self.services.secrets.pop(oid, None)
+            self.services.secret_names.by_name = {
+                k: v for k, v in self.services.secret_names.by_name.items() if v != oid
+            }
+
     async def object_info(self, object_id: str) -> dict:
         return self.services.object_info(object_id)
 
@@ -1520,9 +1561,14 @@ class Scheduler:
     # -- sandboxes ---------------------------------------------------------
     async def sandbox_create(self, **kwargs: Any) -> dict:
         volume_mounts = kwargs.pop("volume_mounts", None) or {}
-        volume_paths = {
-            path: self.volume_service.volume_dir(vid) for path, vid in volume_mounts.items()
-        }
+        volume_paths = {}
+        for path, spec in volume_mounts.items():
+            vid = spec["volume_id"] if isinstance(spec, dict) else spec
+            vol_dir = self.volume_service.volume_dir(vid)
+            if isinstance(spec, dict) and spec.get("sub_path"):
+                vol_dir = os.path.join(vol_dir, spec["sub_path"].strip("/"))
+                os.makedirs(vol_dir, exist_ok=True)
+            volume_paths[path] = vol_dir
         restore_image = kwargs.pop("restore_image_id", None)
         restore_blob = None
         image_fsroot = None
@@ -1583,6 +1629,74 @@ class Scheduler:
         self._extra.setdefault("snapshot_blobs", {})[resp["image_id"]] = resp["blob_id"]
         return resp
 
+    async def sandbox_snapshot_dir(self, sandbox_id: str, path: str) -> dict:
+        """Snapshot one sandbox directory into a new Image whose fsroot holds
+        the tree (parity: reference Sandbox.snapshot_directory,
+        sandbox.py:1643)."""
+        import shutil
+
+        sb = self.sandbox_service._get(sandbox_id)
+        src = os.path.normpath(os.path.join(sb.workdir, path.lstrip("/")))
+        if not os.path.isdir(src):
+            raise InvalidError(f"{path!r} is not a directory in sandbox {sandbox_id}")
+        resp = await self.image_service.get_or_create(
+            [{"kind": "base", "name": "snapshot", "sandbox": sandbox_id, "path": path,
+              "nonce": os.urandom(8).hex()}],
+            build=False,
+        )
+        state = self.image_service.by_id[resp["image_id"]]
+        dst = os.path.join(state.fsroot, path.lstrip("/"))
+        await asyncio.get_running_loop().run_in_executor(
+            None, lambda: shutil.copytree(src, dst, symlinks=True)
+        )
+        state.built = True
+        state.build_log = f"snapshot of {sandbox_id}:{path}"
+        return {"image_id": state.image_id}
+
+    async def sandbox_mount_image(self, sandbox_id: str, path: str, image_id: str) -> None:
+        """Expose an image's filesystem layer at `path` inside the sandbox
+        workdir (parity: Sandbox.mount_image, reference sandbox.py:1548)."""
+        sb = self.sandbox_service._get(sandbox_id)
+        state = self.image_service.by_id.get(image_id)
+        if state is None:
+            raise NotFoundError(f"Image {image_id} not found")
+        link = os.path.normpath(os.path.join(sb.workdir, path.lstrip("/")))
+        src = os.path.join(state.fsroot, path.lstrip("/"))
+        if not os.path.isdir(src):
+            src = state.fsroot
+        os.makedirs(os.path.dirname(link), exist_ok=True)
+        if os.path.islink(link):
+            os.unlink(link)
+        elif os.path.exists(link):
+            raise InvalidError(f"{path!r} already exists in sandbox {sandbox_id}")
+        os.symlink(src, link)
+
+    async def sandbox_unmount_image(self, sandbox_id: str, path: str) -> None:
+        sb = self.sandbox_service._get(sandbox_id)
+        link = os.path.normpath(os.path.join(sb.workdir, path.lstrip("/")))
+        if os.path.islink(link):
+            os.unlink(link)
+
+    async def sandbox_connect_token(
+        self, sandbox_id: str, port: int = 8080, user_metadata: Any = None
+    ) -> dict:
+        """Mint a bearer token for HTTP access to a sandbox-hosted server
+        (parity: SandboxCreateConnectToken, reference sandbox.py:1799).
+        Locally the "proxy" is direct: the URL targets 127.0.0.1:<port>."""
+        if not isinstance(port, int) or not (1 <= port <= 65535):
+            raise InvalidError("port must be between 1 and 65535")
+        self.sandbox_service._get(sandbox_id)  # raises if unknown
+        import secrets as _secrets
+
+        token = _secrets.token_urlsafe(24)
+        self._extra.setdefault("sandbox_connect_tokens", {})[token] = {
+            "sandbox_id": sandbox_id,
+            "port": port,
+            "user_metadata": user_metadata,
+            "created_at": time.time(),
+        }
+        return {"url": f"http://127.0.0.1:{port}", "token": token}
+
     async def sandbox_fs_op(self, sandbox_id: str, op: str, path: str = "", **kwargs: Any) -> Any:
         """Typed remote-FS operations inside a sandbox workdir (parity:
         reference sandbox filesystem API, sandbox_fs.py:68, file_io.py:135)."""
@@ -1637,6 +1751,26 @@ class Scheduler:
 
     async def image_info(self, image_id: str) -> dict:
         return await self.image_service.info(image_id)
+
+    async def image_publish(self, image_id: str, name: str, environment: str = "") -> dict:
+        """Publish a built image under a workspace name; ':latest' implied
+        (parity: reference Image.publish, _image.py:3010)."""
+        if self.image_service.by_id.get(image_id) is None:
+            raise NotFoundError(f"Image {image_id} not found")
+        env = environment or self.default_environment
+        if ":" not in name:
+            name = f"{name}:latest"
+        self._extra.setdefault("image_names", {})[(env, name)] = image_id
+        return {"name": name}
+
+    async def image_from_name(self, name: str, environment: str = "") -> dict:
+        env = environment or self.default_environment
+        if ":" not in name:
+            name = f"{name}:latest"
+        image_id = self._extra.get("image_names", {}).get((env, name))
+        if image_id is None:
+            raise NotFoundError(f"Image '{name}' not found in environment '{env}'")
+        return {"image_id": image_id}
 
     # -- mounts ------------------------------------------------------------
     async def mount_get_or_create(self, manifest: list) -> dict:

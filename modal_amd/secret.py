@@ -121,3 +121,7 @@ class _Secret(_Object, type_kind="secret"):
 
 
 Secret = synchronize_api(_Secret, "Secret")
+
+from .object_manager import install as _install_manager  # noqa: E402
+
+_install_manager(_Secret, Secret, "secret")

@@ -71,6 +71,45 @@ class _Volume(_Object, type_kind="volume"):
     def name(self) -> Any:
         return (getattr(self, "_metadata", None) or {}).get("name")
 
+    # -- mount options (parity: reference volume.py:419 read_only,
+    #    :450 with_mount_options) -----------------------------------------
+    def with_mount_options(
+        self, *, read_only: Any = None, sub_path: Any = None
+    ) -> "_Volume":
+        """A derived handle with per-mount options. `read_only` blocks
+        client-side writes AND makes function/sandbox mounts skip the
+        exit-time commit; `sub_path` scopes the mount to a subdirectory."""
+        base = self
+        opts = dict(getattr(self, "_mount_options", None) or {})
+        if read_only is not None:
+            opts["read_only"] = bool(read_only)
+        if sub_path is not None:
+            opts["sub_path"] = str(sub_path).strip("/")
+
+        async def _load(obj: "_Volume", resolver: Any, existing: Any) -> None:
+            await resolver.load(base)
+            obj._hydrate(base._object_id, base._client, getattr(base, "_metadata", None))
+            obj._mount_options = opts
+
+        derived = _Volume._from_loader(
+            _load, rep=f"{self._rep}.with_mount_options({opts})", deps=lambda: [base]
+        )
+        if self._is_hydrated:
+            derived._hydrate(self._object_id, self._client, getattr(self, "_metadata", None))
+        derived._mount_options = opts
+        return derived
+
+    def read_only(self) -> "_Volume":
+        return self.with_mount_options(read_only=True)
+
+    @property
+    def is_read_only(self) -> bool:
+        return bool((getattr(self, "_mount_options", None) or {}).get("read_only"))
+
+    def _check_writable(self) -> None:
+        if self.is_read_only:
+            raise InvalidError("Volume handle is read-only (with_mount_options)")
+
     @live_method
     async def info(self) -> dict:
         """Name + file count (parity: reference info())."""
@@ -186,6 +225,7 @@ class _Volume(_Object, type_kind="volume"):
     # -- writes ----------------------------------------------------------
     @live_method
     async def remove_file(self, path: str, recursive: bool = False) -> None:
+        self._check_writable()
         await self._client.svc.volume_remove_file(
             volume_id=self.object_id, rel_path=path, recursive=recursive
         )
@@ -205,6 +245,7 @@ class _Volume(_Object, type_kind="volume"):
         await self._client.svc.volume_reload(volume_id=self.object_id)
 
     def batch_upload(self, force: bool = False) -> "_VolumeUploadContextManager":
+        self._check_writable()
         return _VolumeUploadContextManager(self, force=force)
 
     async def _put_data(self, data: bytes, remote_path: str) -> None:
@@ -339,3 +380,7 @@ class _EphemeralVolume:
 
 
 Volume = synchronize_api(_Volume, "Volume")
+
+from .object_manager import install as _install_manager  # noqa: E402
+
+_install_manager(_Volume, Volume, "volume")

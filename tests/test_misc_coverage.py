@@ -451,22 +451,163 @@ def test_sandbox_wait_until_ready_and_tags(client):
     assert sb.returncode == 0
 
 
-def test_image_from_name_deployed(client):
+def test_image_publish_and_from_name(client):
     img = modal.Image.from_scratch()
-    app = modal.App("img-holder")
-
-    @app.function(image=img)
-    def f():
-        return 1
-
-    app.deploy(name="img-holder", client=client)
-    from modal_amd._sync import synchronizer
-
-    synchronizer.run(
-        client.svc.app_set_objects(
-            app_id=app.app_id, objects={"base": (img.object_id, {})}
-        )
-    )
-    got = modal.Image.from_name("img-holder", "base")
+    img.build(client=client)
+    img.publish("base-img")
+    got = modal.Image.from_name("base-img")  # :latest implied
     got.build(client=client)
     assert got.object_id == img.object_id
+    with pytest.raises(Exception):
+        modal.Image.from_name("no-such-img").build(client=client)
+
+
+def test_api_audit_clean():
+    """The AST parity audit (scripts/api_audit.py) must report no missing
+    public members against the reference classes."""
+    import subprocess
+    import sys
+
+    proc = subprocess.run(
+        [sys.executable, "scripts/api_audit.py"],
+        capture_output=True, text=True,
+        cwd=__import__("os").path.dirname(__import__("os").path.dirname(__file__)),
+    )
+    assert proc.returncode == 0, proc.stdout + proc.stderr
+
+
+def test_object_managers(client):
+    modal.Queue.objects.create("mgr-q")
+    with pytest.raises(Exception):
+        modal.Queue.objects.create("mgr-q")
+    modal.Queue.objects.create("mgr-q", allow_existing=True)
+    names = [r["name"] for r in modal.Queue.objects.list()]
+    assert "mgr-q" in names
+    modal.Queue.objects.delete("mgr-q")
+    assert "mgr-q" not in [r["name"] for r in modal.Queue.objects.list()]
+
+    modal.Dict.objects.create("mgr-d")
+    modal.Secret.objects.create("mgr-s")
+    modal.Volume.objects.create("mgr-v")
+    assert [r["name"] for r in modal.Dict.objects.list()] == ["mgr-d"]
+    assert [r["name"] for r in modal.Secret.objects.list()] == ["mgr-s"]
+    assert [r["name"] for r in modal.Volume.objects.list()] == ["mgr-v"]
+    modal.Volume.objects.delete("mgr-v")
+    assert modal.Volume.objects.list() == []
+
+
+def test_volume_mount_options(client, tmp_path):
+    import io
+
+    v = modal.Volume.from_name("opts-v", create_if_missing=True)
+    with v.batch_upload() as b:
+        b.put_file(io.BytesIO(b"deep"), "/sub/inner.txt")
+
+    ro = v.read_only()
+    with pytest.raises(Exception):
+        ro.batch_upload()
+    with pytest.raises(Exception):
+        ro.remove_file("/sub/inner.txt")
+    # reads still work on a read-only handle
+    assert b"".join(ro.read_file("/sub/inner.txt")) == b"deep"
+
+    scoped = v.with_mount_options(sub_path="sub")
+    app = modal.App("vol-opts")
+
+    @app.function(volumes={"/data": scoped})
+    def peek():
+        with open("/data/inner.txt") as f:
+            return f.read()
+
+    with app.run(client=client):
+        assert peek.remote() == "deep"
+
+
+def test_sandbox_filesystem_and_logs(client):
+    sb = modal.Sandbox.create("bash", "-c", "echo marker-out; sleep 0.2", client=client)
+    fs = sb.filesystem
+    fs.write_file("note.txt", b"fs-data")
+    assert fs.exists("note.txt")
+    assert fs.read_file("note.txt") == b"fs-data"
+    assert "note.txt" in fs.list_files(".")
+    sb.wait()
+    logs = sb.logs()
+    out = logs.fetch()
+    assert "marker-out" in out
+
+
+def test_sandbox_snapshot_directory_and_mount_image(client):
+    sb = modal.Sandbox.create("bash", "-c", "mkdir -p proj && echo v1 > proj/file && sleep 600", client=client)
+    sb.wait_until_ready(timeout=10)
+    import time
+
+    deadline = time.time() + 5
+    while not sb.filesystem.exists("proj/file") and time.time() < deadline:
+        time.sleep(0.05)
+    img = sb.snapshot_directory("proj")
+    sb2 = modal.Sandbox.create("sleep", "600", client=client)
+    sb2.mount_image("proj", img)
+    assert sb2.filesystem.read_file("proj/file") == b"v1\n"
+    sb2.unmount_image("proj")
+    assert not sb2.filesystem.exists("proj/file")
+    tok = sb2.create_connect_token(user_metadata={"u": 1}, port=9999)
+    assert tok["token"] and tok["url"].endswith(":9999")
+    sb.detach()
+    sb2.terminate()
+    sb.terminate()
+
+
+def test_function_spec_stub_logs(client):
+    app = modal.App("spec-app")
+
+    @app.function(timeout=30, memory=512)
+    def noisy(x):
+        print(f"noisy says {x}")
+        return x
+
+    impl = getattr(noisy, "_impl", noisy)
+    assert impl.stub is impl.app
+    spec = impl.spec
+    assert spec["timeout"] == 30 and spec["memory"] == 512
+    assert "def noisy" in impl.get_build_def()
+    with app.run(client=client):
+        assert noisy.remote(7) == 7
+        import time
+
+        time.sleep(0.3)
+        from modal_amd._sync import synchronizer
+
+        lines = synchronizer.run(impl.logs().fetch())
+        assert any("noisy says 7" in ln for ln in lines)
+
+
+def test_image_pipe_and_cls_validation():
+    def setup(image, pkg):
+        return image.pip_install(pkg)
+
+    img = modal.Image.debian_slim().pipe(setup, "numpy")
+    impl = getattr(img, "_impl", img)
+    assert any(l.get("kind") == "pip_install" for l in impl._recipe)
+
+    from modal_amd.cls import Cls, parameter
+
+    class Good:
+        x: int = parameter(default=1)
+
+    Cls.validate_construction_mechanism(Good)
+
+    class Bad:
+        x: int = parameter(default=1)
+
+        def __init__(self):
+            pass
+
+    with pytest.raises(Exception):
+        Cls.validate_construction_mechanism(Bad)
+
+    class Unannotated:
+        pass
+
+    Unannotated.y = parameter(default=2)
+    with pytest.raises(Exception):
+        Cls.validate_construction_mechanism(Unannotated)
