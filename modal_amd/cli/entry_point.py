@@ -832,6 +832,108 @@ def cluster_info() -> None:
             click.echo("device mesh: not formed")
 
 
+@entrypoint_cli.group(name="nfs")
+def nfs_cli() -> None:
+    """Manage NetworkFileSystems (parity: reference cli/network_file_system.py)."""
+
+
+@nfs_cli.command(name="list")
+def nfs_list() -> None:
+    import modal_amd as modal
+
+    client = _get_client()
+    rows = synchronizer.run(client.svc.named_objects_list(kind="volume", environment="main"))
+    for r in rows:
+        if r["name"].startswith("nfs/"):
+            click.echo(f'{r["name"][4:]}\t{r["object_id"]}')
+
+
+@nfs_cli.command(name="create")
+@click.argument("name")
+def nfs_create(name: str) -> None:
+    import modal_amd as modal
+
+    modal.NetworkFileSystem.from_name(name, create_if_missing=True).hydrate()
+    click.echo(f"Created NFS '{name}'")
+
+
+@nfs_cli.command(name="put")
+@click.argument("name")
+@click.argument("local_path")
+@click.argument("remote_path", required=False)
+def nfs_put(name: str, local_path: str, remote_path: Optional[str]) -> None:
+    import modal_amd as modal
+
+    nfs = modal.NetworkFileSystem.from_name(name, create_if_missing=True)
+    if os.path.isdir(local_path):
+        nfs.add_local_dir(local_path, remote_path)
+    else:
+        nfs.add_local_file(local_path, remote_path)
+    click.echo("done")
+
+
+@nfs_cli.command(name="get")
+@click.argument("name")
+@click.argument("remote_path")
+@click.argument("local_path", default=".")
+def nfs_get(name: str, remote_path: str, local_path: str) -> None:
+    import modal_amd as modal
+
+    nfs = modal.NetworkFileSystem.from_name(name)
+    dest = local_path
+    if os.path.isdir(dest):
+        dest = os.path.join(dest, os.path.basename(remote_path))
+    with open(dest, "wb") as f:
+        for chunk in nfs.read_file(remote_path):
+            f.write(chunk)
+    click.echo(dest)
+
+
+@nfs_cli.command(name="rm")
+@click.argument("name")
+@click.argument("remote_path")
+def nfs_rm(name: str, remote_path: str) -> None:
+    import modal_amd as modal
+
+    modal.NetworkFileSystem.from_name(name).remove_file(remote_path)
+
+
+@entrypoint_cli.group(name="workspace")
+def workspace_cli() -> None:
+    """Workspace info (parity: reference cli/workspace.py — local: one
+    workspace named after the run dir)."""
+
+
+@workspace_cli.command(name="current")
+def workspace_current() -> None:
+    from ..config import config
+
+    click.echo(config.get("workspace") or "local")
+
+
+@entrypoint_cli.command(name="dashboard")
+def dashboard() -> None:
+    """Print where to look instead of a hosted dashboard: the active run
+    dir (logs, volumes, images) and any live web endpoints."""
+    client = _get_client()
+    click.echo(f"run dir: {client.svc.run_dir}")
+    rows = synchronizer.run(client.svc.app_list())
+    for row in rows:
+        click.echo(f'{row.get("app_id")}\t{row.get("state")}\t{row.get("description") or ""}')
+
+
+@entrypoint_cli.command(name="changelog")
+def changelog() -> None:
+    import modal_amd
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(modal_amd.__file__)))
+    path = os.path.join(root, "CHANGELOG.md")
+    if os.path.exists(path):
+        click.echo(open(path).read())
+    else:
+        click.echo(f"modal-amd {getattr(modal_amd, '__version__', 'dev')} — see git log for changes")
+
+
 def main() -> None:
     try:
         entrypoint_cli(standalone_mode=True)

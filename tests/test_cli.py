@@ -227,3 +227,39 @@ def test_cli_volume_cp_and_rename(client, tmp_path):
     assert result.exit_code == 0, result.output
     vol2 = modal.Volume.from_name("cpvol2")
     assert b"copy-me" == b"".join(vol2.read_file("a.txt"))
+
+
+def test_cli_nfs_roundtrip(client, tmp_path, monkeypatch):
+    from click.testing import CliRunner
+
+    from modal_amd.cli.entry_point import entrypoint_cli
+
+    runner = CliRunner()
+    local = tmp_path / "payload.txt"
+    local.write_text("nfs-cli-data")
+    r = runner.invoke(entrypoint_cli, ["nfs", "create", "cli-nfs"])
+    assert r.exit_code == 0, r.output
+    r = runner.invoke(entrypoint_cli, ["nfs", "put", "cli-nfs", str(local), "p.txt"])
+    assert r.exit_code == 0, r.output
+    r = runner.invoke(entrypoint_cli, ["nfs", "list"])
+    assert "cli-nfs" in r.output
+    out = tmp_path / "back.txt"
+    r = runner.invoke(entrypoint_cli, ["nfs", "get", "cli-nfs", "p.txt", str(out)])
+    assert r.exit_code == 0, r.output
+    assert out.read_text() == "nfs-cli-data"
+    r = runner.invoke(entrypoint_cli, ["nfs", "rm", "cli-nfs", "p.txt"])
+    assert r.exit_code == 0, r.output
+
+
+def test_cli_dashboard_and_workspace(client):
+    from click.testing import CliRunner
+
+    from modal_amd.cli.entry_point import entrypoint_cli
+
+    runner = CliRunner()
+    r = runner.invoke(entrypoint_cli, ["dashboard"])
+    assert r.exit_code == 0 and "run dir:" in r.output
+    r = runner.invoke(entrypoint_cli, ["workspace", "current"])
+    assert r.exit_code == 0 and r.output.strip()
+    r = runner.invoke(entrypoint_cli, ["changelog"])
+    assert r.exit_code == 0
