@@ -55,6 +55,11 @@ async def _prepare_function_options(fn_impl: Any, resolver: Resolver) -> dict:
             options["image_id"] = impl.object_id
         except Exception:
             pass  # image building is best-effort locally
+    proxy = options.pop("proxy", None)
+    if proxy is not None:
+        impl = unwrap(proxy)
+        await resolver.load(impl)
+        options["proxy_url"] = (getattr(impl, "_metadata", None) or {}).get("url")
     schedule = options.get("_schedule")
     if schedule is not None:
         from .schedule import Cron, Period

@@ -101,6 +101,7 @@ class FunctionDef:
     schedule: Optional[dict] = None  # {"cron": "..."} | {"period": seconds}
     image_id: Optional[str] = None
     placement: Optional[dict] = None  # {"gpu_index": i, "gpu_set": [..]} hints
+    proxy_url: Optional[str] = None  # HTTP_PROXY/HTTPS_PROXY for user code
 
     def placement_tag(self) -> str:
         return "gpu" if self.needs_gpu else "any"

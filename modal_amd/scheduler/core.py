@@ -116,6 +116,7 @@ class RPCAdapter:
         "sandbox_from_name", "sandbox_snapshot_fs", "sandbox_fs_op", "sandbox_resize",
         "sandbox_snapshot_dir", "sandbox_mount_image", "sandbox_unmount_image",
         "sandbox_connect_token", "named_objects_list", "named_object_delete", "image_publish", "image_from_name",
+        "proxy_get_or_create", "proxy_stats",
         "image_get_or_create", "image_info", "mount_get_or_create",
         "device_transfer", "tensor_pull_relay",
         "worker_snapshot", "worker_restore", "start_grpc_bridge", "app_get_logs",
@@ -530,6 +531,7 @@ class Scheduler:
             schedule=options.get("schedule"),
             image_id=options.get("image_id"),
             placement=options.get("placement"),
+            proxy_url=options.get("proxy_url"),
         )
         self.functions[fid] = fdef
         if fdef.min_containers or fdef.buffer_containers:
@@ -1222,6 +1224,9 @@ class Scheduler:
                 env.update(self.services.secret_env(secret_id))
             except Exception:
                 pass
+        if fdef.proxy_url:
+            env["HTTP_PROXY"] = env["http_proxy"] = fdef.proxy_url
+            env["HTTPS_PROXY"] = env["https_proxy"] = fdef.proxy_url
         return env
 
     def resolve_function_pythonpaths(self, fdef: FunctionDef) -> list[str]:
@@ -1744,6 +1749,37 @@ class Scheduler:
         if op == "exists":
             return os.path.exists(full)
         raise InvalidError(f"Unknown fs op {op!r}")
+
+    # -- proxies -----------------------------------------------------------
+    async def proxy_get_or_create(self, name: str, environment: str = "") -> dict:
+        """Start (once) the shared local forward proxy and register `name`
+        on it (parity: Proxy.from_name, reference proxy.py:57 — locally all
+        named proxies share one 127.0.0.1 egress point)."""
+        from .localproxy import LocalForwardProxy
+
+        proxy = self._extra.get("local_proxy")
+        if proxy is None:
+            proxy = LocalForwardProxy()
+            await proxy.start()
+            self._extra["local_proxy"] = proxy
+        env = environment or self.default_environment
+        ids = self._extra.setdefault("proxy_ids", {})
+        pid = ids.get((env, name))
+        if pid is None:
+            pid = new_id("tunnel")
+            ids[(env, name)] = pid
+        return {"proxy_id": pid, "url": proxy.url, "port": proxy.port}
+
+    async def proxy_stats(self) -> dict:
+        proxy = self._extra.get("local_proxy")
+        if proxy is None:
+            return {"running": False}
+        return {
+            "running": True,
+            "port": proxy.port,
+            "connections": proxy.connections,
+            "bytes_relayed": proxy.bytes_relayed,
+        }
 
     # -- images ------------------------------------------------------------
     async def image_get_or_create(self, recipe: list) -> dict:

@@ -611,3 +611,85 @@ def test_image_pipe_and_cls_validation():
     Unannotated.y = parameter(default=2)
     with pytest.raises(Exception):
         Cls.validate_construction_mechanism(Unannotated)
+
+
+def test_proxy_routes_function_http_traffic(client):
+    """Proxy.from_name starts a real local forward proxy; a function declared
+    with proxy= sees HTTP(S)_PROXY and its HTTP traffic relays through it."""
+    import http.server
+    import threading
+
+    class H(http.server.BaseHTTPRequestHandler):
+        def do_GET(self):
+            body = b"origin-says-hi"
+            self.send_response(200)
+            self.send_header("Content-Length", str(len(body)))
+            self.end_headers()
+            self.wfile.write(body)
+
+        def log_message(self, *a):
+            pass
+
+    httpd = http.server.HTTPServer(("127.0.0.1", 0), H)
+    port = httpd.server_address[1]
+    t = threading.Thread(target=httpd.serve_forever, daemon=True)
+    t.start()
+    try:
+        p = modal.Proxy.from_name("egress")
+        app = modal.App("proxy-app")
+
+        @app.function(proxy=p)
+        def fetch(url):
+            import os
+            import urllib.request
+
+            proxy_url = os.environ["HTTP_PROXY"]
+            # urllib honors env proxies for absolute http URLs
+            opener = urllib.request.build_opener(
+                urllib.request.ProxyHandler({"http": proxy_url})
+            )
+            with opener.open(url, timeout=10) as resp:
+                return resp.read().decode()
+
+        with app.run(client=client):
+            assert fetch.remote(f"http://127.0.0.1:{port}/x") == "origin-says-hi"
+        from modal_amd._sync import synchronizer
+
+        stats = synchronizer.run(client.svc.proxy_stats())
+        assert stats["running"] and stats["connections"] >= 1
+        assert stats["bytes_relayed"] > 0
+    finally:
+        httpd.shutdown()
+
+
+def test_proxy_connect_tunnel(client):
+    """CONNECT tunneling through the local forward proxy (the https path)."""
+    import socket
+    import threading
+
+    # echo server as the "origin"
+    srv = socket.socket()
+    srv.bind(("127.0.0.1", 0))
+    srv.listen(1)
+    origin_port = srv.getsockname()[1]
+
+    def echo_once():
+        conn, _ = srv.accept()
+        data = conn.recv(1024)
+        conn.sendall(b"echo:" + data)
+        conn.close()
+
+    t = threading.Thread(target=echo_once, daemon=True)
+    t.start()
+
+    from modal_amd._sync import synchronizer
+
+    resp = synchronizer.run(client.svc.proxy_get_or_create(name="tun"))
+    s = socket.create_connection(("127.0.0.1", resp["port"]), timeout=5)
+    s.sendall(f"CONNECT 127.0.0.1:{origin_port} HTTP/1.1\r\n\r\n".encode())
+    reply = s.recv(1024)
+    assert b"200" in reply
+    s.sendall(b"ping")
+    assert s.recv(1024) == b"echo:ping"
+    s.close()
+    srv.close()
