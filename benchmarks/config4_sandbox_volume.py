@@ -64,6 +64,21 @@ def main() -> None:
     rd_s = time.perf_counter() - t0
     assert data == blob, "volume round-trip corrupted payload"
 
+    # whole-file read through read_file_into_fileobj (direct-buffer path)
+    buf = io.BytesIO()
+    t0 = time.perf_counter()
+    n = vol.read_file_into_fileobj("payload.bin", buf)
+    rdinto_s = time.perf_counter() - t0
+    assert n == len(blob) and buf.getvalue() == blob
+
+    # zero-copy into a file (kernel sendfile)
+    import tempfile
+
+    with tempfile.NamedTemporaryFile(dir=".") as tf:
+        t0 = time.perf_counter()
+        vol.read_file_into_fileobj("payload.bin", tf.file)
+        rdfile_s = time.perf_counter() - t0
+
     # -- sandbox reads the mounted volume -------------------------------
     sb2 = modal.Sandbox.create(
         "bash", "-c", "wc -c < data/payload.bin", volumes={"data": vol}
@@ -78,6 +93,8 @@ def main() -> None:
         "sandbox_exec_p50_ms": round(exec_p50_ms, 3),
         "volume_upload_gibps": round(len(blob) / up_s / 2**30, 3),
         "volume_read_gibps": round(len(blob) / rd_s / 2**30, 3),
+        "volume_read_into_gibps": round(len(blob) / rdinto_s / 2**30, 3),
+        "volume_read_sendfile_gibps": round(len(blob) / rdfile_s / 2**30, 3),
         "payload_mb": args.mb,
     }))
 
