@@ -909,7 +909,86 @@ class WorkerRuntime:
 
         app_tok = _app_id_var.set(frt.app_id)
         try:
-            if fn0 is not None and (
+            if fn0 is not None and frt.batch_max_size > 1:
+                # @modal.batched riding the chunk path: the chunk IS the
+                # accumulated batch (no linger needed) — transpose args per
+                # batch_max_size slice, one user call per slice, per-item
+                # outputs flow through the chunk_done aggregation
+                is_async = inspect.iscoroutinefunction(fn0)
+
+                def _transpose(pairs: Any, kw_common: Any, idxs: range) -> tuple:
+                    rows = [
+                        (pairs[ci], kw_common) if kw_common is not None else pairs[ci]
+                        for ci in idxs
+                    ]
+                    if kw_common is not None:
+                        # C2 form: pairs[ci] is the positional tuple
+                        n_args = len(rows[0][0]) if rows else 0
+                        arglists = [[row[0][i] for row in rows] for i in range(n_args)]
+                        kwlists = {k: [v] * len(rows) for k, v in kw_common.items()}
+                    else:
+                        n_args = max((len(a) for a, _ in rows), default=0)
+                        arglists = [
+                            [a[i] if i < len(a) else None for a, _ in rows]
+                            for i in range(n_args)
+                        ]
+                        keys: set = set()
+                        for _, kw in rows:
+                            keys.update(kw)
+                        kwlists = {k: [kw.get(k) for _, kw in rows] for k in keys}
+                    return tuple(arglists), kwlists
+
+                def _store(idxs: range, results: Any) -> None:
+                    if not isinstance(results, (list, tuple)) or len(results) != len(idxs):
+                        raise ValueError(
+                            f"Batched function {frt.name} must return a list of "
+                            f"{len(idxs)} results, got {type(results).__name__}"
+                        )
+                    for ci, r in zip(idxs, results):
+                        values[ci] = r
+
+                try:
+                    pairs, kw_common = await loop.run_in_executor(
+                        self.executor, self._chunk_pairs, msg["chunk_id"]
+                    )
+                    bsz = frt.batch_max_size
+                    for start in range(0, count, bsz):
+                        idxs = range(start, min(start + bsz, count))
+                        try:
+                            arglists, kwlists = _transpose(pairs, kw_common, idxs)
+                            if is_async:
+                                results = await self._execute(frt, fn0, arglists, kwlists)
+                            else:
+
+                                def _call_batch(
+                                    arglists: tuple = arglists, kwlists: dict = kwlists
+                                ) -> Any:
+                                    # executor threads carry their own context
+                                    tok_c = _current_function_call_id.set(call_id)
+                                    tok_a2 = _app_id_var.set(frt.app_id)
+                                    try:
+                                        return fn0(*arglists, **kwlists)
+                                    finally:
+                                        _current_function_call_id.reset(tok_c)
+                                        _app_id_var.reset(tok_a2)
+
+                                results = await loop.run_in_executor(
+                                    self.executor, _call_batch
+                                )
+                            _store(idxs, results)
+                        except BaseException as exc:
+                            data = self._serialize_exception(exc)
+                            err = "".join(
+                                traceback.format_exception_only(type(exc), exc)
+                            ).strip()
+                            for ci in idxs:
+                                errors[ci] = (data, err)
+                except BaseException as exc:
+                    data = self._serialize_exception(exc)
+                    err = "".join(traceback.format_exception_only(type(exc), exc)).strip()
+                    for ci in range(count):
+                        errors.setdefault(ci, (data, err))
+            elif fn0 is not None and (
                 inspect.iscoroutinefunction(fn0) or inspect.isasyncgenfunction(fn0)
             ):
                 # async user function: per-item awaits on this loop

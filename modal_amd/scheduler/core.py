@@ -820,7 +820,6 @@ class Scheduler:
         fast = (
             fdef is not None
             and fdef.cluster_size <= 1
-            and fdef.batch_max_size <= 1
             and not fdef.is_generator
             and not fdef.timeout
             and not fdef.web_config

@@ -161,4 +161,8 @@ def test_outstanding_cap_scales_with_pool(client):
             assert time.time() < deadline
             time.sleep(0.05)
         cap2 = synchronizer.run(probe())
-        assert cap2 >= cap + 256
+        alive2 = sum(1 for w in svc.pool.workers.values() if w.alive)
+        # the cap tracks the formula for the CURRENT pool (alive count may
+        # drift between probes as the autoscaler spawns/reaps workers)
+        assert alive2 > alive
+        assert cap2 >= max(1000, 256 * alive2)
