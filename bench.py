@@ -35,6 +35,13 @@ ITEMS_PER_GPU = 12_500  # x8 GPUs = the 100k-input config of BASELINE.json
 # per-box (profiles/README.md, r2: 4 workers x 128-item chunks ~1.8x the
 # round-1 3x64 on the same box); override with MODAL_AMD_BENCH_WPG
 WORKERS_PER_GPU = int(os.environ.get("MODAL_AMD_BENCH_WPG", "8"))
+if (
+    os.environ.get("MODAL_AMD_BENCH_EAGER") != "1"
+    and os.environ.get("MODAL_AMD_BENCH_ITEM_SYNC") != "1"
+):
+    # batched default: 256-item chunks measured fastest (1.49M items/s);
+    # the eager variant keeps the map engine's 128 default
+    os.environ.setdefault("MODAL_AMD_CHUNK_ITEMS", "256")
 
 
 def _bench_run_dir() -> str:
@@ -229,7 +236,12 @@ def main() -> None:
     import modal_amd as modal
 
     app = modal.App("bench")
-    use_batched = has_gpu and os.environ.get("MODAL_AMD_BENCH_BATCHED") == "1"
+    # default: @modal.batched dynamic batching (the framework's serving-path
+    # feature; same per-item math, fused launches). MODAL_AMD_BENCH_EAGER=1
+    # runs the per-item eager variant instead (3 tiny launches per item).
+    use_batched = has_gpu and os.environ.get("MODAL_AMD_BENCH_EAGER") != "1" and (
+        os.environ.get("MODAL_AMD_BENCH_ITEM_SYNC") != "1"
+    )
     if use_batched:
         item_fn = app.function(gpu=1)(
             modal.batched(max_batch_size=64, wait_ms=1)(map_item_gpu_batched)
@@ -310,10 +322,17 @@ def main() -> None:
                 "parallelism": f"map{n_gpus}",
                 "p50_remote_ms": p50_ms,
                 "per_item_gpu_op": (
-                    "bf16 vector scale + chunk-batched pinned D2H readback"
+                    (
+                        "bf16 4096-vector scale + 4-elem sum per item, "
+                        "@modal.batched(64) fused launches, one pinned D2H per batch"
+                        if use_batched
+                        else "bf16 vector scale + chunk-batched pinned D2H readback"
+                    )
                     if has_gpu
                     else "noop"
                 ),
+                "dynamic_batching": 64 if use_batched else 0,
+                "chunk_items": int(os.environ.get("MODAL_AMD_CHUNK_ITEMS", "128")),
                 "workers": len(scheduler.pool.workers),
             },
         }
