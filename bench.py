@@ -35,13 +35,6 @@ ITEMS_PER_GPU = 12_500  # x8 GPUs = the 100k-input config of BASELINE.json
 # per-box (profiles/README.md, r2: 4 workers x 128-item chunks ~1.8x the
 # round-1 3x64 on the same box); override with MODAL_AMD_BENCH_WPG
 WORKERS_PER_GPU = int(os.environ.get("MODAL_AMD_BENCH_WPG", "8"))
-if (
-    os.environ.get("MODAL_AMD_BENCH_EAGER") != "1"
-    and os.environ.get("MODAL_AMD_BENCH_ITEM_SYNC") != "1"
-):
-    # batched default: 256-item chunks measured fastest (1.49M items/s);
-    # the eager variant keeps the map engine's 128 default
-    os.environ.setdefault("MODAL_AMD_CHUNK_ITEMS", "256")
 
 
 def _bench_run_dir() -> str:
@@ -234,6 +227,16 @@ def main() -> None:
         time.sleep(0.05)
 
     import modal_amd as modal
+
+    if has_gpu and os.environ.get("MODAL_AMD_BENCH_EAGER") != "1" and (
+        os.environ.get("MODAL_AMD_BENCH_ITEM_SYNC") != "1"
+    ):
+        # batched default: 256-item chunks measured fastest at N=1
+        # (1.49M items/s); at N>=4 the scheduler loop carries 4-8x the
+        # chunk rate, so halve it with 512-item chunks (-10% at N=1)
+        os.environ.setdefault(
+            "MODAL_AMD_CHUNK_ITEMS", "256" if n_gpus < 4 else "512"
+        )
 
     app = modal.App("bench")
     # default: @modal.batched dynamic batching (the framework's serving-path

@@ -30,7 +30,12 @@ OUTPUT_FETCH_MAX = 1024
 # items per shared pickle chunk: one C-pickler pass serves ~64 inputs on the
 # client AND one unpickle serves them on the worker (SURVEY §2 row 6's
 # "tensor-aware fast path" generalized to all small map payloads)
-CHUNK_ITEMS = int(os.environ.get("MODAL_AMD_CHUNK_ITEMS", "128"))
+CHUNK_ITEMS = 128  # default; MODAL_AMD_CHUNK_ITEMS overrides per run
+
+
+def _chunk_items() -> int:
+    raw = os.environ.get("MODAL_AMD_CHUNK_ITEMS")
+    return int(raw) if raw else CHUNK_ITEMS
 
 
 class BulkSemaphore:
@@ -246,13 +251,14 @@ async def map_invocation_batches(
             )
 
         try:
+            chunk_items = _chunk_items()
             if fast_zip_iters is not None:
                 from itertools import islice
 
                 iterator = zip(*fast_zip_iters)
                 n_flushed = 0
                 while True:
-                    argsbatch = list(islice(iterator, CHUNK_ITEMS))
+                    argsbatch = list(islice(iterator, chunk_items))
                     if not argsbatch:
                         break
                     # byte-size guard sampled from the first item: split big-
@@ -274,7 +280,7 @@ async def map_invocation_batches(
             if hasattr(input_iter, "__aiter__"):
                 async for item in input_iter:
                     _add(item)
-                    if len(chunk_buf) >= CHUNK_ITEMS or approx_bytes > 4 * 1024 * 1024:
+                    if len(chunk_buf) >= chunk_items or approx_bytes > 4 * 1024 * 1024:
                         approx_bytes = 0
                         await sem.acquire(len(chunk_buf))
                         await flush_chunk()
@@ -285,12 +291,12 @@ async def map_invocation_batches(
 
                 iterator = iter(input_iter)
                 while True:
-                    batch = list(islice(iterator, CHUNK_ITEMS - len(chunk_buf) or CHUNK_ITEMS))
+                    batch = list(islice(iterator, chunk_items - len(chunk_buf) or chunk_items))
                     if not batch and not chunk_buf:
                         break
                     for item in batch:
                         _add(item)
-                    if len(chunk_buf) >= CHUNK_ITEMS or approx_bytes > 4 * 1024 * 1024:
+                    if len(chunk_buf) >= chunk_items or approx_bytes > 4 * 1024 * 1024:
                         approx_bytes = 0
                         await sem.acquire(len(chunk_buf))
                         await flush_chunk()
