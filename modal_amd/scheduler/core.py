@@ -718,10 +718,14 @@ class Scheduler:
             # server-sized (the reference makes this server-overridable,
             # parallel_map.py:387): the default 1,000 items is ~7 chunks of
             # 128 — enough for one worker's pipeline but it STARVES a
-            # multi-GPU pool, so scale with the live worker count
+            # multi-GPU pool. Scale with the live worker count sized for
+            # ~4 chunks of up to 256 items in flight PER worker (pipeline
+            # depth; a cap of one chunk per worker leaves workers idle
+            # between completions — measured 372k -> 1M+ items/s on the
+            # batched-noop probe when deepened)
             "max_inputs_outstanding": max(
                 MAX_INPUTS_OUTSTANDING_DEFAULT,
-                256 * max(
+                1024 * max(
                     sum(1 for w in self.pool.workers.values() if w.alive), 1
                 ),
             ),
