@@ -153,7 +153,7 @@ def test_outstanding_cap_scales_with_pool(client):
 
         cap = synchronizer.run(probe())
         alive = sum(1 for w in client.svc.pool.workers.values() if w.alive)
-        assert cap >= max(1000, 256 * alive)
+        assert cap >= max(1000, 1024 * alive)
         # grow the pool: the cap grows with it
         synchronizer.run(svc.pool.spawn_worker(gpu_index=None))
         deadline = time.time() + 30
@@ -165,4 +165,4 @@ def test_outstanding_cap_scales_with_pool(client):
         # the cap tracks the formula for the CURRENT pool (alive count may
         # drift between probes as the autoscaler spawns/reaps workers)
         assert alive2 > alive
-        assert cap2 >= max(1000, 256 * alive2)
+        assert cap2 >= max(1000, 1024 * alive2)
