@@ -75,7 +75,7 @@ def test_unary_via_protos(grpc_plane):
     item.input.final_input = True
     map_resp = invoke("FunctionMap", map_req, api.FunctionMapResponse)
     assert map_resp.function_call_id.startswith("fc-")
-    assert map_resp.max_inputs_outstanding == 1000
+    assert map_resp.max_inputs_outstanding >= 1000  # server-sized: 1024/worker
     assert len(map_resp.pipelined_inputs) == 1
 
     deadline = time.time() + 30
