@@ -259,12 +259,16 @@ def main() -> None:
     items_per_step = args.items_per_gpu * n_gpus
 
     def run_map_step() -> None:
-        """One map step driven on the framework loop (no per-item sync bridge)."""
+        """One map step driven on the framework loop (no per-item sync
+        bridge); batch-wise consumption (map_batches) skips the per-item
+        flatten — every item is still produced and counted."""
 
         async def _consume() -> None:
             n = 0
-            async for _ in item_fn.map.aio(range(items_per_step), order_outputs=False):
-                n += 1
+            async for batch in item_fn.map_batches.aio(
+                range(items_per_step), order_outputs=False
+            ):
+                n += len(batch)
             assert n == items_per_step
 
         synchronizer.run(_consume())

@@ -780,6 +780,40 @@ def _install_sync_map_methods() -> None:
     setattr(Function, "map", make("map", "map_async_batches", "map_async"))
     setattr(Function, "starmap", make("starmap", "starmap_async_batches", "starmap_async"))
 
+    def make_batches(batches_name: str) -> Any:
+        """fn.map_batches(...): LISTS of results (the engine's native
+        granularity — a bulk consumer skips ~0.5 us/item of per-item
+        async-generator flattening; extension beyond the reference API)."""
+
+        class _MapBatchesDescriptor:
+            def __get__(self, obj: Any, objtype: Any = None) -> Any:
+                if obj is None:
+                    return self
+                impl = obj._impl
+
+                def blocking(*args: Any, **kwargs: Any) -> Any:
+                    agen = getattr(impl, batches_name)(*unwrap(args), **unwrap(kwargs))
+                    for batch in synchronizer.run_generator_sync(agen):
+                        yield wrap(batch)
+
+                async def aio(*args: Any, **kwargs: Any) -> Any:
+                    agen = getattr(impl, batches_name)(*unwrap(args), **unwrap(kwargs))
+                    if synchronizer.in_loop_thread():
+                        async for batch in agen:
+                            yield batch
+                        return
+                    async for batch in synchronizer.run_generator_async(agen):
+                        yield wrap(batch)
+
+                from ._sync import _AioCallable
+
+                return _AioCallable(blocking, aio, getattr(impl, batches_name))
+
+        return _MapBatchesDescriptor()
+
+    setattr(Function, "map_batches", make_batches("map_async_batches"))
+    setattr(Function, "starmap_batches", make_batches("starmap_async_batches"))
+
     class _ForEachDescriptor:
         def __get__(self, obj: Any, objtype: Any = None) -> Any:
             if obj is None:

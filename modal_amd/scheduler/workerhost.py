@@ -29,6 +29,8 @@ from .calls import (
 )
 from .transport import Connection
 
+_CPU_COUNT = os.cpu_count()  # cached: _pool_limit runs per dispatch
+
 if TYPE_CHECKING:
     from .core import Scheduler
 
@@ -305,7 +307,7 @@ class WorkerPool:
                 for i in range(count):
                     await self.spawn_worker(gpu_index=i % n_gpus if n_gpus else None)
             else:
-                count = config.get("worker_count") or min(max((os.cpu_count() or 4) // 2, 1), 8)
+                count = config.get("worker_count") or min(max((_CPU_COUNT or 4) // 2, 1), 8)
                 for _ in range(count):
                     await self.spawn_worker(gpu_index=None)
 
@@ -692,7 +694,7 @@ class WorkerPool:
             return max(configured or 0, n_gpus, 1)
         if configured:
             return configured
-        return min(max((os.cpu_count() or 4) // 2, 1), 8)
+        return min(max((_CPU_COUNT or 4) // 2, 1), 8)
 
     async def _send_def(self, w: WorkerHandle, fdef: FunctionDef) -> None:
         await w.conn.send(

@@ -142,3 +142,38 @@ def test_native_core_asan_clean():
     )
     assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
     assert "ASAN CHECK OK" in out.stdout
+
+
+def test_map_batches_api(client):
+    """fn.map_batches yields lists covering every item exactly once (both
+    sync and .aio forms); starmap_batches matches starmap."""
+    import modal_amd as modal
+
+    app = modal.App("mb")
+
+    @app.function()
+    def sq(x):
+        return x * x
+
+    @app.function()
+    def add(a, b):
+        return a + b
+
+    with app.run(client=client):
+        got = []
+        for batch in sq.map_batches(range(500)):
+            assert isinstance(batch, list)
+            got.extend(batch)
+        assert sorted(got) == sorted(x * x for x in range(500))
+
+        from modal_amd._sync import synchronizer
+
+        async def consume():
+            out = []
+            async for batch in sq.map_batches.aio(range(300), order_outputs=False):
+                out.extend(batch)
+            return out
+
+        assert sorted(synchronizer.run(consume())) == sorted(x * x for x in range(300))
+        flat = [v for b in add.starmap_batches([(1, 2), (3, 4), (5, 6)]) for v in b]
+        assert sorted(flat) == [3, 7, 11]
