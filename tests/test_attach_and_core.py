@@ -73,6 +73,18 @@ with app.run():
     assert double.remote(21) == 42
     out = sorted(double.map(range(20), order_outputs=False))
     assert out == [2 * x for x in range(20)]
+
+@app.function()
+@modal.batched(max_batch_size=8, wait_ms=1)
+def bdouble(xs):
+    return [x * 2 for x in xs]
+
+with app.run():
+    # batched functions over the proxied chunk intake (one-way putc frames)
+    got = sorted(bdouble.map(range(200), order_outputs=False))
+    assert got == sorted(2 * x for x in range(200))
+    flat = [v for b in bdouble.map_batches(range(100)) for v in b]
+    assert flat == [2 * x for x in range(100)]
     with modal.Queue.ephemeral() as q:
         q.put("cross-process")
         assert q.get() == "cross-process"
