@@ -976,6 +976,8 @@ class WorkerRuntime:
                                     self.executor, _call_batch
                                 )
                             _store(idxs, results)
+                        except (asyncio.CancelledError, InputCancellation):
+                            raise  # cancel: abort the chunk, don't mark failed
                         except BaseException as exc:
                             data = self._serialize_exception(exc)
                             err = "".join(
@@ -983,6 +985,8 @@ class WorkerRuntime:
                             ).strip()
                             for ci in idxs:
                                 errors[ci] = (data, err)
+                except (asyncio.CancelledError, InputCancellation):
+                    raise
                 except BaseException as exc:
                     data = self._serialize_exception(exc)
                     err = "".join(traceback.format_exception_only(type(exc), exc)).strip()
@@ -1020,7 +1024,7 @@ class WorkerRuntime:
                 await asyncio.gather(
                     *(run_one(s, min(s + size, count)) for s in range(0, count, size))
                 )
-        except asyncio.CancelledError:
+        except (asyncio.CancelledError, InputCancellation):
             return
         finally:
             _app_id_var.reset(app_tok)
