@@ -452,6 +452,113 @@ py::bytes malz_decompress(py::bytes blob) {
 }
 
 
+
+
+// ---------------------------------------------------------------------------
+// Sandbox supervisor: clone3(CLONE_INTO_CGROUP) + execve.
+//
+// The Python road (asyncio preexec_fn) runs Python between fork and exec in
+// a threaded process — both unsafe and racy for cgroup attachment (the child
+// runs before the parent can write cgroup.procs). clone3 places the child in
+// the target cgroup v2 ATOMICALLY at creation. Parity target: the
+// reference's Go sandbox supervisor, done the Linux-native
+// way.
+// ---------------------------------------------------------------------------
+#include <sys/syscall.h>
+#include <sys/resource.h>
+#include <sched.h>
+#include <signal.h>
+
+namespace {
+
+struct clone_args_compat {
+  uint64_t flags;
+  uint64_t pidfd;
+  uint64_t child_tid;
+  uint64_t parent_tid;
+  uint64_t exit_signal;
+  uint64_t stack;
+  uint64_t stack_size;
+  uint64_t tls;
+  uint64_t set_tid;
+  uint64_t set_tid_size;
+  uint64_t cgroup;
+};
+
+#ifndef CLONE_INTO_CGROUP
+#define CLONE_INTO_CGROUP 0x200000000ULL
+#endif
+
+}  // namespace
+
+// Returns the child pid. Throws on failure. stdio fds are dup2'd onto
+// 0/1/2 in the child (pass -1 to inherit). The child setsid()s so the whole
+// tree is killable as one process group.
+static int64_t spawn_supervised(const std::vector<std::string>& argv,
+                                const std::string& cwd,
+                                const std::vector<std::string>& env,
+                                const std::string& cgroup_dir,
+                                int64_t rlimit_as_mib,
+                                int stdin_fd, int stdout_fd, int stderr_fd) {
+  if (argv.empty()) throw std::runtime_error("empty argv");
+  int cg_fd = -1;
+  if (!cgroup_dir.empty()) {
+    cg_fd = ::open(cgroup_dir.c_str(), O_DIRECTORY | O_RDONLY | O_CLOEXEC);
+    if (cg_fd < 0)
+      throw std::runtime_error("open cgroup dir: " + std::string(strerror(errno)));
+  }
+
+  std::vector<char*> cargv;
+  cargv.reserve(argv.size() + 1);
+  for (const auto& a : argv) cargv.push_back(const_cast<char*>(a.c_str()));
+  cargv.push_back(nullptr);
+  std::vector<char*> cenv;
+  cenv.reserve(env.size() + 1);
+  for (const auto& e : env) cenv.push_back(const_cast<char*>(e.c_str()));
+  cenv.push_back(nullptr);
+
+  clone_args_compat ca{};
+  ca.exit_signal = SIGCHLD;
+  if (cg_fd >= 0) {
+    ca.flags = CLONE_INTO_CGROUP;
+    ca.cgroup = (uint64_t)cg_fd;
+  }
+
+  long pid = syscall(SYS_clone3, &ca, sizeof(ca));
+  if (pid < 0 && cg_fd >= 0 && (errno == EINVAL || errno == ENOSYS || errno == EPERM ||
+                                errno == EBADF || errno == EOPNOTSUPP)) {
+    // kernel/cgroup too old or not delegated: plain clone3; the caller's
+    // Python-side attach (child writes cgroup.procs) remains the fallback
+    ::close(cg_fd);
+    cg_fd = -1;
+    throw std::runtime_error("clone3(CLONE_INTO_CGROUP) unavailable: " +
+                             std::string(strerror(errno)));
+  }
+  if (pid < 0) {
+    if (cg_fd >= 0) ::close(cg_fd);
+    throw std::runtime_error("clone3: " + std::string(strerror(errno)));
+  }
+  if (pid == 0) {
+    // child: async-signal-safe calls only
+    ::setsid();
+    if (stdin_fd >= 0) ::dup2(stdin_fd, 0);
+    if (stdout_fd >= 0) ::dup2(stdout_fd, 1);
+    if (stderr_fd >= 0) ::dup2(stderr_fd, 2);
+    for (int fd = 3; fd < 1024; ++fd) ::close(fd);
+    if (!cwd.empty() && ::chdir(cwd.c_str()) != 0) _exit(126);
+    if (rlimit_as_mib > 0) {
+      struct rlimit rl;
+      rl.rlim_cur = rl.rlim_max = (rlim_t)rlimit_as_mib << 20;
+      ::setrlimit(RLIMIT_AS, &rl);
+    }
+    ::execve(cargv[0], cargv.data(), cenv.data());
+    _exit(127);
+  }
+  if (cg_fd >= 0) ::close(cg_fd);
+  return (int64_t)pid;
+}
+
+
 PYBIND11_MODULE(_core, m) {
   m.doc() = "modal_amd native core: shm ring transport + batch framing";
   py::class_<ShmRing>(m, "ShmRing")
@@ -465,4 +572,8 @@ PYBIND11_MODULE(_core, m) {
   m.def("malz_compress", &malz_compress, py::arg("data"), py::arg("min_gain") = 0.95);
   m.def("malz_decompress", &malz_decompress);
   m.def("unpack_payloads", &unpack_payloads);
+  m.def("spawn_supervised", &spawn_supervised, py::arg("argv"), py::arg("cwd"),
+        py::arg("env"), py::arg("cgroup_dir") = "", py::arg("rlimit_as_mib") = 0,
+        py::arg("stdin_fd") = -1, py::arg("stdout_fd") = -1, py::arg("stderr_fd") = -1,
+        "clone3(CLONE_INTO_CGROUP) + execve supervisor (atomic cgroup placement)");
 }

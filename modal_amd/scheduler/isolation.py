@@ -109,6 +109,7 @@ class CgroupBox:
         self.cpu = cpu
         self.pids_max = pids_max
         self._dirs: list[str] = []  # controller dirs to attach/cleanup
+        self.v2_dir: Optional[str] = None  # set when the box is cgroup v2
 
     def create(self) -> bool:
         caps = capabilities()
@@ -137,6 +138,7 @@ class CgroupBox:
                 if self.cpu:
                     self._write(d, "cpu.max", f"{int(self.cpu * 100000)} 100000")
                 self._dirs.append(d)
+                self.v2_dir = d
                 made = True
             except OSError:
                 pass

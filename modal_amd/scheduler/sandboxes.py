@@ -350,25 +350,50 @@ class SandboxService:
             state.ns_root = ns_meta["mnt"]
             state.ns_ready = ns_meta["ready"]
 
-        rlimit_fallback = _resource_preexec(cpu, None if cg is not None else memory)
+        # native spawner first: clone3(CLONE_INTO_CGROUP) places the child in
+        # its cgroup v2 ATOMICALLY, and the child runs only C between clone
+        # and exec (no Python preexec_fn in a threaded process). Python
+        # fallback covers cgroup v1 boxes and MODAL_AMD_PY_SPAWN=1.
+        from . import supervisor as _sup
 
-        def preexec() -> None:
-            if rlimit_fallback is not None:
-                rlimit_fallback()  # includes setsid
-            else:
-                os.setsid()
-            if cg is not None:
-                cg.attach_pid_in_child()  # inherited by the whole subtree
-
-        proc = await asyncio.create_subprocess_exec(
-            *args,
-            cwd=state.workdir,
-            env=full_env,
-            stdin=asyncio.subprocess.PIPE,
-            stdout=asyncio.subprocess.PIPE,
-            stderr=asyncio.subprocess.PIPE,
-            preexec_fn=preexec,
+        cxx_ok = (
+            os.environ.get("MODAL_AMD_PY_SPAWN") != "1"
+            and _sup.available()
+            and (cg is None or cg.v2_dir is not None)
         )
+        proc = None
+        if cxx_ok:
+            try:
+                proc = await _sup.spawn(
+                    list(args),
+                    cwd=state.workdir,
+                    env=full_env,
+                    cgroup_dir=(cg.v2_dir if cg is not None else "") or "",
+                    rlimit_as_mib=int(memory) if (memory and cg is None) else 0,
+                    cpu=cpu,
+                )
+            except Exception:
+                proc = None  # e.g. clone3 blocked by seccomp: take the Python road
+        if proc is None:
+            rlimit_fallback = _resource_preexec(cpu, None if cg is not None else memory)
+
+            def preexec() -> None:
+                if rlimit_fallback is not None:
+                    rlimit_fallback()  # includes setsid
+                else:
+                    os.setsid()
+                if cg is not None:
+                    cg.attach_pid_in_child()  # inherited by the whole subtree
+
+            proc = await asyncio.create_subprocess_exec(
+                *args,
+                cwd=state.workdir,
+                env=full_env,
+                stdin=asyncio.subprocess.PIPE,
+                stdout=asyncio.subprocess.PIPE,
+                stderr=asyncio.subprocess.PIPE,
+                preexec_fn=preexec,
+            )
         await state.main.attach(proc)
         if timeout:
             state.timeout = timeout

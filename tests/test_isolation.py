@@ -149,3 +149,67 @@ def test_sandbox_sees_image_layer(client):
     sb.wait()
     assert sb.returncode == 0, sb.stderr.read()
     assert "from-image" in sb.stdout.read()
+
+
+def test_cxx_supervisor_spawn_unit():
+    """The native clone3 spawner: stdio pipes, exit codes, signals, setsid."""
+    import asyncio
+
+    from modal_amd.scheduler import supervisor
+
+    if not supervisor.available():
+        pytest.skip("native core without spawn_supervised")
+
+    async def main():
+        p = await supervisor.spawn(["bash", "-c", "read l; echo ok:$l; exit 5"])
+        p.stdin.write(b"x\n")
+        await p.stdin.drain()
+        p.stdin.close()
+        assert await p.stdout.read() == b"ok:x\n"
+        assert await p.wait() == 5
+        p2 = await supervisor.spawn(["sleep", "30"])
+        assert os.getpgid(p2.pid) == p2.pid  # setsid: own process group
+        p2.terminate()
+        assert await p2.wait() == -15
+
+    asyncio.run(main())
+
+
+def test_sandbox_uses_cxx_spawner(client):
+    """Sandboxes created through the native path behave identically
+    (stdio capture, returncode, termination)."""
+    import modal_amd as modal
+
+    sb = modal.Sandbox.create("bash", "-c", "echo native-spawn; exit 7", client=client)
+    rc = sb.wait(raise_on_termination=False)
+    assert rc == 7
+    assert "native-spawn" in sb.stdout.read()
+
+
+def test_cxx_supervisor_cgroup_attach(client):
+    """With delegated cgroup v2, the child lands in its box atomically."""
+    from modal_amd.scheduler.isolation import CgroupBox, capabilities
+
+    if not capabilities().get("cgv2"):
+        pytest.skip("no cgroup v2 delegation")
+    cg = CgroupBox("cxx-attach-test", memory_mib=256)
+    if not cg.create() or cg.v2_dir is None:
+        pytest.skip("cgroup v2 box creation not permitted")
+    import asyncio
+
+    from modal_amd.scheduler import supervisor
+
+    async def main():
+        p = await supervisor.spawn(
+            ["bash", "-c", "cat /proc/self/cgroup; sleep 0.1"],
+            cgroup_dir=cg.v2_dir,
+        )
+        out = await p.stdout.read()
+        await p.wait()
+        return out.decode()
+
+    try:
+        out = asyncio.run(main())
+        assert "cxx-attach-test" in out, out
+    finally:
+        cg.cleanup()
