@@ -356,10 +356,14 @@ class SandboxService:
         # fallback covers cgroup v1 boxes and MODAL_AMD_PY_SPAWN=1.
         from . import supervisor as _sup
 
+        # the native spawner pays a full fork of this (torch-loaded) process
+        # (~1 ms of page-table copy; CPython's no-preexec road uses vfork),
+        # so take it only when its atomic cgroup placement matters
         cxx_ok = (
             os.environ.get("MODAL_AMD_PY_SPAWN") != "1"
             and _sup.available()
-            and (cg is None or cg.v2_dir is not None)
+            and cg is not None
+            and cg.v2_dir is not None
         )
         proc = None
         if cxx_ok:
@@ -486,7 +490,8 @@ class SandboxService:
             if sb.cgroup is not None and getattr(sb.cgroup, "v2_dir", None):
                 cgdir = sb.cgroup.v2_dir
             proc = None
-            if os.environ.get("MODAL_AMD_PY_SPAWN") != "1" and _sup.available():
+            # native only when joining the sandbox cgroup (else vfork wins)
+            if cgdir and os.environ.get("MODAL_AMD_PY_SPAWN") != "1" and _sup.available():
                 try:
                     proc = await _sup.spawn(
                         list(argv), cwd=cwd or "", env=full_env, cgroup_dir=cgdir
