@@ -258,3 +258,24 @@ def test_batched_map_rides_chunks(client):
         assert len(bad) == 16  # only the batch containing 40 fails
         assert sum(1 for r in res if not isinstance(r, Exception)) == 48
         assert double.remote(5) == 10  # unary path still batches singles
+
+
+@pytest.mark.parametrize(
+    "batch,chunk",
+    [(3, 8), (8, 8), (13, 8), (64, 16), (7, 64), (100, 64), (64, 256)],
+)
+def test_batched_chunk_geometry(client, monkeypatch, batch, chunk):
+    """Protocol geometry sweep: batch size vs chunk size in every relation
+    (batch < chunk, =, >, non-divisible) keeps map exactly-once + ordered."""
+    monkeypatch.setenv("MODAL_AMD_CHUNK_ITEMS", str(chunk))
+    app = modal.App(f"geo-{batch}-{chunk}")
+
+    @app.function()
+    @modal.batched(max_batch_size=batch, wait_ms=1)
+    def f(xs):
+        assert len(xs) <= batch
+        return [x * 3 for x in xs]
+
+    with app.run(client=client):
+        n = 311  # prime: never divides evenly into batches or chunks
+        assert list(f.map(range(n))) == [x * 3 for x in range(n)]
