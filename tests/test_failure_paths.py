@@ -382,3 +382,39 @@ def test_bigarg_xfer_survives_worker_death(client, run_dir):
     assert os.path.exists(marker), "kill never triggered (fixture too lucky)"
     leftovers = os.listdir(os.path.join(run_dir, "xfer"))
     assert leftovers == [], f"xfer files leaked after worker death: {leftovers}"
+
+
+def test_batched_chunk_worker_death_requeues(client, run_dir):
+    """A worker killed mid-batched-chunk: the chunk group requeues and every
+    item still arrives exactly once (the batched path shares the range
+    protocol's redelivery)."""
+    app = modal.App("batch-death")
+
+    @app.function()
+    @modal.batched(max_batch_size=16, wait_ms=1)
+    def slowish(xs):
+        import time as _t
+
+        _t.sleep(0.02)
+        return [x + 100 for x in xs]
+
+    with app.run(client=client):
+        import threading
+
+        svc = client.svc
+
+        def killer():
+            time.sleep(0.15)
+            for w in list(svc.pool.workers.values()):
+                if w.alive and w.proc is not None:
+                    try:
+                        w.proc.kill()
+                    except Exception:
+                        pass
+                    break
+
+        t = threading.Thread(target=killer, daemon=True)
+        t.start()
+        got = sorted(slowish.map(range(400), order_outputs=False))
+        assert got == [x + 100 for x in range(400)]
+        t.join()
